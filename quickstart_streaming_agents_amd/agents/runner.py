@@ -1,0 +1,229 @@
+"""AI_RUN_AGENT / AI_TOOL_INVOKE: the per-record agent episode machine.
+
+Semantics contract (LAB1-Walkthrough.md:155-181, LAB3:396-447, LAB4:330-384):
+``CREATE AGENT ... USING MODEL m USING PROMPT p [USING TOOLS t] WITH
+('max_iterations'='10', 'max_consecutive_failures'='2')`` and per-row
+``AI_RUN_AGENT(agent, user_prompt[, key][, debug])`` -> ``(status, response)``:
+a reason -> tool -> observe loop bounded by the iteration cap, aborting after
+N consecutive tool failures, with the free-text response post-parsed by
+REGEXP_EXTRACT (agents/parse.py).
+
+Episodes are *generators* so thousands of concurrent per-record episodes can
+share batched GPU decode: an episode yields ("llm", prompt) or
+("tool", name, args) requests and is resumed with the result.  The serving
+engine (models/serve.py) gangs the "llm" requests into continuous-batching
+decode steps while "tool" requests run on an I/O thread pool — the GPU never
+waits on MCP round trips.
+
+Action selection is pluggable: ``ToolCallPolicy`` parses tool-call syntax
+from model output (for instruction-tuned weights); the lab harnesses install
+deterministic policies (labs/pipelines.py) because the north-star benchmark
+runs random-init weights — the LLM decode is real compute, the act decisions
+are scripted and documented as such.
+"""
+
+from __future__ import annotations
+
+import json
+import re
+from dataclasses import dataclass, field
+from typing import Any, Callable, Generator
+
+
+@dataclass
+class ToolDef:
+    name: str
+    description: str = ""
+
+
+@dataclass
+class ToolSet:
+    """CREATE TOOL ... WITH ('type'='mcp','allowed_tools'='...','request_timeout'=N)."""
+    name: str
+    connection: str = ""
+    allowed_tools: tuple[str, ...] = ()
+    request_timeout_s: float = 30.0
+
+    def allows(self, tool: str) -> bool:
+        return not self.allowed_tools or tool in self.allowed_tools
+
+
+@dataclass
+class AgentSpec:
+    """CREATE AGENT definition."""
+    name: str
+    model: str
+    prompt: str
+    tools: ToolSet | None = None
+    max_iterations: int = 10
+    max_consecutive_failures: int = 2
+    options: dict = field(default_factory=dict)
+
+
+# --- episode actions a policy can return -----------------------------------
+
+@dataclass
+class ToolCall:
+    name: str
+    arguments: dict
+
+
+@dataclass
+class Finish:
+    response: str
+
+
+@dataclass
+class Continue:
+    note: str = ""
+
+
+Action = "ToolCall | Finish | Continue"
+
+
+class ToolCallPolicy:
+    """Parse ``TOOL_CALL {json}`` syntax from model output; finish otherwise."""
+
+    TOOL_RE = re.compile(r"TOOL_CALL\s*:?\s*(\{[\s\S]*?\})", re.IGNORECASE)
+
+    def __call__(self, text: str, iteration: int, ctx: dict) -> Any:
+        m = self.TOOL_RE.search(text or "")
+        if m:
+            try:
+                call = json.loads(m.group(1))
+                return ToolCall(call.get("name", ""),
+                                call.get("arguments", {}) or {})
+            except json.JSONDecodeError:
+                return Continue("unparseable tool call")
+        return Finish(text)
+
+
+@dataclass
+class EpisodeResult:
+    status: str            # SUCCESS | FAILED
+    response: str
+    iterations: int = 0
+    tool_calls: int = 0
+    failures: int = 0
+    trace: list[dict] = field(default_factory=list)
+
+
+def episode(agent: AgentSpec, user_prompt: str,
+            policy: Callable[[str, int, dict], Any] | None = None,
+            max_new_tokens: int = 128, debug: bool = False,
+            ) -> Generator[tuple, Any, EpisodeResult]:
+    """The AI_RUN_AGENT loop as a resumable generator.
+
+    Yields ("llm", prompt, max_new_tokens) and ("tool", name, arguments);
+    caller .send()s the string result back.  Tool errors are sent as
+    exceptions via .throw() or as ("__error__", msg) sentinel strings.
+    Returns EpisodeResult via StopIteration.value.
+    """
+    policy = policy or ToolCallPolicy()
+    transcript: list[str] = []
+    ctx: dict = {"agent": agent, "observations": [], "user_prompt": user_prompt}
+    consecutive_failures = 0
+    tool_calls = 0
+    trace: list[dict] = []
+    last_text = ""
+
+    for it in range(agent.max_iterations):
+        prompt = _build_prompt(agent, user_prompt, transcript)
+        text = yield ("llm", prompt, max_new_tokens)
+        last_text = text or ""
+        if debug:
+            trace.append({"iteration": it, "model_output": last_text})
+        action = policy(last_text, it, ctx)
+        if isinstance(action, Finish):
+            return EpisodeResult("SUCCESS", action.response, it + 1, tool_calls,
+                                 0, trace)
+        if isinstance(action, Continue):
+            transcript.append(f"[thought] {action.note or last_text[:200]}")
+            continue
+        if isinstance(action, ToolCall):
+            if agent.tools is not None and not agent.tools.allows(action.name):
+                result = f"__error__ tool {action.name} not allowed"
+            else:
+                result = yield ("tool", action.name, action.arguments)
+            tool_calls += 1
+            if isinstance(result, str) and result.startswith("__error__"):
+                consecutive_failures += 1
+                if debug:
+                    trace.append({"iteration": it, "tool": action.name,
+                                  "error": result})
+                if consecutive_failures >= agent.max_consecutive_failures:
+                    return EpisodeResult(
+                        "FAILED",
+                        f"aborted after {consecutive_failures} consecutive "
+                        f"tool failures: {result}",
+                        it + 1, tool_calls, consecutive_failures, trace)
+                transcript.append(f"[tool {action.name} failed] {result}")
+            else:
+                consecutive_failures = 0
+                ctx["observations"].append((action.name, result))
+                if debug:
+                    trace.append({"iteration": it, "tool": action.name,
+                                  "result": str(result)[:500]})
+                transcript.append(f"[observation from {action.name}] {result}")
+    return EpisodeResult("FAILED", last_text, agent.max_iterations, tool_calls,
+                         consecutive_failures, trace)
+
+
+def _build_prompt(agent: AgentSpec, user_prompt: str,
+                  transcript: list[str]) -> str:
+    parts = [agent.prompt, "", user_prompt]
+    if transcript:
+        parts.append("")
+        parts.extend(transcript)
+    return "\n".join(parts)
+
+
+# ---------------------------------------------------------------------------
+# Synchronous drivers (CPU tests / single-record paths)
+# ---------------------------------------------------------------------------
+
+
+def drive_episode(ep: Generator, llm: Callable[[str, int], str],
+                  tool: Callable[[str, dict], str]) -> EpisodeResult:
+    """Run one episode to completion with direct (unbatched) calls."""
+    try:
+        req = ep.send(None)
+        while True:
+            if req[0] == "llm":
+                result = llm(req[1], req[2])
+            elif req[0] == "tool":
+                try:
+                    result = tool(req[1], req[2])
+                except Exception as e:
+                    result = f"__error__ {e}"
+            else:
+                raise ValueError(f"unknown request {req[0]!r}")
+            req = ep.send(result)
+    except StopIteration as stop:
+        return stop.value
+
+
+def ai_tool_invoke(model_llm: Callable[[str, int], str],
+                   tool: Callable[[str, dict], str], prompt: str,
+                   tool_descriptions: dict[str, str],
+                   debug: bool = False) -> dict:
+    """AI_TOOL_INVOKE(model, prompt, MAP[], MAP[tool->desc], MAP[debug]):
+    one LLM call + at most one tool round trip (LAB1-Walkthrough.md:80-92).
+    Returns {tool -> result} plus the model response."""
+    tool_block = "\n".join(f"- {name}: {desc}"
+                           for name, desc in tool_descriptions.items())
+    text = model_llm(f"{prompt}\n\nAvailable tools:\n{tool_block}", 128)
+    out: dict[str, str] = {"response": text}
+    m = ToolCallPolicy.TOOL_RE.search(text or "")
+    if m:
+        try:
+            call = json.loads(m.group(1))
+            name = call.get("name", "")
+            if name in tool_descriptions:
+                try:
+                    out[name] = tool(name, call.get("arguments", {}) or {})
+                except Exception as e:
+                    out[name] = f"__error__ {e}"
+        except json.JSONDecodeError:
+            pass
+    return out

@@ -1,0 +1,175 @@
+"""Topic / partition abstraction (the Kafka-shaped edge of the engine).
+
+An in-process, optionally file-backed log: topics with N partitions, each an
+append-only sequence of (key, value, timestamp_ms) records with offsets.
+Mirrors the behavior the reference's publishers and tests rely on:
+
+- purge-then-publish recovery (reference lab4_datagen.py:294-304 purges the
+  claims topic + downstream topics before re-publishing),
+- message counts from watermark offsets without consuming
+  (testing/helpers/kafka_helper.py:88-119),
+- single-partition publishing for watermark determinism
+  (publish_lab1_data.py:264 pins partition=0).
+
+Values are raw bytes (Confluent Avro wire format via wire.avro) or any
+Python object for in-memory pipelines; the engine's Avro boundary
+encodes/decodes at the edge.
+"""
+
+from __future__ import annotations
+
+import threading
+from dataclasses import dataclass
+from typing import Any
+
+from .avro import Schema, deserialize, serialize
+from .registry import SchemaRegistry
+
+
+@dataclass(frozen=True)
+class Record:
+    topic: str
+    partition: int
+    offset: int
+    timestamp_ms: int
+    key: Any
+    value: Any
+
+
+class Partition:
+    __slots__ = ("_records", "_lock")
+
+    def __init__(self) -> None:
+        self._records: list[Record] = []
+        self._lock = threading.Lock()
+
+    def append(self, rec: Record) -> int:
+        with self._lock:
+            object.__setattr__(rec, "offset", len(self._records))
+            self._records.append(rec)
+            return rec.offset
+
+    @property
+    def end_offset(self) -> int:
+        return len(self._records)
+
+    def read(self, start: int, max_count: int | None = None) -> list[Record]:
+        end = len(self._records)
+        if max_count is not None:
+            end = min(end, start + max_count)
+        return self._records[start:end]
+
+    def truncate_all(self) -> None:
+        with self._lock:
+            self._records.clear()
+
+
+class Topic:
+    def __init__(self, name: str, num_partitions: int = 1):
+        self.name = name
+        self.partitions = [Partition() for _ in range(num_partitions)]
+
+    @property
+    def num_partitions(self) -> int:
+        return len(self.partitions)
+
+    def append(self, value: Any, key: Any = None, timestamp_ms: int = 0,
+               partition: int | None = None) -> Record:
+        if partition is None:
+            partition = (hash(key) % len(self.partitions)) if key is not None else 0
+        rec = Record(self.name, partition, -1, timestamp_ms, key, value)
+        self.partitions[partition].append(rec)
+        return rec
+
+    def message_count(self) -> int:
+        """Count from end offsets without consuming (kafka_helper.py:88-119)."""
+        return sum(p.end_offset for p in self.partitions)
+
+    def purge(self) -> None:
+        """delete_records at the latest watermark == drop everything."""
+        for p in self.partitions:
+            p.truncate_all()
+
+    def read_all(self) -> list[Record]:
+        out: list[Record] = []
+        for p in self.partitions:
+            out.extend(p.read(0))
+        out.sort(key=lambda r: (r.timestamp_ms, r.partition, r.offset))
+        return out
+
+
+class Broker:
+    """Holds topics + a schema registry; the process-local 'cluster'."""
+
+    def __init__(self, registry: SchemaRegistry | None = None):
+        self.topics: dict[str, Topic] = {}
+        self.registry = registry or SchemaRegistry()
+        self._lock = threading.Lock()
+
+    def create_topic(self, name: str, num_partitions: int = 1,
+                     if_not_exists: bool = True) -> Topic:
+        with self._lock:
+            t = self.topics.get(name)
+            if t is not None:
+                if if_not_exists:
+                    return t
+                raise ValueError(f"topic exists: {name}")
+            t = Topic(name, num_partitions)
+            self.topics[name] = t
+            return t
+
+    def topic(self, name: str) -> Topic:
+        return self.topics[name]
+
+    def delete_topic(self, name: str) -> None:
+        with self._lock:
+            self.topics.pop(name, None)
+
+
+class AvroProducer:
+    """Produce Python dicts as Confluent-Avro wire bytes onto a topic."""
+
+    def __init__(self, broker: Broker, topic: str, value_schema: Schema | str | dict,
+                 key_schema: Schema | str | dict | None = None):
+        self.broker = broker
+        self.topic = broker.create_topic(topic)
+        self.value_schema = value_schema if isinstance(value_schema, Schema) else Schema(value_schema)
+        self.value_schema_id = broker.registry.register(f"{topic}-value", self.value_schema)
+        self.key_schema = None
+        self.key_schema_id = None
+        if key_schema is not None:
+            self.key_schema = key_schema if isinstance(key_schema, Schema) else Schema(key_schema)
+            self.key_schema_id = broker.registry.register(f"{topic}-key", self.key_schema)
+
+    def produce(self, value: dict, key: Any = None, timestamp_ms: int = 0,
+                partition: int | None = None) -> Record:
+        raw_v = serialize(self.value_schema, self.value_schema_id, value)
+        raw_k = key
+        if self.key_schema is not None and key is not None:
+            raw_k = serialize(self.key_schema, self.key_schema_id, key)
+        return self.topic.append(raw_v, key=raw_k, timestamp_ms=timestamp_ms,
+                                 partition=partition)
+
+
+class AvroConsumer:
+    """Consume wire-format records back into dicts (Avro-or-passthrough tolerant)."""
+
+    def __init__(self, broker: Broker, topic: str, value_schema: Schema | str | dict):
+        self.broker = broker
+        self.topic_name = topic
+        self.value_schema = value_schema if isinstance(value_schema, Schema) else Schema(value_schema)
+        self._offsets: dict[int, int] = {}
+
+    def poll(self, max_count: int | None = None) -> list[tuple[Record, Any]]:
+        topic = self.broker.topic(self.topic_name)
+        out: list[tuple[Record, Any]] = []
+        for pi, p in enumerate(topic.partitions):
+            start = self._offsets.get(pi, 0)
+            recs = p.read(start, max_count)
+            for r in recs:
+                v = r.value
+                if isinstance(v, (bytes, bytearray)):
+                    _, v = deserialize(self.value_schema, bytes(v))
+                out.append((r, v))
+            self._offsets[pi] = start + len(recs)
+        return out

@@ -1,0 +1,150 @@
+"""Windows, watermarks, anomaly detection, joins: semantics + determinism."""
+
+import numpy as np
+
+from quickstart_streaming_agents_amd.runtime.anomaly import (
+    AnomalyDetector, ar_forecast, normal_quantile,
+)
+from quickstart_streaming_agents_amd.runtime.joins import (
+    TTLTable, enrich_join, interval_join,
+)
+from quickstart_streaming_agents_amd.runtime.windows import (
+    TumblingWindows, aggregate,
+)
+
+MIN5 = 5 * 60 * 1000
+
+
+def _rows(zone_counts, window, base=0):
+    rows = []
+    for zone, n in zone_counts.items():
+        for i in range(n):
+            rows.append({"pickup_zone": zone,
+                         "request_ts": base + window * MIN5 + i * 1000})
+    return rows
+
+
+def test_tumbling_window_close_on_watermark():
+    tw = TumblingWindows(MIN5, lambda r: r["pickup_zone"],
+                         lambda r: r["request_ts"], watermark_delay_ms=5000)
+    closed = tw.feed(_rows({"A": 3}, 0))
+    assert closed == []  # watermark has not passed window end
+    closed = tw.feed(_rows({"A": 2}, 1))  # ts in window 1 advances watermark
+    # window 1 max ts = 5min+1000 -> wm just past 5min end? max= 300000+1000;
+    # wm = 301000-5000 = 296000 < 300000 so window 0 still open
+    assert closed == []
+    closed = tw.feed([{"pickup_zone": "A", "request_ts": MIN5 + 10_000}])
+    assert len(closed) == 1
+    assert closed[0].window_start == 0 and len(closed[0].rows) == 3
+    assert closed[0].window_time == MIN5 - 1
+
+
+def test_late_rows_dropped():
+    tw = TumblingWindows(MIN5, lambda r: r["pickup_zone"],
+                         lambda r: r["request_ts"], watermark_delay_ms=5000)
+    tw.feed([{"pickup_zone": "A", "request_ts": 3 * MIN5}])
+    closed = tw.feed([{"pickup_zone": "A", "request_ts": 1000}])  # very late
+    assert all(w.window_start != 0 for w in closed)
+    assert tw._late_dropped == 1
+
+
+def test_flush_closes_all_and_aggregates():
+    tw = TumblingWindows(MIN5, lambda r: r["pickup_zone"],
+                         lambda r: r["request_ts"])
+    tw.feed(_rows({"A": 4, "B": 2}, 0))
+    panes = tw.flush()
+    rows = aggregate(panes, {"request_count": len})
+    assert {(r["key"], r["request_count"]) for r in rows} == {("A", 4), ("B", 2)}
+
+
+def test_normal_quantile():
+    assert abs(normal_quantile(0.975) - 1.959964) < 1e-4
+    assert abs(normal_quantile(0.9995) - 3.290527) < 1e-3
+    assert abs(normal_quantile(0.5)) < 1e-12
+
+
+def test_ar_forecast_linear_trend():
+    # AR on a clean linear trend must forecast the next point closely.
+    hist = np.arange(1.0, 41.0)
+    f, s, dof = ar_forecast(hist, 4)
+    # ridge shrinkage tolerates a small mean-reversion bias on a pure ramp
+    assert abs(f - 41.0) < 2.0
+    assert s < 2.0
+    assert dof >= 30
+
+
+def test_detector_min_training_gate():
+    det = AnomalyDetector(min_training_size=5, confidence_percentage=99.0)
+    for i in range(5):
+        r = det.update("k", 100.0)
+        assert not r.is_anomaly  # still training
+    r = det.update("k", 100.0)
+    assert not r.is_anomaly
+    r = det.update("k", 10_000.0)
+    assert r.is_anomaly and r.forecast_value < 10_000.0
+
+
+def test_detector_spike_detection_with_jitter():
+    rng = np.random.default_rng(0)
+    det = AnomalyDetector(min_training_size=20, max_training_size=500,
+                          confidence_percentage=99.9)
+    flagged = []
+    for i in range(100):
+        v = 15.0 + rng.normal(0, 1.5)
+        if i == 99:
+            v = 75.0
+        flagged.append(det.update("z", float(v)).is_anomaly)
+    assert flagged[99] is True
+    assert sum(flagged[:99]) == 0  # no false positives on the steady stream
+
+
+def test_detector_max_training_window():
+    det = AnomalyDetector(min_training_size=3, max_training_size=10)
+    for i in range(50):
+        det.update("k", float(i % 7))
+    assert len(det._history["k"]) == 10
+
+
+def test_ttl_table_eviction():
+    t = TTLTable(lambda r: r["id"], ttl_ms=1000)
+    t.upsert({"id": "a", "v": 1}, ts_ms=0)
+    assert t.get("a", 500) == {"id": "a", "v": 1}
+    assert t.get("a", 2000) is None  # expired
+
+
+def test_enrich_join_lab1_shape():
+    customers = TTLTable(lambda r: r["customer_id"], ttl_ms=3_600_000)
+    products = TTLTable(lambda r: r["product_id"], ttl_ms=3_600_000)
+    customers.upsert({"customer_id": "C1", "customer_email": "a@x.com",
+                      "customer_name": "A", "state": "CA"}, 0)
+    products.upsert({"product_id": "P1", "product_name": "AirPods Pro",
+                     "price": 249.0, "department": "Electronics"}, 0)
+    orders = [{"order_id": "O1", "customer_id": "C1", "product_id": "P1",
+               "price": 249.0, "order_ts": 100}]
+    out = enrich_join(orders, lambda r: r["order_ts"],
+                      [(customers, lambda r: r["customer_id"], None),
+                       (products, lambda r: r["product_id"], None)])
+    assert len(out) == 1
+    assert out[0]["customer_email"] == "a@x.com"
+    assert out[0]["product_name"] == "AirPods Pro"
+    # missing dimension -> row held back
+    orders2 = [{"order_id": "O2", "customer_id": "CX", "product_id": "P1",
+                "price": 1.0, "order_ts": 100}]
+    assert enrich_join(orders2, lambda r: r["order_ts"],
+                       [(customers, lambda r: r["customer_id"], None)]) == []
+
+
+def test_interval_join_lab4_shape():
+    H6 = 6 * 3600 * 1000
+    anomalies = [{"city": "Naples", "window_time": H6 - 1, "is_anomaly": True}]
+    claims = [
+        {"claim_id": "C1", "city": "Naples", "claim_timestamp": 100},
+        {"claim_id": "C2", "city": "Naples", "claim_timestamp": H6 + 100},  # after
+        {"claim_id": "C3", "city": "Tampa", "claim_timestamp": 100},  # wrong key
+    ]
+    out = interval_join(claims, anomalies,
+                        lambda c: c["claim_timestamp"],
+                        lambda a: a["window_time"],
+                        lambda c: c["city"], lambda a: a["city"],
+                        lower_ms=-H6, upper_ms=0)
+    assert [r["claim_id"] for r in out] == ["C1"]

@@ -1,0 +1,108 @@
+"""Avro codec, Confluent wire framing, registry, topic log."""
+
+import io
+
+import pytest
+
+from quickstart_streaming_agents_amd.wire import (
+    AvroConsumer, AvroProducer, Broker, Schema, deserialize, peek_schema_id,
+    serialize,
+)
+from quickstart_streaming_agents_amd.wire.avro import _read_long, _write_long
+from quickstart_streaming_agents_amd.labs import schemas
+
+
+def test_zigzag_roundtrip():
+    for n in [0, 1, -1, 63, -64, 64, 127, 128, 1 << 20, -(1 << 20),
+              1736966400000, -(1 << 40)]:
+        out = io.BytesIO()
+        _write_long(out, n)
+        assert _read_long(io.BytesIO(out.getvalue())) == n
+
+
+def test_known_zigzag_bytes():
+    # Avro spec examples: 0->00, -1->01, 1->02, -2->03, 2->04
+    for n, expected in [(0, b"\x00"), (-1, b"\x01"), (1, b"\x02"),
+                        (-2, b"\x03"), (2, b"\x04"), (-64, b"\x7f"),
+                        (64, b"\x80\x01")]:
+        out = io.BytesIO()
+        _write_long(out, n)
+        assert out.getvalue() == expected
+
+
+def test_record_roundtrip_orders():
+    s = Schema(schemas.ORDERS)
+    rec = {"order_id": "ORD-00001", "customer_id": "CUST-001",
+           "product_id": "PROD-001", "price": 249.0, "order_ts": 1736966400000}
+    payload = serialize(s, 7, rec)
+    assert payload[0] == 0
+    assert peek_schema_id(payload) == 7
+    sid, decoded = deserialize(s, payload)
+    assert sid == 7
+    assert decoded == rec
+
+
+def test_union_null_fields():
+    s = Schema(schemas.CLAIMS)
+    claim = {"claim_id": "CLM-1", "applicant_name": None, "city": "Naples",
+             "is_primary_residence": "Yes", "damage_assessed": None,
+             "claim_amount": "12000.00", "has_insurance": None,
+             "insurance_amount": None, "claim_narrative": "flooded",
+             "assessment_date": None, "disaster_date": None,
+             "previous_claims_count": None, "last_claim_date": None,
+             "assessment_source": None, "shared_account": None,
+             "shared_phone": None, "claim_timestamp": 1000}
+    _, decoded = deserialize(s, serialize(s, 1, claim))
+    assert decoded == claim
+
+
+def test_array_and_optional_array():
+    s = Schema(schemas.DOCUMENTS)
+    doc = {"document_id": "DOC-1", "title": "T", "chunk": "c" * 10,
+           "pages": None, "section_reference": "S1",
+           "fraud_categories": ["duplicate", "inflated"],
+           "policy_keywords": [], "char_count": 10}
+    _, decoded = deserialize(s, serialize(s, 2, doc))
+    assert decoded["fraud_categories"] == ["duplicate", "inflated"]
+    assert decoded["policy_keywords"] == []
+    assert decoded["char_count"] == 10
+
+
+def test_float_array_embedding():
+    s = Schema({"type": "array", "items": "float"})
+    vec = [0.5, -1.25, 3.0]
+    out = io.BytesIO()
+    s.write(out, vec)
+    assert s.read(io.BytesIO(out.getvalue())) == vec
+
+
+def test_registry_dedup_and_subjects():
+    b = Broker()
+    p1 = AvroProducer(b, "orders", schemas.ORDERS)
+    p2 = AvroProducer(b, "orders", schemas.ORDERS)
+    assert p1.value_schema_id == p2.value_schema_id
+    assert "orders-value" in b.registry.subjects()
+
+
+def test_topic_counts_purge_and_consumer():
+    b = Broker()
+    prod = AvroProducer(b, "orders", schemas.ORDERS)
+    for i in range(5):
+        prod.produce({"order_id": f"O{i}", "customer_id": "C", "product_id": "P",
+                      "price": 1.0, "order_ts": i * 1000},
+                     key=f"O{i}", timestamp_ms=i * 1000, partition=0)
+    t = b.topic("orders")
+    assert t.message_count() == 5
+    cons = AvroConsumer(b, "orders", schemas.ORDERS)
+    got = cons.poll()
+    assert len(got) == 5
+    assert got[0][1]["order_id"] == "O0"
+    assert cons.poll() == []  # offsets advanced
+    t.purge()
+    assert t.message_count() == 0
+
+
+def test_bad_wire_format_rejected():
+    s = Schema(schemas.QUERIES)
+    with pytest.raises(ValueError):
+        deserialize(s, b"\x01\x00\x00\x00\x07junk")
