@@ -1,0 +1,68 @@
+"""Episode scheduler: batch LLM requests across concurrent agent episodes.
+
+The reference's AI_RUN_AGENT runs one managed-LLM round trip per iteration
+per record.  Here thousands of per-record episodes run
+concurrently and their ("llm", prompt) requests are ganged into batched
+model calls while ("tool", ...) requests run on an I/O thread pool — the
+model (and later the GPU decode engine, models/serve.py) always sees large
+batches, and episodes blocked on tool I/O never stall the batch.
+"""
+
+from __future__ import annotations
+
+from concurrent.futures import ThreadPoolExecutor
+from typing import Any, Callable, Generator
+
+from .runner import EpisodeResult
+
+
+def run_episodes(episodes: list[Generator],
+                 llm_batch: Callable[[list[str], list[int]], list[str]],
+                 tool: Callable[[str, dict], str],
+                 max_tool_workers: int = 16) -> list[EpisodeResult]:
+    """Drive episodes to completion with ganged LLM batches.
+
+    llm_batch(prompts, max_new_tokens_list) -> list of generated texts.
+    tool(name, args) -> result text (exceptions become __error__ results).
+    """
+    results: dict[int, EpisodeResult] = {}
+    # (episode index -> pending request), advanced in rounds
+    pending: dict[int, tuple] = {}
+
+    def _advance(idx: int, send_value: Any) -> None:
+        try:
+            req = episodes[idx].send(send_value)
+            pending[idx] = req
+        except StopIteration as stop:
+            results[idx] = stop.value
+            pending.pop(idx, None)
+
+    for idx in range(len(episodes)):
+        _advance(idx, None)
+
+    with ThreadPoolExecutor(max_workers=max_tool_workers) as pool:
+        while pending:
+            llm_ids = [i for i, r in pending.items() if r[0] == "llm"]
+            tool_ids = [i for i, r in pending.items() if r[0] == "tool"]
+            # Tools first (I/O overlaps nothing on CPU; on GPU the engine
+            # overlaps decode with these futures).
+            tool_futs = {}
+            for i in tool_ids:
+                _, name, args = pending[i]
+                tool_futs[i] = pool.submit(_safe_tool, tool, name, args)
+            if llm_ids:
+                prompts = [pending[i][1] for i in llm_ids]
+                maxtoks = [pending[i][2] for i in llm_ids]
+                texts = llm_batch(prompts, maxtoks)
+                for i, text in zip(llm_ids, texts):
+                    _advance(i, text)
+            for i, fut in tool_futs.items():
+                _advance(i, fut.result())
+    return [results[i] for i in range(len(episodes))]
+
+
+def _safe_tool(tool: Callable[[str, dict], str], name: str, args: dict) -> str:
+    try:
+        return tool(name, args)
+    except Exception as e:
+        return f"__error__ {e}"

@@ -321,6 +321,48 @@ def lab2_documents(seed: int = 42, n_chunks: int = 64) -> list[dict]:
     return out
 
 
+_FEMA_SECTIONS = [
+    ("Eligibility", "Assistance is limited to the applicant's primary "
+     "residence; secondary homes are ineligible for housing assistance.",
+     ["ineligible", "residence"]),
+    ("Duplication of Benefits", "Assistance may not duplicate insurance "
+     "payouts; insured losses must first be claimed against the policy.",
+     ["duplicate", "insurance"]),
+    ("Documentation", "Claims require an inspection or verifiable damage "
+     "documentation; undocumented claims require additional records.",
+     ["documentation"]),
+    ("Fraud Indicators", "Shared bank accounts or phone numbers across "
+     "multiple claims, and amounts far exceeding assessed damage, are "
+     "fraud indicators subject to denial and referral.",
+     ["fraud", "shared", "inflated"]),
+    ("Award Limits", "Awards are capped at the assessed damage less any "
+     "insurance amount; partial approvals apply when coverage overlaps.",
+     ["partial", "limits"]),
+]
+
+
+def lab4_policy_docs(seed: int = 42, n_chunks: int = 40) -> list[dict]:
+    """FEMA-policy-style chunks with the lab4 metadata schema
+    (fraud_categories / policy_keywords — lab4 main.tf:270-290)."""
+    rng = random.Random(seed)
+    out = []
+    for i in range(n_chunks):
+        title, base, cats = _FEMA_SECTIONS[i % len(_FEMA_SECTIONS)]
+        chunk = (f"{base} (guidance {i // len(_FEMA_SECTIONS) + 1}, ref "
+                 + "".join(rng.choice("0123456789") for _ in range(6)) + ")")
+        out.append({
+            "document_id": f"POLICY-{i:04d}",
+            "title": f"FEMA IHP {title}",
+            "chunk": chunk,
+            "pages": f"{10 + i}-{11 + i}",
+            "section_reference": f"IHP-{i % len(_FEMA_SECTIONS) + 1}.{i:02d}",
+            "fraud_categories": cats,
+            "policy_keywords": [w.lower() for w in title.split()],
+            "char_count": len(chunk),
+        })
+    return out
+
+
 def publish_lab2(broker: Broker, seed: int = 42, n_chunks: int = 64) -> None:
     broker.create_topic("documents").purge()
     pd = AvroProducer(broker, "documents", schemas.DOCUMENTS)
