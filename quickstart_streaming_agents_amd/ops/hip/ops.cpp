@@ -1,0 +1,234 @@
+// Torch bindings for the qsa MI355X (gfx950) kernels.  HIP-native: no CUDA
+// naming, no compatibility shims — this extension only builds for ROCm.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#define CHK(x) TORCH_CHECK(x, #x)
+#define CHK_DEV(t) TORCH_CHECK((t).is_cuda(), #t " must be on the GPU")
+#define CHK_CONT(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+#define CHK_BF16(t) TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16")
+#define CHK_F32(t) TORCH_CHECK((t).scalar_type() == at::kFloat, #t " must be f32")
+#define CHK_I32(t) TORCH_CHECK((t).scalar_type() == at::kInt, #t " must be i32")
+
+static inline const unsigned short* u16(const torch::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr());
+}
+static inline unsigned short* u16m(torch::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+static inline hipStream_t cur_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+// ---- launcher decls (defined in the .hip files) ---------------------------
+extern "C" void qsa_rmsnorm_launch(const unsigned short*, const unsigned short*,
+                                   unsigned short*, unsigned short*, long long,
+                                   int, float, hipStream_t);
+extern "C" void qsa_swiglu_launch(const unsigned short*, const unsigned short*,
+                                  unsigned short*, long long, int, hipStream_t);
+extern "C" void qsa_rope_launch(unsigned short*, unsigned short*, const float*,
+                                const float*, const int*, int, int, int, int,
+                                hipStream_t);
+extern "C" void qsa_softmax_rows_launch(float*, int, int, int, int,
+                                        hipStream_t);
+extern "C" void qsa_paged_attn_decode_launch(const unsigned short*,
+                                             const unsigned short*,
+                                             const unsigned short*, const int*,
+                                             const int*, unsigned short*, float,
+                                             int, int, int, int, int,
+                                             hipStream_t);
+extern "C" void qsa_kv_append_launch(const unsigned short*,
+                                     const unsigned short*, unsigned short*,
+                                     unsigned short*, const int*, const int*,
+                                     int, int, int, int, hipStream_t);
+extern "C" void qsa_kv_scatter_launch(const unsigned short*,
+                                      const unsigned short*, unsigned short*,
+                                      unsigned short*, const int*, int, int,
+                                      int, hipStream_t);
+extern "C" void qsa_topk_launch(const float*, const float*, float*, int*,
+                                float*, int*, int, int, int, int, int,
+                                hipStream_t);
+extern "C" void qsa_window_agg_launch(const long long*, const int*,
+                                      const float*, int*, float*, long long,
+                                      long long, int, long long, int,
+                                      hipStream_t);
+extern "C" void qsa_anomaly_batch_launch(const float*, const int*, float*,
+                                         float*, int*, int, int, int,
+                                         hipStream_t);
+
+// ---------------------------------------------------------------------------
+
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
+  CHK_DEV(x); CHK_CONT(x); CHK_BF16(x); CHK_BF16(w); CHK_CONT(w);
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "H % 8 == 0");
+  const long long rows = x.numel() / H;
+  auto y = torch::empty_like(x);
+  qsa_rmsnorm_launch(u16(x), u16(w), u16m(y), nullptr, rows, H, (float)eps,
+                     cur_stream());
+  return y;
+}
+
+torch::Tensor rmsnorm_residual(torch::Tensor x, torch::Tensor res,
+                               torch::Tensor w, double eps) {
+  CHK_DEV(x); CHK_CONT(x); CHK_BF16(x); CHK_BF16(res); CHK_CONT(res);
+  CHK_BF16(w); CHK_CONT(w);
+  const int H = x.size(-1);
+  TORCH_CHECK(H % 8 == 0, "H % 8 == 0");
+  TORCH_CHECK(res.sizes() == x.sizes(), "residual shape mismatch");
+  const long long rows = x.numel() / H;
+  auto y = torch::empty_like(x);
+  qsa_rmsnorm_launch(u16(x), u16(w), u16m(y), u16m(res), rows, H, (float)eps,
+                     cur_stream());
+  return y;
+}
+
+torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
+  CHK_DEV(gate); CHK_CONT(gate); CHK_BF16(gate); CHK_BF16(up); CHK_CONT(up);
+  TORCH_CHECK(gate.numel() == up.numel(), "shape mismatch");
+  TORCH_CHECK(gate.numel() % 8 == 0, "numel % 8 == 0");
+  auto y = torch::empty_like(gate);
+  const long long n = gate.numel();
+  const int blocks = (int)std::min<long long>((n / 8 + 255) / 256, 8192);
+  qsa_swiglu_launch(u16(gate), u16(up), u16m(y), n, blocks, cur_stream());
+  return y;
+}
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
+                  torch::Tensor sin_t, torch::Tensor pos) {
+  CHK_DEV(q); CHK_CONT(q); CHK_BF16(q); CHK_BF16(k); CHK_CONT(k);
+  CHK_F32(cos_t); CHK_F32(sin_t); CHK_I32(pos);
+  const int B = q.size(0), QH = q.size(1), D = q.size(2);
+  const int KVH = k.size(1);
+  TORCH_CHECK(k.size(0) == B && k.size(2) == D, "k shape");
+  qsa_rope_launch(u16m(q), u16m(k), cos_t.data_ptr<float>(),
+                  sin_t.data_ptr<float>(), pos.data_ptr<int>(), B, QH, KVH, D,
+                  cur_stream());
+}
+
+void softmax_rows_(torch::Tensor scores, long col_offset, bool causal) {
+  CHK_DEV(scores); CHK_CONT(scores); CHK_F32(scores);
+  const int rows = scores.size(0), cols = scores.size(1);
+  qsa_softmax_rows_launch(scores.data_ptr<float>(), rows, cols,
+                          (int)col_offset, causal ? 1 : 0, cur_stream());
+}
+
+torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor kc,
+                                torch::Tensor vc, torch::Tensor block_table,
+                                torch::Tensor seq_lens, double scale) {
+  CHK_DEV(q); CHK_CONT(q); CHK_BF16(q); CHK_BF16(kc); CHK_BF16(vc);
+  CHK_CONT(kc); CHK_CONT(vc); CHK_I32(block_table); CHK_I32(seq_lens);
+  CHK_CONT(block_table);
+  const int B = q.size(0), QH = q.size(1), D = q.size(2);
+  const int KVH = kc.size(1);
+  TORCH_CHECK(D == 128 || D == 64, "D must be 64/128");
+  TORCH_CHECK(QH % KVH == 0, "GQA requires QH % KVH == 0");
+  TORCH_CHECK(QH / KVH <= 8, "GQA ratio <= 8");
+  TORCH_CHECK(kc.size(2) == D / 8 && kc.size(3) == 64 && kc.size(4) == 8,
+              "K cache layout [P, KVH, D/8, 64, 8]");
+  TORCH_CHECK(vc.size(2) == 64 && vc.size(3) == D,
+              "V cache layout [P, KVH, 64, D]");
+  const int max_pages = block_table.size(1);
+  auto out = torch::empty_like(q);
+  qsa_paged_attn_decode_launch(u16(q), u16(kc), u16(vc),
+                               block_table.data_ptr<int>(),
+                               seq_lens.data_ptr<int>(), u16m(out),
+                               (float)scale, B, QH, KVH, max_pages, D,
+                               cur_stream());
+  return out;
+}
+
+void kv_append(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kc,
+               torch::Tensor vc, torch::Tensor block_table,
+               torch::Tensor seq_lens) {
+  CHK_DEV(knew); CHK_CONT(knew); CHK_BF16(knew); CHK_BF16(vnew);
+  CHK_I32(block_table); CHK_I32(seq_lens);
+  const int B = knew.size(0), KVH = knew.size(1), D = knew.size(2);
+  qsa_kv_append_launch(u16(knew), u16(vnew), u16m(kc), u16m(vc),
+                       block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
+                       B, KVH, D, (int)block_table.size(1), cur_stream());
+}
+
+void kv_scatter(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kc,
+                torch::Tensor vc, torch::Tensor slots) {
+  CHK_DEV(knew); CHK_CONT(knew); CHK_BF16(knew); CHK_BF16(vnew); CHK_I32(slots);
+  const int T = knew.size(0), KVH = knew.size(1), D = knew.size(2);
+  if (T == 0) return;
+  qsa_kv_scatter_launch(u16(knew), u16(vnew), u16m(kc), u16m(vc),
+                        slots.data_ptr<int>(), T, KVH, D, cur_stream());
+}
+
+std::vector<torch::Tensor> topk_cosine(torch::Tensor queries,
+                                       torch::Tensor docs, long k) {
+  CHK_DEV(queries); CHK_CONT(queries); CHK_F32(queries); CHK_F32(docs);
+  CHK_CONT(docs);
+  const int Q = queries.size(0), D = queries.size(1), N = docs.size(0);
+  TORCH_CHECK(docs.size(1) == D, "dim mismatch");
+  TORCH_CHECK(k >= 1 && k <= 16, "k in [1,16]");
+  const int nblk = std::max(1, (N + 2047) / 2048);
+  auto opts_f = queries.options();
+  auto opts_i = queries.options().dtype(at::kInt);
+  auto cand_s = torch::empty({Q, nblk, k}, opts_f);
+  auto cand_i = torch::empty({Q, nblk, k}, opts_i);
+  auto out_s = torch::empty({Q, k}, opts_f);
+  auto out_i = torch::empty({Q, k}, opts_i);
+  qsa_topk_launch(queries.data_ptr<float>(), docs.data_ptr<float>(),
+                  cand_s.data_ptr<float>(), cand_i.data_ptr<int>(),
+                  out_s.data_ptr<float>(), out_i.data_ptr<int>(), Q, N, D,
+                  (int)k, nblk, cur_stream());
+  return {out_s, out_i};
+}
+
+std::vector<torch::Tensor> window_agg(torch::Tensor ts, torch::Tensor key,
+                                      c10::optional<torch::Tensor> value,
+                                      long t0, long win_ms, long nwin,
+                                      long nkeys) {
+  CHK_DEV(ts); CHK_CONT(ts); CHK_I32(key);
+  TORCH_CHECK(ts.scalar_type() == at::kLong, "ts must be i64");
+  const long long n = ts.numel();
+  auto counts = torch::zeros({nkeys, nwin}, key.options());
+  auto sums = torch::zeros({nkeys, nwin}, ts.options().dtype(at::kFloat));
+  const float* vptr = nullptr;
+  if (value.has_value()) {
+    CHK_F32(value.value());
+    vptr = value.value().data_ptr<float>();
+  }
+  const int blocks = (int)std::min<long long>((n + 255) / 256, 4096);
+  qsa_window_agg_launch(reinterpret_cast<const long long*>(ts.data_ptr<int64_t>()),
+                        key.data_ptr<int>(), vptr,
+                        counts.data_ptr<int>(), sums.data_ptr<float>(), t0,
+                        win_ms, (int)nwin, n, std::max(blocks, 1),
+                        cur_stream());
+  return {counts, sums};
+}
+
+std::vector<torch::Tensor> anomaly_batch(torch::Tensor series,
+                                         torch::Tensor lengths, long order) {
+  CHK_DEV(series); CHK_CONT(series); CHK_F32(series); CHK_I32(lengths);
+  const int K = series.size(0), Tmax = series.size(1);
+  auto fc = torch::empty({K}, series.options());
+  auto se = torch::empty({K}, series.options());
+  auto dof = torch::empty({K}, series.options().dtype(at::kInt));
+  qsa_anomaly_batch_launch(series.data_ptr<float>(), lengths.data_ptr<int>(),
+                           fc.data_ptr<float>(), se.data_ptr<float>(),
+                           dof.data_ptr<int>(), K, Tmax, (int)order,
+                           cur_stream());
+  return {fc, se, dof};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16)");
+  m.def("rmsnorm_residual", &rmsnorm_residual,
+        "fused residual-add + RMSNorm (bf16; residual updated in place)");
+  m.def("swiglu", &swiglu, "fused silu(gate)*up (bf16)");
+  m.def("rope_inplace", &rope_inplace, "rotary embedding in place (bf16)");
+  m.def("softmax_rows_", &softmax_rows_, "row softmax in place (f32)");
+  m.def("paged_attn_decode", &paged_attn_decode,
+        "paged-attention decode (bf16, GQA, page=64)");
+  m.def("kv_append", &kv_append, "append one step's k/v to the paged cache");
+  m.def("kv_scatter", &kv_scatter, "scatter prefill k/v by slot ids");
+  m.def("topk_cosine", &topk_cosine, "exact cosine top-k over the HBM index");
+  m.def("window_agg", &window_agg, "segmented (key, window) count/sum");
+  m.def("anomaly_batch", &anomaly_batch, "batched AR+ridge anomaly scorer");
+}

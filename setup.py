@@ -1,0 +1,43 @@
+"""In-tree build of the qsa_hip extension for MI355X (gfx950).
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+The built .so lands inside quickstart_streaming_agents_amd/ so the gpurun
+snapshot carries it (it is git-ignored; history stays source-only).
+hipcc cross-compiles gfx950 without a GPU present.
+"""
+
+import os
+
+from setuptools import setup
+
+os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+HIP_DIR = os.path.join(ROOT, "quickstart_streaming_agents_amd", "ops", "hip")
+
+sources = [
+    os.path.join(HIP_DIR, "ops.cpp"),
+    os.path.join(HIP_DIR, "elementwise.hip"),
+    os.path.join(HIP_DIR, "paged_attn.hip"),
+    os.path.join(HIP_DIR, "topk_cosine.hip"),
+    os.path.join(HIP_DIR, "streaming.hip"),
+]
+
+setup(
+    name="qsa-hip",
+    version="0.1.0",
+    ext_modules=[
+        CUDAExtension(
+            name="quickstart_streaming_agents_amd.qsa_hip",
+            sources=sources,
+            extra_compile_args={
+                "cxx": ["-O3", "-std=c++17"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+            },
+        )
+    ],
+    cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
+)

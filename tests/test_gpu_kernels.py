@@ -1,0 +1,187 @@
+"""Numerics: each HIP kernel vs its plain-PyTorch fp32 reference.
+
+All tests are @pytest.mark.gpu (MI355X only).  Tolerances account for
+bf16 I/O (rel ~1e-2 on normalized magnitudes).
+"""
+
+import numpy as np
+import pytest
+import torch
+
+from quickstart_streaming_agents_amd.ops import cpu_ref
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from quickstart_streaming_agents_amd.ops import ext as get_ext
+    return get_ext()
+
+
+def dev():
+    return "cuda:0"
+
+
+def test_rmsnorm(ext):
+    torch.manual_seed(0)
+    x = torch.randn(33, 4096, dtype=torch.bfloat16, device=dev())
+    w = torch.randn(4096, dtype=torch.bfloat16, device=dev())
+    y = ext.rmsnorm(x, w, 1e-5)
+    ref = cpu_ref.rmsnorm_ref(x.cpu(), w.cpu())
+    torch.testing.assert_close(y.float().cpu(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_rmsnorm_residual(ext):
+    torch.manual_seed(1)
+    x = torch.randn(17, 4096, dtype=torch.bfloat16, device=dev())
+    res = torch.randn(17, 4096, dtype=torch.bfloat16, device=dev())
+    w = torch.randn(4096, dtype=torch.bfloat16, device=dev())
+    ref_y, ref_res = cpu_ref.rmsnorm_ref(x.cpu(), w.cpu(), residual=res.cpu())
+    y = ext.rmsnorm_residual(x, res, w, 1e-5)
+    torch.testing.assert_close(y.float().cpu(), ref_y, atol=5e-2, rtol=5e-2)
+    torch.testing.assert_close(res.float().cpu(), ref_res, atol=3e-2, rtol=3e-2)
+
+
+def test_swiglu(ext):
+    torch.manual_seed(2)
+    g = torch.randn(64, 14336, dtype=torch.bfloat16, device=dev())
+    u = torch.randn(64, 14336, dtype=torch.bfloat16, device=dev())
+    y = ext.swiglu(g, u)
+    ref = cpu_ref.swiglu_ref(g.cpu(), u.cpu())
+    torch.testing.assert_close(y.float().cpu(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_rope(ext):
+    torch.manual_seed(3)
+    B, QH, KVH, D = 9, 8, 2, 128
+    cos_t, sin_t = cpu_ref.rope_tables(512, D)
+    q = torch.randn(B, QH, D, dtype=torch.bfloat16, device=dev())
+    k = torch.randn(B, KVH, D, dtype=torch.bfloat16, device=dev())
+    pos = torch.randint(0, 512, (B,), dtype=torch.int32, device=dev())
+    q_ref = cpu_ref.rope_ref(q.cpu(), pos.cpu(), cos_t, sin_t)
+    k_ref = cpu_ref.rope_ref(k.cpu(), pos.cpu(), cos_t, sin_t)
+    ext.rope_inplace(q, k, cos_t.to(dev()), sin_t.to(dev()), pos)
+    torch.testing.assert_close(q.float().cpu(), q_ref, atol=3e-2, rtol=3e-2)
+    torch.testing.assert_close(k.float().cpu(), k_ref, atol=3e-2, rtol=3e-2)
+
+
+def test_softmax_rows(ext):
+    torch.manual_seed(4)
+    s = torch.randn(32, 1000, dtype=torch.float32, device=dev()) * 4
+    ref = cpu_ref.softmax_rows_ref(s.cpu(), col_offset=500, causal=True)
+    ext.softmax_rows_(s, 500, True)
+    torch.testing.assert_close(s.cpu(), ref, atol=1e-5, rtol=1e-4)
+
+
+def _make_paged_kv(B, KVH, D, seq_lens, npages_total):
+    torch.manual_seed(7)
+    kc = torch.randn(npages_total, KVH, D // 8, 64, 8, dtype=torch.bfloat16,
+                     device=dev())
+    vc = torch.randn(npages_total, KVH, 64, D, dtype=torch.bfloat16, device=dev())
+    max_pages = max((s + 63) // 64 for s in seq_lens)
+    bt = torch.zeros(B, max_pages, dtype=torch.int32, device=dev())
+    nxt = 0
+    for b, s in enumerate(seq_lens):
+        for p in range((s + 63) // 64):
+            bt[b, p] = nxt % npages_total
+            nxt += 1
+    return kc, vc, bt
+
+
+def test_paged_attn_decode(ext):
+    B, QH, KVH, D = 4, 32, 8, 128
+    seq_lens = [1, 64, 129, 500]
+    kc, vc, bt = _make_paged_kv(B, KVH, D, seq_lens, 64)
+    q = torch.randn(B, QH, D, dtype=torch.bfloat16, device=dev())
+    sl = torch.tensor(seq_lens, dtype=torch.int32, device=dev())
+    scale = 1.0 / (D ** 0.5)
+    out = ext.paged_attn_decode(q, kc, vc, bt, sl, scale)
+    ref = cpu_ref.paged_attn_ref(q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                                 sl.cpu(), scale)
+    torch.testing.assert_close(out.float().cpu(), ref, atol=2e-2, rtol=2e-2)
+
+
+def test_kv_append_then_attend(ext):
+    B, QH, KVH, D = 2, 8, 2, 128
+    seq_lens = [65, 120]
+    kc, vc, bt = _make_paged_kv(B, KVH, D, seq_lens, 16)
+    knew = torch.randn(B, KVH, D, dtype=torch.bfloat16, device=dev())
+    vnew = torch.randn(B, KVH, D, dtype=torch.bfloat16, device=dev())
+    sl = torch.tensor(seq_lens, dtype=torch.int32, device=dev())
+    ext.kv_append(knew, vnew, kc, vc, bt, sl)
+    # verify the append landed where the reference expects
+    for b, s in enumerate(seq_lens):
+        pos = s - 1
+        page = int(bt[b, pos // 64])
+        pin = pos % 64
+        got_v = vc[page, :, pin, :].float().cpu()
+        torch.testing.assert_close(got_v, vnew[b].float().cpu(), atol=1e-3,
+                                   rtol=1e-3)
+        got_k = kc[page, :, :, pin, :].reshape(KVH, D).float().cpu()
+        torch.testing.assert_close(got_k, knew[b].float().cpu(), atol=1e-3,
+                                   rtol=1e-3)
+
+
+def test_kv_scatter(ext):
+    KVH, D = 2, 128
+    T = 100
+    kc = torch.zeros(4, KVH, D // 8, 64, 8, dtype=torch.bfloat16, device=dev())
+    vc = torch.zeros(4, KVH, 64, D, dtype=torch.bfloat16, device=dev())
+    knew = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev())
+    vnew = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev())
+    slots = torch.arange(T, dtype=torch.int32, device=dev()) + 28
+    ext.kv_scatter(knew, vnew, kc, vc, slots)
+    t = 40
+    slot = int(slots[t])
+    got = vc[slot // 64, :, slot % 64, :].float().cpu()
+    torch.testing.assert_close(got, vnew[t].float().cpu(), atol=1e-3, rtol=1e-3)
+
+
+def test_topk_cosine(ext):
+    torch.manual_seed(5)
+    Q, N, D, k = 7, 20000, 1536, 3
+    docs = torch.randn(N, D, device=dev())
+    docs = docs / docs.norm(dim=-1, keepdim=True)
+    qs = torch.randn(Q, D, device=dev())
+    qs = qs / qs.norm(dim=-1, keepdim=True)
+    s, i = ext.topk_cosine(qs, docs, k)
+    rs, ri = cpu_ref.topk_cosine_ref(qs.cpu(), docs.cpu(), k)
+    torch.testing.assert_close(s.cpu(), rs, atol=1e-4, rtol=1e-4)
+    assert torch.equal(i.cpu(), ri)
+
+
+def test_window_agg(ext):
+    torch.manual_seed(6)
+    n, nkeys, nwin = 50000, 7, 288
+    win = 300_000
+    ts = torch.randint(0, nwin * win, (n,), dtype=torch.int64, device=dev())
+    key = torch.randint(0, nkeys, (n,), dtype=torch.int32, device=dev())
+    val = torch.rand(n, dtype=torch.float32, device=dev()) * 100
+    counts, sums = ext.window_agg(ts, key, val, 0, win, nwin, nkeys)
+    rc, rs = cpu_ref.window_agg_ref(ts.cpu(), key.cpu(), val.cpu(), 0, win,
+                                    nwin, nkeys)
+    assert torch.equal(counts.cpu(), rc)
+    torch.testing.assert_close(sums.cpu(), rs, atol=1e-1, rtol=1e-4)
+
+
+def test_anomaly_batch_matches_cpu(ext):
+    from quickstart_streaming_agents_amd.runtime.anomaly import ar_forecast
+    rng = np.random.default_rng(0)
+    K, Tmax = 16, 400
+    series = np.zeros((K, Tmax), dtype=np.float32)
+    lengths = np.zeros(K, dtype=np.int32)
+    for kk in range(K):
+        n = int(rng.integers(10, Tmax))
+        lengths[kk] = n
+        base = rng.uniform(10, 1000)
+        series[kk, :n] = base + rng.normal(0, base * 0.02, n).astype(np.float32)
+    s_t = torch.from_numpy(series).to(dev())
+    l_t = torch.from_numpy(lengths).to(dev())
+    fc, se, dof = ext.anomaly_batch(s_t, l_t, 4)
+    for kk in range(K):
+        f_ref, se_ref, dof_ref = ar_forecast(
+            series[kk, :lengths[kk]].astype(np.float64), 4)
+        assert dof[kk].item() == dof_ref
+        assert abs(fc[kk].item() - f_ref) < max(3e-3 * abs(f_ref), 1e-2)
+        assert abs(se[kk].item() - se_ref) < max(0.05 * se_ref, 1e-2)
