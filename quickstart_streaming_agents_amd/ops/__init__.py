@@ -11,6 +11,7 @@ from __future__ import annotations
 _ext = None
 _import_error: Exception | None = None
 try:
+    import torch  # noqa: F401  (loads libc10/libtorch the extension links)
     from .. import qsa_hip as _ext  # built in-tree by setup.py
 except ImportError as e:  # pragma: no cover - exercised only when unbuilt
     _import_error = e

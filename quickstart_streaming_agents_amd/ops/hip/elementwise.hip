@@ -133,10 +133,16 @@ qsa_rope_kernel(unsigned short* __restrict__ q, unsigned short* __restrict__ k,
 // ---------------------------------------------------------------------------
 __global__ void __launch_bounds__(256)
 qsa_softmax_rows_kernel(float* __restrict__ scores, int rows, int cols,
-                        int col_offset, int causal) {
+                        int col_offset, int causal, int row_mod,
+                        const int* __restrict__ row_limits) {
   const int row = blockIdx.x;
   float* r = scores + (long long)row * cols;
-  const int limit = causal ? min(cols, row + col_offset + 1) : cols;
+  int limit = cols;
+  if (causal) {
+    const int pos = row_mod > 0 ? row % row_mod : row;
+    limit = min(cols, pos + col_offset + 1);
+  }
+  if (row_limits) limit = min(limit, row_limits[row]);
   __shared__ float scratch[8];
 
   float mx = -3.0e38f;
@@ -193,7 +199,9 @@ extern "C" void qsa_rope_launch(unsigned short* q, unsigned short* k,
 
 extern "C" void qsa_softmax_rows_launch(float* scores, int rows, int cols,
                                         int col_offset, int causal,
+                                        int row_mod, const int* row_limits,
                                         hipStream_t stream) {
   hipLaunchKernelGGL(qsa_softmax_rows_kernel, dim3(rows), dim3(256), 0, stream,
-                     scores, rows, cols, col_offset, causal);
+                     scores, rows, cols, col_offset, causal, row_mod,
+                     row_limits);
 }

@@ -30,8 +30,8 @@ extern "C" void qsa_swiglu_launch(const unsigned short*, const unsigned short*,
 extern "C" void qsa_rope_launch(unsigned short*, unsigned short*, const float*,
                                 const float*, const int*, int, int, int, int,
                                 hipStream_t);
-extern "C" void qsa_softmax_rows_launch(float*, int, int, int, int,
-                                        hipStream_t);
+extern "C" void qsa_softmax_rows_launch(float*, int, int, int, int, int,
+                                        const int*, hipStream_t);
 extern "C" void qsa_paged_attn_decode_launch(const unsigned short*,
                                              const unsigned short*,
                                              const unsigned short*, const int*,
@@ -107,11 +107,18 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
                   cur_stream());
 }
 
-void softmax_rows_(torch::Tensor scores, long col_offset, bool causal) {
+void softmax_rows_(torch::Tensor scores, long col_offset, bool causal,
+                   long row_mod, c10::optional<torch::Tensor> row_limits) {
   CHK_DEV(scores); CHK_CONT(scores); CHK_F32(scores);
   const int rows = scores.size(0), cols = scores.size(1);
+  const int* rl = nullptr;
+  if (row_limits.has_value()) {
+    CHK_I32(row_limits.value());
+    rl = row_limits.value().data_ptr<int>();
+  }
   qsa_softmax_rows_launch(scores.data_ptr<float>(), rows, cols,
-                          (int)col_offset, causal ? 1 : 0, cur_stream());
+                          (int)col_offset, causal ? 1 : 0, (int)row_mod, rl,
+                          cur_stream());
 }
 
 torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor kc,
@@ -223,7 +230,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused residual-add + RMSNorm (bf16; residual updated in place)");
   m.def("swiglu", &swiglu, "fused silu(gate)*up (bf16)");
   m.def("rope_inplace", &rope_inplace, "rotary embedding in place (bf16)");
-  m.def("softmax_rows_", &softmax_rows_, "row softmax in place (f32)");
+  m.def("softmax_rows_", &softmax_rows_, "row softmax in place (f32)",
+        py::arg("scores"), py::arg("col_offset") = 0, py::arg("causal") = false,
+        py::arg("row_mod") = 0, py::arg("row_limits") = py::none());
   m.def("paged_attn_decode", &paged_attn_decode,
         "paged-attention decode (bf16, GQA, page=64)");
   m.def("kv_append", &kv_append, "append one step's k/v to the paged cache");

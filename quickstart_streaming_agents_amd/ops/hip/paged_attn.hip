@@ -83,6 +83,7 @@ qsa_paged_attn_decode(const unsigned short* __restrict__ q,   // [B, QH, D]
       const float m_new = fmaxf(m, pmax);
       float alpha = __expf(m - m_new);  // m==-inf -> exp(-inf)=0 ok
       if (m <= -3.0e38f) alpha = 0.f;
+      m = m_new;
       const float p = valid ? __expf(sc - m_new) : 0.f;
       s = s * alpha + wave_reduce_sum(p);
       o0 *= alpha;
