@@ -1,0 +1,194 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Lab1 price-match agent decisions/sec on MI355X.
+
+Measures the BASELINE.json headline metric — agent decisions/sec (+ p50
+end-to-end latency) for the Lab1 price-match pipeline — on N GPUs of one
+node, weak scaling (per-GPU work fixed): each rank runs the full pipeline
+on its own shard of the order stream (DP over stream partitions), with the
+agent LLM (Llama-3-8B bf16, random init), the continuous-batching decode
+engine, the paged-attention HIP kernel, and real MCP tool round trips
+against a local stub server.  Synthetic data (deterministic datagen),
+no network.
+
+One decision = one order through AI_RUN_AGENT(price_match_agent):
+LLM turn -> http_get (competitor page) -> LLM turn -> send_email/finish ->
+final 3-section response + REGEXP_EXTRACT parse.
+
+Reference implied operating point: ~0.011 decisions/sec (1 order / 90-120 s,
+BASELINE.md).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+ROOT = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, ROOT)
+
+BASELINE_DECISIONS_PER_SEC = 0.011  # BASELINE.md implied operating point
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--batch", type=int, default=24,
+                   help="decisions per step per GPU")
+    p.add_argument("--model", default="llama3-8b")
+    p.add_argument("--decode-tokens", type=int, default=64,
+                   help="decode tokens per agent LLM turn")
+    p.add_argument("--max-seq-len", type=int, default=2048)
+    p.add_argument("--stub-llm", action="store_true",
+                   help="CPU contract check: stub LLM instead of the GPU engine")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        backend = "nccl" if (torch.cuda.is_available() and not args.stub_llm) \
+            else "gloo"
+        dist.init_process_group(backend=backend)
+
+    use_gpu = torch.cuda.is_available() and not args.stub_llm
+    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+    if use_gpu:
+        torch.cuda.set_device(device)
+
+    from quickstart_streaming_agents_amd.agents.mcp import McpClient, StubMcpServer
+    from quickstart_streaming_agents_amd.agents.runner import AgentSpec, ToolSet, episode
+    from quickstart_streaming_agents_amd.agents.schedule import run_episodes
+    from quickstart_streaming_agents_amd.labs import datagen, pipelines
+
+    # ---- synthetic lab1 shard for this rank ------------------------------
+    n_orders = args.batch * (args.steps + args.warmup)
+    products = datagen.lab1_products()
+    customers = datagen.lab1_customers(seed=42 + rank)
+    enriched = []
+    for i in range(n_orders):
+        p = products[i % len(products)]
+        c = customers[i % len(customers)]
+        enriched.append({
+            "order_id": f"ORD-{rank:02d}-{i:06d}",
+            "product_name": p["product_name"],
+            "product_id": p["product_id"],
+            "order_price": p["price"],
+            "customer_id": c["customer_id"],
+            "customer_email": c["customer_email"],
+            "order_ts": i * 1000,
+        })
+
+    server = StubMcpServer().start()
+    client = McpClient(server.mcp_endpoint)
+    tool_fn = pipelines.mcp_tool_fn(client)
+    competitor_url = f"{server.base_url}/competitor"
+
+    # ---- model backend ---------------------------------------------------
+    if args.stub_llm:
+        llm = pipelines.StubLLM()
+    else:
+        from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                                  LlamaModel)
+        from quickstart_streaming_agents_amd.models.serve import Engine, EngineLLM
+        model = LlamaModel(LlamaConfig.preset(args.model), device=device)
+        engine = Engine(model, max_batch=args.batch,
+                        max_seq_len=args.max_seq_len)
+        llm = EngineLLM(engine)
+
+    tools = ToolSet("lab1_remote_mcp", allowed_tools=("http_get", "send_email"))
+    agent = AgentSpec("price_match_agent", "remote_mcp_model",
+                      pipelines.LAB1_AGENT_PROMPT, tools,
+                      max_iterations=10, max_consecutive_failures=2)
+
+    def run_step(step_idx: int) -> int:
+        orders = enriched[step_idx * args.batch:(step_idx + 1) * args.batch]
+        eps = [episode(agent,
+                       pipelines.lab1_user_prompt(o, competitor_url,
+                                                  o["customer_email"]),
+                       policy=pipelines.Lab1PriceMatchPolicy(
+                           o, competitor_url, o["customer_email"]),
+                       max_new_tokens=args.decode_tokens)
+               for o in orders]
+        results = run_episodes(eps, llm, tool_fn)
+        ok = sum(r.status == "SUCCESS" for r in results)
+        assert ok == len(orders), f"{len(orders) - ok} episodes failed"
+        return ok
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize()
+        if dist is not None:
+            dist.barrier()
+
+    for w in range(args.warmup):
+        run_step(w)
+    sync()
+
+    step_times = []
+    t0 = time.perf_counter()
+    for s in range(args.steps):
+        ts = time.perf_counter()
+        run_step(args.warmup + s)
+        if use_gpu:
+            torch.cuda.synchronize()
+        step_times.append(time.perf_counter() - ts)
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX elapsed over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        if dist.get_backend() == "nccl":
+            t = t.to(device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_decisions = args.batch * args.steps * world
+    value = total_decisions / elapsed
+    p50_ms = statistics.median(step_times) * 1000.0
+
+    if rank == 0:
+        out = {
+            "metric": "lab1_price_match_agent_decisions_per_sec",
+            "value": round(value, 4),
+            "unit": "decisions/s",
+            "n_gpus": world if use_gpu else args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 2),
+            "p50_e2e_ms": round(p50_ms, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / BASELINE_DECISIONS_PER_SEC, 1),
+            "dtype": "bf16" if not args.stub_llm else "stub",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * world,
+                "seq_len": args.max_seq_len,
+                "decode_tokens_per_turn": args.decode_tokens,
+                "parallelism": f"dp{world}",
+            },
+        }
+        print(json.dumps(out))
+    server.stop()
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -100,7 +100,7 @@ class Lab1PriceMatchPolicy:
             return ToolCall("http_get", {"url": self.competitor_url})
         price = _extract_competitor_price(obs["http_get"],
                                           self.order["product_name"])
-        our_price = float(self.order["price"])
+        our_price = float(self.order.get("order_price", self.order.get("price")))
         if price is None:
             return Finish("Competitor Price:\nNot found\n\nDecision:\nNO_MATCH"
                           "\n\nSummary:\nNo matching product found on the "
@@ -152,12 +152,40 @@ def lab1_enriched_orders(broker: Broker, state_ttl_ms: int = 3_600_000) -> list[
 
 
 def lab1_user_prompt(order: dict, competitor_url: str, email_recipient: str) -> str:
+    """Per-order user prompt carrying the URL, product, price and the full
+    email template — mirroring the reference's CONCAT user prompt shape
+    (LAB1-Walkthrough.md:208-254), so prompt token lengths are realistic."""
+    price = float(order.get("order_price", order.get("price", 0.0)))
+    oid = order["order_id"]
     return (
         f"COMPETITOR URL: {competitor_url}\n"
         f"PRODUCT NAME: {order['product_name']}\n"
-        f"OUR ORDER PRICE: ${float(order['order_price']):.2f}\n"
+        f"OUR ORDER PRICE: ${price:.2f}\n"
         f"EMAIL RECIPIENT: {email_recipient}\n"
-        f"ORDER ID: {order['order_id']}")
+        f"EMAIL SUBJECT: Price Match Applied - Order #{oid}\n\n"
+        "EMAIL BODY TEMPLATE:\n"
+        f"Subject: Your Price Match Has Been Applied - Order #{oid}\n\n"
+        "Dear Valued Customer,\n\n"
+        "Good news: we found a better price for your recent purchase and "
+        "have automatically applied a price match.\n\n"
+        "ORDER DETAILS:\n"
+        f"  - Order Number: #{oid}\n"
+        f"  - Product: {order['product_name']}\n\n"
+        "PRICE MATCH DETAILS:\n"
+        f"  - Original Price: ${price:.2f}\n"
+        "  - Competitor Price Found: $[INSERT_COMPETITOR_PRICE]\n"
+        "  - Your Savings: $[INSERT_SAVINGS]\n\n"
+        "ACTION TAKEN:\n"
+        "We processed a price-match refund of $[INSERT_SAVINGS] to your "
+        "original payment method; expect the credit within 3-5 business "
+        "days.\n\n"
+        "WHY WE DO THIS:\n"
+        "Our automated price-matching system continuously monitors "
+        "competitor prices so you always get the best deal.\n\n"
+        "Thank you for shopping with us.\n"
+        "Customer Success Team\n"
+        "---\n"
+        "This is an automated message from the price matching system.")
 
 
 def lab1_run(broker: Broker, llm_batch, tool_fn, competitor_url: str,
