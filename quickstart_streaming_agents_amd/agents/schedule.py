@@ -36,16 +36,21 @@ def run_episodes(episodes: list[Generator],
         except StopIteration as stop:
             results[idx] = stop.value
             pending.pop(idx, None)
+            if hasattr(llm_batch, "release"):
+                llm_batch.release(idx)  # free the conversation's KV prefix
 
     for idx in range(len(episodes)):
         _advance(idx, None)
+
+    # conversation-aware backends (EngineLLM) take conv ids for KV prefix
+    # reuse across an episode's turns; plain callables take 2 args.
+    conv_aware = hasattr(llm_batch, "release")
 
     with ThreadPoolExecutor(max_workers=max_tool_workers) as pool:
         while pending:
             llm_ids = [i for i, r in pending.items() if r[0] == "llm"]
             tool_ids = [i for i, r in pending.items() if r[0] == "tool"]
-            # Tools first (I/O overlaps nothing on CPU; on GPU the engine
-            # overlaps decode with these futures).
+            # Tools first (their futures overlap the LLM batch below).
             tool_futs = {}
             for i in tool_ids:
                 _, name, args = pending[i]
@@ -53,11 +58,16 @@ def run_episodes(episodes: list[Generator],
             if llm_ids:
                 prompts = [pending[i][1] for i in llm_ids]
                 maxtoks = [pending[i][2] for i in llm_ids]
-                texts = llm_batch(prompts, maxtoks)
+                if conv_aware:
+                    texts = llm_batch(prompts, maxtoks, llm_ids)
+                else:
+                    texts = llm_batch(prompts, maxtoks)
                 for i, text in zip(llm_ids, texts):
                     _advance(i, text)
             for i, fut in tool_futs.items():
                 _advance(i, fut.result())
+    if hasattr(llm_batch, "release_all"):
+        llm_batch.release_all()
     return [results[i] for i in range(len(episodes))]
 
 

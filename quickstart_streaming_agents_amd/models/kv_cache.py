@@ -58,6 +58,15 @@ class PagedKVCache:
             pages.append(self._free.pop())
         self._seq_len[seq_id] = new_len
 
+    def truncate(self, seq_id: int, new_len: int) -> None:
+        """Roll a sequence back to new_len positions (prefix reuse across
+        agent turns: decode-token KV beyond the shared prompt is discarded)."""
+        pages = self._seq_pages[seq_id]
+        keep = max(self.pages_for(new_len), 1)
+        while len(pages) > keep:
+            self._free.append(pages.pop())
+        self._seq_len[seq_id] = new_len
+
     def free(self, seq_id: int) -> None:
         pages = self._seq_pages.pop(seq_id, [])
         self._free.extend(reversed(pages))

@@ -139,18 +139,48 @@ class LlamaModel:
         final_h = e.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
         return F.linear(final_h, self.lm_head).float()
 
-    @torch.no_grad()
-    def forward_prefill(self, tokens: torch.Tensor, kv: PagedKVCache,
-                        seq_id: int) -> torch.Tensor:
-        """Prefill ONE sequence's prompt [t] i64; writes the paged cache and
-        returns logits for the last position [vocab]."""
+    def _gather_kv(self, kc_l: torch.Tensor, vc_l: torch.Tensor,
+                   pages: list[int], n: int):
+        """Gather positions [0, n) of a sequence from the paged cache ->
+        (K [KVH, n, D], V [KVH, n, D]) contiguous for the prefill GEMMs."""
         c = self.cfg
-        t = tokens.shape[0]
+        idx = torch.tensor(pages, dtype=torch.int64, device=self.device)
+        kpg = kc_l.index_select(0, idx)        # [np, KVH, D/8, 64, 8]
+        kseq = kpg.permute(1, 0, 3, 2, 4).reshape(
+            c.n_kv_heads, len(pages) * 64, c.d_head)[:, :n]
+        vpg = vc_l.index_select(0, idx)        # [np, KVH, 64, D]
+        vseq = vpg.permute(1, 0, 2, 3).reshape(
+            c.n_kv_heads, len(pages) * 64, c.d_head)[:, :n]
+        return kseq, vseq
+
+    @torch.no_grad()
+    def forward_prefill_batch(self, items: list[tuple[torch.Tensor, int, int]],
+                              kv: PagedKVCache) -> torch.Tensor:
+        """Batched (and incremental) prefill.
+
+        items: (new_tokens [t_i] i64, seq_id, start_pos).  start_pos > 0 means
+        the sequence's first start_pos positions are already in the paged
+        cache (prefix reuse across agent turns).  All per-token projections
+        run as ONE fused GEMM over the concatenation; only attention loops
+        per sequence.  Returns last-position logits [n_items, vocab].
+        """
+        c = self.cfg
         e = ext()
-        positions = torch.arange(t, dtype=torch.int32, device=self.device)
-        slots = kv.slot_ids(seq_id, 0, t)
+        dev = self.device
+        lens = [int(t.shape[0]) for t, _, _ in items]
+        offs = [0]
+        for t in lens:
+            offs.append(offs[-1] + t)
+        T = offs[-1]
+        tokens = torch.cat([t for t, _, _ in items])
+        positions = torch.cat([
+            torch.arange(s, s + n, dtype=torch.int32, device=dev)
+            for (_, _, s), n in zip(items, lens)]).contiguous()
+        slots = torch.cat([
+            kv.slot_ids(sid, s, n)
+            for (_, sid, s), n in zip(items, lens)]).contiguous()
+
         res = self.embed.index_select(0, tokens).contiguous()
-        R = c.n_q_heads // c.n_kv_heads
         h = None
         mlp_out = None
         for li, L in enumerate(self.layers):
@@ -159,21 +189,29 @@ class LlamaModel:
             else:
                 h = e.rmsnorm_residual(mlp_out, res, L["attn_norm"], c.norm_eps)
             qkv = F.linear(h, L["wqkv"])
-            q, k, v = self._split_qkv(qkv, t)
+            q, k, v = self._split_qkv(qkv, T)
             e.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
             e.kv_scatter(k, v, kv.k[li], kv.v[li], slots)
-            # attention: [QH, t, t] f32 scores, causal softmax kernel, PV
-            qf = q.permute(1, 0, 2)                             # [QH, t, D]
-            kf = k.permute(1, 0, 2)                             # [KVH, t, D]
-            vf = v.permute(1, 0, 2)
-            kf = kf.repeat_interleave(R, dim=0)                 # [QH, t, D]
-            vf = vf.repeat_interleave(R, dim=0)
-            scores = (torch.bmm(qf.float(), kf.float().transpose(1, 2))
-                      * self.scale).reshape(c.n_q_heads * t, t).contiguous()
-            e.softmax_rows_(scores, 0, True, t, None)
-            attn = torch.bmm(scores.reshape(c.n_q_heads, t, t),
-                             vf.float()).to(self.dtype)          # [QH, t, D]
-            attn = attn.permute(1, 0, 2).reshape(t, -1).contiguous()
+            attn = torch.empty(T, c.n_q_heads * c.d_head, dtype=self.dtype,
+                               device=dev)
+            for (tt, sid, start), n, off in zip(items, lens, offs):
+                ctx = start + n
+                if start == 0:
+                    kseq = k[off:off + n].permute(1, 0, 2)
+                    vseq = v[off:off + n].permute(1, 0, 2)
+                else:
+                    kseq, vseq = self._gather_kv(kv.k[li], kv.v[li],
+                                                 kv._seq_pages[sid], ctx)
+                qf = q[off:off + n].permute(1, 0, 2).float()   # [QH, n, D]
+                R = c.n_q_heads // c.n_kv_heads
+                kf = kseq.float().repeat_interleave(R, dim=0)  # [QH, ctx, D]
+                vf = vseq.float().repeat_interleave(R, dim=0)
+                scores = (torch.bmm(qf, kf.transpose(1, 2)) * self.scale) \
+                    .reshape(c.n_q_heads * n, ctx).contiguous()
+                e.softmax_rows_(scores, start, True, n, None)
+                a = torch.bmm(scores.reshape(c.n_q_heads, n, ctx), vf)
+                attn[off:off + n] = a.permute(1, 0, 2).reshape(n, -1) \
+                    .to(self.dtype)
             o = F.linear(attn, L["wo"])
             h = e.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             gu = F.linear(h, L["wgu"])
@@ -181,4 +219,12 @@ class LlamaModel:
                            gu[:, c.ffn:].contiguous())
             mlp_out = F.linear(act, L["wdown"])
         final_h = e.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
-        return F.linear(final_h[-1:], self.lm_head).float()[0]
+        last = torch.tensor([offs[i] + lens[i] - 1 for i in range(len(items))],
+                            dtype=torch.int64, device=dev)
+        return F.linear(final_h.index_select(0, last), self.lm_head).float()
+
+    @torch.no_grad()
+    def forward_prefill(self, tokens: torch.Tensor, kv: PagedKVCache,
+                        seq_id: int) -> torch.Tensor:
+        """Single-sequence prefill (wrapper over the batch path)."""
+        return self.forward_prefill_batch([(tokens, seq_id, 0)], kv)[0]

@@ -1,11 +1,17 @@
 """Continuous-batching serving engine for the agent LLM.
 
 Sequences are admitted into the running decode batch as KV pages free up
-(vLLM-style): each step() admits pending prompts (per-sequence prefill via
-the GEMM+causal-softmax path), then runs ONE fused decode step for every
-running sequence through the paged-attention kernel.  Thousands of agent
-episodes (agents/schedule.py) gang their LLM turns into these batches; tool
-round trips run on host threads between turns, so the GPU stays busy.
+(vLLM-style): each step() admits pending prompts via ONE batched prefill
+(per-token projections fused across sequences; attention per sequence),
+then runs ONE fused decode step for every running sequence through the
+paged-attention kernel.  Thousands of agent episodes (agents/schedule.py)
+gang their LLM turns into these batches; tool round trips run on host
+threads between turns, so the GPU stays busy.
+
+Multi-turn agent episodes reuse their KV prefix: a conversation keeps its
+sequence alive between turns, the engine rolls the cache back to the shared
+prompt prefix and prefills only the delta (observation text) — turn N of an
+episode costs O(new tokens), not O(whole transcript).
 
 Greedy sampling keeps the benchmark deterministic.
 """
@@ -25,29 +31,34 @@ from .tokenizer import HashTokenizer
 @dataclass
 class Sequence:
     seq_id: int
-    prompt: list[int]
+    prompt: list[int]            # full prompt tokens for this turn
     max_new_tokens: int
+    cached_len: int = 0          # prefix positions already in the KV cache
     out_tokens: list[int] = field(default_factory=list)
     done: bool = False
+    keep_alive: bool = False     # conversation: keep KV after completion
 
 
 @dataclass
 class EngineStats:
     prefill_tokens: int = 0
+    cached_prefix_tokens: int = 0
     decode_tokens: int = 0
     decode_steps: int = 0
     prefills: int = 0
+    prefill_batches: int = 0
     wall_s: float = 0.0
 
 
 class Engine:
     def __init__(self, model: LlamaModel, kv_pages: int | None = None,
                  max_batch: int = 256, max_seq_len: int = 4096,
-                 eos_id: int | None = None):
+                 eos_id: int | None = None, prefill_batch_tokens: int = 16384):
         self.model = model
         self.max_batch = max_batch
         self.max_seq_len = max_seq_len
         self.eos_id = eos_id  # None -> run to max_new_tokens (random weights)
+        self.prefill_batch_tokens = prefill_batch_tokens
         if kv_pages is None:
             kv_pages = max_batch * ((max_seq_len + 63) // 64) + 64
         self.kv = model.new_kv_cache(kv_pages)
@@ -57,28 +68,77 @@ class Engine:
         self.stats = EngineStats()
 
     # ------------------------------------------------------------------
-    def submit(self, prompt_tokens: list[int], max_new_tokens: int) -> Sequence:
-        seq = Sequence(self._next_id, list(prompt_tokens[: self.max_seq_len - 1]),
-                       max_new_tokens)
+    def submit(self, prompt_tokens: list[int], max_new_tokens: int,
+               continue_from: Sequence | None = None,
+               keep_alive: bool = False) -> Sequence:
+        """Queue a prompt.  continue_from: a completed keep-alive sequence
+        whose prompt is a prefix of this one — its KV prefix is reused."""
+        prompt = list(prompt_tokens[: self.max_seq_len - 1])
+        if continue_from is not None and not continue_from.keep_alive:
+            continue_from = None
+        if continue_from is not None:
+            # at least one token must prefill (last-position logits)
+            shared = min(len(continue_from.prompt), len(prompt) - 1)
+            # shared prefix = the previous turn's prompt (its decode-token KV
+            # is rolled back); bail to fresh prefill on any mismatch
+            if prompt[:shared] != continue_from.prompt[:shared]:
+                shared = 0
+            if shared > 0:
+                seq = Sequence(continue_from.seq_id, prompt, max_new_tokens,
+                               cached_len=shared, keep_alive=keep_alive)
+                self.kv.truncate(seq.seq_id, shared)
+                self._next_id = max(self._next_id, seq.seq_id + 1)
+                self.pending.append(seq)
+                return seq
+            self.kv.free(continue_from.seq_id)
+        seq = Sequence(self._next_id, prompt, max_new_tokens,
+                       keep_alive=keep_alive)
         self._next_id += 1
         self.pending.append(seq)
         return seq
 
+    def release(self, seq: Sequence) -> None:
+        """Free a keep-alive conversation's KV."""
+        if seq.keep_alive:
+            seq.keep_alive = False
+            self.kv.free(seq.seq_id)
+
+    # ------------------------------------------------------------------
     def _admit(self) -> None:
-        while self.pending and len(self.running) < self.max_batch:
+        batch_items: list[tuple[torch.Tensor, int, int]] = []
+        admitted: list[Sequence] = []
+        new_tokens = 0
+        while self.pending and len(self.running) + len(admitted) < self.max_batch:
             seq = self.pending[0]
-            need = self.kv.pages_for(len(seq.prompt) + seq.max_new_tokens)
+            total = len(seq.prompt) + seq.max_new_tokens
+            have = len(self.kv._seq_pages.get(seq.seq_id, ())) \
+                if seq.cached_len else 0
+            need = self.kv.pages_for(total) - have
             if need > self.kv.free_pages:
                 break
+            delta = len(seq.prompt) - seq.cached_len
+            if batch_items and new_tokens + delta > self.prefill_batch_tokens:
+                break
             self.pending.pop(0)
-            self.kv.allocate(seq.seq_id, len(seq.prompt))
-            toks = torch.tensor(seq.prompt, dtype=torch.int64,
-                                device=self.model.device)
-            logits = self.model.forward_prefill(toks, self.kv, seq.seq_id)
-            first = int(torch.argmax(logits).item())
-            seq.out_tokens.append(first)
-            self.stats.prefill_tokens += len(seq.prompt)
+            if seq.cached_len:
+                self.kv.extend(seq.seq_id, len(seq.prompt))
+            else:
+                self.kv.allocate(seq.seq_id, len(seq.prompt))
+            toks = torch.tensor(seq.prompt[seq.cached_len:],
+                                dtype=torch.int64, device=self.model.device)
+            batch_items.append((toks, seq.seq_id, seq.cached_len))
+            admitted.append(seq)
+            new_tokens += delta
+            self.stats.prefill_tokens += delta
+            self.stats.cached_prefix_tokens += seq.cached_len
             self.stats.prefills += 1
+        if not admitted:
+            return
+        logits = self.model.forward_prefill_batch(batch_items, self.kv)
+        first = torch.argmax(logits, dim=-1).tolist()
+        self.stats.prefill_batches += 1
+        for seq, tok in zip(admitted, first):
+            seq.out_tokens.append(int(tok))
             self.running.append(seq)
             self._maybe_finish(seq)
 
@@ -88,15 +148,25 @@ class Engine:
                 and seq.out_tokens[-1] == self.eos_id):
             seq.done = True
 
+    def _retire(self) -> None:
+        done = [s for s in self.running if s.done]
+        for s in done:
+            if not s.keep_alive:
+                self.kv.free(s.seq_id)
+            else:
+                # roll back to the prompt: the next turn extends the prompt,
+                # not the decoded tokens
+                self.kv.truncate(s.seq_id, len(s.prompt))
+        self.running = [s for s in self.running if not s.done]
+
     def step(self) -> int:
-        """Admit + one decode step; returns number of running sequences."""
+        """Admit + one decode step; returns remaining work count."""
         self._admit()
         batch = [s for s in self.running if not s.done]
         if not batch:
-            self.running = [s for s in self.running if not s.done]
-            return 0
+            self._retire()
+            return len(self.running) + len(self.pending)
         dev = self.model.device
-        # the token decoded this step is the last sampled one
         tokens = torch.tensor([s.out_tokens[-1] for s in batch],
                               dtype=torch.int64, device=dev)
         positions = torch.tensor(
@@ -114,10 +184,7 @@ class Engine:
             self._maybe_finish(s)
         self.stats.decode_tokens += len(batch)
         self.stats.decode_steps += 1
-        done = [s for s in self.running if s.done]
-        for s in done:
-            self.kv.free(s.seq_id)
-        self.running = [s for s in self.running if not s.done]
+        self._retire()
         return len(self.running) + len(self.pending)
 
     def run_to_completion(self) -> None:
@@ -136,14 +203,38 @@ class Engine:
 
 
 class EngineLLM:
-    """llm_batch adapter for the lab pipelines: text in, text out."""
+    """llm_batch adapter for the lab pipelines: text in, text out.
+
+    When the scheduler passes conversation ids (agents/schedule.py), each
+    conversation keeps its engine sequence alive between turns and only the
+    prompt delta is prefassed (KV prefix reuse)."""
 
     def __init__(self, engine: Engine, tokenizer: HashTokenizer | None = None):
         self.engine = engine
         self.tokenizer = tokenizer or HashTokenizer(
             engine.model.cfg.vocab_size)
+        self._convs: dict = {}
 
-    def __call__(self, prompts: list[str], max_new_tokens: list[int]) -> list[str]:
-        enc = [self.tokenizer.encode(p) for p in prompts]
-        outs = self.engine.generate_batch(enc, list(max_new_tokens))
-        return [self.tokenizer.decode(o) for o in outs]
+    def __call__(self, prompts: list[str], max_new_tokens: list[int],
+                 conv_ids: list | None = None) -> list[str]:
+        seqs = []
+        for i, (p, m) in enumerate(zip(prompts, max_new_tokens)):
+            conv = conv_ids[i] if conv_ids is not None else None
+            prev = self._convs.get(conv) if conv is not None else None
+            enc = self.tokenizer.encode(p)
+            seq = self.engine.submit(enc, m, continue_from=prev,
+                                     keep_alive=conv is not None)
+            if conv is not None:
+                self._convs[conv] = seq
+            seqs.append(seq)
+        self.engine.run_to_completion()
+        return [self.tokenizer.decode(s.out_tokens) for s in seqs]
+
+    def release(self, conv_id) -> None:
+        seq = self._convs.pop(conv_id, None)
+        if seq is not None:
+            self.engine.release(seq)
+
+    def release_all(self) -> None:
+        for conv in list(self._convs):
+            self.release(conv)

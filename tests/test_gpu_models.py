@@ -66,6 +66,48 @@ def test_engine_continuous_admission(tiny):
     assert e.kv.free_pages == 20  # everything freed
 
 
+def test_prefix_reuse_matches_fresh(tiny):
+    """A multi-turn conversation with KV prefix reuse must produce the same
+    greedy tokens as fresh full prefills of each turn's prompt."""
+    from quickstart_streaming_agents_amd.models.serve import Engine, EngineLLM
+    turn1 = "system prompt here\nuser question about prices"
+    turn2 = turn1 + "\n[observation from http_get] competitor page says $209"
+    turn3 = turn2 + "\n[observation from send_email] email sent ok"
+
+    e_fresh = Engine(tiny, max_batch=4, max_seq_len=512)
+    llm_fresh = EngineLLM(e_fresh)
+    fresh = [llm_fresh([t], [8])[0] for t in (turn1, turn2, turn3)]
+
+    e_conv = Engine(tiny, max_batch=4, max_seq_len=512)
+    llm_conv = EngineLLM(e_conv)
+    conv = [llm_conv([t], [8], ["c0"])[0] for t in (turn1, turn2, turn3)]
+    llm_conv.release("c0")
+
+    assert fresh == conv
+    assert e_conv.stats.cached_prefix_tokens > 0  # reuse actually happened
+    assert e_conv.stats.prefill_tokens < e_fresh.stats.prefill_tokens
+    assert e_conv.kv.free_pages == e_conv.kv.n_pages
+
+
+def test_batched_prefill_matches_single(tiny):
+    """forward_prefill_batch over several sequences == per-sequence calls."""
+    import torch as T
+    prompts = [[1, 5, 9, 200, 17, 88], [1, 77, 31, 4], [1, 2, 3]]
+    kv_a = tiny.new_kv_cache(16)
+    items = []
+    for sid, p in enumerate(prompts):
+        kv_a.allocate(sid, len(p))
+        items.append((T.tensor(p, dtype=T.int64, device="cuda:0"), sid, 0))
+    batch_logits = tiny.forward_prefill_batch(items, kv_a)
+    for sid, p in enumerate(prompts):
+        kv_b = tiny.new_kv_cache(16)
+        kv_b.allocate(0, len(p))
+        solo = tiny.forward_prefill(
+            T.tensor(p, dtype=T.int64, device="cuda:0"), kv_b, 0)
+        T.testing.assert_close(batch_logits[sid], solo, atol=5e-2, rtol=5e-2)
+        assert int(batch_logits[sid].argmax()) == int(solo.argmax())
+
+
 def test_encoder_contract():
     from quickstart_streaming_agents_amd.models.encoder import EmbeddingEncoder
     enc = EmbeddingEncoder(device="cuda:0")
