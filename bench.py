@@ -185,6 +185,13 @@ def main():
             },
         }
         print(json.dumps(out))
+        if not args.stub_llm:
+            st = engine.stats
+            print(f"[bench] prefill_tokens={st.prefill_tokens} "
+                  f"cached_prefix_tokens={st.cached_prefix_tokens} "
+                  f"prefill_batches={st.prefill_batches} "
+                  f"decode_tokens={st.decode_tokens} "
+                  f"decode_steps={st.decode_steps}", file=sys.stderr)
     server.stop()
     if dist is not None:
         dist.destroy_process_group()

@@ -202,16 +202,18 @@ class LlamaModel:
                 else:
                     kseq, vseq = self._gather_kv(kv.k[li], kv.v[li],
                                                  kv._seq_pages[sid], ctx)
-                qf = q[off:off + n].permute(1, 0, 2).float()   # [QH, n, D]
+                # GQA grouped: [KVH, R*n, D] x [KVH, ctx, D] -> no K/V copies
                 R = c.n_q_heads // c.n_kv_heads
-                kf = kseq.float().repeat_interleave(R, dim=0)  # [QH, ctx, D]
-                vf = vseq.float().repeat_interleave(R, dim=0)
+                qf = q[off:off + n].permute(1, 0, 2).reshape(
+                    c.n_kv_heads, R * n, c.d_head).float()
+                kf = kseq.float()                              # [KVH, ctx, D]
+                vf = vseq.float()
                 scores = (torch.bmm(qf, kf.transpose(1, 2)) * self.scale) \
                     .reshape(c.n_q_heads * n, ctx).contiguous()
                 e.softmax_rows_(scores, start, True, n, None)
-                a = torch.bmm(scores.reshape(c.n_q_heads, n, ctx), vf)
-                attn[off:off + n] = a.permute(1, 0, 2).reshape(n, -1) \
-                    .to(self.dtype)
+                a = torch.bmm(scores.reshape(c.n_kv_heads, R * n, ctx), vf)
+                attn[off:off + n] = a.reshape(c.n_q_heads, n, c.d_head) \
+                    .permute(1, 0, 2).reshape(n, -1).to(self.dtype)
             o = F.linear(attn, L["wo"])
             h = e.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             gu = F.linear(h, L["wgu"])

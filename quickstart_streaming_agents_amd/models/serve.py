@@ -66,6 +66,86 @@ class Engine:
         self.pending: list[Sequence] = []
         self.running: list[Sequence] = []
         self.stats = EngineStats()
+        # hipGraph-captured decode step (launch-bound otherwise: ~300
+        # kernel/GEMM launches per step across 32 layers)
+        self.use_graph = torch.cuda.is_available()
+        self._graph = None
+        self._gbuf: dict = {}
+
+    # ---- hipGraph decode -------------------------------------------------
+    MAX_RUN = 512  # on-device token-history depth per graph run
+
+    def _ensure_graph(self) -> None:
+        if self._graph is not None:
+            return
+        dev = self.model.device
+        B = self.max_batch
+        cap = (self.max_seq_len + 63) // 64
+        gb = {
+            "tokens": torch.zeros(B, dtype=torch.int64, device=dev),
+            "block_table": torch.zeros(B, cap, dtype=torch.int32, device=dev),
+            "seq_lens": torch.zeros(B, dtype=torch.int32, device=dev),
+            "active": torch.zeros(B, dtype=torch.int32, device=dev),
+            "ctr": torch.zeros(1, dtype=torch.int64, device=dev),
+            "hist": torch.zeros(self.MAX_RUN, B, dtype=torch.int64, device=dev),
+        }
+
+        def body():
+            # self-feeding decode step: advance seq_lens in-graph, append the
+            # step's k/v at seq_lens-1, attend, sample, record, feed back.
+            gb["seq_lens"].add_(gb["active"])
+            positions = (gb["seq_lens"] - 1).clamp(min=0).int()
+            logits = self.model.forward_decode(
+                gb["tokens"], self.kv, gb["block_table"], gb["seq_lens"],
+                positions)
+            nxt = torch.argmax(logits, dim=-1)
+            gb["hist"].index_copy_(0, gb["ctr"], nxt.unsqueeze(0))
+            gb["ctr"].add_(1)
+            gb["tokens"].copy_(nxt)
+
+        # warm up the exact captured ops (allocator + rocBLAS algo selection)
+        for _ in range(2):
+            body()
+        gb["ctr"].zero_()
+        gb["seq_lens"].zero_()
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            body()
+        self._graph = g
+        self._gbuf = gb
+
+    def _decode_run_graph(self, batch: list[Sequence], run: int) -> None:
+        """Run `run` decode steps for `batch` with zero host round trips:
+        pages are pre-extended, the graph self-feeds, one sync at the end."""
+        self._ensure_graph()
+        gb = self._gbuf
+        n = len(batch)
+        dev = self.model.device
+        for s in batch:  # pre-extend pages for the whole run
+            self.kv.extend(s.seq_id, len(s.prompt) + len(s.out_tokens) + run)
+            # bookkeeping length stays at the pre-run value for seq_lens
+            self.kv._seq_len[s.seq_id] = len(s.prompt) + len(s.out_tokens)
+        seq_ids = [s.seq_id for s in batch]
+        bt = self.kv.block_table(seq_ids)
+        gb["block_table"][:n, :bt.shape[1]] = bt
+        gb["seq_lens"][:n] = self.kv.seq_lens_tensor(seq_ids)
+        gb["tokens"][:n] = torch.tensor([s.out_tokens[-1] for s in batch],
+                                        dtype=torch.int64, device=dev)
+        gb["active"][:n] = 1
+        if n < self.max_batch:
+            gb["seq_lens"][n:] = 0
+            gb["active"][n:] = 0
+        gb["ctr"].zero_()
+        for _ in range(run):
+            self._graph.replay()
+        hist = gb["hist"][:run, :n].t().tolist()  # one sync
+        for s, toks in zip(batch, hist):
+            s.out_tokens.extend(int(t) for t in toks)
+            self.kv._seq_len[s.seq_id] = len(s.prompt) + len(s.out_tokens)
+            self._maybe_finish(s)
+        self.stats.decode_tokens += n * run
+        self.stats.decode_steps += run
 
     # ------------------------------------------------------------------
     def submit(self, prompt_tokens: list[int], max_new_tokens: int,
@@ -167,18 +247,22 @@ class Engine:
             self._retire()
             return len(self.running) + len(self.pending)
         dev = self.model.device
-        tokens = torch.tensor([s.out_tokens[-1] for s in batch],
-                              dtype=torch.int64, device=dev)
-        positions = torch.tensor(
-            [len(s.prompt) + len(s.out_tokens) - 1 for s in batch],
-            dtype=torch.int32, device=dev)
         for s in batch:
             self.kv.extend(s.seq_id, len(s.prompt) + len(s.out_tokens))
-        seq_ids = [s.seq_id for s in batch]
-        bt = self.kv.block_table(seq_ids)
-        sl = self.kv.seq_lens_tensor(seq_ids)
-        logits = self.model.forward_decode(tokens, self.kv, bt, sl, positions)
-        nxt = torch.argmax(logits, dim=-1).tolist()
+        if self.use_graph:
+            nxt = self._decode_graph(batch)
+        else:
+            tokens = torch.tensor([s.out_tokens[-1] for s in batch],
+                                  dtype=torch.int64, device=dev)
+            positions = torch.tensor(
+                [len(s.prompt) + len(s.out_tokens) - 1 for s in batch],
+                dtype=torch.int32, device=dev)
+            seq_ids = [s.seq_id for s in batch]
+            bt = self.kv.block_table(seq_ids)
+            sl = self.kv.seq_lens_tensor(seq_ids)
+            logits = self.model.forward_decode(tokens, self.kv, bt, sl,
+                                               positions)
+            nxt = torch.argmax(logits, dim=-1).tolist()
         for s, tok in zip(batch, nxt):
             s.out_tokens.append(int(tok))
             self._maybe_finish(s)
@@ -189,8 +273,27 @@ class Engine:
 
     def run_to_completion(self) -> None:
         t0 = time.perf_counter()
-        while self.pending or self.running:
-            self.step()
+        if self.use_graph:
+            while self.pending or self.running:
+                before = len(self.pending)
+                self._admit()
+                batch = [s for s in self.running if not s.done]
+                if not batch:
+                    self._retire()
+                    if self.pending and len(self.pending) == before and \
+                            not self.running:
+                        raise MemoryError(
+                            "decode stalled: pending prompts cannot be "
+                            "admitted (KV pages exhausted?)")
+                    continue
+                run = min(min(s.max_new_tokens - len(s.out_tokens)
+                              for s in batch), self.MAX_RUN)
+                if run > 0:
+                    self._decode_run_graph(batch, run)
+                self._retire()
+        else:
+            while self.pending or self.running:
+                self.step()
         if torch.cuda.is_available():
             torch.cuda.synchronize()
         self.stats.wall_s += time.perf_counter() - t0
