@@ -202,18 +202,22 @@ class LlamaModel:
                 else:
                     kseq, vseq = self._gather_kv(kv.k[li], kv.v[li],
                                                  kv._seq_pages[sid], ctx)
-                # GQA grouped: [KVH, R*n, D] x [KVH, ctx, D] -> no K/V copies
+                # GQA grouped: [KVH, R*n, D] x [KVH, ctx, D]; bf16 MFMA
+                # GEMMs (f32 bmm is 1/16 the MFMA rate on CDNA4)
                 R = c.n_q_heads // c.n_kv_heads
                 qf = q[off:off + n].permute(1, 0, 2).reshape(
-                    c.n_kv_heads, R * n, c.d_head).float()
-                kf = kseq.float()                              # [KVH, ctx, D]
-                vf = vseq.float()
-                scores = (torch.bmm(qf, kf.transpose(1, 2)) * self.scale) \
+                    c.n_kv_heads, R * n, c.d_head)
+                kf = kseq.contiguous()                         # [KVH, ctx, D]
+                vf = vseq.contiguous()
+                scores = (torch.bmm(qf, kf.transpose(1, 2)).float()
+                          * self.scale) \
                     .reshape(c.n_q_heads * n, ctx).contiguous()
                 e.softmax_rows_(scores, start, True, n, None)
-                a = torch.bmm(scores.reshape(c.n_kv_heads, R * n, ctx), vf)
+                probs = scores.reshape(c.n_kv_heads, R * n, ctx) \
+                    .to(self.dtype)
+                a = torch.bmm(probs, vf)
                 attn[off:off + n] = a.reshape(c.n_q_heads, n, c.d_head) \
-                    .permute(1, 0, 2).reshape(n, -1).to(self.dtype)
+                    .permute(1, 0, 2).reshape(n, -1)
             o = F.linear(attn, L["wo"])
             h = e.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             gu = F.linear(h, L["wgu"])

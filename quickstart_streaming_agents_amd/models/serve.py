@@ -53,7 +53,7 @@ class EngineStats:
 class Engine:
     def __init__(self, model: LlamaModel, kv_pages: int | None = None,
                  max_batch: int = 256, max_seq_len: int = 4096,
-                 eos_id: int | None = None, prefill_batch_tokens: int = 16384):
+                 eos_id: int | None = None, prefill_batch_tokens: int = 32768):
         self.model = model
         self.max_batch = max_batch
         self.max_seq_len = max_seq_len
@@ -276,7 +276,13 @@ class Engine:
         if self.use_graph:
             while self.pending or self.running:
                 before = len(self.pending)
-                self._admit()
+                # drain the whole pending queue (possibly several prefill
+                # batches) before decoding: decode runs want the full batch
+                while self.pending:
+                    n_before = len(self.pending)
+                    self._admit()
+                    if len(self.pending) == n_before:
+                        break
                 batch = [s for s in self.running if not s.done]
                 if not batch:
                     self._retire()
