@@ -191,7 +191,8 @@ def main():
                   f"cached_prefix_tokens={st.cached_prefix_tokens} "
                   f"prefill_batches={st.prefill_batches} "
                   f"decode_tokens={st.decode_tokens} "
-                  f"decode_steps={st.decode_steps}", file=sys.stderr)
+                  f"decode_steps={st.decode_steps} "
+                  f"prefill_s={st.prefill_s:.2f} decode_s={st.decode_s:.2f}", file=sys.stderr)
     server.stop()
     if dist is not None:
         dist.destroy_process_group()

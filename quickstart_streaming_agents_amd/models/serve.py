@@ -48,6 +48,12 @@ class EngineStats:
     prefills: int = 0
     prefill_batches: int = 0
     wall_s: float = 0.0
+    prefill_s: float = 0.0
+    decode_s: float = 0.0
+
+
+import os as _os
+_TIMING = _os.environ.get("QSA_TIMING", "") == "1"
 
 
 class Engine:
@@ -137,8 +143,14 @@ class Engine:
             gb["seq_lens"][n:] = 0
             gb["active"][n:] = 0
         gb["ctr"].zero_()
+        if _TIMING:
+            torch.cuda.synchronize()
+            _t0 = time.perf_counter()
         for _ in range(run):
             self._graph.replay()
+        if _TIMING:
+            torch.cuda.synchronize()
+            self.stats.decode_s += time.perf_counter() - _t0
         hist = gb["hist"][:run, :n].t().tolist()  # one sync
         for s, toks in zip(batch, hist):
             s.out_tokens.extend(int(t) for t in toks)
@@ -214,7 +226,13 @@ class Engine:
             self.stats.prefills += 1
         if not admitted:
             return
+        if _TIMING:
+            torch.cuda.synchronize()
+            _t0 = time.perf_counter()
         logits = self.model.forward_prefill_batch(batch_items, self.kv)
+        if _TIMING:
+            torch.cuda.synchronize()
+            self.stats.prefill_s += time.perf_counter() - _t0
         first = torch.argmax(logits, dim=-1).tolist()
         self.stats.prefill_batches += 1
         for seq, tok in zip(admitted, first):
