@@ -127,7 +127,7 @@ class EmbeddingEncoder:
             o = F.linear(attn, Ly["wo"])
             h = e.rmsnorm_residual(o, res, Ly["mlp_norm"], c.norm_eps)
             gu = F.linear(h, Ly["wgu"])
-            act = e.swiglu(gu[:, :c.ffn].contiguous(), gu[:, c.ffn:].contiguous())
+            act = e.swiglu(gu[:, :c.ffn], gu[:, c.ffn:])
             mlp_out = F.linear(act, Ly["wdown"])
         final_h = e.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
         # mean-pool valid positions, project, L2-normalize

@@ -125,19 +125,24 @@ class LlamaModel:
             else:
                 h = e.rmsnorm_residual(mlp_out, res, L["attn_norm"], c.norm_eps)
             qkv = F.linear(h, L["wqkv"])
-            q, k, v = self._split_qkv(qkv, B)
+            # strided [B, H, D] views straight into the fused qkv buffer —
+            # the kernels take row strides, no contiguous() copies
+            qd = c.n_q_heads * c.d_head
+            kd = c.n_kv_heads * c.d_head
+            q = qkv[:, :qd].view(B, c.n_q_heads, c.d_head)
+            k = qkv[:, qd:qd + kd].view(B, c.n_kv_heads, c.d_head)
+            v = qkv[:, qd + kd:].view(B, c.n_kv_heads, c.d_head)
             e.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
             e.kv_append(k, v, kv.k[li], kv.v[li], block_table, seq_lens)
             attn = e.paged_attn_decode(q, kv.k[li], kv.v[li], block_table,
                                        seq_lens, self.scale)
-            o = F.linear(attn.reshape(B, -1), L["wo"])
+            o = F.linear(attn.view(B, -1), L["wo"])
             h = e.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             gu = F.linear(h, L["wgu"])
-            act = e.swiglu(gu[:, :c.ffn].contiguous(),
-                           gu[:, c.ffn:].contiguous())
+            act = e.swiglu(gu[:, :c.ffn], gu[:, c.ffn:])
             mlp_out = F.linear(act, L["wdown"])
         final_h = e.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
-        return F.linear(final_h, self.lm_head).float()
+        return F.linear(final_h, self.lm_head)
 
     def _gather_kv(self, kc_l: torch.Tensor, vc_l: torch.Tensor,
                    pages: list[int], n: int):
@@ -221,13 +226,12 @@ class LlamaModel:
             o = F.linear(attn, L["wo"])
             h = e.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             gu = F.linear(h, L["wgu"])
-            act = e.swiglu(gu[:, :c.ffn].contiguous(),
-                           gu[:, c.ffn:].contiguous())
+            act = e.swiglu(gu[:, :c.ffn], gu[:, c.ffn:])
             mlp_out = F.linear(act, L["wdown"])
         final_h = e.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
         last = torch.tensor([offs[i] + lens[i] - 1 for i in range(len(items))],
                             dtype=torch.int64, device=dev)
-        return F.linear(final_h.index_select(0, last), self.lm_head).float()
+        return F.linear(final_h.index_select(0, last), self.lm_head)
 
     @torch.no_grad()
     def forward_prefill(self, tokens: torch.Tensor, kv: PagedKVCache,

@@ -78,12 +78,16 @@ qsa_rmsnorm_kernel(const unsigned short* __restrict__ x,
 __global__ void __launch_bounds__(256)
 qsa_swiglu_kernel(const unsigned short* __restrict__ gate,
                   const unsigned short* __restrict__ up,
-                  unsigned short* __restrict__ y, long long n) {
+                  unsigned short* __restrict__ y, long long rows,
+                  long long cols, long long in_stride) {
   long long i = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
   const long long stride = (long long)gridDim.x * blockDim.x * 8;
+  const long long n = rows * cols;
   for (; i + 7 < n; i += stride) {
-    uint4 g = *reinterpret_cast<const uint4*>(gate + i);
-    uint4 u = *reinterpret_cast<const uint4*>(up + i);
+    const long long r = i / cols, cidx = i % cols;
+    const long long off = r * in_stride + cidx;
+    uint4 g = *reinterpret_cast<const uint4*>(gate + off);
+    uint4 u = *reinterpret_cast<const uint4*>(up + off);
     unsigned int gp[4] = {g.x, g.y, g.z, g.w};
     unsigned int upk[4] = {u.x, u.y, u.z, u.w};
     uint4 outv;
@@ -108,7 +112,8 @@ qsa_swiglu_kernel(const unsigned short* __restrict__ gate,
 __global__ void
 qsa_rope_kernel(unsigned short* __restrict__ q, unsigned short* __restrict__ k,
                 const float* __restrict__ cos_t, const float* __restrict__ sin_t,
-                const int* __restrict__ pos, int B, int QH, int KVH, int D) {
+                const int* __restrict__ pos, int B, int QH, int KVH, int D,
+                long long qstride, long long kstride) {
   const int bh = blockIdx.x;
   const int nheads = QH + KVH;
   const int b = bh / nheads;
@@ -117,8 +122,8 @@ qsa_rope_kernel(unsigned short* __restrict__ q, unsigned short* __restrict__ k,
   const int d = threadIdx.x;
   if (d >= half) return;
   unsigned short* base =
-      (h < QH) ? q + ((long long)b * QH + h) * D
-               : k + ((long long)b * KVH + (h - QH)) * D;
+      (h < QH) ? q + (long long)b * qstride + (long long)h * D
+               : k + (long long)b * kstride + (long long)(h - QH) * D;
   const long long toff = (long long)pos[b] * half + d;
   const float c = cos_t[toff], s = sin_t[toff];
   const float x0 = bf16_to_f32(base[d]);
@@ -184,17 +189,21 @@ extern "C" void qsa_rmsnorm_launch(const unsigned short* x,
 
 extern "C" void qsa_swiglu_launch(const unsigned short* gate,
                                   const unsigned short* up, unsigned short* y,
-                                  long long n, int blocks, hipStream_t stream) {
+                                  long long rows, long long cols,
+                                  long long in_stride, int blocks,
+                                  hipStream_t stream) {
   hipLaunchKernelGGL(qsa_swiglu_kernel, dim3(blocks), dim3(256), 0, stream,
-                     gate, up, y, n);
+                     gate, up, y, rows, cols, in_stride);
 }
 
 extern "C" void qsa_rope_launch(unsigned short* q, unsigned short* k,
                                 const float* cos_t, const float* sin_t,
                                 const int* pos, int B, int QH, int KVH, int D,
+                                long long qstride, long long kstride,
                                 hipStream_t stream) {
   hipLaunchKernelGGL(qsa_rope_kernel, dim3(B * (QH + KVH)), dim3(D / 2), 0,
-                     stream, q, k, cos_t, sin_t, pos, B, QH, KVH, D);
+                     stream, q, k, cos_t, sin_t, pos, B, QH, KVH, D, qstride,
+                     kstride);
 }
 
 extern "C" void qsa_softmax_rows_launch(float* scores, int rows, int cols,

@@ -26,10 +26,11 @@ extern "C" void qsa_rmsnorm_launch(const unsigned short*, const unsigned short*,
                                    unsigned short*, unsigned short*, long long,
                                    int, float, hipStream_t);
 extern "C" void qsa_swiglu_launch(const unsigned short*, const unsigned short*,
-                                  unsigned short*, long long, int, hipStream_t);
+                                  unsigned short*, long long, long long,
+                                  long long, int, hipStream_t);
 extern "C" void qsa_rope_launch(unsigned short*, unsigned short*, const float*,
                                 const float*, const int*, int, int, int, int,
-                                hipStream_t);
+                                long long, long long, hipStream_t);
 extern "C" void qsa_softmax_rows_launch(float*, int, int, int, int, int,
                                         const int*, hipStream_t);
 extern "C" void qsa_paged_attn_decode_launch(const unsigned short*,
@@ -37,11 +38,12 @@ extern "C" void qsa_paged_attn_decode_launch(const unsigned short*,
                                              const unsigned short*, const int*,
                                              const int*, unsigned short*, float,
                                              int, int, int, int, int,
-                                             hipStream_t);
+                                             long long, hipStream_t);
 extern "C" void qsa_kv_append_launch(const unsigned short*,
                                      const unsigned short*, unsigned short*,
                                      unsigned short*, const int*, const int*,
-                                     int, int, int, int, hipStream_t);
+                                     int, int, int, int, long long,
+                                     hipStream_t);
 extern "C" void qsa_kv_scatter_launch(const unsigned short*,
                                       const unsigned short*, unsigned short*,
                                       unsigned short*, const int*, int, int,
@@ -85,26 +87,40 @@ torch::Tensor rmsnorm_residual(torch::Tensor x, torch::Tensor res,
 }
 
 torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
-  CHK_DEV(gate); CHK_CONT(gate); CHK_BF16(gate); CHK_BF16(up); CHK_CONT(up);
-  TORCH_CHECK(gate.numel() == up.numel(), "shape mismatch");
-  TORCH_CHECK(gate.numel() % 8 == 0, "numel % 8 == 0");
-  auto y = torch::empty_like(gate);
-  const long long n = gate.numel();
+  // gate/up: [rows, cols] views with equal row stride (e.g. halves of the
+  // fused gate_up GEMM output); cols % 8 == 0, element-contiguous rows.
+  CHK_DEV(gate); CHK_BF16(gate); CHK_BF16(up);
+  TORCH_CHECK(gate.dim() == 2 && up.dim() == 2, "2-D");
+  TORCH_CHECK(gate.sizes() == up.sizes(), "shape mismatch");
+  TORCH_CHECK(gate.stride(1) == 1 && up.stride(1) == 1, "rows contiguous");
+  TORCH_CHECK(gate.stride(0) == up.stride(0), "row strides must match");
+  const long long rows = gate.size(0), cols = gate.size(1);
+  TORCH_CHECK(cols % 8 == 0, "cols % 8 == 0");
+  auto y = torch::empty({rows, cols}, gate.options());
+  const long long n = rows * cols;
   const int blocks = (int)std::min<long long>((n / 8 + 255) / 256, 8192);
-  qsa_swiglu_launch(u16(gate), u16(up), u16m(y), n, blocks, cur_stream());
+  qsa_swiglu_launch(u16(gate), u16(up), u16m(y), rows, cols, gate.stride(0),
+                    blocks, cur_stream());
   return y;
+}
+
+static inline void chk_hd_strided(const torch::Tensor& t, int D,
+                                  const char* name) {
+  TORCH_CHECK(t.dim() == 3 && t.stride(2) == 1 && t.stride(1) == D,
+              name, " must be a [B, H, D] view with contiguous heads");
 }
 
 void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cos_t,
                   torch::Tensor sin_t, torch::Tensor pos) {
-  CHK_DEV(q); CHK_CONT(q); CHK_BF16(q); CHK_BF16(k); CHK_CONT(k);
+  CHK_DEV(q); CHK_BF16(q); CHK_BF16(k);
   CHK_F32(cos_t); CHK_F32(sin_t); CHK_I32(pos);
   const int B = q.size(0), QH = q.size(1), D = q.size(2);
   const int KVH = k.size(1);
   TORCH_CHECK(k.size(0) == B && k.size(2) == D, "k shape");
+  chk_hd_strided(q, D, "q"); chk_hd_strided(k, D, "k");
   qsa_rope_launch(u16m(q), u16m(k), cos_t.data_ptr<float>(),
                   sin_t.data_ptr<float>(), pos.data_ptr<int>(), B, QH, KVH, D,
-                  cur_stream());
+                  q.stride(0), k.stride(0), cur_stream());
 }
 
 void softmax_rows_(torch::Tensor scores, long col_offset, bool causal,
@@ -124,10 +140,11 @@ void softmax_rows_(torch::Tensor scores, long col_offset, bool causal,
 torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor kc,
                                 torch::Tensor vc, torch::Tensor block_table,
                                 torch::Tensor seq_lens, double scale) {
-  CHK_DEV(q); CHK_CONT(q); CHK_BF16(q); CHK_BF16(kc); CHK_BF16(vc);
+  CHK_DEV(q); CHK_BF16(q); CHK_BF16(kc); CHK_BF16(vc);
   CHK_CONT(kc); CHK_CONT(vc); CHK_I32(block_table); CHK_I32(seq_lens);
   CHK_CONT(block_table);
   const int B = q.size(0), QH = q.size(1), D = q.size(2);
+  chk_hd_strided(q, D, "q");
   const int KVH = kc.size(1);
   TORCH_CHECK(D == 128 || D == 64, "D must be 64/128");
   TORCH_CHECK(QH % KVH == 0, "GQA requires QH % KVH == 0");
@@ -137,24 +154,27 @@ torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor kc,
   TORCH_CHECK(vc.size(2) == 64 && vc.size(3) == D,
               "V cache layout [P, KVH, 64, D]");
   const int max_pages = block_table.size(1);
-  auto out = torch::empty_like(q);
+  auto out = torch::empty({B, QH, D}, q.options());
   qsa_paged_attn_decode_launch(u16(q), u16(kc), u16(vc),
                                block_table.data_ptr<int>(),
                                seq_lens.data_ptr<int>(), u16m(out),
                                (float)scale, B, QH, KVH, max_pages, D,
-                               cur_stream());
+                               q.stride(0), cur_stream());
   return out;
 }
 
 void kv_append(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kc,
                torch::Tensor vc, torch::Tensor block_table,
                torch::Tensor seq_lens) {
-  CHK_DEV(knew); CHK_CONT(knew); CHK_BF16(knew); CHK_BF16(vnew);
+  CHK_DEV(knew); CHK_BF16(knew); CHK_BF16(vnew);
   CHK_I32(block_table); CHK_I32(seq_lens);
   const int B = knew.size(0), KVH = knew.size(1), D = knew.size(2);
+  chk_hd_strided(knew, D, "knew"); chk_hd_strided(vnew, D, "vnew");
+  TORCH_CHECK(knew.stride(0) == vnew.stride(0), "k/v row strides must match");
   qsa_kv_append_launch(u16(knew), u16(vnew), u16m(kc), u16m(vc),
                        block_table.data_ptr<int>(), seq_lens.data_ptr<int>(),
-                       B, KVH, D, (int)block_table.size(1), cur_stream());
+                       B, KVH, D, (int)block_table.size(1), knew.stride(0),
+                       cur_stream());
 }
 
 void kv_scatter(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kc,

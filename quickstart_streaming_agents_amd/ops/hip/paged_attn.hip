@@ -27,13 +27,14 @@
 
 template <int D>
 __global__ void __launch_bounds__(256)
-qsa_paged_attn_decode(const unsigned short* __restrict__ q,   // [B, QH, D]
+qsa_paged_attn_decode(const unsigned short* __restrict__ q,   // [B, QH, D] (row stride qstride)
                       const unsigned short* __restrict__ kc,  // [P, KVH, D/8, 64, 8]
                       const unsigned short* __restrict__ vc,  // [P, KVH, 64, D]
                       const int* __restrict__ block_table,    // [B, max_pages]
                       const int* __restrict__ seq_lens,       // [B]
                       unsigned short* __restrict__ out,       // [B, QH, D]
-                      float scale, int B, int QH, int KVH, int max_pages) {
+                      float scale, int B, int QH, int KVH, int max_pages,
+                      long long qstride) {
   const int b = blockIdx.x / KVH;
   const int kvh = blockIdx.x % KVH;
   const int wave = threadIdx.x / QSA_WAVE;
@@ -49,22 +50,32 @@ qsa_paged_attn_decode(const unsigned short* __restrict__ q,   // [B, QH, D]
     // ---- q packed bf16x2 into regs: D/2 uints (64 for D=128) ----
     unsigned int qpk[D / 2];  // D bf16 = D/2 packed uints
     const unsigned int* qsrc = reinterpret_cast<const unsigned int*>(
-        q + ((long long)b * QH + qh) * D);
+        q + (long long)b * qstride + (long long)qh * D);
 #pragma unroll
     for (int i = 0; i < D / 2; ++i) qpk[i] = qsrc[i];
 
+    // PV decomposition: lane = pg * DG + dg — dim-group dg owns 8 output
+    // dims, position-group pg covers PG=64/DG positions per iteration, so a
+    // V read is 64 lanes x 16 B = 1 KiB fully coalesced and the page's PV
+    // takes PAGE/PG wide iterations instead of 64 scalar ones.
+    constexpr int DG = D / 8;        // lanes per output row (16 for D=128)
+    constexpr int PG = QSA_WAVE / DG;  // positions per iteration (4)
+    const int dg = lane % DG;
+    const int pg = lane / DG;
+
     float m = -3.0e38f, s = 0.f;
-    float o0 = 0.f, o1 = 0.f;  // lane-owned dims 2*lane, 2*lane+1
+    float o8[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o8[j] = 0.f;
 
     for (int pi = 0; pi < npages; ++pi) {
       const int page = btab[pi];
       const int pos = pi * QSA_PAGE + lane;
       const bool valid = pos < seqlen;
-      // ---- score: dot(q, K[:, pos]) over d-slices of 8 ----
+      // ---- score: dot(q, K[:, pos]) over d-slices of 8 (lane = position)
       float sc = 0.f;
       const uint4* kbase = reinterpret_cast<const uint4*>(
           kc + ((((long long)page * KVH + kvh) * (D / 8)) * QSA_PAGE) * 8);
-      // slice d0 stride in uint4 units: QSA_PAGE (64 lanes * 16B)
 #pragma unroll
       for (int d0 = 0; d0 < D / 8; ++d0) {
         uint4 kv4 = kbase[d0 * QSA_PAGE + lane];
@@ -86,28 +97,43 @@ qsa_paged_attn_decode(const unsigned short* __restrict__ q,   // [B, QH, D]
       m = m_new;
       const float p = valid ? __expf(sc - m_new) : 0.f;
       s = s * alpha + wave_reduce_sum(p);
-      o0 *= alpha;
-      o1 *= alpha;
-      // ---- PV: lane owns dims (2l, 2l+1); prob via shfl broadcast ----
-      const unsigned short* vbase =
-          vc + (((long long)page * KVH + kvh) * QSA_PAGE) * D;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o8[j] *= alpha;
+      // ---- PV: PG positions per 16-B-per-lane iteration ----
+      const uint4* vbase = reinterpret_cast<const uint4*>(
+          vc + (((long long)page * KVH + kvh) * QSA_PAGE) * D);
       const int nvalid = min(seqlen - pi * QSA_PAGE, QSA_PAGE);
+      const int niter = (nvalid + PG - 1) / PG;
 #pragma unroll 4
-      for (int t = 0; t < nvalid; ++t) {
-        const float pt = __shfl(p, t, QSA_WAVE);
-        if (lane * 2 < D) {
-          const unsigned int vpk = *reinterpret_cast<const unsigned int*>(
-              vbase + (long long)t * D + lane * 2);
-          float2 vf = bf16x2_to_f32x2(vpk);
-          o0 = fmaf(pt, vf.x, o0);
-          o1 = fmaf(pt, vf.y, o1);
+      for (int it = 0; it < niter; ++it) {
+        const int t = it * PG + pg;
+        const float pt = __shfl(p, t, QSA_WAVE);  // p==0 beyond nvalid
+        const uint4 v4 = vbase[(long long)t * DG + dg];
+        unsigned int vv[4] = {v4.x, v4.y, v4.z, v4.w};
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float2 vf = bf16x2_to_f32x2(vv[j]);
+          o8[2 * j] = fmaf(pt, vf.x, o8[2 * j]);
+          o8[2 * j + 1] = fmaf(pt, vf.y, o8[2 * j + 1]);
         }
       }
     }
+    // butterfly-reduce the PG position-group partials (same dg lanes)
+#pragma unroll
+    for (int off = DG; off < QSA_WAVE; off <<= 1) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o8[j] += __shfl_xor(o8[j], off, QSA_WAVE);
+    }
     const float inv = (s > 0.f) ? 1.f / s : 0.f;
-    unsigned int* orow = reinterpret_cast<unsigned int*>(
-        out + ((long long)b * QH + qh) * D);
-    if (lane * 2 < D) orow[lane] = f32x2_to_bf16x2(o0 * inv, o1 * inv);
+    if (pg == 0) {
+      uint4* orow = reinterpret_cast<uint4*>(out + ((long long)b * QH + qh) * D);
+      uint4 packed;
+      packed.x = f32x2_to_bf16x2(o8[0] * inv, o8[1] * inv);
+      packed.y = f32x2_to_bf16x2(o8[2] * inv, o8[3] * inv);
+      packed.z = f32x2_to_bf16x2(o8[4] * inv, o8[5] * inv);
+      packed.w = f32x2_to_bf16x2(o8[6] * inv, o8[7] * inv);
+      orow[dg] = packed;
+    }
   }
 }
 
@@ -115,17 +141,17 @@ extern "C" void qsa_paged_attn_decode_launch(
     const unsigned short* q, const unsigned short* kc,
     const unsigned short* vc, const int* block_table, const int* seq_lens,
     unsigned short* out, float scale, int B, int QH, int KVH, int max_pages,
-    int D, hipStream_t stream) {
+    int D, long long qstride, hipStream_t stream) {
   dim3 grid(B * KVH);
   dim3 block(256);
   if (D == 128) {
     hipLaunchKernelGGL((qsa_paged_attn_decode<128>), grid, block, 0, stream,
                        q, kc, vc, block_table, seq_lens, out, scale, B, QH,
-                       KVH, max_pages);
+                       KVH, max_pages, qstride);
   } else if (D == 64) {
     hipLaunchKernelGGL((qsa_paged_attn_decode<64>), grid, block, 0, stream,
                        q, kc, vc, block_table, seq_lens, out, scale, B, QH,
-                       KVH, max_pages);
+                       KVH, max_pages, qstride);
   }
 }
 
@@ -135,11 +161,11 @@ extern "C" void qsa_paged_attn_decode_launch(
 // One block per (b, kvh); D threads.
 // ---------------------------------------------------------------------------
 __global__ void
-qsa_kv_append(const unsigned short* __restrict__ knew,  // [B, KVH, D]
-              const unsigned short* __restrict__ vnew,  // [B, KVH, D]
+qsa_kv_append(const unsigned short* __restrict__ knew,  // [B, KVH, D] (row stride kvstride)
+              const unsigned short* __restrict__ vnew,
               unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
               const int* __restrict__ block_table, const int* __restrict__ seq_lens,
-              int B, int KVH, int D, int max_pages) {
+              int B, int KVH, int D, int max_pages, long long kvstride) {
   const int b = blockIdx.x / KVH;
   const int kvh = blockIdx.x % KVH;
   const int d = threadIdx.x;
@@ -148,8 +174,8 @@ qsa_kv_append(const unsigned short* __restrict__ knew,  // [B, KVH, D]
   if (pos < 0) return;
   const int page = block_table[(long long)b * max_pages + pos / QSA_PAGE];
   const int pin = pos % QSA_PAGE;
-  const unsigned short kv = knew[((long long)b * KVH + kvh) * D + d];
-  const unsigned short vv = vnew[((long long)b * KVH + kvh) * D + d];
+  const unsigned short kv = knew[(long long)b * kvstride + (long long)kvh * D + d];
+  const unsigned short vv = vnew[(long long)b * kvstride + (long long)kvh * D + d];
   // K layout [page, kvh, D/8, 64, 8]
   kc[((((long long)page * KVH + kvh) * (D / 8) + d / 8) * QSA_PAGE + pin) * 8 +
      d % 8] = kv;
@@ -184,9 +210,11 @@ extern "C" void qsa_kv_append_launch(const unsigned short* knew,
                                      unsigned short* kc, unsigned short* vc,
                                      const int* block_table,
                                      const int* seq_lens, int B, int KVH,
-                                     int D, int max_pages, hipStream_t stream) {
+                                     int D, int max_pages, long long kvstride,
+                                     hipStream_t stream) {
   hipLaunchKernelGGL(qsa_kv_append, dim3(B * KVH), dim3(D), 0, stream, knew,
-                     vnew, kc, vc, block_table, seq_lens, B, KVH, D, max_pages);
+                     vnew, kc, vc, block_table, seq_lens, B, KVH, D, max_pages,
+                     kvstride);
 }
 
 extern "C" void qsa_kv_scatter_launch(const unsigned short* knew,
