@@ -17,7 +17,7 @@ from dataclasses import dataclass
 import torch
 import torch.nn.functional as F
 
-from ..ops import ext
+from ..ops import dispatch as D
 from ..vector.index import EMBED_DIM
 from .tokenizer import HashTokenizer
 
@@ -79,7 +79,6 @@ class EmbeddingEncoder:
         if not texts:
             return np.zeros((0, self.cfg.out_dim), dtype=np.float32)
         c = self.cfg
-        e = ext()
         enc = [self.tokenizer.encode(t)[: c.max_pos] or [HashTokenizer.BOS]
                for t in texts]
         n = len(enc)
@@ -102,15 +101,15 @@ class EmbeddingEncoder:
         T = n * L
         for li, Ly in enumerate(self.layers):
             if li == 0:
-                h = e.rmsnorm(res, Ly["attn_norm"], c.norm_eps)
+                h = D.rmsnorm(res, Ly["attn_norm"], c.norm_eps)
             else:
-                h = e.rmsnorm_residual(mlp_out, res, Ly["attn_norm"], c.norm_eps)
+                h = D.rmsnorm_residual(mlp_out, res, Ly["attn_norm"], c.norm_eps)
             qkv = F.linear(h, Ly["wqkv"])
             q = qkv[:, :c.hidden].reshape(T, c.n_heads, c.d_head).contiguous()
             k = qkv[:, c.hidden:2 * c.hidden].reshape(T, c.n_heads,
                                                       c.d_head).contiguous()
             v = qkv[:, 2 * c.hidden:].reshape(T, c.n_heads, c.d_head).contiguous()
-            e.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
+            D.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
             # [n*heads, L, D]
             qh = q.reshape(n, L, c.n_heads, c.d_head).permute(0, 2, 1, 3) \
                 .reshape(n * c.n_heads, L, c.d_head)
@@ -120,16 +119,16 @@ class EmbeddingEncoder:
                 .reshape(n * c.n_heads, L, c.d_head)
             scores = (torch.bmm(qh.float(), kh.float().transpose(1, 2))
                       * self.scale).reshape(n * c.n_heads * L, L).contiguous()
-            e.softmax_rows_(scores, 0, False, 0, row_limits)
+            D.softmax_rows_(scores, 0, False, 0, row_limits)
             attn = torch.bmm(scores.reshape(n * c.n_heads, L, L), vh.float())
             attn = attn.reshape(n, c.n_heads, L, c.d_head).permute(0, 2, 1, 3) \
                 .reshape(T, c.hidden).to(self.dtype).contiguous()
             o = F.linear(attn, Ly["wo"])
-            h = e.rmsnorm_residual(o, res, Ly["mlp_norm"], c.norm_eps)
+            h = D.rmsnorm_residual(o, res, Ly["mlp_norm"], c.norm_eps)
             gu = F.linear(h, Ly["wgu"])
-            act = e.swiglu(gu[:, :c.ffn], gu[:, c.ffn:])
+            act = D.swiglu(gu[:, :c.ffn], gu[:, c.ffn:])
             mlp_out = F.linear(act, Ly["wdown"])
-        final_h = e.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
+        final_h = D.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
         # mean-pool valid positions, project, L2-normalize
         hs = final_h.reshape(n, L, c.hidden).float()
         mask = (torch.arange(L, device=self.device).unsqueeze(0)

@@ -12,6 +12,18 @@ Decode path (the hot path): fully fused per layer —
   paged_attn_decode -> o GEMM -> rmsnorm_residual -> merged gate_up GEMM ->
   swiglu -> down GEMM
 Prefill: per-sequence chunk with GEMM scores + the causal softmax kernel.
+
+Tensor parallelism (Megatron-style, RCCL over xGMI): attention heads and
+FFN columns shard across the TP group; wo / wdown are row-parallel with ONE
+all-reduce each per layer.  Weight init is deterministic — every rank
+generates the identical full tensor from the shared seed and slices its
+shard, so TP=N matches TP=1 numerics.  8-way xGMI is point-to-point
+(7 links x ~153 GB/s), so decode-sized all-reduces use the default RCCL
+algorithm which picks direct p2p for small messages.
+
+Ops route through ops.dispatch: HIP kernels on cuda tensors (fail-loud if
+the extension is missing), reference semantics on CPU so the engine and
+the TP path run under multi-process gloo tests.
 """
 
 from __future__ import annotations
@@ -21,7 +33,7 @@ from dataclasses import dataclass
 import torch
 import torch.nn.functional as F
 
-from ..ops import ext
+from ..ops import dispatch as D
 from .kv_cache import PagedKVCache
 
 
@@ -48,41 +60,74 @@ class LlamaConfig:
                        n_q_heads=32, n_kv_heads=8, d_head=64, ffn=8192)
         if name == "tiny":
             return cls(name=name, vocab_size=2048, hidden=256, n_layers=2,
-                       n_q_heads=4, n_kv_heads=1, d_head=64, ffn=512,
+                       n_q_heads=4, n_kv_heads=2, d_head=64, ffn=512,
                        max_pos=2048)
         raise ValueError(f"unknown preset {name}")
 
 
 class LlamaModel:
     def __init__(self, cfg: LlamaConfig, device: str = "cuda",
-                 dtype=torch.bfloat16, seed: int = 0):
+                 dtype=torch.bfloat16, seed: int = 0,
+                 tp_rank: int = 0, tp_size: int = 1, tp_group=None):
         self.cfg = cfg
         self.device = device
         self.dtype = dtype
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        self.tp_group = tp_group
+        c = cfg
+        assert c.n_q_heads % tp_size == 0, "q heads must divide TP"
+        assert c.n_kv_heads % tp_size == 0, "kv heads must divide TP"
+        assert c.ffn % tp_size == 0, "ffn must divide TP"
+        self.n_q = c.n_q_heads // tp_size       # local q heads
+        self.n_kv = c.n_kv_heads // tp_size     # local kv heads
+        self.ffn_local = c.ffn // tp_size
         gen_dev = device if str(device).startswith("cuda") else "cpu"
         g = torch.Generator(device=gen_dev).manual_seed(seed)
 
         def w(*shape, std=0.02):
             # generate on-device: 8B-class random init in seconds, no 32 GB
-            # host spike
+            # host spike. Full tensor on every rank -> deterministic shards.
             t = torch.randn(*shape, generator=g, dtype=torch.float32,
                             device=gen_dev) * std
             return t.to(device=device, dtype=dtype)
 
-        c = cfg
-        self.qkv_dim = (c.n_q_heads + 2 * c.n_kv_heads) * c.d_head
+        def rows(t: torch.Tensor, n_shards: int) -> torch.Tensor:
+            """This rank's contiguous row shard (column-parallel weight)."""
+            if n_shards == 1:
+                return t
+            sz = t.shape[0] // n_shards
+            return t[self.tp_rank * sz:(self.tp_rank + 1) * sz].contiguous()
+
+        def cols(t: torch.Tensor, n_shards: int) -> torch.Tensor:
+            """This rank's contiguous col shard (row-parallel weight)."""
+            if n_shards == 1:
+                return t
+            sz = t.shape[1] // n_shards
+            return t[:, self.tp_rank * sz:(self.tp_rank + 1) * sz].contiguous()
+
+        self.qkv_dim = (self.n_q + 2 * self.n_kv) * c.d_head
         self.embed = w(c.vocab_size, c.hidden)
         self.lm_head = w(c.vocab_size, c.hidden)
         self.final_norm = torch.ones(c.hidden, device=device, dtype=dtype)
         self.layers = []
         for _ in range(c.n_layers):
+            wq_full = w(c.n_q_heads * c.d_head, c.hidden)
+            wk_full = w(c.n_kv_heads * c.d_head, c.hidden)
+            wv_full = w(c.n_kv_heads * c.d_head, c.hidden)
+            wqkv = torch.cat([rows(wq_full, tp_size), rows(wk_full, tp_size),
+                              rows(wv_full, tp_size)], dim=0).contiguous()
+            wo = cols(w(c.hidden, c.n_q_heads * c.d_head), tp_size)
+            wgu_full = w(2 * c.ffn, c.hidden)
+            wgu = torch.cat([rows(wgu_full[:c.ffn], tp_size),
+                             rows(wgu_full[c.ffn:], tp_size)],
+                            dim=0).contiguous()
+            wdown = cols(w(c.hidden, c.ffn, std=0.02 / (2 * c.n_layers) ** 0.5),
+                         tp_size)
             self.layers.append({
                 "attn_norm": torch.ones(c.hidden, device=device, dtype=dtype),
                 "mlp_norm": torch.ones(c.hidden, device=device, dtype=dtype),
-                "wqkv": w(self.qkv_dim, c.hidden),
-                "wo": w(c.hidden, c.n_q_heads * c.d_head),
-                "wgu": w(2 * c.ffn, c.hidden),
-                "wdown": w(c.hidden, c.ffn, std=0.02 / (2 * c.n_layers) ** 0.5),
+                "wqkv": wqkv, "wo": wo, "wgu": wgu, "wdown": wdown,
             })
         # rope tables
         half = c.d_head // 2
@@ -94,17 +139,23 @@ class LlamaModel:
         self.scale = 1.0 / (c.d_head ** 0.5)
 
     def new_kv_cache(self, n_pages: int) -> PagedKVCache:
-        return PagedKVCache(self.cfg.n_layers, self.cfg.n_kv_heads,
-                            self.cfg.d_head, n_pages, self.device, self.dtype)
+        return PagedKVCache(self.cfg.n_layers, self.n_kv, self.cfg.d_head,
+                            n_pages, self.device, self.dtype)
+
+    def _tp_all_reduce(self, t: torch.Tensor) -> torch.Tensor:
+        if self.tp_size > 1:
+            import torch.distributed as dist
+            dist.all_reduce(t, group=self.tp_group)
+        return t
 
     # ------------------------------------------------------------------
     def _split_qkv(self, qkv: torch.Tensor, n: int):
         c = self.cfg
-        qd = c.n_q_heads * c.d_head
-        kd = c.n_kv_heads * c.d_head
-        q = qkv[:, :qd].reshape(n, c.n_q_heads, c.d_head).contiguous()
-        k = qkv[:, qd:qd + kd].reshape(n, c.n_kv_heads, c.d_head).contiguous()
-        v = qkv[:, qd + kd:].reshape(n, c.n_kv_heads, c.d_head).contiguous()
+        qd = self.n_q * c.d_head
+        kd = self.n_kv * c.d_head
+        q = qkv[:, :qd].reshape(n, self.n_q, c.d_head).contiguous()
+        k = qkv[:, qd:qd + kd].reshape(n, self.n_kv, c.d_head).contiguous()
+        v = qkv[:, qd + kd:].reshape(n, self.n_kv, c.d_head).contiguous()
         return q, k, v
 
     @torch.no_grad()
@@ -115,33 +166,33 @@ class LlamaModel:
         token (its k/v is appended at position seq_lens-1)."""
         c = self.cfg
         B = tokens.shape[0]
-        e = ext()
         res = self.embed.index_select(0, tokens).contiguous()
         h = None
         mlp_out = None
         for li, L in enumerate(self.layers):
             if li == 0:
-                h = e.rmsnorm(res, L["attn_norm"], c.norm_eps)
+                h = D.rmsnorm(res, L["attn_norm"], c.norm_eps)
             else:
-                h = e.rmsnorm_residual(mlp_out, res, L["attn_norm"], c.norm_eps)
+                h = D.rmsnorm_residual(mlp_out, res, L["attn_norm"],
+                                       c.norm_eps)
             qkv = F.linear(h, L["wqkv"])
             # strided [B, H, D] views straight into the fused qkv buffer —
             # the kernels take row strides, no contiguous() copies
-            qd = c.n_q_heads * c.d_head
-            kd = c.n_kv_heads * c.d_head
-            q = qkv[:, :qd].view(B, c.n_q_heads, c.d_head)
-            k = qkv[:, qd:qd + kd].view(B, c.n_kv_heads, c.d_head)
-            v = qkv[:, qd + kd:].view(B, c.n_kv_heads, c.d_head)
-            e.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
-            e.kv_append(k, v, kv.k[li], kv.v[li], block_table, seq_lens)
-            attn = e.paged_attn_decode(q, kv.k[li], kv.v[li], block_table,
+            qd = self.n_q * c.d_head
+            kd = self.n_kv * c.d_head
+            q = qkv[:, :qd].view(B, self.n_q, c.d_head)
+            k = qkv[:, qd:qd + kd].view(B, self.n_kv, c.d_head)
+            v = qkv[:, qd + kd:].view(B, self.n_kv, c.d_head)
+            D.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
+            D.kv_append(k, v, kv.k[li], kv.v[li], block_table, seq_lens)
+            attn = D.paged_attn_decode(q, kv.k[li], kv.v[li], block_table,
                                        seq_lens, self.scale)
-            o = F.linear(attn.view(B, -1), L["wo"])
-            h = e.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
+            o = self._tp_all_reduce(F.linear(attn.view(B, -1), L["wo"]))
+            h = D.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             gu = F.linear(h, L["wgu"])
-            act = e.swiglu(gu[:, :c.ffn], gu[:, c.ffn:])
-            mlp_out = F.linear(act, L["wdown"])
-        final_h = e.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
+            act = D.swiglu(gu[:, :self.ffn_local], gu[:, self.ffn_local:])
+            mlp_out = self._tp_all_reduce(F.linear(act, L["wdown"]))
+        final_h = D.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
         return F.linear(final_h, self.lm_head)
 
     def _gather_kv(self, kc_l: torch.Tensor, vc_l: torch.Tensor,
@@ -152,10 +203,10 @@ class LlamaModel:
         idx = torch.tensor(pages, dtype=torch.int64, device=self.device)
         kpg = kc_l.index_select(0, idx)        # [np, KVH, D/8, 64, 8]
         kseq = kpg.permute(1, 0, 3, 2, 4).reshape(
-            c.n_kv_heads, len(pages) * 64, c.d_head)[:, :n]
+            self.n_kv, len(pages) * 64, c.d_head)[:, :n]
         vpg = vc_l.index_select(0, idx)        # [np, KVH, 64, D]
         vseq = vpg.permute(1, 0, 2, 3).reshape(
-            c.n_kv_heads, len(pages) * 64, c.d_head)[:, :n]
+            self.n_kv, len(pages) * 64, c.d_head)[:, :n]
         return kseq, vseq
 
     @torch.no_grad()
@@ -170,7 +221,6 @@ class LlamaModel:
         per sequence.  Returns last-position logits [n_items, vocab].
         """
         c = self.cfg
-        e = ext()
         dev = self.device
         lens = [int(t.shape[0]) for t, _, _ in items]
         offs = [0]
@@ -190,14 +240,15 @@ class LlamaModel:
         mlp_out = None
         for li, L in enumerate(self.layers):
             if li == 0:
-                h = e.rmsnorm(res, L["attn_norm"], c.norm_eps)
+                h = D.rmsnorm(res, L["attn_norm"], c.norm_eps)
             else:
-                h = e.rmsnorm_residual(mlp_out, res, L["attn_norm"], c.norm_eps)
+                h = D.rmsnorm_residual(mlp_out, res, L["attn_norm"],
+                                       c.norm_eps)
             qkv = F.linear(h, L["wqkv"])
             q, k, v = self._split_qkv(qkv, T)
-            e.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
-            e.kv_scatter(k, v, kv.k[li], kv.v[li], slots)
-            attn = torch.empty(T, c.n_q_heads * c.d_head, dtype=self.dtype,
+            D.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
+            D.kv_scatter(k, v, kv.k[li], kv.v[li], slots)
+            attn = torch.empty(T, self.n_q * c.d_head, dtype=self.dtype,
                                device=dev)
             for (tt, sid, start), n, off in zip(items, lens, offs):
                 ctx = start + n
@@ -209,26 +260,26 @@ class LlamaModel:
                                                  kv._seq_pages[sid], ctx)
                 # GQA grouped: [KVH, R*n, D] x [KVH, ctx, D]; bf16 MFMA
                 # GEMMs (f32 bmm is 1/16 the MFMA rate on CDNA4)
-                R = c.n_q_heads // c.n_kv_heads
+                R = self.n_q // self.n_kv
                 qf = q[off:off + n].permute(1, 0, 2).reshape(
-                    c.n_kv_heads, R * n, c.d_head)
+                    self.n_kv, R * n, c.d_head)
                 kf = kseq.contiguous()                         # [KVH, ctx, D]
                 vf = vseq.contiguous()
                 scores = (torch.bmm(qf, kf.transpose(1, 2)).float()
                           * self.scale) \
-                    .reshape(c.n_q_heads * n, ctx).contiguous()
-                e.softmax_rows_(scores, start, True, n, None)
-                probs = scores.reshape(c.n_kv_heads, R * n, ctx) \
+                    .reshape(self.n_q * n, ctx).contiguous()
+                D.softmax_rows_(scores, start, True, n, None)
+                probs = scores.reshape(self.n_kv, R * n, ctx) \
                     .to(self.dtype)
                 a = torch.bmm(probs, vf)
-                attn[off:off + n] = a.reshape(c.n_q_heads, n, c.d_head) \
+                attn[off:off + n] = a.reshape(self.n_q, n, c.d_head) \
                     .permute(1, 0, 2).reshape(n, -1)
-            o = F.linear(attn, L["wo"])
-            h = e.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
+            o = self._tp_all_reduce(F.linear(attn, L["wo"]))
+            h = D.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             gu = F.linear(h, L["wgu"])
-            act = e.swiglu(gu[:, :c.ffn], gu[:, c.ffn:])
-            mlp_out = F.linear(act, L["wdown"])
-        final_h = e.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
+            act = D.swiglu(gu[:, :self.ffn_local], gu[:, self.ffn_local:])
+            mlp_out = self._tp_all_reduce(F.linear(act, L["wdown"]))
+        final_h = D.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
         last = torch.tensor([offs[i] + lens[i] - 1 for i in range(len(items))],
                             dtype=torch.int64, device=dev)
         return F.linear(final_h.index_select(0, last), self.lm_head)

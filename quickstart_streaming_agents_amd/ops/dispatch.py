@@ -1,0 +1,146 @@
+"""Device dispatch for the op layer.
+
+Same call surface as the qsa_hip extension (ops/hip/ops.cpp). CUDA tensors
+go to the hand-written HIP/CDNA4 kernels — fail-loud if the extension is
+missing (ops/__init__.py).  CPU tensors run reference implementations with
+IDENTICAL semantics (including in-place mutation contracts), which lets the
+engine / model / parallel layers run under multi-process gloo tests in
+CPU-only CI.  A CUDA tensor NEVER falls back to the CPU path.
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import ext
+
+
+def _cuda(t: torch.Tensor) -> bool:
+    return t.is_cuda
+
+
+# ---- elementwise ----------------------------------------------------------
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+    if _cuda(x):
+        return ext().rmsnorm(x, w, eps)
+    xf = x.float()
+    y = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps) * w.float()
+    return y.to(x.dtype)
+
+
+def rmsnorm_residual(x: torch.Tensor, res: torch.Tensor, w: torch.Tensor,
+                     eps: float) -> torch.Tensor:
+    """res <- res + x (in place); returns rmsnorm(res) * w."""
+    if _cuda(x):
+        return ext().rmsnorm_residual(x, res, w, eps)
+    res.add_(x)
+    rf = res.float()
+    y = rf * torch.rsqrt(rf.pow(2).mean(-1, keepdim=True) + eps) * w.float()
+    return y.to(x.dtype)
+
+
+def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    if _cuda(gate):
+        return ext().swiglu(gate, up)
+    return (torch.nn.functional.silu(gate.float()) * up.float()).to(gate.dtype)
+
+
+def rope_inplace(q: torch.Tensor, k: torch.Tensor, cos_t: torch.Tensor,
+                 sin_t: torch.Tensor, pos: torch.Tensor) -> None:
+    if _cuda(q):
+        ext().rope_inplace(q, k, cos_t, sin_t, pos)
+        return
+    for t in (q, k):
+        D = t.shape[-1]
+        half = D // 2
+        c = cos_t[pos.long()].unsqueeze(1)
+        s = sin_t[pos.long()].unsqueeze(1)
+        tf = t.float()
+        x0, x1 = tf[..., :half], tf[..., half:]
+        t.copy_(torch.cat([x0 * c - x1 * s, x0 * s + x1 * c],
+                          dim=-1).to(t.dtype))
+
+
+def softmax_rows_(scores: torch.Tensor, col_offset: int = 0,
+                  causal: bool = False, row_mod: int = 0,
+                  row_limits: torch.Tensor | None = None) -> None:
+    if _cuda(scores):
+        ext().softmax_rows_(scores, col_offset, causal, row_mod, row_limits)
+        return
+    rows, cols = scores.shape
+    limit = torch.full((rows,), cols, dtype=torch.int64)
+    if causal:
+        r = torch.arange(rows)
+        pos = r % row_mod if row_mod > 0 else r
+        limit = torch.minimum(limit, pos + col_offset + 1)
+    if row_limits is not None:
+        limit = torch.minimum(limit, row_limits.long())
+    col = torch.arange(cols).unsqueeze(0)
+    masked = scores.masked_fill(col >= limit.unsqueeze(1), float("-inf"))
+    out = torch.softmax(masked, dim=-1)
+    scores.copy_(torch.nan_to_num(out, nan=0.0))
+
+
+# ---- paged KV -------------------------------------------------------------
+
+def kv_append(knew: torch.Tensor, vnew: torch.Tensor, kc: torch.Tensor,
+              vc: torch.Tensor, block_table: torch.Tensor,
+              seq_lens: torch.Tensor) -> None:
+    """Write row b's k/v at position seq_lens[b]-1 of its paged sequence."""
+    if _cuda(knew):
+        ext().kv_append(knew, vnew, kc, vc, block_table, seq_lens)
+        return
+    B, KVH, D = knew.shape
+    for b in range(B):
+        n = int(seq_lens[b])
+        if n <= 0:
+            continue
+        page = int(block_table[b, (n - 1) // 64])
+        off = (n - 1) % 64
+        # K layout [P, KVH, D/8, 64, 8]
+        kc[page, :, :, off, :] = knew[b].reshape(KVH, D // 8, 8)
+        vc[page, :, off, :] = vnew[b]
+
+
+def kv_scatter(knew: torch.Tensor, vnew: torch.Tensor, kc: torch.Tensor,
+               vc: torch.Tensor, slots: torch.Tensor) -> None:
+    """Write row t at global slot ids (page*64 + offset)."""
+    if _cuda(knew):
+        ext().kv_scatter(knew, vnew, kc, vc, slots)
+        return
+    T, KVH, D = knew.shape
+    for t in range(T):
+        s = int(slots[t])
+        page, off = s // 64, s % 64
+        kc[page, :, :, off, :] = knew[t].reshape(KVH, D // 8, 8)
+        vc[page, :, off, :] = vnew[t]
+
+
+def paged_attn_decode(q: torch.Tensor, kc: torch.Tensor, vc: torch.Tensor,
+                      block_table: torch.Tensor, seq_lens: torch.Tensor,
+                      scale: float) -> torch.Tensor:
+    if _cuda(q):
+        return ext().paged_attn_decode(q, kc, vc, block_table, seq_lens,
+                                       scale)
+    from .cpu_ref import paged_attn_ref
+    return paged_attn_ref(q, kc, vc, block_table, seq_lens,
+                          scale).to(q.dtype)
+
+
+# ---- retrieval / streaming ------------------------------------------------
+
+def topk_cosine(queries: torch.Tensor, docs: torch.Tensor, k: int):
+    if _cuda(queries):
+        return ext().topk_cosine(queries, docs, k)
+    from .cpu_ref import topk_cosine_ref
+    return topk_cosine_ref(queries, docs, k)
+
+
+def window_agg(ts: torch.Tensor, key: torch.Tensor,
+               value: torch.Tensor | None, t0: int, win_ms: int, nwin: int,
+               nkeys: int):
+    if _cuda(ts):
+        return ext().window_agg(ts, key, value, t0, win_ms, nwin, nkeys)
+    from .cpu_ref import window_agg_ref
+    return window_agg_ref(ts, key, value, t0, win_ms, nwin, nkeys)
