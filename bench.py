@@ -101,10 +101,9 @@ def main():
     if args.stub_llm:
         llm = pipelines.StubLLM()
     else:
-        from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
-                                                                  LlamaModel)
+        from quickstart_streaming_agents_amd.models import build_model
         from quickstart_streaming_agents_amd.models.serve import Engine, EngineLLM
-        model = LlamaModel(LlamaConfig.preset(args.model), device=device)
+        model = build_model(args.model, device=device)
         engine = Engine(model, max_batch=args.batch,
                         max_seq_len=args.max_seq_len)
         llm = EngineLLM(engine)

@@ -118,17 +118,13 @@ class LlamaModel:
             wqkv = torch.cat([rows(wq_full, tp_size), rows(wk_full, tp_size),
                               rows(wv_full, tp_size)], dim=0).contiguous()
             wo = cols(w(c.hidden, c.n_q_heads * c.d_head), tp_size)
-            wgu_full = w(2 * c.ffn, c.hidden)
-            wgu = torch.cat([rows(wgu_full[:c.ffn], tp_size),
-                             rows(wgu_full[c.ffn:], tp_size)],
-                            dim=0).contiguous()
-            wdown = cols(w(c.hidden, c.ffn, std=0.02 / (2 * c.n_layers) ** 0.5),
-                         tp_size)
-            self.layers.append({
+            layer = {
                 "attn_norm": torch.ones(c.hidden, device=device, dtype=dtype),
                 "mlp_norm": torch.ones(c.hidden, device=device, dtype=dtype),
-                "wqkv": wqkv, "wo": wo, "wgu": wgu, "wdown": wdown,
-            })
+                "wqkv": wqkv, "wo": wo,
+            }
+            layer.update(self._ffn_weights(w, rows, cols))
+            self.layers.append(layer)
         # rope tables
         half = c.d_head // 2
         inv = 1.0 / (c.rope_theta **
@@ -137,6 +133,19 @@ class LlamaModel:
         self.rope_cos = ang.cos().float().to(device)
         self.rope_sin = ang.sin().float().to(device)
         self.scale = 1.0 / (c.d_head ** 0.5)
+
+    def _ffn_weights(self, w, rows, cols) -> dict:
+        """Per-layer FFN weights; the generator order is part of the
+        deterministic-init contract (same seed => same full tensors on
+        every rank, sliced per shard)."""
+        c = self.cfg
+        wgu_full = w(2 * c.ffn, c.hidden)
+        wgu = torch.cat([rows(wgu_full[:c.ffn], self.tp_size),
+                         rows(wgu_full[c.ffn:], self.tp_size)],
+                        dim=0).contiguous()
+        wdown = cols(w(c.hidden, c.ffn, std=0.02 / (2 * c.n_layers) ** 0.5),
+                     self.tp_size)
+        return {"wgu": wgu, "wdown": wdown}
 
     def new_kv_cache(self, n_pages: int) -> PagedKVCache:
         return PagedKVCache(self.cfg.n_layers, self.n_kv, self.cfg.d_head,
@@ -189,11 +198,16 @@ class LlamaModel:
                                        seq_lens, self.scale)
             o = self._tp_all_reduce(F.linear(attn.view(B, -1), L["wo"]))
             h = D.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
-            gu = F.linear(h, L["wgu"])
-            act = D.swiglu(gu[:, :self.ffn_local], gu[:, self.ffn_local:])
-            mlp_out = self._tp_all_reduce(F.linear(act, L["wdown"]))
+            mlp_out = self._ffn(L, h)
         final_h = D.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
         return F.linear(final_h, self.lm_head)
+
+    def _ffn(self, L: dict, h: torch.Tensor) -> torch.Tensor:
+        """Dense SwiGLU FFN (TP row/col-parallel). Mixtral overrides with
+        the routed MoE (models/mixtral.py)."""
+        gu = F.linear(h, L["wgu"])
+        act = D.swiglu(gu[:, :self.ffn_local], gu[:, self.ffn_local:])
+        return self._tp_all_reduce(F.linear(act, L["wdown"]))
 
     def _gather_kv(self, kc_l: torch.Tensor, vc_l: torch.Tensor,
                    pages: list[int], n: int):
@@ -276,9 +290,7 @@ class LlamaModel:
                     .permute(1, 0, 2).reshape(n, -1)
             o = self._tp_all_reduce(F.linear(attn, L["wo"]))
             h = D.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
-            gu = F.linear(h, L["wgu"])
-            act = D.swiglu(gu[:, :self.ffn_local], gu[:, self.ffn_local:])
-            mlp_out = self._tp_all_reduce(F.linear(act, L["wdown"]))
+            mlp_out = self._ffn(L, h)
         final_h = D.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
         last = torch.tensor([offs[i] + lens[i] - 1 for i in range(len(items))],
                             dtype=torch.int64, device=dev)
