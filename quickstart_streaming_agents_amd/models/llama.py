@@ -133,6 +133,17 @@ class LlamaModel:
         self.rope_cos = ang.cos().float().to(device)
         self.rope_sin = ang.sin().float().to(device)
         self.scale = 1.0 / (c.d_head ** 0.5)
+        # decode-path weight packing: fragment-major copies for the
+        # weight-streaming skinny GEMM (288 GB HBM affords the duplicate)
+        self.lm_head_f = None
+        if str(device).startswith("cuda") and dtype == torch.bfloat16:
+            for L in self.layers:
+                for key in ("wqkv", "wo", "wgu", "wdown"):
+                    wt = L.get(key)
+                    if wt is not None and D.can_pack_weight(*wt.shape):
+                        L[key + "_f"] = D.pack_weight_frag(wt)
+            if D.can_pack_weight(*self.lm_head.shape):
+                self.lm_head_f = D.pack_weight_frag(self.lm_head)
 
     def _ffn_weights(self, w, rows, cols) -> dict:
         """Per-layer FFN weights; the generator order is part of the
@@ -150,6 +161,14 @@ class LlamaModel:
     def new_kv_cache(self, n_pages: int) -> PagedKVCache:
         return PagedKVCache(self.cfg.n_layers, self.n_kv, self.cfg.d_head,
                             n_pages, self.device, self.dtype)
+
+    def _linear(self, x: torch.Tensor, w: torch.Tensor,
+                wf: torch.Tensor | None = None) -> torch.Tensor:
+        """Projection: decode-sized cuda batches go through the
+        weight-streaming skinny-GEMM kernel, everything else rocBLAS."""
+        if wf is not None and x.is_cuda and x.shape[0] <= 32:
+            return D.skinny_linear(x, wf, w.shape[0], w.shape[1])
+        return F.linear(x, w)
 
     def _tp_all_reduce(self, t: torch.Tensor) -> torch.Tensor:
         if self.tp_size > 1:
@@ -184,7 +203,7 @@ class LlamaModel:
             else:
                 h = D.rmsnorm_residual(mlp_out, res, L["attn_norm"],
                                        c.norm_eps)
-            qkv = F.linear(h, L["wqkv"])
+            qkv = self._linear(h, L["wqkv"], L.get("wqkv_f"))
             # strided [B, H, D] views straight into the fused qkv buffer —
             # the kernels take row strides, no contiguous() copies
             qd = self.n_q * c.d_head
@@ -196,7 +215,8 @@ class LlamaModel:
             D.kv_append(k, v, kv.k[li], kv.v[li], block_table, seq_lens)
             attn = D.paged_attn_decode(q, kv.k[li], kv.v[li], block_table,
                                        seq_lens, self.scale)
-            o = self._tp_all_reduce(F.linear(attn.view(B, -1), L["wo"]))
+            o = self._tp_all_reduce(
+                self._linear(attn.view(B, -1), L["wo"], L.get("wo_f")))
             h = D.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             mlp_out = self._ffn(L, h)
         final_h = D.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
@@ -205,9 +225,10 @@ class LlamaModel:
     def _ffn(self, L: dict, h: torch.Tensor) -> torch.Tensor:
         """Dense SwiGLU FFN (TP row/col-parallel). Mixtral overrides with
         the routed MoE (models/mixtral.py)."""
-        gu = F.linear(h, L["wgu"])
+        gu = self._linear(h, L["wgu"], L.get("wgu_f"))
         act = D.swiglu(gu[:, :self.ffn_local], gu[:, self.ffn_local:])
-        return self._tp_all_reduce(F.linear(act, L["wdown"]))
+        return self._tp_all_reduce(
+            self._linear(act, L["wdown"], L.get("wdown_f")))
 
     def _gather_kv(self, kc_l: torch.Tensor, vc_l: torch.Tensor,
                    pages: list[int], n: int):
@@ -294,7 +315,8 @@ class LlamaModel:
         final_h = D.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
         last = torch.tensor([offs[i] + lens[i] - 1 for i in range(len(items))],
                             dtype=torch.int64, device=dev)
-        return F.linear(final_h.index_select(0, last), self.lm_head)
+        return self._linear(final_h.index_select(0, last), self.lm_head,
+                            self.lm_head_f)
 
     @torch.no_grad()
     def forward_prefill(self, tokens: torch.Tensor, kv: PagedKVCache,

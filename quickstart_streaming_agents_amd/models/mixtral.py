@@ -98,6 +98,20 @@ class MixtralModel(LlamaModel):
             return F.linear(act, wdown)
 
         if self.ep_size == 1:
+            if h.shape[0] <= 64 and h.is_cuda:
+                # decode-sized batch: dense compute-all-experts.  Top-2/8 on
+                # 24+ tokens touches every expert's weights anyway (the MoE
+                # FFN is weight-bandwidth-bound, FLOPs are incidental), and
+                # this path is hipGraph-capture-safe: no .nonzero()/host
+                # sync, routing becomes a masked weighted sum.
+                gates = torch.zeros_like(probs)  # [T, E]
+                gates.scatter_(1, top_idx, top_w)
+                out = torch.zeros(h.shape, dtype=torch.float32,
+                                  device=h.device)
+                for e in range(c.n_experts):
+                    y = expert_fn(e, h)
+                    out += gates[:, e:e + 1] * y.float()
+                return out.to(h.dtype)
             return self._dispatch.run(h, top_idx, top_w, expert_fn)
         # TP group == EP group: h is identical on all ranks after the
         # attention all-reduce, so each rank dispatches only its token

@@ -128,6 +128,30 @@ def paged_attn_decode(q: torch.Tensor, kc: torch.Tensor, vc: torch.Tensor,
                           scale).to(q.dtype)
 
 
+# ---- skinny decode GEMM ---------------------------------------------------
+
+def can_pack_weight(n: int, k: int) -> bool:
+    return n % 64 == 0 and k % 256 == 0
+
+
+def pack_weight_frag(w: torch.Tensor) -> torch.Tensor:
+    """[N,K] bf16 -> fragment-major stream layout for skinny_linear (GPU)."""
+    if w.is_cuda:
+        return ext().pack_weight_frag(w)
+    n, k = w.shape
+    return w.reshape(n // 16, 16, k // 32, 32).permute(0, 2, 1, 3).contiguous()
+
+
+def skinny_linear(x: torch.Tensor, wf: torch.Tensor, n: int,
+                  k: int) -> torch.Tensor:
+    """x[M,K] @ W^T via the weight-streaming decode kernel (M <= 32)."""
+    if x.is_cuda:
+        return ext().skinny_gemm(x, wf, n, k)
+    w = wf.reshape(n // 16, k // 32, 16, 32).permute(0, 2, 1, 3) \
+        .reshape(n, k)
+    return (x.float() @ w.float().T).to(x.dtype)
+
+
 # ---- retrieval / streaming ------------------------------------------------
 
 def topk_cosine(queries: torch.Tensor, docs: torch.Tensor, k: int):

@@ -39,6 +39,10 @@ extern "C" void qsa_paged_attn_decode_launch(const unsigned short*,
                                              const int*, unsigned short*, float,
                                              int, int, int, int, int,
                                              long long, hipStream_t);
+extern "C" void qsa_paged_attn_decode_split_launch(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const int*, const int*, float*, float*, unsigned short*, float, int, int,
+    int, int, int, long long, int, hipStream_t);
 extern "C" void qsa_kv_append_launch(const unsigned short*,
                                      const unsigned short*, unsigned short*,
                                      unsigned short*, const int*, const int*,
@@ -48,6 +52,12 @@ extern "C" void qsa_kv_scatter_launch(const unsigned short*,
                                       const unsigned short*, unsigned short*,
                                       unsigned short*, const int*, int, int,
                                       int, hipStream_t);
+extern "C" void qsa_skinny_gemm_launch(const unsigned short*,
+                                       const unsigned short*, unsigned short*,
+                                       float*, int, int, long long, long long,
+                                       int, hipStream_t);
+extern "C" void qsa_f32_to_bf16_launch(const float*, unsigned short*,
+                                       long long, hipStream_t);
 extern "C" void qsa_topk_launch(const float*, const float*, float*, int*,
                                 float*, int*, int, int, int, int, int,
                                 hipStream_t);
@@ -155,11 +165,26 @@ torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor kc,
               "V cache layout [P, KVH, 64, D]");
   const int max_pages = block_table.size(1);
   auto out = torch::empty({B, QH, D}, q.options());
-  qsa_paged_attn_decode_launch(u16(q), u16(kc), u16(vc),
-                               block_table.data_ptr<int>(),
-                               seq_lens.data_ptr<int>(), u16m(out),
-                               (float)scale, B, QH, KVH, max_pages, D,
-                               q.stride(0), cur_stream());
+  // flash-decoding context split: fill the 256-CU chip (>=~640 WGs) when
+  // B*KVH alone underfills it; NS=1 keeps the single-pass kernel.
+  int ns = (int)std::min<long long>(16, std::max<long long>(
+      1, (640 + (long long)B * KVH - 1) / ((long long)B * KVH)));
+  if (ns > 1) {
+    auto opts_f = q.options().dtype(at::kFloat);
+    auto part_o = torch::empty({B, QH, ns, D}, opts_f);
+    auto part_ml = torch::empty({B, QH, ns, 2}, opts_f);
+    qsa_paged_attn_decode_split_launch(
+        u16(q), u16(kc), u16(vc), block_table.data_ptr<int>(),
+        seq_lens.data_ptr<int>(), part_o.data_ptr<float>(),
+        part_ml.data_ptr<float>(), u16m(out), (float)scale, B, QH, KVH,
+        max_pages, D, q.stride(0), ns, cur_stream());
+  } else {
+    qsa_paged_attn_decode_launch(u16(q), u16(kc), u16(vc),
+                                 block_table.data_ptr<int>(),
+                                 seq_lens.data_ptr<int>(), u16m(out),
+                                 (float)scale, B, QH, KVH, max_pages, D,
+                                 q.stride(0), cur_stream());
+  }
   return out;
 }
 
@@ -184,6 +209,45 @@ void kv_scatter(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kc,
   if (T == 0) return;
   qsa_kv_scatter_launch(u16(knew), u16(vnew), u16m(kc), u16m(vc),
                         slots.data_ptr<int>(), T, KVH, D, cur_stream());
+}
+
+torch::Tensor pack_weight_frag(torch::Tensor w) {
+  // [N, K] bf16 -> MFMA-fragment-major [N/16, K/32, 16, 32] for the
+  // skinny-GEMM weight stream (one wave reads one contiguous 1 KiB block).
+  CHK_BF16(w); CHK_CONT(w);
+  const long long N = w.size(0), K = w.size(1);
+  TORCH_CHECK(N % 16 == 0 && K % 32 == 0, "pack needs N%16==0, K%32==0");
+  return w.reshape({N / 16, 16, K / 32, 32})
+      .permute({0, 2, 1, 3}).contiguous();
+}
+
+torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor wf, long N, long K) {
+  // C[M,N] = a[M,K] @ W^T with W pre-packed by pack_weight_frag.
+  CHK_DEV(a); CHK_BF16(a); CHK_BF16(wf); CHK_CONT(wf);
+  TORCH_CHECK(a.dim() == 2 && a.stride(1) == 1, "a rows must be contiguous");
+  const int M = a.size(0);
+  TORCH_CHECK(M >= 1 && M <= 32, "skinny_gemm: M in [1,32]");
+  TORCH_CHECK(a.size(1) == K, "K mismatch");
+  TORCH_CHECK(K % 256 == 0 && N % 64 == 0, "K%256==0, N%64==0");
+  TORCH_CHECK(wf.numel() == (long long)N * K, "wf size");
+  auto out = torch::empty({M, (long long)N}, a.options());
+  const long long nblk = N / 64;
+  int splitk = 1;
+  if (nblk < 256)
+    splitk = (int)std::min<long long>(
+        {8, K / 256, (512 + nblk - 1) / nblk});
+  if (splitk > 1) {
+    auto cf = torch::zeros({M, (long long)N},
+                           a.options().dtype(at::kFloat));
+    qsa_skinny_gemm_launch(u16(a), u16(wf), nullptr, cf.data_ptr<float>(), M,
+                           (int)N, K, a.stride(0), splitk, cur_stream());
+    qsa_f32_to_bf16_launch(cf.data_ptr<float>(), u16m(out),
+                           (long long)M * N, cur_stream());
+  } else {
+    qsa_skinny_gemm_launch(u16(a), u16(wf), u16m(out), nullptr, M, (int)N, K,
+                           a.stride(0), 1, cur_stream());
+  }
+  return out;
 }
 
 std::vector<torch::Tensor> topk_cosine(torch::Tensor queries,
@@ -257,6 +321,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "paged-attention decode (bf16, GQA, page=64)");
   m.def("kv_append", &kv_append, "append one step's k/v to the paged cache");
   m.def("kv_scatter", &kv_scatter, "scatter prefill k/v by slot ids");
+  m.def("pack_weight_frag", &pack_weight_frag,
+        "repack [N,K] bf16 into MFMA-fragment-major for skinny_gemm");
+  m.def("skinny_gemm", &skinny_gemm,
+        "decode-batch GEMM (M<=32) on the packed weight stream");
   m.def("topk_cosine", &topk_cosine, "exact cosine top-k over the HBM index");
   m.def("window_agg", &window_agg, "segmented (key, window) count/sum");
   m.def("anomaly_batch", &anomaly_batch, "batched AR+ridge anomaly scorer");

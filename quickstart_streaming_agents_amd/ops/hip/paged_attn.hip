@@ -137,6 +137,151 @@ qsa_paged_attn_decode(const unsigned short* __restrict__ q,   // [B, QH, D] (row
   }
 }
 
+// ---------------------------------------------------------------------------
+// Flash-decoding context split: B*KVH workgroups underfill the 256-CU chip
+// (e.g. B=24, KVH=8 -> 192 WGs, 4 waves each = terrible occupancy and the
+// kernel runs latency-bound at ~8% of the HBM roofline).  Split the context
+// into NS chunks -> grid B*KVH*NS (>=512 WGs, 2-4 blocks/CU), each workgroup
+// produces an UNNORMALIZED partial (o, m, l) over its page range, and a tiny
+// reduce kernel merges the NS partials per (b, head) row:
+//   M = max m_s;  O = sum_s exp(m_s-M) o_s;  L = sum_s exp(m_s-M) l_s;
+//   out = O / L.
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ void __launch_bounds__(256)
+qsa_paged_attn_decode_split(const unsigned short* __restrict__ q,
+                            const unsigned short* __restrict__ kc,
+                            const unsigned short* __restrict__ vc,
+                            const int* __restrict__ block_table,
+                            const int* __restrict__ seq_lens,
+                            float* __restrict__ part_o,    // [B, QH, NS, D]
+                            float* __restrict__ part_ml,   // [B, QH, NS, 2]
+                            float scale, int B, int QH, int KVH, int max_pages,
+                            long long qstride, int NS) {
+  const int split = blockIdx.x % NS;
+  const int bk = blockIdx.x / NS;
+  const int b = bk / KVH;
+  const int kvh = bk % KVH;
+  const int wave = threadIdx.x / QSA_WAVE;
+  const int lane = threadIdx.x % QSA_WAVE;
+  const int R = QH / KVH;
+  const int seqlen = seq_lens[b];
+  const int npages = (seqlen + QSA_PAGE - 1) / QSA_PAGE;
+  const int chunk = (npages + NS - 1) / NS;
+  const int p0 = split * chunk;
+  const int p1 = min(npages, p0 + chunk);
+  const int* btab = block_table + (long long)b * max_pages;
+
+  for (int r = wave; r < R; r += 4) {
+    const int qh = kvh * R + r;
+    float* pml = part_ml + (((long long)b * QH + qh) * NS + split) * 2;
+    float* po = part_o + (((long long)b * QH + qh) * NS + split) * D;
+    if (seqlen <= 0 || p0 >= npages) {
+      if (lane == 0) { pml[0] = -3.0e38f; pml[1] = 0.f; }
+      continue;
+    }
+    unsigned int qpk[D / 2];
+    const unsigned int* qsrc = reinterpret_cast<const unsigned int*>(
+        q + (long long)b * qstride + (long long)qh * D);
+#pragma unroll
+    for (int i = 0; i < D / 2; ++i) qpk[i] = qsrc[i];
+
+    constexpr int DG = D / 8;
+    constexpr int PG = QSA_WAVE / DG;
+    const int dg = lane % DG;
+    const int pg = lane / DG;
+
+    float m = -3.0e38f, s = 0.f;
+    float o8[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o8[j] = 0.f;
+
+    for (int pi = p0; pi < p1; ++pi) {
+      const int page = btab[pi];
+      const int pos = pi * QSA_PAGE + lane;
+      const bool valid = pos < seqlen;
+      float sc = 0.f;
+      const uint4* kbase = reinterpret_cast<const uint4*>(
+          kc + ((((long long)page * KVH + kvh) * (D / 8)) * QSA_PAGE) * 8);
+#pragma unroll
+      for (int d0 = 0; d0 < D / 8; ++d0) {
+        uint4 kv4 = kbase[d0 * QSA_PAGE + lane];
+        unsigned int kk[4] = {kv4.x, kv4.y, kv4.z, kv4.w};
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float2 kf = bf16x2_to_f32x2(kk[j]);
+          float2 qf = bf16x2_to_f32x2(qpk[d0 * 4 + j]);
+          sc = fmaf(kf.x, qf.x, sc);
+          sc = fmaf(kf.y, qf.y, sc);
+        }
+      }
+      sc = valid ? sc * scale : -3.0e38f;
+      const float pmax = wave_reduce_max(sc);
+      const float m_new = fmaxf(m, pmax);
+      float alpha = __expf(m - m_new);
+      if (m <= -3.0e38f) alpha = 0.f;
+      m = m_new;
+      const float p = valid ? __expf(sc - m_new) : 0.f;
+      s = s * alpha + wave_reduce_sum(p);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o8[j] *= alpha;
+      const uint4* vbase = reinterpret_cast<const uint4*>(
+          vc + (((long long)page * KVH + kvh) * QSA_PAGE) * D);
+      const int nvalid = min(seqlen - pi * QSA_PAGE, QSA_PAGE);
+      const int niter = (nvalid + PG - 1) / PG;
+#pragma unroll 4
+      for (int it = 0; it < niter; ++it) {
+        const int t = it * PG + pg;
+        const float pt = __shfl(p, t, QSA_WAVE);
+        const uint4 v4 = vbase[(long long)t * DG + dg];
+        unsigned int vv[4] = {v4.x, v4.y, v4.z, v4.w};
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float2 vf = bf16x2_to_f32x2(vv[j]);
+          o8[2 * j] = fmaf(pt, vf.x, o8[2 * j]);
+          o8[2 * j + 1] = fmaf(pt, vf.y, o8[2 * j + 1]);
+        }
+      }
+    }
+#pragma unroll
+    for (int off = DG; off < QSA_WAVE; off <<= 1) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o8[j] += __shfl_xor(o8[j], off, QSA_WAVE);
+    }
+    if (pg == 0) {
+      // unnormalized partial in f32 (reduce kernel applies 1/L)
+      float4* prow = reinterpret_cast<float4*>(po);
+      prow[dg * 2] = make_float4(o8[0], o8[1], o8[2], o8[3]);
+      prow[dg * 2 + 1] = make_float4(o8[4], o8[5], o8[6], o8[7]);
+    }
+    if (lane == 0) { pml[0] = m; pml[1] = s; }
+  }
+}
+
+template <int D>
+__global__ void __launch_bounds__(128)
+qsa_attn_reduce(const float* __restrict__ part_o,   // [rows, NS, D]
+                const float* __restrict__ part_ml,  // [rows, NS, 2]
+                unsigned short* __restrict__ out,   // [rows, D]
+                int NS) {
+  const long long row = blockIdx.x;
+  const int d = threadIdx.x;  // D threads
+  const float* ml = part_ml + row * NS * 2;
+  float M = -3.0e38f;
+  for (int s = 0; s < NS; ++s)
+    if (ml[2 * s + 1] > 0.f) M = fmaxf(M, ml[2 * s]);
+  float L = 0.f, acc = 0.f;
+  for (int s = 0; s < NS; ++s) {
+    const float ls = ml[2 * s + 1];
+    if (ls <= 0.f) continue;
+    const float w = __expf(ml[2 * s] - M);
+    L += w * ls;
+    acc += w * part_o[(row * NS + s) * D + d];
+  }
+  const float v = (L > 0.f) ? acc / L : 0.f;
+  out[row * D + d] = f32_to_bf16(v);
+}
+
 extern "C" void qsa_paged_attn_decode_launch(
     const unsigned short* q, const unsigned short* kc,
     const unsigned short* vc, const int* block_table, const int* seq_lens,
@@ -152,6 +297,29 @@ extern "C" void qsa_paged_attn_decode_launch(
     hipLaunchKernelGGL((qsa_paged_attn_decode<64>), grid, block, 0, stream,
                        q, kc, vc, block_table, seq_lens, out, scale, B, QH,
                        KVH, max_pages, qstride);
+  }
+}
+
+extern "C" void qsa_paged_attn_decode_split_launch(
+    const unsigned short* q, const unsigned short* kc,
+    const unsigned short* vc, const int* block_table, const int* seq_lens,
+    float* part_o, float* part_ml, unsigned short* out, float scale, int B,
+    int QH, int KVH, int max_pages, int D, long long qstride, int NS,
+    hipStream_t stream) {
+  dim3 grid(B * KVH * NS);
+  dim3 block(256);
+  if (D == 128) {
+    hipLaunchKernelGGL((qsa_paged_attn_decode_split<128>), grid, block, 0,
+                       stream, q, kc, vc, block_table, seq_lens, part_o,
+                       part_ml, scale, B, QH, KVH, max_pages, qstride, NS);
+    hipLaunchKernelGGL((qsa_attn_reduce<128>), dim3(B * QH), dim3(128), 0,
+                       stream, part_o, part_ml, out, NS);
+  } else if (D == 64) {
+    hipLaunchKernelGGL((qsa_paged_attn_decode_split<64>), grid, block, 0,
+                       stream, q, kc, vc, block_table, seq_lens, part_o,
+                       part_ml, scale, B, QH, KVH, max_pages, qstride, NS);
+    hipLaunchKernelGGL((qsa_attn_reduce<64>), dim3(B * QH), dim3(64), 0,
+                       stream, part_o, part_ml, out, NS);
   }
 }
 

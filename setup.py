@@ -22,6 +22,7 @@ sources = [
     os.path.join(HIP_DIR, "ops.cpp"),
     os.path.join(HIP_DIR, "elementwise.hip"),
     os.path.join(HIP_DIR, "paged_attn.hip"),
+    os.path.join(HIP_DIR, "skinny_gemm.hip"),
     os.path.join(HIP_DIR, "topk_cosine.hip"),
     os.path.join(HIP_DIR, "streaming.hip"),
 ]

@@ -185,3 +185,35 @@ def test_anomaly_batch_matches_cpu(ext):
         assert dof[kk].item() == dof_ref
         assert abs(fc[kk].item() - f_ref) < max(3e-3 * abs(f_ref), 1e-2)
         assert abs(se[kk].item() - se_ref) < max(0.05 * se_ref, 1e-2)
+
+
+def test_skinny_gemm_matches_linear():
+    """Weight-streaming decode GEMM vs fp32 reference: both split-K (small
+    N) and single-pass (large N) paths, ragged M."""
+    import torch.nn.functional as F
+    from quickstart_streaming_agents_amd.ops import ext
+    torch.manual_seed(3)
+    for M, N, K in [(24, 4096, 4096), (1, 512, 512), (32, 1024, 256),
+                    (17, 28672, 512)]:
+        a = torch.randn(M, K, device="cuda:0", dtype=torch.bfloat16) * 0.5
+        w = torch.randn(N, K, device="cuda:0", dtype=torch.bfloat16) * 0.02
+        wf = ext().pack_weight_frag(w)
+        out = ext().skinny_gemm(a, wf, N, K)
+        ref = a.float() @ w.float().T
+        err = (out.float() - ref).abs().max().item()
+        scale = ref.abs().max().item() + 1e-6
+        assert err / scale < 2e-2, f"M{M} N{N} K{K}: rel err {err/scale}"
+
+
+def test_skinny_gemm_strided_rows():
+    """Row-strided activations (views into a larger buffer) work."""
+    from quickstart_streaming_agents_amd.ops import ext
+    torch.manual_seed(4)
+    buf = torch.randn(8, 1024, device="cuda:0", dtype=torch.bfloat16)
+    a = buf[:, :512]
+    w = torch.randn(256, 512, device="cuda:0", dtype=torch.bfloat16) * 0.05
+    wf = ext().pack_weight_frag(w)
+    out = ext().skinny_gemm(a.contiguous(), wf, 256, 512)
+    ref = a.float() @ w.float().T
+    assert (out.float() - ref).abs().max().item() < 2e-2 * (
+        ref.abs().max().item() + 1e-6)
