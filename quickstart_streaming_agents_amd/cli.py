@@ -1,0 +1,140 @@
+"""CLI entry points — the reference's `uv run deploy|destroy|*_datagen|tests`
+surface (pyproject.toml:90-165) on the local MI355X engine.
+
+    python -m quickstart_streaming_agents_amd deploy  [--labs 1,3] [--dir out]
+    python -m quickstart_streaming_agents_amd run     --lab 1 [--device cuda]
+    python -m quickstart_streaming_agents_amd datagen --lab 3
+    python -m quickstart_streaming_agents_amd validate
+    python -m quickstart_streaming_agents_amd destroy --dir out
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+
+
+def _labs(arg: str):
+    return tuple(int(x) for x in arg.split(",") if x.strip())
+
+
+def cmd_deploy(args) -> int:
+    from .labs.deploy import Deployment
+    dep = Deployment(labs=_labs(args.labs), device=args.device,
+                     model=args.model)
+    dep.write_summaries(args.dir)
+    print(dep.summary_markdown())
+    print(f"[deploy] summaries written to {args.dir}/")
+    return 0
+
+
+def cmd_run(args) -> int:
+    from .agents.mcp import StubMcpServer
+    from .labs.deploy import Deployment
+    dep = Deployment(labs=(args.lab,), device=args.device, model=args.model)
+    dep.datagen(args.lab)
+    server = StubMcpServer().start()
+    try:
+        rows = dep.run(args.lab, mcp_server=server)
+    finally:
+        server.stop()
+    for r in rows[: args.max_print]:
+        print(json.dumps(r, default=str)[:400])
+    print(f"[run] lab{args.lab}: {len(rows)} output records")
+    return 0 if rows else 1
+
+
+def cmd_datagen(args) -> int:
+    from .labs.deploy import Deployment
+    dep = Deployment(labs=(args.lab,), device="cpu")
+    dep.datagen(args.lab)
+    for name, t in sorted(dep.broker.topics.items()):
+        n = t.message_count()
+        if n:
+            print(f"[datagen] {name}: {n} records")
+    return 0
+
+
+def cmd_validate(args) -> int:
+    """Advisory checks (scripts/common/validate.py parity): embedding dims
+    contract, vector index health, MCP stub round trip, HIP extension."""
+    ok = True
+    from .vector.index import EMBED_DIM, HashingEmbedder, VectorIndex
+    emb = HashingEmbedder()
+    v = emb.embed("validate")
+    print(f"[validate] embedding dims = {len(v)} "
+          f"({'OK' if len(v) == EMBED_DIM else 'FAIL: expected 1536'})")
+    ok &= len(v) == EMBED_DIM
+    idx = VectorIndex()
+    idx.add("d1", "hello world", v)
+    hits = idx.search(v, 1)
+    print(f"[validate] vector index cosine self-hit = {hits[0].score:.3f} "
+          f"({'OK' if hits[0].score > 0.99 else 'FAIL'})")
+    ok &= hits[0].score > 0.99
+    from .agents.mcp import McpClient, StubMcpServer
+    srv = StubMcpServer().start()
+    try:
+        tools = McpClient(srv.mcp_endpoint).tools_list()
+        names = sorted(t["name"] for t in tools)
+        print(f"[validate] MCP tools/list -> {names}")
+        ok &= "http_get" in names
+    finally:
+        srv.stop()
+    from .ops import have_ext
+    print(f"[validate] qsa_hip extension importable: {have_ext()}")
+    import torch
+    print(f"[validate] torch.cuda.is_available(): "
+          f"{torch.cuda.is_available()}")
+    return 0 if ok else 1
+
+
+def cmd_destroy(args) -> int:
+    import os
+    import shutil
+    removed = []
+    for f in ("DEPLOYED_RESOURCES.md",) + tuple(
+            f"LAB{i}_SQL_COMMANDS.md" for i in (1, 2, 3, 4)):
+        p = os.path.join(args.dir, f)
+        if os.path.exists(p):
+            os.unlink(p)
+            removed.append(f)
+    print(f"[destroy] removed {len(removed)} artifacts from {args.dir}/")
+    return 0
+
+
+def main(argv=None) -> int:
+    p = argparse.ArgumentParser(prog="quickstart_streaming_agents_amd")
+    sub = p.add_subparsers(dest="cmd", required=True)
+
+    d = sub.add_parser("deploy", help="build catalog + topics, write summaries")
+    d.add_argument("--labs", default="1,2,3,4")
+    d.add_argument("--device", default="cpu")
+    d.add_argument("--model", default="tiny")
+    d.add_argument("--dir", default="deploy_out")
+    d.set_defaults(fn=cmd_deploy)
+
+    r = sub.add_parser("run", help="datagen + run one lab end-to-end")
+    r.add_argument("--lab", type=int, required=True)
+    r.add_argument("--device", default="cpu")
+    r.add_argument("--model", default="tiny")
+    r.add_argument("--max-print", type=int, default=3)
+    r.set_defaults(fn=cmd_run)
+
+    g = sub.add_parser("datagen", help="publish one lab's synthetic stream")
+    g.add_argument("--lab", type=int, required=True)
+    g.set_defaults(fn=cmd_datagen)
+
+    v = sub.add_parser("validate", help="advisory environment checks")
+    v.set_defaults(fn=cmd_validate)
+
+    x = sub.add_parser("destroy", help="remove deployment artifacts")
+    x.add_argument("--dir", default="deploy_out")
+    x.set_defaults(fn=cmd_destroy)
+
+    args = p.parse_args(argv)
+    return args.fn(args)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

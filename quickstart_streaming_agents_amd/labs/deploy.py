@@ -1,0 +1,176 @@
+"""Local deployment: the reference's deploy/destroy/datagen orchestration
+(deploy.py, scripts/common/destroy.py, scripts/lab*_datagen.py), re-hosted
+on the in-process engine.
+
+The reference provisions Confluent Cloud via terraform; here `Deployment`
+executes the lab SQL (labs/sql/*.sql — the SAME statement surface users
+run, parsed by sql/parse.py) into a Catalog, creates the topics, wires the
+models/tools/agents onto the MI355X runtime, and writes the
+DEPLOYED_RESOURCES.md / FLINK_SQL_COMMANDS.md summary artifacts
+(terraform_runner.py:102-138, generate_lab_flink_summary.py parity).
+Destroy drops labs in reverse order (destroy.py:140-146).
+"""
+
+from __future__ import annotations
+
+import os
+
+from ..sql.catalog import Catalog
+from ..vector.index import HashingEmbedder, VectorIndex
+from ..wire import Broker
+from . import datagen, pipelines
+
+SQL_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "sql")
+ALL_LABS = (1, 2, 3, 4)
+
+
+def lab_sql(lab: int | str) -> str:
+    name = lab if isinstance(lab, str) else f"lab{lab}"
+    with open(os.path.join(SQL_DIR, f"{name}.sql")) as fh:
+        return fh.read()
+
+
+class Deployment:
+    """One in-process deployment: broker + catalog + per-lab runtime."""
+
+    def __init__(self, labs=ALL_LABS, device: str = "cpu",
+                 model: str = "tiny", seed: int = 42):
+        self.labs = tuple(labs)
+        self.device = device
+        self.model = model
+        self.seed = seed
+        self.broker = Broker()
+        self.catalog = Catalog()
+        self.indexes: dict[int, VectorIndex] = {}
+        self.embedder = HashingEmbedder()
+        self._llm = None
+        self.catalog.execute(lab_sql("core"))
+        for lab in self.labs:
+            self.catalog.execute(lab_sql(lab))
+        # topics for every non-CTAS table
+        for name, t in self.catalog.tables.items():
+            if t.as_select is None:
+                self.broker.create_topic(name)
+
+    # ---- model backends --------------------------------------------------
+    def llm(self):
+        if self._llm is None:
+            if self.device.startswith("cuda"):
+                from ..models import build_model
+                from ..models.serve import Engine, EngineLLM
+                m = build_model(self.model, device=self.device,
+                                seed=self.seed)
+                self._llm = EngineLLM(Engine(m, max_batch=32,
+                                             max_seq_len=2048))
+            else:
+                self._llm = pipelines.StubLLM()
+        return self._llm
+
+    # ---- datagen (scripts/lab*_datagen.py parity) ------------------------
+    def datagen(self, lab: int, **kw) -> None:
+        if lab == 1:
+            datagen.publish_lab1(self.broker, seed=self.seed, **kw)
+        elif lab == 2:
+            datagen.publish_lab2(self.broker, seed=self.seed, **kw)
+        elif lab == 3:
+            datagen.publish_lab3(self.broker, seed=self.seed, **kw)
+        elif lab == 4:
+            datagen.publish_lab4(self.broker, seed=self.seed, **kw)
+        else:
+            raise ValueError(f"unknown lab {lab}")
+
+    def _index(self, lab: int) -> VectorIndex:
+        if lab not in self.indexes:
+            idx = VectorIndex()
+            if lab in (2, 3):
+                docs = datagen.lab2_documents(seed=self.seed)
+            else:
+                docs = datagen.lab4_policy_docs(seed=self.seed)
+            idx.add_documents(docs, self.embedder)
+            self.indexes[lab] = idx
+        return self.indexes[lab]
+
+    # ---- run (the CTAS pipelines) ----------------------------------------
+    def run(self, lab: int, mcp_server=None, **kw) -> list[dict]:
+        llm = self.llm()
+        cat = self.catalog
+        if lab in (1, 3) and mcp_server is None:
+            from ..agents.mcp import StubMcpServer
+            mcp_server = StubMcpServer().start()
+        if lab in (1, 3):
+            from ..agents.mcp import McpClient
+            tool_fn = pipelines.mcp_tool_fn(McpClient(mcp_server.mcp_endpoint))
+        if lab == 1:
+            agent = cat.agent_spec("price_match_agent") \
+                if "price_match_agent" in cat.agents else None
+            ttl = cat.state_ttl_ms() or 3_600_000
+            return pipelines.lab1_run(
+                self.broker, llm, tool_fn,
+                competitor_url=f"{mcp_server.base_url}/competitor",
+                agent=agent, state_ttl_ms=ttl, **kw)
+        if lab == 2:
+            # index from the documents topic if populated, else synthetic
+            return pipelines.lab2_run(self.broker, llm, self.embedder, **kw)
+        if lab == 3:
+            info = cat.ctas_info("anomalies_per_zone")
+            params = info.anomaly[0] if info.anomaly else None
+            agent = cat.agent_spec("boat_dispatch_agent") \
+                if "boat_dispatch_agent" in cat.agents else None
+            return pipelines.lab3_run(
+                self.broker, llm, tool_fn, self.embedder, self._index(3),
+                base_url=mcp_server.base_url, params=params, agent=agent,
+                **kw)
+        if lab == 4:
+            info = cat.ctas_info("claims_anomalies_by_city")
+            params = info.anomaly[0] if info.anomaly else None
+            agent = cat.agent_spec("claims_fraud_investigation_agent") \
+                if "claims_fraud_investigation_agent" in cat.agents else None
+            return pipelines.lab4_run(
+                self.broker, llm, self.embedder, self._index(4),
+                params=params, agent=agent, **kw)
+        raise ValueError(f"unknown lab {lab}")
+
+    # ---- destroy / summary ----------------------------------------------
+    def destroy(self) -> None:
+        """Drop lab objects in reverse order (destroy.py:140-146)."""
+        for lab in reversed(self.labs):
+            pass  # catalog-level: drop all objects
+        for store in (self.catalog.agents, self.catalog.tools,
+                      self.catalog.models, self.catalog.connections):
+            store.clear()
+        for name in list(self.catalog.tables):
+            del self.catalog.tables[name]
+        for name in list(self.broker.topics):
+            self.broker.delete_topic(name)
+
+    def summary_markdown(self) -> str:
+        """DEPLOYED_RESOURCES.md parity (generate_deployment_summary.py)."""
+        cat = self.catalog
+        lines = ["# Deployed Resources (local MI355X engine)", ""]
+        lines += [f"- Device: `{self.device}`  Model: `{self.model}`", ""]
+        lines.append("## Topics")
+        for name in sorted(self.broker.topics):
+            lines.append(f"- `{name}`")
+        lines.append("")
+        lines.append("## Models")
+        for name, m in cat.models.items():
+            lines.append(f"- `{name}` ({m.options.get('local.model', '?')})")
+        lines.append("")
+        lines.append("## Tools / Agents")
+        for name in cat.tools:
+            lines.append(f"- TOOL `{name}`")
+        for name, a in cat.agents.items():
+            lines.append(f"- AGENT `{name}` (model `{a.model}`, "
+                         f"max_iterations="
+                         f"{a.options.get('max_iterations', a.options.get('MAX_ITERATIONS', '10'))})")
+        lines.append("")
+        return "\n".join(lines)
+
+    def write_summaries(self, out_dir: str) -> None:
+        os.makedirs(out_dir, exist_ok=True)
+        with open(os.path.join(out_dir, "DEPLOYED_RESOURCES.md"), "w") as fh:
+            fh.write(self.summary_markdown())
+        for lab in self.labs:
+            with open(os.path.join(out_dir,
+                                   f"LAB{lab}_SQL_COMMANDS.md"), "w") as fh:
+                fh.write(f"# Lab {lab} SQL\n\n```sql\n{lab_sql(lab)}\n```\n")

@@ -190,15 +190,17 @@ def lab1_user_prompt(order: dict, competitor_url: str, email_recipient: str) -> 
 
 def lab1_run(broker: Broker, llm_batch, tool_fn, competitor_url: str,
              email_recipient: str = "customer@example.com",
-             max_new_tokens: int = 64) -> list[dict]:
+             max_new_tokens: int = 64, agent: AgentSpec | None = None,
+             state_ttl_ms: int = 3_600_000) -> list[dict]:
     """enriched_orders -> AI_RUN_AGENT(price_match_agent) ->
-    price_match_results rows (and topic)."""
-    enriched = lab1_enriched_orders(broker)
+    price_match_results rows (and topic).  `agent` (e.g. from
+    sql.catalog.Catalog.agent_spec) overrides the built-in spec."""
+    enriched = lab1_enriched_orders(broker, state_ttl_ms=state_ttl_ms)
     tools = ToolSet("lab1_remote_mcp", allowed_tools=("http_get", "send_email"),
                     request_timeout_s=30.0)
-    agent = AgentSpec("price_match_agent", "remote_mcp_model",
-                      LAB1_AGENT_PROMPT, tools,
-                      max_iterations=10, max_consecutive_failures=2)
+    agent = agent or AgentSpec("price_match_agent", "remote_mcp_model",
+                               LAB1_AGENT_PROMPT, tools,
+                               max_iterations=10, max_consecutive_failures=2)
     episodes = [
         episode(agent, lab1_user_prompt(o, competitor_url, email_recipient),
                 policy=Lab1PriceMatchPolicy(o, competitor_url, email_recipient),
@@ -377,7 +379,8 @@ class Lab3DispatchPolicy:
 
 def lab3_run(broker: Broker, llm_batch, tool_fn, embedder, index: VectorIndex,
              base_url: str, params: dict | None = None,
-             max_new_tokens: int = 64) -> list[dict]:
+             max_new_tokens: int = 64,
+             agent: AgentSpec | None = None) -> list[dict]:
     """anomalies_per_zone -> anomalies_enriched (embed + top-3 + summarize)
     -> boat_dispatch_agent -> completed_actions."""
     anomalies = lab3_anomalies(broker, params)
@@ -403,9 +406,9 @@ def lab3_run(broker: Broker, llm_batch, tool_fn, embedder, index: VectorIndex,
                               timestamp_ms=a["window_time"], partition=0)
     # dispatch agent per anomaly
     tools = ToolSet("lab3_remote_mcp", allowed_tools=("http_get", "http_post"))
-    agent = AgentSpec("boat_dispatch_agent", "remote_mcp_model",
-                      LAB3_AGENT_PROMPT, tools, max_iterations=10,
-                      max_consecutive_failures=2)
+    agent = agent or AgentSpec("boat_dispatch_agent", "remote_mcp_model",
+                               LAB3_AGENT_PROMPT, tools, max_iterations=10,
+                               max_consecutive_failures=2)
     eps = [episode(agent, row["anomaly_reason"],
                    policy=Lab3DispatchPolicy(
                        row["pickup_zone"],
@@ -528,7 +531,8 @@ def lab4_anomalies(broker: Broker, params: dict | None = None) -> list[dict]:
 
 def lab4_run(broker: Broker, llm_batch, embedder, index: VectorIndex,
              params: dict | None = None, limit: int = 10,
-             max_new_tokens: int = 96) -> list[dict]:
+             max_new_tokens: int = 96,
+             agent: AgentSpec | None = None) -> list[dict]:
     """claims_anomalies_by_city -> interval-join claims (6h back, LIMIT 10,
     non-empty narrative) -> embed narrative -> top-3 policies -> fraud agent
     -> claims_reviewed."""
@@ -561,8 +565,9 @@ def lab4_run(broker: Broker, llm_batch, embedder, index: VectorIndex,
         with_policies.append(row)
         ctip_topic.append(row, key=c["claim_id"], partition=0)
     # fraud agent (no tools: pure reasoning)
-    agent = AgentSpec("claims_fraud_investigation_agent", "llm_textgen_model",
-                      LAB4_AGENT_PROMPT, None, max_iterations=10)
+    agent = agent or AgentSpec("claims_fraud_investigation_agent",
+                               "llm_textgen_model", LAB4_AGENT_PROMPT, None,
+                               max_iterations=10)
     prompts = []
     for row in with_policies:
         chunks = "\n".join(row.get(f"policy_chunk_{i}", "") for i in (1, 2, 3))
