@@ -220,7 +220,7 @@ class LlamaModel:
             h = D.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             mlp_out = self._ffn(L, h)
         final_h = D.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
-        return F.linear(final_h, self.lm_head)
+        return self._linear(final_h, self.lm_head, self.lm_head_f)
 
     def _ffn(self, L: dict, h: torch.Tensor) -> torch.Tensor:
         """Dense SwiGLU FFN (TP row/col-parallel). Mixtral overrides with

@@ -75,14 +75,22 @@ class MixtralModel(LlamaModel):
         per_rank = c.n_experts // self.ep_size
         my_lo = (self.tp_rank % self.ep_size) * per_rank
         experts = {}
+        packed = {}
+        pack = (str(self.device).startswith("cuda")
+                and self.dtype == torch.bfloat16)
         for e in range(c.n_experts):
             wgu = w(2 * c.ffn, c.hidden)
             wdown = w(c.hidden, c.ffn, std=0.02 / (2 * c.n_layers) ** 0.5)
             if my_lo <= e < my_lo + per_rank or self.ep_size == 1:
                 experts[e] = (wgu, wdown)
+                if pack and D.can_pack_weight(*wgu.shape) and \
+                        D.can_pack_weight(*wdown.shape):
+                    packed[e] = (D.pack_weight_frag(wgu),
+                                 D.pack_weight_frag(wdown))
             else:
                 del wgu, wdown
-        return {"router": router, "experts": experts}
+        return {"router": router, "experts": experts,
+                "experts_f": packed}
 
     def _ffn(self, L: dict, h: torch.Tensor) -> torch.Tensor:
         c = self._moe_cfg
@@ -93,6 +101,11 @@ class MixtralModel(LlamaModel):
 
         def expert_fn(e: int, rows: torch.Tensor) -> torch.Tensor:
             wgu, wdown = L["experts"][e]
+            pk = L.get("experts_f", {}).get(e)
+            if pk is not None and rows.is_cuda and rows.shape[0] <= 32:
+                gu = D.skinny_linear(rows, pk[0], *wgu.shape)
+                act = D.swiglu(gu[:, :c.ffn], gu[:, c.ffn:])
+                return D.skinny_linear(act, pk[1], *wdown.shape)
             gu = F.linear(rows, wgu)
             act = D.swiglu(gu[:, :c.ffn], gu[:, c.ffn:])
             return F.linear(act, wdown)

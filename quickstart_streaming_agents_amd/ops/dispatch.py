@@ -131,7 +131,7 @@ def paged_attn_decode(q: torch.Tensor, kc: torch.Tensor, vc: torch.Tensor,
 # ---- skinny decode GEMM ---------------------------------------------------
 
 def can_pack_weight(n: int, k: int) -> bool:
-    return n % 64 == 0 and k % 256 == 0
+    return n % 16 == 0 and k % 256 == 0
 
 
 def pack_weight_frag(w: torch.Tensor) -> torch.Tensor:

@@ -54,10 +54,8 @@ extern "C" void qsa_kv_scatter_launch(const unsigned short*,
                                       int, hipStream_t);
 extern "C" void qsa_skinny_gemm_launch(const unsigned short*,
                                        const unsigned short*, unsigned short*,
-                                       float*, int, int, long long, long long,
-                                       int, hipStream_t);
-extern "C" void qsa_f32_to_bf16_launch(const float*, unsigned short*,
-                                       long long, hipStream_t);
+                                       int, int, long long, long long,
+                                       hipStream_t);
 extern "C" void qsa_topk_launch(const float*, const float*, float*, int*,
                                 float*, int*, int, int, int, int, int,
                                 hipStream_t);
@@ -228,25 +226,11 @@ torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor wf, long N, long K) {
   const int M = a.size(0);
   TORCH_CHECK(M >= 1 && M <= 32, "skinny_gemm: M in [1,32]");
   TORCH_CHECK(a.size(1) == K, "K mismatch");
-  TORCH_CHECK(K % 256 == 0 && N % 64 == 0, "K%256==0, N%64==0");
+  TORCH_CHECK(K % 256 == 0 && N % 16 == 0, "K%256==0, N%16==0");
   TORCH_CHECK(wf.numel() == (long long)N * K, "wf size");
   auto out = torch::empty({M, (long long)N}, a.options());
-  const long long nblk = N / 64;
-  int splitk = 1;
-  if (nblk < 256)
-    splitk = (int)std::min<long long>(
-        {8, K / 256, (512 + nblk - 1) / nblk});
-  if (splitk > 1) {
-    auto cf = torch::zeros({M, (long long)N},
-                           a.options().dtype(at::kFloat));
-    qsa_skinny_gemm_launch(u16(a), u16(wf), nullptr, cf.data_ptr<float>(), M,
-                           (int)N, K, a.stride(0), splitk, cur_stream());
-    qsa_f32_to_bf16_launch(cf.data_ptr<float>(), u16m(out),
-                           (long long)M * N, cur_stream());
-  } else {
-    qsa_skinny_gemm_launch(u16(a), u16(wf), u16m(out), nullptr, M, (int)N, K,
-                           a.stride(0), 1, cur_stream());
-  }
+  qsa_skinny_gemm_launch(u16(a), u16(wf), u16m(out), M, (int)N, K,
+                         a.stride(0), cur_stream());
   return out;
 }
 

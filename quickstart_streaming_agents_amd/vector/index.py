@@ -144,6 +144,31 @@ class VectorIndex:
             self._torch_device = device
         return self._torch_matrix
 
+    def search_batch_gpu(self, queries: np.ndarray, k: int = 3,
+                         device: str = "cuda") -> list[list[SearchHit]]:
+        """Exact cosine top-k on the HBM-resident matrix via the HIP
+        tiled-dot-product/top-k kernel (ops/hip/topk_cosine.hip)."""
+        import torch
+
+        from ..ops import dispatch as D
+        if not len(self):
+            return [[] for _ in range(len(queries))]
+        q = np.asarray(queries, dtype=np.float32)
+        norms = np.linalg.norm(q, axis=1, keepdims=True)
+        q = q / np.maximum(norms, 1e-12)
+        docs = self.to_torch(device)
+        qt = torch.from_numpy(q).to(device=device, dtype=torch.float32)
+        k_eff = min(k, len(self))
+        scores, idx = D.topk_cosine(qt.contiguous(), docs, k_eff)
+        scores = scores.cpu().numpy()
+        idx = idx.cpu().numpy()
+        out = []
+        for qi in range(len(q)):
+            out.append([SearchHit(self.ids[i], self.chunks[i],
+                                  float(scores[qi, j]), self.metadata[i])
+                        for j, i in enumerate(idx[qi])])
+        return out
+
     @staticmethod
     def mongo_score(cosine: float) -> float:
         return (1.0 + cosine) / 2.0
