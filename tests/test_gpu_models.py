@@ -172,3 +172,25 @@ def test_mixtral_gpu_engine():
     assert all(len(o) == 6 for o in outs)
     eng2 = Engine(model, max_batch=4, max_seq_len=256)
     assert eng2.generate_batch([[1, 5, 9, 13], [2, 4, 6]], [6, 6]) == outs
+
+
+def test_vector_index_gpu_search_matches_cpu():
+    import numpy as np
+    from quickstart_streaming_agents_amd.vector.index import (HashingEmbedder,
+                                                              VectorIndex)
+    emb = HashingEmbedder()
+    idx = VectorIndex()
+    rng = np.random.default_rng(11)
+    for i in range(300):
+        words = " ".join(rng.choice(
+            ["window", "join", "agent", "kafka", "state"], size=5).tolist())
+        idx.add(f"d{i}", words, emb.embed(words))
+    queries = np.stack([emb.embed("agent state window"),
+                        emb.embed("kafka join")])
+    cpu = idx.search_batch(queries, 3)
+    gpu = idx.search_batch_gpu(queries, 3, device="cuda:0")
+    for c_hits, g_hits in zip(cpu, gpu):
+        assert [h.document_id for h in c_hits] == \
+            [h.document_id for h in g_hits]
+        for c, g in zip(c_hits, g_hits):
+            assert abs(c.score - g.score) < 1e-4
