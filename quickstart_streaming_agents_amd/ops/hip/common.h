@@ -66,3 +66,10 @@ __device__ __forceinline__ float block_reduce_sum(float v, float* scratch) {
 
 #define QSA_CHECK(cond, msg)                                            \
   TORCH_CHECK(cond, "qsa_hip: ", msg)
+
+typedef __attribute__((ext_vector_type(2))) __bf16 qsa_bf16x2_t;
+__device__ __forceinline__ qsa_bf16x2_t as_bf16x2(unsigned int u) {
+  union { unsigned int u; qsa_bf16x2_t v; } c;
+  c.u = u;
+  return c.v;
+}

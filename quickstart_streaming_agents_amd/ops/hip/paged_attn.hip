@@ -81,12 +81,9 @@ qsa_paged_attn_decode(const unsigned short* __restrict__ q,   // [B, QH, D] (row
         uint4 kv4 = kbase[d0 * QSA_PAGE + lane];
         unsigned int kk[4] = {kv4.x, kv4.y, kv4.z, kv4.w};
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          float2 kf = bf16x2_to_f32x2(kk[j]);
-          float2 qf = bf16x2_to_f32x2(qpk[d0 * 4 + j]);
-          sc = fmaf(kf.x, qf.x, sc);
-          sc = fmaf(kf.y, qf.y, sc);
-        }
+        for (int j = 0; j < 4; ++j)   // v_dot2_f32_bf16: 2 MACs/instr
+          sc = __builtin_amdgcn_fdot2_f32_bf16(
+              as_bf16x2(kk[j]), as_bf16x2(qpk[d0 * 4 + j]), sc, false);
       }
       sc = valid ? sc * scale : -3.0e38f;
       // ---- online softmax over this page ----
@@ -208,12 +205,9 @@ qsa_paged_attn_decode_split(const unsigned short* __restrict__ q,
         uint4 kv4 = kbase[d0 * QSA_PAGE + lane];
         unsigned int kk[4] = {kv4.x, kv4.y, kv4.z, kv4.w};
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          float2 kf = bf16x2_to_f32x2(kk[j]);
-          float2 qf = bf16x2_to_f32x2(qpk[d0 * 4 + j]);
-          sc = fmaf(kf.x, qf.x, sc);
-          sc = fmaf(kf.y, qf.y, sc);
-        }
+        for (int j = 0; j < 4; ++j)   // v_dot2_f32_bf16: 2 MACs/instr
+          sc = __builtin_amdgcn_fdot2_f32_bf16(
+              as_bf16x2(kk[j]), as_bf16x2(qpk[d0 * 4 + j]), sc, false);
       }
       sc = valid ? sc * scale : -3.0e38f;
       const float pmax = wave_reduce_max(sc);

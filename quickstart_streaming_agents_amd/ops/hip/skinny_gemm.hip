@@ -43,14 +43,17 @@ qsa_skinny_gemm(const unsigned short* __restrict__ A,   // [M, K] row stride lda
   f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
   f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
 
-  // A fragment source for this lane: row mt*16 + (lane&15),
-  // k-offset (lane>>4)*8 within each 32-wide k-step.
+  // A stages through LDS per SUPER-chunk of 1024 k (the 4 waves' chunks):
+  // cooperative coalesced fill once, low-latency ds_read fragments after —
+  // dependent per-k-step global A loads were the v2 kernel's stall.
+  // Row pad 8 elements (16 B) spreads the 16-row fragment read groups
+  // across banks.
+  __shared__ unsigned short As[32][1024 + 8];
+
+  // A fragment LDS address pieces: row mt*16 + (lane&15),
+  // k-offset = wave*256 + s*32 + (lane>>4)*8 within the super-chunk.
   const int arow = lane & 15;
   const int akoff = (lane >> 4) * 8;
-  const bool row0_ok = arow < M;
-  const bool row1_ok = 16 + arow < M;
-  const unsigned short* a0base = A + (long long)arow * lda + akoff;
-  const unsigned short* a1base = A + (long long)(16 + arow) * lda + akoff;
 
   // W stream: block (nt, kk) at ((nt*(K/32) + kk) * 512) elements; this
   // lane's 16 B at (lane&15)*32 + (lane>>4)*8 inside the block.
@@ -58,21 +61,62 @@ qsa_skinny_gemm(const unsigned short* __restrict__ A,   // [M, K] row stride lda
       Wf + (long long)nt * (K >> 5) * 512 +
       (long long)((lane & 15) * 32 + (lane >> 4) * 8);
 
-  const bf16x8 zero8 = {0, 0, 0, 0, 0, 0, 0, 0};
-  for (long long c = wave; c < kchunks; c += 4) {
-    const long long k0 = c * QSA_KCH;
+  const long long nsuper = kchunks / 4;   // K % 1024 == 0 (K % 256 == 0 and
+  const long long ktail = nsuper * 4;     // tail chunks handled separately)
+  for (long long sc = 0; sc < nsuper; ++sc) {
+    // ---- cooperative stage: A[0:32][sc*1024 : +1024] (64 KiB) ----------
+    {
+      const int tid = (int)threadIdx.x;
+      // 2048 pieces of 16 B; thread t fills pieces t, t+256, ...
+#pragma unroll
+      for (int p = 0; p < 8; ++p) {
+        const int piece = tid + p * 256;
+        const int row = piece >> 6;            // 64 pieces per row
+        const int off16 = piece & 63;
+        uint4 v = make_uint4(0, 0, 0, 0);
+        if (row < M)
+          v = *reinterpret_cast<const uint4*>(
+              A + (long long)row * lda + sc * 1024 + off16 * 8);
+        *reinterpret_cast<uint4*>(&As[row][off16 * 8]) = v;
+      }
+    }
+    __syncthreads();
+    const long long k0 = sc * 1024 + wave * QSA_KCH;
 #pragma unroll
     for (int s = 0; s < 8; ++s) {
       const long long kk = (k0 >> 5) + s;
       const bf16x8 w = *reinterpret_cast<const bf16x8*>(
           __builtin_assume_aligned(wbase + kk * 512, 16));
-      const long long ak = k0 + s * 32;
-      const bf16x8 a0 = row0_ok
-          ? *reinterpret_cast<const bf16x8*>(a0base + ak) : zero8;
-      const bf16x8 a1 = row1_ok
-          ? *reinterpret_cast<const bf16x8*>(a1base + ak) : zero8;
+      const int la = wave * QSA_KCH + s * 32 + akoff;
+      const bf16x8 a0 = *reinterpret_cast<const bf16x8*>(&As[arow][la]);
+      const bf16x8 a1 = *reinterpret_cast<const bf16x8*>(&As[16 + arow][la]);
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, w, acc0, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, w, acc1, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+  // ---- K tail (kchunks % 4 != 0): direct global fragments --------------
+  if (ktail < kchunks) {
+    const bf16x8 zero8 = {0, 0, 0, 0, 0, 0, 0, 0};
+    const bool row0_ok = arow < M;
+    const bool row1_ok = 16 + arow < M;
+    const unsigned short* a0base = A + (long long)arow * lda + akoff;
+    const unsigned short* a1base = A + (long long)(16 + arow) * lda + akoff;
+    for (long long c = ktail + wave; c < kchunks; c += 4) {
+      const long long k0 = c * QSA_KCH;
+#pragma unroll
+      for (int s = 0; s < 8; ++s) {
+        const long long kk = (k0 >> 5) + s;
+        const bf16x8 w = *reinterpret_cast<const bf16x8*>(
+            __builtin_assume_aligned(wbase + kk * 512, 16));
+        const long long ak = k0 + s * 32;
+        const bf16x8 a0 = row0_ok
+            ? *reinterpret_cast<const bf16x8*>(a0base + ak) : zero8;
+        const bf16x8 a1 = row1_ok
+            ? *reinterpret_cast<const bf16x8*>(a1base + ak) : zero8;
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, w, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, w, acc1, 0, 0, 0);
+      }
     }
   }
 
