@@ -56,6 +56,11 @@ extern "C" void qsa_skinny_gemm_launch(const unsigned short*,
                                        const unsigned short*, unsigned short*,
                                        int, int, long long, long long,
                                        hipStream_t);
+extern "C" void qsa_skinny_gemm_probe_launch(const unsigned short*,
+                                             const unsigned short*,
+                                             unsigned short*, int, int,
+                                             long long, long long, int, int,
+                                             int, hipStream_t);
 extern "C" void qsa_topk_launch(const float*, const float*, float*, int*,
                                 float*, int*, int, int, int, int, int,
                                 hipStream_t);
@@ -63,6 +68,7 @@ extern "C" void qsa_window_agg_launch(const long long*, const int*,
                                       const float*, int*, float*, long long,
                                       long long, int, long long, int,
                                       hipStream_t);
+void register_avro(pybind11::module_&);  // avro_codec.cpp
 extern "C" void qsa_anomaly_batch_launch(const float*, const int*, float*,
                                          float*, int*, int, int, int,
                                          hipStream_t);
@@ -235,6 +241,17 @@ torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor wf, long N, long K) {
   return out;
 }
 
+torch::Tensor skinny_gemm_probe(torch::Tensor a, torch::Tensor wf, long N,
+                                long K, long waves, long nt, long variant) {
+  CHK_DEV(a); CHK_BF16(a); CHK_BF16(wf); CHK_CONT(wf);
+  const int M = a.size(0);
+  auto out = torch::empty({M, (long long)N}, a.options());
+  qsa_skinny_gemm_probe_launch(u16(a), u16(wf), u16m(out), M, (int)N, K,
+                               a.stride(0), (int)waves, (int)nt,
+                               (int)variant, cur_stream());
+  return out;
+}
+
 std::vector<torch::Tensor> topk_cosine(torch::Tensor queries,
                                        torch::Tensor docs, long k) {
   CHK_DEV(queries); CHK_CONT(queries); CHK_F32(queries); CHK_F32(docs);
@@ -310,7 +327,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "repack [N,K] bf16 into MFMA-fragment-major for skinny_gemm");
   m.def("skinny_gemm", &skinny_gemm,
         "decode-batch GEMM (M<=32) on the packed weight stream");
+  m.def("skinny_gemm_probe", &skinny_gemm_probe,
+        "ablation probe: waves/nt/variant sweep");
   m.def("topk_cosine", &topk_cosine, "exact cosine top-k over the HBM index");
   m.def("window_agg", &window_agg, "segmented (key, window) count/sum");
   m.def("anomaly_batch", &anomaly_batch, "batched AR+ridge anomaly scorer");
+  register_avro(m);
 }

@@ -126,6 +126,18 @@ class Broker:
             self.topics.pop(name, None)
 
 
+def _native_codec(schema: Schema):
+    """The C++ batch codec (ops/hip/avro_codec.cpp) when the extension is
+    built — byte-identical to the Python codec (tests assert it)."""
+    try:
+        from ..ops import ext, have_ext
+        if have_ext():
+            return ext().AvroCodec(schema.defn)
+    except Exception:
+        pass
+    return None
+
+
 class AvroProducer:
     """Produce Python dicts as Confluent-Avro wire bytes onto a topic."""
 
@@ -135,6 +147,7 @@ class AvroProducer:
         self.topic = broker.create_topic(topic)
         self.value_schema = value_schema if isinstance(value_schema, Schema) else Schema(value_schema)
         self.value_schema_id = broker.registry.register(f"{topic}-value", self.value_schema)
+        self._codec = _native_codec(self.value_schema)
         self.key_schema = None
         self.key_schema_id = None
         if key_schema is not None:
@@ -143,7 +156,10 @@ class AvroProducer:
 
     def produce(self, value: dict, key: Any = None, timestamp_ms: int = 0,
                 partition: int | None = None) -> Record:
-        raw_v = serialize(self.value_schema, self.value_schema_id, value)
+        if self._codec is not None:
+            raw_v = bytes(self._codec.serialize(self.value_schema_id, value))
+        else:
+            raw_v = serialize(self.value_schema, self.value_schema_id, value)
         raw_k = key
         if self.key_schema is not None and key is not None:
             raw_k = serialize(self.key_schema, self.key_schema_id, key)
@@ -158,6 +174,7 @@ class AvroConsumer:
         self.broker = broker
         self.topic_name = topic
         self.value_schema = value_schema if isinstance(value_schema, Schema) else Schema(value_schema)
+        self._codec = _native_codec(self.value_schema)
         self._offsets: dict[int, int] = {}
 
     def poll(self, max_count: int | None = None) -> list[tuple[Record, Any]]:
@@ -169,7 +186,10 @@ class AvroConsumer:
             for r in recs:
                 v = r.value
                 if isinstance(v, (bytes, bytearray)):
-                    _, v = deserialize(self.value_schema, bytes(v))
+                    if self._codec is not None:
+                        _, v = self._codec.deserialize(bytes(v))
+                    else:
+                        _, v = deserialize(self.value_schema, bytes(v))
                 out.append((r, v))
             self._offsets[pi] = start + len(recs)
         return out

@@ -20,6 +20,7 @@ HIP_DIR = os.path.join(ROOT, "quickstart_streaming_agents_amd", "ops", "hip")
 
 sources = [
     os.path.join(HIP_DIR, "ops.cpp"),
+    os.path.join(HIP_DIR, "avro_codec.cpp"),
     os.path.join(HIP_DIR, "elementwise.hip"),
     os.path.join(HIP_DIR, "paged_attn.hip"),
     os.path.join(HIP_DIR, "skinny_gemm.hip"),

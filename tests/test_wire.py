@@ -106,3 +106,39 @@ def test_bad_wire_format_rejected():
     s = Schema(schemas.QUERIES)
     with pytest.raises(ValueError):
         deserialize(s, b"\x01\x00\x00\x00\x07junk")
+
+
+def test_native_codec_byte_identical():
+    """C++ AvroCodec (ops/hip/avro_codec.cpp) must be byte-identical to the
+    Python codec on every lab schema and round-trip every record."""
+    import pytest
+
+    from quickstart_streaming_agents_amd.labs import schemas as S
+    from quickstart_streaming_agents_amd.ops import ext, have_ext
+    from quickstart_streaming_agents_amd.wire.avro import Schema, serialize
+    if not have_ext():
+        pytest.skip("extension not built")
+    samples = {
+        "ORDERS": {"order_id": "O1", "customer_id": "C1", "product_id": "P1",
+                   "price": 12.5, "order_ts": 1726000000000},
+        "DOCUMENTS": {"document_id": "d1", "chunk": "hello é",
+                      "title": None, "pages": "1-2",
+                      "section_reference": None,
+                      "fraud_categories": ["a", "b"],
+                      "policy_keywords": [], "char_count": 7},
+    }
+    for name, rec in samples.items():
+        defn = getattr(S, name)
+        codec = ext().AvroCodec(defn)
+        py = Schema(defn)
+        # fill any missing optional fields with None
+        full = {f["name"]: rec.get(f["name"]) for f in defn["fields"]}
+        full.update(rec)
+        assert bytes(codec.serialize(9, full)) == serialize(py, 9, full)
+        sid, back = codec.deserialize(serialize(py, 9, full))
+        assert sid == 9
+        for k, v in full.items():
+            if isinstance(v, float):
+                assert abs(back[k] - v) < 1e-9
+            else:
+                assert back[k] == v

@@ -49,6 +49,16 @@ def main():
         print(f"{name:8s} N={N:6d} K={K:6d}: rocblas {ms_blas*1e3:8.1f} us  "
               f"skinny {ms_sk*1e3:8.1f} us  floor {floor*1e3:8.1f} us  "
               f"(skinny {floor/ms_sk*100:.0f}% SOL)")
+        if os.environ.get("QSA_SWEEP") == "1":
+            for waves, nt, var, tag in [
+                    (4, 1, 0, "w4 nt  full"), (4, 0, 0, "w4 pln full"),
+                    (8, 1, 0, "w8 nt  full"), (8, 0, 0, "w8 pln full"),
+                    (2, 1, 0, "w2 nt  full"), (1, 1, 0, "w1 nt  full"),
+                    (4, 1, 1, "w4 nt  Wonly"), (4, 1, 2, "w4 nt  Aonly"),
+                    (8, 1, 1, "w8 nt  Wonly")]:
+                ms = t_ms(lambda: e.skinny_gemm_probe(a, wf, N, K, waves,
+                                                      nt, var))
+                print(f"    {tag:14s} {ms*1e3:8.1f} us ({floor/ms*100:.0f}% SOL)")
         del a, w, wf
 
     # paged attention at bench geometry
