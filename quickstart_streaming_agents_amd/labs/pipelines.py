@@ -307,6 +307,27 @@ def lab3_surge_query(zone: str, window_time: int, request_count: int,
             f"{forecast:.0f} expected)?")
 
 
+
+
+def _detect_series(det: AnomalyDetector, rows: list[dict],
+                   value_fn) -> list:
+    """Per-row anomaly results in row order.  On a GPU host every
+    (key, window) step scores in one batched HIP kernel launch
+    (runtime/anomaly.py batch_results_gpu); CPU falls back to the
+    sequential reference — identical results (GPU test asserts)."""
+    import torch
+    if not torch.cuda.is_available():
+        return [det.update(r["key"], float(value_fn(r))) for r in rows]
+    series: dict[str, list[float]] = {}
+    idx: list[tuple[str, int]] = []
+    for r in rows:
+        k = r["key"]
+        series.setdefault(k, [])
+        idx.append((k, len(series[k])))
+        series[k].append(float(value_fn(r)))
+    per_key = det.batch_results_gpu(series)
+    return [per_key[k][i] for k, i in idx]
+
 def lab3_anomalies(broker: Broker, params: dict | None = None) -> list[dict]:
     """5-min TUMBLE per pickup_zone + ML_DETECT_ANOMALIES; keep
     is_anomaly AND request_count > upper_bound (LAB3:99-198)."""
@@ -321,8 +342,8 @@ def lab3_anomalies(broker: Broker, params: dict | None = None) -> list[dict]:
     det = AnomalyDetector.from_json_params(params)
     out = []
     apz_topic = broker.create_topic("anomalies_per_zone")
-    for r in rows:
-        res = det.update(r["key"], float(r["request_count"]))
+    results = _detect_series(det, rows, lambda r: r["request_count"])
+    for r, res in zip(rows, results):
         if res.is_anomaly and r["request_count"] > res.upper_bound:
             row = {
                 "pickup_zone": r["key"],
@@ -510,8 +531,8 @@ def lab4_anomalies(broker: Broker, params: dict | None = None) -> list[dict]:
     det = AnomalyDetector.from_json_params(params)
     out = []
     topic = broker.create_topic("claims_anomalies_by_city")
-    for r in rows:
-        res = det.update(r["key"], r["total_claim_amount"])
+    results = _detect_series(det, rows, lambda r: r["total_claim_amount"])
+    for r, res in zip(rows, results):
         if res.is_anomaly and r["total_claim_amount"] > res.upper_bound:
             row = {
                 "city": r["key"],

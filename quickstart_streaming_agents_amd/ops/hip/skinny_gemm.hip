@@ -46,12 +46,17 @@ qsa_skinny_gemm_t(const unsigned short* __restrict__ A,   // [M,K] stride lda
   f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
   f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
 
+  // Out-of-batch rows CLAMP to row M-1 and load garbage that is never
+  // stored (per-row MFMA outputs are independent; the store is guarded).
+  // A conditional `ok ? load : zero` here would be the guide's .s-level
+  // trap (c): hipcc branches around each load and drains vmcnt(0) per
+  // element — measured 1.5-2.5x on this kernel.
   const int arow = lane & 15;
   const int akoff = (lane >> 4) * 8;
-  const bool row0_ok = arow < M;
-  const bool row1_ok = 16 + arow < M;
-  const unsigned short* a0base = A + (long long)arow * lda + akoff;
-  const unsigned short* a1base = A + (long long)(16 + arow) * lda + akoff;
+  const int r0 = min(arow, M - 1);
+  const int r1 = min(16 + arow, M - 1);
+  const unsigned short* a0base = A + (long long)r0 * lda + akoff;
+  const unsigned short* a1base = A + (long long)r1 * lda + akoff;
 
   // W stream: block (nt, kk) at ((nt*(K/32) + kk) * 512) elements; this
   // lane's 16 B at (lane&15)*32 + (lane>>4)*8 inside the block.
@@ -78,8 +83,8 @@ qsa_skinny_gemm_t(const unsigned short* __restrict__ A,   // [M,K] stride lda
       const long long ak = k0 + s * 32;
       bf16x8 a0 = zero8, a1 = zero8;
       if (VARIANT != 1) {
-        if (row0_ok) a0 = *reinterpret_cast<const bf16x8*>(a0base + ak);
-        if (row1_ok) a1 = *reinterpret_cast<const bf16x8*>(a1base + ak);
+        a0 = *reinterpret_cast<const bf16x8*>(a0base + ak);
+        a1 = *reinterpret_cast<const bf16x8*>(a1base + ak);
       }
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, w, acc0, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, w, acc1, 0, 0, 0);

@@ -154,3 +154,57 @@ class AnomalyDetector:
 
     def series_results(self, key: str, values: list[float]) -> list[AnomalyResult]:
         return [self.update(key, v) for v in values]
+
+    # ------------------------------------------------------------------
+    def batch_results_gpu(self, series_by_key: dict[str, list[float]],
+                          device: str = "cuda") -> dict[str, list["AnomalyResult"]]:
+        """Every (key, window-step) scored in ONE kernel launch: step i of a
+        key becomes its own ragged row (the rolling history before i,
+        max_training-trimmed) of the [rows, Tmax] matrix fed to the batched
+        AR HIP kernel (ops/hip/streaming.hip); the tolerance-band math then
+        vectorizes on host.  Mutates per-key history exactly like the
+        sequential path, and matches it (GPU test asserts)."""
+        import torch
+
+        from ..ops import dispatch as D
+        jobs = []          # (key, step_idx, prefix)
+        out: dict[str, list[AnomalyResult]] = {}
+        for key, values in series_by_key.items():
+            hist = self._history.setdefault(key, [])
+            out[key] = [None] * len(values)   # type: ignore[list-item]
+            for i, v in enumerate(values):
+                if len(hist) < self.min_training:
+                    out[key][i] = AnomalyResult(float(v), float("inf"),
+                                                float("-inf"), False)
+                else:
+                    jobs.append((key, i,
+                                 np.asarray(hist[-self.max_training:],
+                                            dtype=np.float32)))
+                hist.append(float(v))
+            if len(hist) > self.max_training:
+                del hist[: len(hist) - self.max_training]
+        if jobs:
+            tmax = max(len(p) for _, _, p in jobs)
+            mat = np.zeros((len(jobs), tmax), dtype=np.float32)
+            lens = np.zeros(len(jobs), dtype=np.int32)
+            for r, (_, _, p) in enumerate(jobs):
+                mat[r, :len(p)] = p
+                lens[r] = len(p)
+            mt = torch.from_numpy(mat).to(device)
+            lt = torch.from_numpy(lens).to(device)
+            fc, se, dof = D.ext().anomaly_batch(mt, lt, self.order)
+            fc = fc.cpu().numpy()
+            se = se.cpu().numpy()
+            dof = np.maximum(dof.cpu().numpy(), 1).astype(np.float64)
+            zt = self.z + (self.z ** 3 + self.z) / (4.0 * dof)
+            h = 2.0 / (9.0 * dof)
+            chi2_low = dof * (1.0 - h - 0.8416 * np.sqrt(h)) ** 3
+            se_adj = se * np.sqrt(dof / np.maximum(chi2_low, 1e-9))
+            upper = fc + zt * se_adj
+            lower = fc - zt * se_adj
+            for r, (key, i, _) in enumerate(jobs):
+                v = series_by_key[key][i]
+                out[key][i] = AnomalyResult(
+                    float(fc[r]), float(upper[r]), float(lower[r]),
+                    bool(v > upper[r] or v < lower[r]))
+        return out

@@ -194,3 +194,22 @@ def test_vector_index_gpu_search_matches_cpu():
             [h.document_id for h in g_hits]
         for c, g in zip(c_hits, g_hits):
             assert abs(c.score - g.score) < 1e-4
+
+
+def test_lab3_lab4_gpu_anomaly_matches_cpu_and_invariants():
+    """The batched GPU anomaly path must reproduce the sequential CPU
+    reference on the real lab datagen — and hence the determinism
+    contracts (French Quarter only / Naples only)."""
+    from quickstart_streaming_agents_amd.labs import datagen, pipelines
+    from quickstart_streaming_agents_amd.wire import Broker
+
+    b1 = Broker()
+    datagen.publish_lab3(b1, seed=42)
+    rows = pipelines.lab3_anomalies(b1)      # GPU path (cuda available)
+    assert 1 <= len(rows) <= 2
+    assert all(r["pickup_zone"] == "French Quarter" for r in rows)
+
+    b2 = Broker()
+    datagen.publish_lab4(b2, seed=42)
+    rows4 = pipelines.lab4_anomalies(b2)
+    assert [r["city"] for r in rows4] == ["Naples"]
