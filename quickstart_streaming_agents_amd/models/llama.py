@@ -137,12 +137,17 @@ class LlamaModel:
         # weight-streaming skinny GEMM (288 GB HBM affords the duplicate)
         self.lm_head_f = None
         if str(device).startswith("cuda") and dtype == torch.bfloat16:
+            def want(shape):
+                # only shapes where skinny beats rocBLAS (kernel_bench):
+                # duplicating the big-N weights buys nothing
+                return D.can_pack_weight(*shape) and \
+                    shape[0] * shape[1] <= 6144 * 4096
             for L in self.layers:
                 for key in ("wqkv", "wo", "wgu", "wdown"):
                     wt = L.get(key)
-                    if wt is not None and D.can_pack_weight(*wt.shape):
+                    if wt is not None and want(wt.shape):
                         L[key + "_f"] = D.pack_weight_frag(wt)
-            if D.can_pack_weight(*self.lm_head.shape):
+            if want(self.lm_head.shape):
                 self.lm_head_f = D.pack_weight_frag(self.lm_head)
 
     def _ffn_weights(self, w, rows, cols) -> dict:
@@ -166,8 +171,12 @@ class LlamaModel:
                 wf: torch.Tensor | None = None) -> torch.Tensor:
         """Projection: decode-sized cuda batches go through the
         weight-streaming skinny-GEMM kernel, everything else rocBLAS."""
-        if wf is not None and x.is_cuda and x.shape[0] <= 32:
-            return D.skinny_linear(x, wf, w.shape[0], w.shape[1])
+        n, k = w.shape
+        if wf is not None and x.is_cuda and x.shape[0] <= 32 and \
+                n * k <= 6144 * 4096:
+            # skinny wins the small decode shapes (qkv/wo); rocBLAS is at
+            # the bandwidth floor for the big-N ones (kernel_bench sweep)
+            return D.skinny_linear(x, wf, n, k)
         return F.linear(x, w)
 
     def _tp_all_reduce(self, t: torch.Tensor) -> torch.Tensor:

@@ -84,7 +84,8 @@ class MixtralModel(LlamaModel):
             if my_lo <= e < my_lo + per_rank or self.ep_size == 1:
                 experts[e] = (wgu, wdown)
                 if pack and D.can_pack_weight(*wgu.shape) and \
-                        D.can_pack_weight(*wdown.shape):
+                        D.can_pack_weight(*wdown.shape) and \
+                        wgu.numel() <= 6144 * 4096:
                     packed[e] = (D.pack_weight_frag(wgu),
                                  D.pack_weight_frag(wdown))
             else:

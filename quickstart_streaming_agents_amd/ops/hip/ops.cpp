@@ -171,8 +171,8 @@ torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor kc,
   auto out = torch::empty({B, QH, D}, q.options());
   // flash-decoding context split: fill the 256-CU chip (>=~640 WGs) when
   // B*KVH alone underfills it; NS=1 keeps the single-pass kernel.
-  int ns = (int)std::min<long long>(16, std::max<long long>(
-      1, (1024 + (long long)B * KVH - 1) / ((long long)B * KVH)));
+  int ns = (int)std::min<long long>(32, std::max<long long>(
+      1, (2048 + (long long)B * KVH - 1) / ((long long)B * KVH)));
   ns = std::min(ns, std::max(1, max_pages));
   if (ns > 1) {
     auto opts_f = q.options().dtype(at::kFloat);

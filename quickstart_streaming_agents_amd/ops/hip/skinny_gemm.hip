@@ -123,8 +123,10 @@ extern "C" void qsa_skinny_gemm_launch(const unsigned short* A,
                                        unsigned short* Cbf, int M, int N,
                                        long long K, long long lda,
                                        hipStream_t stream) {
-  hipLaunchKernelGGL((qsa_skinny_gemm_t<4, true, 0>), dim3(N / 16), dim3(256),
-                     0, stream, A, Wf, Cbf, M, N, K, lda);
+  // 8 waves, plain (non-nt) W loads: the measured best full-kernel config
+  // (tools/kernel_bench.py sweep; nt helps only the W-isolated stream).
+  hipLaunchKernelGGL((qsa_skinny_gemm_t<8, false, 0>), dim3(N / 16),
+                     dim3(512), 0, stream, A, Wf, Cbf, M, N, K, lda);
 }
 
 // Ablation/tuning probe for tools/kernel_bench.py.
