@@ -220,8 +220,8 @@ class LlamaModel:
             q = qkv[:, :qd].view(B, self.n_q, c.d_head)
             k = qkv[:, qd:qd + kd].view(B, self.n_kv, c.d_head)
             v = qkv[:, qd + kd:].view(B, self.n_kv, c.d_head)
-            D.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
-            D.kv_append(k, v, kv.k[li], kv.v[li], block_table, seq_lens)
+            D.rope_kv_append(q, k, v, kv.k[li], kv.v[li], self.rope_cos,
+                             self.rope_sin, block_table, seq_lens)
             attn = D.paged_attn_decode(q, kv.k[li], kv.v[li], block_table,
                                        seq_lens, self.scale)
             o = self._tp_all_reduce(

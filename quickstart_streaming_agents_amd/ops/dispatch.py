@@ -103,6 +103,21 @@ def kv_append(knew: torch.Tensor, vnew: torch.Tensor, kc: torch.Tensor,
         vc[page, :, off, :] = vnew[b]
 
 
+def rope_kv_append(q, k, v, kc, vc, cos_t, sin_t, block_table,
+                   seq_lens) -> None:
+    """Fused decode-step rope(q,k in place semantics) + cache append.
+    The fused kernel rotates q in place and writes rotated k (+v) straight
+    to the paged cache; CPU reference composes the two reference ops."""
+    if _cuda(q):
+        ext().rope_kv_append(q, k, v, kc, vc, cos_t, sin_t, block_table,
+                             seq_lens)
+        return
+    # positions = seq_lens - 1 (clamped)
+    pos = (seq_lens.long() - 1).clamp(min=0).int()
+    rope_inplace(q, k, cos_t, sin_t, pos)
+    kv_append(k, v, kc, vc, block_table, seq_lens)
+
+
 def kv_scatter(knew: torch.Tensor, vnew: torch.Tensor, kc: torch.Tensor,
                vc: torch.Tensor, slots: torch.Tensor) -> None:
     """Write row t at global slot ids (page*64 + offset)."""

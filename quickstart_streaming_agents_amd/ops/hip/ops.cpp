@@ -48,6 +48,14 @@ extern "C" void qsa_kv_append_launch(const unsigned short*,
                                      unsigned short*, const int*, const int*,
                                      int, int, int, int, long long,
                                      hipStream_t);
+extern "C" void qsa_rope_kv_append_launch(unsigned short*,
+                                          const unsigned short*,
+                                          const unsigned short*,
+                                          unsigned short*, unsigned short*,
+                                          const float*, const float*,
+                                          const int*, const int*, int, int,
+                                          int, int, int, long long, long long,
+                                          hipStream_t);
 extern "C" void qsa_kv_scatter_launch(const unsigned short*,
                                       const unsigned short*, unsigned short*,
                                       unsigned short*, const int*, int, int,
@@ -207,6 +215,26 @@ void kv_append(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kc,
                        cur_stream());
 }
 
+void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                    torch::Tensor kc, torch::Tensor vc, torch::Tensor cos_t,
+                    torch::Tensor sin_t, torch::Tensor block_table,
+                    torch::Tensor seq_lens) {
+  CHK_DEV(q); CHK_BF16(q); CHK_BF16(k); CHK_BF16(v);
+  CHK_F32(cos_t); CHK_F32(sin_t); CHK_I32(block_table); CHK_I32(seq_lens);
+  CHK_CONT(block_table);
+  const int B = q.size(0), QH = q.size(1), D = q.size(2);
+  const int KVH = k.size(1);
+  chk_hd_strided(q, D, "q"); chk_hd_strided(k, D, "k");
+  chk_hd_strided(v, D, "v");
+  TORCH_CHECK(k.stride(0) == v.stride(0), "k/v strides must match");
+  qsa_rope_kv_append_launch(u16m(q), u16(k), u16(v), u16m(kc), u16m(vc),
+                            cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+                            block_table.data_ptr<int>(),
+                            seq_lens.data_ptr<int>(), B, QH, KVH, D,
+                            (int)block_table.size(1), q.stride(0),
+                            k.stride(0), cur_stream());
+}
+
 void kv_scatter(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kc,
                 torch::Tensor vc, torch::Tensor slots) {
   CHK_DEV(knew); CHK_CONT(knew); CHK_BF16(knew); CHK_BF16(vnew); CHK_I32(slots);
@@ -322,6 +350,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("paged_attn_decode", &paged_attn_decode,
         "paged-attention decode (bf16, GQA, page=64)");
   m.def("kv_append", &kv_append, "append one step's k/v to the paged cache");
+  m.def("rope_kv_append", &rope_kv_append,
+        "fused decode rope + paged-cache append");
   m.def("kv_scatter", &kv_scatter, "scatter prefill k/v by slot ids");
   m.def("pack_weight_frag", &pack_weight_frag,
         "repack [N,K] bf16 into MFMA-fragment-major for skinny_gemm");

@@ -367,6 +367,67 @@ qsa_kv_scatter(const unsigned short* __restrict__ knew,  // [T, KVH, D]
   vc[((page * KVH + kvh) * QSA_PAGE + pin) * D + d] = vv;
 }
 
+// Fused decode-step RoPE + KV append: rotates q in place, rotates k and
+// writes it (and v) STRAIGHT into the paged cache — one kernel instead of
+// rope + kv_append, and k never round-trips through HBM in its pre-cache
+// form.  One block per (b, head) over QH q-heads + KVH kv-heads.
+__global__ void
+qsa_rope_kv_append(unsigned short* __restrict__ q,        // [B, QH, D]
+                   const unsigned short* __restrict__ k,  // [B, KVH, D]
+                   const unsigned short* __restrict__ v,  // [B, KVH, D]
+                   unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
+                   const float* __restrict__ cos_t, const float* __restrict__ sin_t,
+                   const int* __restrict__ block_table,
+                   const int* __restrict__ seq_lens,
+                   int B, int QH, int KVH, int D, int max_pages,
+                   long long qstride, long long kvstride) {
+  const int nheads = QH + KVH;
+  const int b = blockIdx.x / nheads;
+  const int h = blockIdx.x % nheads;
+  const int half = D / 2;
+  const int d = threadIdx.x;
+  const int pos = seq_lens[b] - 1;
+  if (pos < 0) return;
+  const long long toff = (long long)pos * half + (d % half);
+  const float c = cos_t[toff], s = sin_t[toff];
+  if (h < QH) {
+    if (d >= half) return;
+    unsigned short* base = q + (long long)b * qstride + (long long)h * D;
+    const float x0 = bf16_to_f32(base[d]);
+    const float x1 = bf16_to_f32(base[d + half]);
+    base[d] = f32_to_bf16(x0 * c - x1 * s);
+    base[d + half] = f32_to_bf16(x0 * s + x1 * c);
+    return;
+  }
+  if (d >= D) return;
+  const int kvh = h - QH;
+  const int page = block_table[(long long)b * max_pages + pos / QSA_PAGE];
+  const int pin = pos % QSA_PAGE;
+  // k: rope pair (d, d+half) computed per thread d < half; threads
+  // [half, D) carry the rotated upper half (recomputed: cheap VALU)
+  const unsigned short* kbase =
+      k + (long long)b * kvstride + (long long)kvh * D;
+  const int dl = d % half;
+  const float x0 = bf16_to_f32(kbase[dl]);
+  const float x1 = bf16_to_f32(kbase[dl + half]);
+  const float kr = (d < half) ? (x0 * c - x1 * s) : (x0 * s + x1 * c);
+  kc[((((long long)page * KVH + kvh) * (D / 8) + d / 8) * QSA_PAGE + pin) * 8 +
+     d % 8] = f32_to_bf16(kr);
+  vc[(((long long)page * KVH + kvh) * QSA_PAGE + pin) * D + d] =
+      v[(long long)b * kvstride + (long long)kvh * D + d];
+}
+
+extern "C" void qsa_rope_kv_append_launch(
+    unsigned short* q, const unsigned short* k, const unsigned short* v,
+    unsigned short* kc, unsigned short* vc, const float* cos_t,
+    const float* sin_t, const int* block_table, const int* seq_lens, int B,
+    int QH, int KVH, int D, int max_pages, long long qstride,
+    long long kvstride, hipStream_t stream) {
+  hipLaunchKernelGGL(qsa_rope_kv_append, dim3(B * (QH + KVH)), dim3(D), 0,
+                     stream, q, k, v, kc, vc, cos_t, sin_t, block_table,
+                     seq_lens, B, QH, KVH, D, max_pages, qstride, kvstride);
+}
+
 extern "C" void qsa_kv_append_launch(const unsigned short* knew,
                                      const unsigned short* vnew,
                                      unsigned short* kc, unsigned short* vc,
