@@ -103,6 +103,24 @@ def cmd_destroy(args) -> int:
     return 0
 
 
+def cmd_tests(args) -> int:
+    import os
+    import subprocess
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [sys.executable, "-m", "pytest", "-q"]
+    if args.quick:
+        cmd += ["tests/test_wire.py", "tests/test_sql.py",
+                "tests/test_datagen.py"]
+    else:
+        cmd += ["tests"]
+    cmd += ["-m", "gpu" if args.gpu else "not gpu"]
+    if args.resume:
+        cmd.append("--lf")
+    if args.k:
+        cmd += ["-k", args.k]
+    return subprocess.call(cmd, cwd=root)
+
+
 def main(argv=None) -> int:
     p = argparse.ArgumentParser(prog="quickstart_streaming_agents_amd")
     sub = p.add_subparsers(dest="cmd", required=True)
@@ -131,6 +149,16 @@ def main(argv=None) -> int:
     x = sub.add_parser("destroy", help="remove deployment artifacts")
     x.add_argument("--dir", default="deploy_out")
     x.set_defaults(fn=cmd_destroy)
+
+    t = sub.add_parser("tests", help="run the test suite "
+                       "(scripts/run_tests.py parity)")
+    t.add_argument("--quick", action="store_true",
+                   help="fast preflight subset (wire/sql/datagen)")
+    t.add_argument("--gpu", action="store_true", help="GPU-marked tests")
+    t.add_argument("--resume", action="store_true",
+                   help="continue from the last failure (pytest --lf)")
+    t.add_argument("-k", default=None, help="pytest -k expression")
+    t.set_defaults(fn=cmd_tests)
 
     args = p.parse_args(argv)
     return args.fn(args)
