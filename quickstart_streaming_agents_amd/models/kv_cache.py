@@ -1,8 +1,9 @@
 """Paged KV cache for the decode engine (page = 64 positions).
 
 Layouts match the HIP decode kernel (ops/hip/paged_attn.hip):
-  K: [pages, KVH, D/8, 64, 8]  (d-major x8 -> coalesced score reads)
-  V: [pages, KVH, 64, D]       (position-major -> coalesced PV reads)
+  K: [pages, KVH, D/8, 64, 8]  (d-major x8 -> coalesced MFMA A-frag reads)
+  V: [pages, KVH, D, 64]       (transposed, pos minor -> coalesced MFMA
+                                B-frag reads in the PV pass)
 
 One allocator spans all layers: page p of layer l lives in that layer's
 tensors, sharing the page-id space so a sequence has ONE page list used by
@@ -27,7 +28,7 @@ class PagedKVCache:
         self.k = [torch.zeros(n_pages, kvh, d_head // 8, self.PAGE, 8,
                               dtype=dtype, device=device)
                   for _ in range(n_layers)]
-        self.v = [torch.zeros(n_pages, kvh, self.PAGE, d_head, dtype=dtype,
+        self.v = [torch.zeros(n_pages, kvh, d_head, self.PAGE, dtype=dtype,
                               device=device)
                   for _ in range(n_layers)]
         self._free = list(range(n_pages - 1, -1, -1))

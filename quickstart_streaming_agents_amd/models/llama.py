@@ -248,8 +248,8 @@ class LlamaModel:
         kpg = kc_l.index_select(0, idx)        # [np, KVH, D/8, 64, 8]
         kseq = kpg.permute(1, 0, 3, 2, 4).reshape(
             self.n_kv, len(pages) * 64, c.d_head)[:, :n]
-        vpg = vc_l.index_select(0, idx)        # [np, KVH, 64, D]
-        vseq = vpg.permute(1, 0, 2, 3).reshape(
+        vpg = vc_l.index_select(0, idx)        # [np, KVH, D, 64]
+        vseq = vpg.permute(1, 0, 3, 2).reshape(
             self.n_kv, len(pages) * 64, c.d_head)[:, :n]
         return kseq, vseq
 

@@ -49,14 +49,14 @@ def rope_ref(x: torch.Tensor, pos: torch.Tensor, cos_t: torch.Tensor,
 def paged_attn_ref(q: torch.Tensor, kc: torch.Tensor, vc: torch.Tensor,
                    block_table: torch.Tensor, seq_lens: torch.Tensor,
                    scale: float) -> torch.Tensor:
-    """q [B, QH, D]; kc [P, KVH, D/8, 64, 8]; vc [P, KVH, 64, D]."""
+    """q [B, QH, D]; kc [P, KVH, D/8, 64, 8]; vc [P, KVH, D, 64]."""
     B, QH, D = q.shape
     KVH = kc.shape[1]
     R = QH // KVH
     out = torch.zeros_like(q, dtype=torch.float32)
     # un-page K into [P, KVH, 64, D]
     kn = kc.permute(0, 1, 3, 2, 4).reshape(kc.shape[0], KVH, 64, D).float()
-    vn = vc.float()
+    vn = vc.permute(0, 1, 3, 2).float()   # -> [P, KVH, 64, D]
     for b in range(B):
         n = int(seq_lens[b])
         if n == 0:

@@ -98,9 +98,9 @@ def kv_append(knew: torch.Tensor, vnew: torch.Tensor, kc: torch.Tensor,
             continue
         page = int(block_table[b, (n - 1) // 64])
         off = (n - 1) % 64
-        # K layout [P, KVH, D/8, 64, 8]
+        # K layout [P, KVH, D/8, 64, 8]; V layout [P, KVH, D, 64]
         kc[page, :, :, off, :] = knew[b].reshape(KVH, D // 8, 8)
-        vc[page, :, off, :] = vnew[b]
+        vc[page, :, :, off] = vnew[b]
 
 
 def rope_kv_append(q, k, v, kc, vc, cos_t, sin_t, block_table,
@@ -129,7 +129,7 @@ def kv_scatter(knew: torch.Tensor, vnew: torch.Tensor, kc: torch.Tensor,
         s = int(slots[t])
         page, off = s // 64, s % 64
         kc[page, :, :, off, :] = knew[t].reshape(KVH, D // 8, 8)
-        vc[page, :, off, :] = vnew[t]
+        vc[page, :, :, off] = vnew[t]
 
 
 def paged_attn_decode(q: torch.Tensor, kc: torch.Tensor, vc: torch.Tensor,

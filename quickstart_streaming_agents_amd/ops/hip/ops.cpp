@@ -33,13 +33,7 @@ extern "C" void qsa_rope_launch(unsigned short*, unsigned short*, const float*,
                                 long long, long long, hipStream_t);
 extern "C" void qsa_softmax_rows_launch(float*, int, int, int, int, int,
                                         const int*, hipStream_t);
-extern "C" void qsa_paged_attn_decode_launch(const unsigned short*,
-                                             const unsigned short*,
-                                             const unsigned short*, const int*,
-                                             const int*, unsigned short*, float,
-                                             int, int, int, int, int,
-                                             long long, hipStream_t);
-extern "C" void qsa_paged_attn_decode_split_launch(
+extern "C" void qsa_paged_attn_mfma_launch(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const int*, const int*, float*, float*, unsigned short*, float, int, int,
     int, int, int, long long, int, hipStream_t);
@@ -170,34 +164,26 @@ torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor kc,
   const int KVH = kc.size(1);
   TORCH_CHECK(D == 128 || D == 64, "D must be 64/128");
   TORCH_CHECK(QH % KVH == 0, "GQA requires QH % KVH == 0");
-  TORCH_CHECK(QH / KVH <= 8, "GQA ratio <= 8");
+  TORCH_CHECK(QH / KVH <= 16, "GQA ratio <= 16");
   TORCH_CHECK(kc.size(2) == D / 8 && kc.size(3) == 64 && kc.size(4) == 8,
               "K cache layout [P, KVH, D/8, 64, 8]");
-  TORCH_CHECK(vc.size(2) == 64 && vc.size(3) == D,
-              "V cache layout [P, KVH, 64, D]");
+  TORCH_CHECK(vc.size(2) == D && vc.size(3) == 64,
+              "V cache layout [P, KVH, D, 64] (transposed)");
   const int max_pages = block_table.size(1);
   auto out = torch::empty({B, QH, D}, q.options());
-  // flash-decoding context split: fill the 256-CU chip (>=~640 WGs) when
-  // B*KVH alone underfills it; NS=1 keeps the single-pass kernel.
+  // flash-decoding split count: fill the 256-CU chip (one WAVE per
+  // (b, kvh, split); 4 splits share a workgroup)
   int ns = (int)std::min<long long>(32, std::max<long long>(
       1, (2048 + (long long)B * KVH - 1) / ((long long)B * KVH)));
   ns = std::min(ns, std::max(1, max_pages));
-  if (ns > 1) {
-    auto opts_f = q.options().dtype(at::kFloat);
-    auto part_o = torch::empty({B, QH, ns, D}, opts_f);
-    auto part_ml = torch::empty({B, QH, ns, 2}, opts_f);
-    qsa_paged_attn_decode_split_launch(
-        u16(q), u16(kc), u16(vc), block_table.data_ptr<int>(),
-        seq_lens.data_ptr<int>(), part_o.data_ptr<float>(),
-        part_ml.data_ptr<float>(), u16m(out), (float)scale, B, QH, KVH,
-        max_pages, D, q.stride(0), ns, cur_stream());
-  } else {
-    qsa_paged_attn_decode_launch(u16(q), u16(kc), u16(vc),
-                                 block_table.data_ptr<int>(),
-                                 seq_lens.data_ptr<int>(), u16m(out),
-                                 (float)scale, B, QH, KVH, max_pages, D,
-                                 q.stride(0), cur_stream());
-  }
+  auto opts_f = q.options().dtype(at::kFloat);
+  auto part_o = torch::empty({B, QH, ns, D}, opts_f);
+  auto part_ml = torch::empty({B, QH, ns, 2}, opts_f);
+  qsa_paged_attn_mfma_launch(
+      u16(q), u16(kc), u16(vc), block_table.data_ptr<int>(),
+      seq_lens.data_ptr<int>(), part_o.data_ptr<float>(),
+      part_ml.data_ptr<float>(), u16m(out), (float)scale, B, QH, KVH,
+      max_pages, D, q.stride(0), ns, cur_stream());
   return out;
 }
 

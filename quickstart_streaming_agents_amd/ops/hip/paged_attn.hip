@@ -1,255 +1,222 @@
-// Paged-attention DECODE kernel for CDNA4 (gfx950) — the K4 hot op.
+// Paged-attention DECODE for CDNA4 (gfx950) — the K4 hot op, on MFMA.
 //
 // Replaces the reference's managed-LLM call (ML_PREDICT 'llm_textgen_model',
 // terraform/core/main.tf:461) with on-GPU batched decode attention over a
 // paged KV cache resident in 288 GB HBM3E.
 //
-// Design (one workgroup per (batch b, kv head)):
-//   - 4 waves; wave w owns GQA query head r = w, w+4, ... (R = QH/KVH).
-//     All waves walk the same pages in the same order, so the 4x K/V re-read
-//     is served by the CU's L1/L2 (page K slab = 16 KB, V = 16 KB).
-//   - q for the wave's head lives packed bf16x2 in 64 VGPRs (no LDS traffic
-//     in the score loop).
-//   - K cache layout [page, kvh, D/8, PAGE=64, 8] ("d-major x8"): for a
-//     fixed d-slice, lane p reads positions p contiguously -> one fully
-//     coalesced 1 KiB wave transaction per 16 B slice.
-//   - V cache layout [page, kvh, PAGE, D] (position-major): in the PV pass
-//     lane l owns output dims (2l, 2l+1); reading V[pos, 2l..2l+1] is a
-//     coalesced 256 B row per position.
-//   - online softmax per wave; probs broadcast lane->lane via __shfl
-//     (no LDS); each wave writes its head's output row independently.
+// Design (flash-decoding, one WAVE per (batch b, kv head, context split)):
+//   - All R GQA query heads of a kv head compute TOGETHER on the matrix
+//     cores: scores via mfma_f32_16x16x32_bf16 with A = K-tile
+//     [16 pos x 32 k] and B = Q [32 k x 16 heads] (heads pad to 16), PV
+//     via A = P [16 heads x 32 pos] (packed bf16, redistributed with 8
+//     cross-lane shuffles) and B = V^T [32 pos x 16 dims].  A VALU
+//     decode kernel pays ~16 VALU per 16 B of KV; this pays ~1 MFMA per
+//     1 KiB plus the online-softmax tail.
+//   - K cache layout [page, kvh, D/8, 64, 8] (d-major x8): the A-fragment
+//     read (lane = pos x k-slice) is one fully coalesced 1 KiB wave load.
+//   - V cache layout [page, kvh, D, 64] (TRANSPOSED, pos minor): the PV
+//     B-fragment read (lane = dim x pos-slice) is also one coalesced 1 KiB
+//     wave load.  Every KV byte is read exactly ONCE per (b, kvh) — the
+//     per-head VALU design re-read it R times.
+//   - Context splits fill the chip (grid = B*KVH*NS >= ~2k waves);
+//     unnormalized partials (o, m, l) merge in qsa_attn_reduce.
 //
-// Every KV byte is read once per wave, coalesced — the kernel targets the
-// HBM roofline, which is what decode attention is bound by.
+// The online softmax runs per head-column in registers: column max/sum
+// reduce with two shfl_xor hops (the 4 hi-lane groups of a column).
 #include "common.h"
 
 #define QSA_PAGE 64
 
+using bf16x8_a = __attribute__((ext_vector_type(8))) short;
+using f32x4_a = __attribute__((ext_vector_type(4))) float;
+
 template <int D>
 __global__ void __launch_bounds__(256)
-qsa_paged_attn_decode(const unsigned short* __restrict__ q,   // [B, QH, D] (row stride qstride)
-                      const unsigned short* __restrict__ kc,  // [P, KVH, D/8, 64, 8]
-                      const unsigned short* __restrict__ vc,  // [P, KVH, 64, D]
-                      const int* __restrict__ block_table,    // [B, max_pages]
-                      const int* __restrict__ seq_lens,       // [B]
-                      unsigned short* __restrict__ out,       // [B, QH, D]
-                      float scale, int B, int QH, int KVH, int max_pages,
-                      long long qstride) {
-  const int b = blockIdx.x / KVH;
-  const int kvh = blockIdx.x % KVH;
-  const int wave = threadIdx.x / QSA_WAVE;
-  const int lane = threadIdx.x % QSA_WAVE;
-  const int R = QH / KVH;
-  const int seqlen = seq_lens[b];
-  if (seqlen <= 0) return;
-  const int npages = (seqlen + QSA_PAGE - 1) / QSA_PAGE;
-  const int* btab = block_table + (long long)b * max_pages;
-
-  for (int r = wave; r < R; r += 4) {
-    const int qh = kvh * R + r;
-    // ---- q packed bf16x2 into regs: D/2 uints (64 for D=128) ----
-    unsigned int qpk[D / 2];  // D bf16 = D/2 packed uints
-    const unsigned int* qsrc = reinterpret_cast<const unsigned int*>(
-        q + (long long)b * qstride + (long long)qh * D);
-#pragma unroll
-    for (int i = 0; i < D / 2; ++i) qpk[i] = qsrc[i];
-
-    // PV decomposition: lane = pg * DG + dg — dim-group dg owns 8 output
-    // dims, position-group pg covers PG=64/DG positions per iteration, so a
-    // V read is 64 lanes x 16 B = 1 KiB fully coalesced and the page's PV
-    // takes PAGE/PG wide iterations instead of 64 scalar ones.
-    constexpr int DG = D / 8;        // lanes per output row (16 for D=128)
-    constexpr int PG = QSA_WAVE / DG;  // positions per iteration (4)
-    const int dg = lane % DG;
-    const int pg = lane / DG;
-
-    float m = -3.0e38f, s = 0.f;
-    float o8[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) o8[j] = 0.f;
-
-    for (int pi = 0; pi < npages; ++pi) {
-      const int page = btab[pi];
-      const int pos = pi * QSA_PAGE + lane;
-      const bool valid = pos < seqlen;
-      // ---- score: dot(q, K[:, pos]) over d-slices of 8 (lane = position)
-      float sc = 0.f;
-      const uint4* kbase = reinterpret_cast<const uint4*>(
-          kc + ((((long long)page * KVH + kvh) * (D / 8)) * QSA_PAGE) * 8);
-#pragma unroll
-      for (int d0 = 0; d0 < D / 8; ++d0) {
-        uint4 kv4 = kbase[d0 * QSA_PAGE + lane];
-        unsigned int kk[4] = {kv4.x, kv4.y, kv4.z, kv4.w};
-#pragma unroll
-        for (int j = 0; j < 4; ++j)   // v_dot2_f32_bf16: 2 MACs/instr
-          sc = __builtin_amdgcn_fdot2_f32_bf16(
-              as_bf16x2(kk[j]), as_bf16x2(qpk[d0 * 4 + j]), sc, false);
-      }
-      sc = valid ? sc * scale : -3.0e38f;
-      // ---- online softmax over this page ----
-      const float pmax = wave_reduce_max(sc);
-      const float m_new = fmaxf(m, pmax);
-      float alpha = __expf(m - m_new);  // m==-inf -> exp(-inf)=0 ok
-      if (m <= -3.0e38f) alpha = 0.f;
-      m = m_new;
-      const float p = valid ? __expf(sc - m_new) : 0.f;
-      s = s * alpha + wave_reduce_sum(p);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) o8[j] *= alpha;
-      // ---- PV: PG positions per 16-B-per-lane iteration ----
-      const uint4* vbase = reinterpret_cast<const uint4*>(
-          vc + (((long long)page * KVH + kvh) * QSA_PAGE) * D);
-      const int nvalid = min(seqlen - pi * QSA_PAGE, QSA_PAGE);
-      const int niter = (nvalid + PG - 1) / PG;
-#pragma unroll 4
-      for (int it = 0; it < niter; ++it) {
-        const int t = it * PG + pg;
-        const float pt = __shfl(p, t, QSA_WAVE);  // p==0 beyond nvalid
-        const uint4 v4 = vbase[(long long)t * DG + dg];
-        unsigned int vv[4] = {v4.x, v4.y, v4.z, v4.w};
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          float2 vf = bf16x2_to_f32x2(vv[j]);
-          o8[2 * j] = fmaf(pt, vf.x, o8[2 * j]);
-          o8[2 * j + 1] = fmaf(pt, vf.y, o8[2 * j + 1]);
-        }
-      }
-    }
-    // butterfly-reduce the PG position-group partials (same dg lanes)
-#pragma unroll
-    for (int off = DG; off < QSA_WAVE; off <<= 1) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) o8[j] += __shfl_xor(o8[j], off, QSA_WAVE);
-    }
-    const float inv = (s > 0.f) ? 1.f / s : 0.f;
-    if (pg == 0) {
-      uint4* orow = reinterpret_cast<uint4*>(out + ((long long)b * QH + qh) * D);
-      uint4 packed;
-      packed.x = f32x2_to_bf16x2(o8[0] * inv, o8[1] * inv);
-      packed.y = f32x2_to_bf16x2(o8[2] * inv, o8[3] * inv);
-      packed.z = f32x2_to_bf16x2(o8[4] * inv, o8[5] * inv);
-      packed.w = f32x2_to_bf16x2(o8[6] * inv, o8[7] * inv);
-      orow[dg] = packed;
-    }
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Flash-decoding context split: B*KVH workgroups underfill the 256-CU chip
-// (e.g. B=24, KVH=8 -> 192 WGs, 4 waves each = terrible occupancy and the
-// kernel runs latency-bound at ~8% of the HBM roofline).  Split the context
-// into NS chunks -> grid B*KVH*NS (>=512 WGs, 2-4 blocks/CU), each workgroup
-// produces an UNNORMALIZED partial (o, m, l) over its page range, and a tiny
-// reduce kernel merges the NS partials per (b, head) row:
-//   M = max m_s;  O = sum_s exp(m_s-M) o_s;  L = sum_s exp(m_s-M) l_s;
-//   out = O / L.
-// ---------------------------------------------------------------------------
-template <int D>
-__global__ void __launch_bounds__(256)
-qsa_paged_attn_decode_split(const unsigned short* __restrict__ q,
-                            const unsigned short* __restrict__ kc,
-                            const unsigned short* __restrict__ vc,
-                            const int* __restrict__ block_table,
-                            const int* __restrict__ seq_lens,
-                            float* __restrict__ part_o,    // [B, QH, NS, D]
-                            float* __restrict__ part_ml,   // [B, QH, NS, 2]
-                            float scale, int B, int QH, int KVH, int max_pages,
-                            long long qstride, int NS) {
-  const int split = blockIdx.x % NS;
-  const int bk = blockIdx.x / NS;
+qsa_paged_attn_mfma(const unsigned short* __restrict__ q,   // [B, QH, D]
+                    const unsigned short* __restrict__ kc,  // [P, KVH, D/8, 64, 8]
+                    const unsigned short* __restrict__ vc,  // [P, KVH, D, 64]
+                    const int* __restrict__ block_table,    // [B, max_pages]
+                    const int* __restrict__ seq_lens,       // [B]
+                    float* __restrict__ part_o,              // [B, QH, NS, D]
+                    float* __restrict__ part_ml,             // [B, QH, NS, 2]
+                    float scale, int B, int QH, int KVH, int max_pages,
+                    long long qstride, int NS) {
+  constexpr int KSTEPS = D / 32;      // QK^T k-steps (4 for D=128)
+  constexpr int DTILES = D / 16;      // PV d-tiles (8 for D=128)
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int nsb = (NS + 3) / 4;       // split-blocks per (b, kvh)
+  const int bk = blockIdx.x / nsb;
+  const int split = (blockIdx.x % nsb) * 4 + wave;
   const int b = bk / KVH;
   const int kvh = bk % KVH;
-  const int wave = threadIdx.x / QSA_WAVE;
-  const int lane = threadIdx.x % QSA_WAVE;
   const int R = QH / KVH;
+  if (split >= NS) return;
+
   const int seqlen = seq_lens[b];
   const int npages = (seqlen + QSA_PAGE - 1) / QSA_PAGE;
   const int chunk = (npages + NS - 1) / NS;
   const int p0 = split * chunk;
   const int p1 = min(npages, p0 + chunk);
-  const int* btab = block_table + (long long)b * max_pages;
+  const int col = lane & 15;          // head col (scores) / dim col (PV)
+  const int hi = lane >> 4;           // 4 hi-lane groups
 
-  for (int r = wave; r < R; r += 4) {
-    const int qh = kvh * R + r;
-    float* pml = part_ml + (((long long)b * QH + qh) * NS + split) * 2;
-    float* po = part_o + (((long long)b * QH + qh) * NS + split) * D;
-    if (seqlen <= 0 || p0 >= npages) {
-      if (lane == 0) { pml[0] = -3.0e38f; pml[1] = 0.f; }
-      continue;
-    }
-    unsigned int qpk[D / 2];
-    const unsigned int* qsrc = reinterpret_cast<const unsigned int*>(
-        q + (long long)b * qstride + (long long)qh * D);
-#pragma unroll
-    for (int i = 0; i < D / 2; ++i) qpk[i] = qsrc[i];
-
-    constexpr int DG = D / 8;
-    constexpr int PG = QSA_WAVE / DG;
-    const int dg = lane % DG;
-    const int pg = lane / DG;
-
-    float m = -3.0e38f, s = 0.f;
-    float o8[8];
-#pragma unroll
-    for (int j = 0; j < 8; ++j) o8[j] = 0.f;
-
-    for (int pi = p0; pi < p1; ++pi) {
-      const int page = btab[pi];
-      const int pos = pi * QSA_PAGE + lane;
-      const bool valid = pos < seqlen;
-      float sc = 0.f;
-      const uint4* kbase = reinterpret_cast<const uint4*>(
-          kc + ((((long long)page * KVH + kvh) * (D / 8)) * QSA_PAGE) * 8);
-#pragma unroll
-      for (int d0 = 0; d0 < D / 8; ++d0) {
-        uint4 kv4 = kbase[d0 * QSA_PAGE + lane];
-        unsigned int kk[4] = {kv4.x, kv4.y, kv4.z, kv4.w};
-#pragma unroll
-        for (int j = 0; j < 4; ++j)   // v_dot2_f32_bf16: 2 MACs/instr
-          sc = __builtin_amdgcn_fdot2_f32_bf16(
-              as_bf16x2(kk[j]), as_bf16x2(qpk[d0 * 4 + j]), sc, false);
-      }
-      sc = valid ? sc * scale : -3.0e38f;
-      const float pmax = wave_reduce_max(sc);
-      const float m_new = fmaxf(m, pmax);
-      float alpha = __expf(m - m_new);
-      if (m <= -3.0e38f) alpha = 0.f;
-      m = m_new;
-      const float p = valid ? __expf(sc - m_new) : 0.f;
-      s = s * alpha + wave_reduce_sum(p);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) o8[j] *= alpha;
-      const uint4* vbase = reinterpret_cast<const uint4*>(
-          vc + (((long long)page * KVH + kvh) * QSA_PAGE) * D);
-      const int nvalid = min(seqlen - pi * QSA_PAGE, QSA_PAGE);
-      const int niter = (nvalid + PG - 1) / PG;
-#pragma unroll 4
-      for (int it = 0; it < niter; ++it) {
-        const int t = it * PG + pg;
-        const float pt = __shfl(p, t, QSA_WAVE);
-        const uint4 v4 = vbase[(long long)t * DG + dg];
-        unsigned int vv[4] = {v4.x, v4.y, v4.z, v4.w};
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          float2 vf = bf16x2_to_f32x2(vv[j]);
-          o8[2 * j] = fmaf(pt, vf.x, o8[2 * j]);
-          o8[2 * j + 1] = fmaf(pt, vf.y, o8[2 * j + 1]);
-        }
-      }
-    }
-#pragma unroll
-    for (int off = DG; off < QSA_WAVE; off <<= 1) {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) o8[j] += __shfl_xor(o8[j], off, QSA_WAVE);
-    }
-    if (pg == 0) {
-      // unnormalized partial in f32 (reduce kernel applies 1/L)
-      float4* prow = reinterpret_cast<float4*>(po);
-      prow[dg * 2] = make_float4(o8[0], o8[1], o8[2], o8[3]);
-      prow[dg * 2 + 1] = make_float4(o8[4], o8[5], o8[6], o8[7]);
-    }
-    if (lane == 0) { pml[0] = m; pml[1] = s; }
+  // every (head) slot writes its partial, real or empty
+  float* pml_base = part_ml + (((long long)b * QH + kvh * R) * NS + split) * 2;
+  if (seqlen <= 0 || p0 >= npages) {
+    if (lane < R)
+      *reinterpret_cast<float2*>(pml_base + (long long)lane * NS * 2) =
+          make_float2(-3.0e38f, 0.f);
+    return;
   }
+
+  // ---- Q B-fragments: lane holds Q[head=col][k = ks*32 + hi*8 + e] ----
+  const int qh_l = kvh * R + min(col, R - 1);   // clamp: pad heads compute
+  const unsigned short* qrow = q + (long long)b * qstride + (long long)qh_l * D;
+  bf16x8_a qf[KSTEPS];
+#pragma unroll
+  for (int ks = 0; ks < KSTEPS; ++ks)
+    qf[ks] = *reinterpret_cast<const bf16x8_a*>(qrow + ks * 32 + hi * 8);
+
+  float m = -3.0e38f, lsum = 0.f;
+  f32x4_a o_acc[DTILES];
+#pragma unroll
+  for (int dt = 0; dt < DTILES; ++dt)
+    o_acc[dt] = (f32x4_a){0.f, 0.f, 0.f, 0.f};
+
+  const int* btab = block_table + (long long)b * max_pages;
+  for (int pi = p0; pi < p1; ++pi) {
+    const int page = btab[pi];
+    const unsigned short* kbase =
+        kc + ((((long long)page * KVH + kvh) * (D / 8)) * QSA_PAGE) * 8;
+    const unsigned short* vbase =
+        vc + (((long long)page * KVH + kvh) * D) * QSA_PAGE;
+    const int nvalid = min(seqlen - pi * QSA_PAGE, QSA_PAGE);
+
+    // ---- scores: 4 pos-tiles x KSTEPS MFMAs ---------------------------
+    // A-frag: lane holds K[pos = pt*16 + col][k = ks*32 + hi*8 + e];
+    // K layout element (pos, k) at ((k/8)*64 + pos)*8 + k%8 ->
+    // 16 B at ((ks*4 + hi)*64 + pos)*8.
+    f32x4_a sc[4];
+#pragma unroll
+    for (int pt = 0; pt < 4; ++pt) {
+      f32x4_a acc = {0.f, 0.f, 0.f, 0.f};
+      const int pos = pt * 16 + col;
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        const bf16x8_a kf = *reinterpret_cast<const bf16x8_a*>(
+            kbase + (((long long)(ks * 4 + hi) * QSA_PAGE) + pos) * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[ks], acc,
+                                                      0, 0, 0);
+      }
+      sc[pt] = acc;
+    }
+    // scale + mask invalid positions; C layout: row pos = hi*4 + r,
+    // col head = lane&15 -> lane's r-th value is position pt*16 + hi*4 + r
+    float pagemax = -3.0e38f;
+#pragma unroll
+    for (int pt = 0; pt < 4; ++pt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int pos = pt * 16 + hi * 4 + r;
+        float v = (pos < nvalid) ? sc[pt][r] * scale : -3.0e38f;
+        sc[pt][r] = v;
+        pagemax = fmaxf(pagemax, v);
+      }
+    }
+    // column (per-head) max across the 4 hi groups
+    pagemax = fmaxf(pagemax, __shfl_xor(pagemax, 16, QSA_WAVE));
+    pagemax = fmaxf(pagemax, __shfl_xor(pagemax, 32, QSA_WAVE));
+    const float m_new = fmaxf(m, pagemax);
+    float alpha = __expf(m - m_new);
+    if (m <= -3.0e38f) alpha = 0.f;
+    m = m_new;
+#pragma unroll
+    for (int dt = 0; dt < DTILES; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha;
+    }
+    // P = exp(sc - m) (invalid -> 0), packed bf16x2 per tile quad
+    unsigned int ppk[8];   // tile pt -> 2 uints (4 bf16 = quads r0..3)
+    float psum = 0.f;
+#pragma unroll
+    for (int pt = 0; pt < 4; ++pt) {
+      float p0f = 0.f, p1f = 0.f, p2f = 0.f, p3f = 0.f;
+      if (sc[pt][0] > -1.0e38f) p0f = __expf(sc[pt][0] - m_new);
+      if (sc[pt][1] > -1.0e38f) p1f = __expf(sc[pt][1] - m_new);
+      if (sc[pt][2] > -1.0e38f) p2f = __expf(sc[pt][2] - m_new);
+      if (sc[pt][3] > -1.0e38f) p3f = __expf(sc[pt][3] - m_new);
+      psum += p0f + p1f + p2f + p3f;
+      ppk[pt * 2] = f32x2_to_bf16x2(p0f, p1f);
+      ppk[pt * 2 + 1] = f32x2_to_bf16x2(p2f, p3f);
+    }
+    psum += __shfl_xor(psum, 16, QSA_WAVE);
+    psum += __shfl_xor(psum, 32, QSA_WAVE);
+    lsum = lsum * alpha + psum;
+
+    // ---- PV: two 32-pos halves ---------------------------------------
+    // A-frag needs lane (head=col, hi) to hold P[head][pos = hi*8 + e]:
+    // packed pairs come from source lanes (same col) with hi' = (2*hi)&3
+    // and (2*hi+1)&3, tile = half*2 + (hi>>1).
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      // lane (col, hi) needs P[head=col][pos = half*32 + hi*8 + e]:
+      // tile T = half*2 + (hi>>1), quads (hi&1)*2 and (hi&1)*2+1 — i.e.
+      // source lanes (col, g) with g = (2*hi)&3 and (2*hi+1)&3.  __shfl
+      // evaluates the value expression in EACH lane, so the register
+      // index must be compile-time: shuffle both candidate tiles'
+      // packed regs and select by hi (guide rule 20: runtime-indexed
+      // arrays spill to scratch; and a runtime index would read the
+      // SOURCE lane's differently-computed tile).
+      const int src_a = col + (((2 * hi) & 3) << 4);
+      const int src_b = col + (((2 * hi + 1) & 3) << 4);
+      const int tA = half * 2;           // tile for hi 0,1
+      const int tB = half * 2 + 1;       // tile for hi 2,3
+      const unsigned int a0A = __shfl(ppk[tA * 2], src_a, QSA_WAVE);
+      const unsigned int a1A = __shfl(ppk[tA * 2 + 1], src_a, QSA_WAVE);
+      const unsigned int b0A = __shfl(ppk[tA * 2], src_b, QSA_WAVE);
+      const unsigned int b1A = __shfl(ppk[tA * 2 + 1], src_b, QSA_WAVE);
+      const unsigned int a0B = __shfl(ppk[tB * 2], src_a, QSA_WAVE);
+      const unsigned int a1B = __shfl(ppk[tB * 2 + 1], src_a, QSA_WAVE);
+      const unsigned int b0B = __shfl(ppk[tB * 2], src_b, QSA_WAVE);
+      const unsigned int b1B = __shfl(ppk[tB * 2 + 1], src_b, QSA_WAVE);
+      const bool lo = hi < 2;
+      bf16x8_a pa;
+      unsigned int* pa_u = reinterpret_cast<unsigned int*>(&pa);
+      pa_u[0] = lo ? a0A : a0B;
+      pa_u[1] = lo ? a1A : a1B;
+      pa_u[2] = lo ? b0A : b0B;
+      pa_u[3] = lo ? b1A : b1B;
+#pragma unroll
+      for (int dt = 0; dt < DTILES; ++dt) {
+        // B-frag: lane holds V^T[pos = half*32 + hi*8 + e][d = dt*16+col]
+        // = vc[.., d, pos]: 16 B at (d*64 + half*32 + hi*8)
+        const bf16x8_a vf = *reinterpret_cast<const bf16x8_a*>(
+            vbase + ((long long)(dt * 16 + col) * QSA_PAGE) + half * 32 +
+            hi * 8);
+        o_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vf,
+                                                            o_acc[dt],
+                                                            0, 0, 0);
+      }
+    }
+  }
+
+  // ---- store partials: O C-layout row = head = hi*4 + r, col = dim ----
+  // lane (hi, col): holds O[head hi*4+r][d = dt*16 + col]
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int head = hi * 4 + r;
+    if (head < R) {
+      float* po = part_o +
+          (((long long)b * QH + kvh * R + head) * NS + split) * D;
+#pragma unroll
+      for (int dt = 0; dt < DTILES; ++dt)
+        po[dt * 16 + col] = o_acc[dt][r];
+    }
+  }
+  if (lane < R)
+    *reinterpret_cast<float2*>(pml_base + (long long)lane * NS * 2) =
+        make_float2(m, lsum);
 }
 
 template <int D>
@@ -276,101 +243,37 @@ qsa_attn_reduce(const float* __restrict__ part_o,   // [rows, NS, D]
   out[row * D + d] = f32_to_bf16(v);
 }
 
-extern "C" void qsa_paged_attn_decode_launch(
-    const unsigned short* q, const unsigned short* kc,
-    const unsigned short* vc, const int* block_table, const int* seq_lens,
-    unsigned short* out, float scale, int B, int QH, int KVH, int max_pages,
-    int D, long long qstride, hipStream_t stream) {
-  dim3 grid(B * KVH);
-  dim3 block(256);
-  if (D == 128) {
-    hipLaunchKernelGGL((qsa_paged_attn_decode<128>), grid, block, 0, stream,
-                       q, kc, vc, block_table, seq_lens, out, scale, B, QH,
-                       KVH, max_pages, qstride);
-  } else if (D == 64) {
-    hipLaunchKernelGGL((qsa_paged_attn_decode<64>), grid, block, 0, stream,
-                       q, kc, vc, block_table, seq_lens, out, scale, B, QH,
-                       KVH, max_pages, qstride);
-  }
-}
-
-extern "C" void qsa_paged_attn_decode_split_launch(
+extern "C" void qsa_paged_attn_mfma_launch(
     const unsigned short* q, const unsigned short* kc,
     const unsigned short* vc, const int* block_table, const int* seq_lens,
     float* part_o, float* part_ml, unsigned short* out, float scale, int B,
     int QH, int KVH, int max_pages, int D, long long qstride, int NS,
     hipStream_t stream) {
-  dim3 grid(B * KVH * NS);
+  const int nsb = (NS + 3) / 4;
+  dim3 grid(B * KVH * nsb);
   dim3 block(256);
   if (D == 128) {
-    hipLaunchKernelGGL((qsa_paged_attn_decode_split<128>), grid, block, 0,
-                       stream, q, kc, vc, block_table, seq_lens, part_o,
-                       part_ml, scale, B, QH, KVH, max_pages, qstride, NS);
+    hipLaunchKernelGGL((qsa_paged_attn_mfma<128>), grid, block, 0, stream,
+                       q, kc, vc, block_table, seq_lens, part_o, part_ml,
+                       scale, B, QH, KVH, max_pages, qstride, NS);
     hipLaunchKernelGGL((qsa_attn_reduce<128>), dim3(B * QH), dim3(128), 0,
                        stream, part_o, part_ml, out, NS);
   } else if (D == 64) {
-    hipLaunchKernelGGL((qsa_paged_attn_decode_split<64>), grid, block, 0,
-                       stream, q, kc, vc, block_table, seq_lens, part_o,
-                       part_ml, scale, B, QH, KVH, max_pages, qstride, NS);
+    hipLaunchKernelGGL((qsa_paged_attn_mfma<64>), grid, block, 0, stream,
+                       q, kc, vc, block_table, seq_lens, part_o, part_ml,
+                       scale, B, QH, KVH, max_pages, qstride, NS);
     hipLaunchKernelGGL((qsa_attn_reduce<64>), dim3(B * QH), dim3(64), 0,
                        stream, part_o, part_ml, out, NS);
   }
 }
 
 // ---------------------------------------------------------------------------
-// KV-cache append: scatter the step's new k/v [B, KVH, D] into the paged
-// cache at position seq_lens[b]-1 (called after RoPE, before attention).
-// One block per (b, kvh); D threads.
+// KV-cache writers.  K layout [page, kvh, D/8, 64, 8]; V layout
+// [page, kvh, D, 64] (transposed for the PV B-fragment stream).
 // ---------------------------------------------------------------------------
-__global__ void
-qsa_kv_append(const unsigned short* __restrict__ knew,  // [B, KVH, D] (row stride kvstride)
-              const unsigned short* __restrict__ vnew,
-              unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
-              const int* __restrict__ block_table, const int* __restrict__ seq_lens,
-              int B, int KVH, int D, int max_pages, long long kvstride) {
-  const int b = blockIdx.x / KVH;
-  const int kvh = blockIdx.x % KVH;
-  const int d = threadIdx.x;
-  if (d >= D) return;
-  const int pos = seq_lens[b] - 1;
-  if (pos < 0) return;
-  const int page = block_table[(long long)b * max_pages + pos / QSA_PAGE];
-  const int pin = pos % QSA_PAGE;
-  const unsigned short kv = knew[(long long)b * kvstride + (long long)kvh * D + d];
-  const unsigned short vv = vnew[(long long)b * kvstride + (long long)kvh * D + d];
-  // K layout [page, kvh, D/8, 64, 8]
-  kc[((((long long)page * KVH + kvh) * (D / 8) + d / 8) * QSA_PAGE + pin) * 8 +
-     d % 8] = kv;
-  // V layout [page, kvh, 64, D]
-  vc[(((long long)page * KVH + kvh) * QSA_PAGE + pin) * D + d] = vv;
-}
-
-// Prefill bulk variant: scatter T tokens' k/v [T, KVH, D] given their
-// (seq position) mapping to pages via per-token slot ids precomputed on host:
-// slot[t] = page * 64 + offset.
-__global__ void
-qsa_kv_scatter(const unsigned short* __restrict__ knew,  // [T, KVH, D]
-               const unsigned short* __restrict__ vnew,
-               unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
-               const int* __restrict__ slots,  // [T]
-               int T, int KVH, int D) {
-  const long long t = blockIdx.x / KVH;
-  const int kvh = blockIdx.x % KVH;
-  const int d = threadIdx.x;
-  if (t >= T || d >= D) return;
-  const int slot = slots[t];
-  const long long page = slot / QSA_PAGE;
-  const int pin = slot % QSA_PAGE;
-  const unsigned short kv = knew[(t * KVH + kvh) * D + d];
-  const unsigned short vv = vnew[(t * KVH + kvh) * D + d];
-  kc[(((page * KVH + kvh) * (D / 8) + d / 8) * QSA_PAGE + pin) * 8 + d % 8] = kv;
-  vc[((page * KVH + kvh) * QSA_PAGE + pin) * D + d] = vv;
-}
 
 // Fused decode-step RoPE + KV append: rotates q in place, rotates k and
-// writes it (and v) STRAIGHT into the paged cache — one kernel instead of
-// rope + kv_append, and k never round-trips through HBM in its pre-cache
-// form.  One block per (b, head) over QH q-heads + KVH kv-heads.
+// writes it (and v) STRAIGHT into the paged cache.
 __global__ void
 qsa_rope_kv_append(unsigned short* __restrict__ q,        // [B, QH, D]
                    const unsigned short* __restrict__ k,  // [B, KVH, D]
@@ -403,8 +306,6 @@ qsa_rope_kv_append(unsigned short* __restrict__ q,        // [B, QH, D]
   const int kvh = h - QH;
   const int page = block_table[(long long)b * max_pages + pos / QSA_PAGE];
   const int pin = pos % QSA_PAGE;
-  // k: rope pair (d, d+half) computed per thread d < half; threads
-  // [half, D) carry the rotated upper half (recomputed: cheap VALU)
   const unsigned short* kbase =
       k + (long long)b * kvstride + (long long)kvh * D;
   const int dl = d % half;
@@ -413,7 +314,7 @@ qsa_rope_kv_append(unsigned short* __restrict__ q,        // [B, QH, D]
   const float kr = (d < half) ? (x0 * c - x1 * s) : (x0 * s + x1 * c);
   kc[((((long long)page * KVH + kvh) * (D / 8) + d / 8) * QSA_PAGE + pin) * 8 +
      d % 8] = f32_to_bf16(kr);
-  vc[(((long long)page * KVH + kvh) * QSA_PAGE + pin) * D + d] =
+  vc[(((long long)page * KVH + kvh) * D + d) * QSA_PAGE + pin] =
       v[(long long)b * kvstride + (long long)kvh * D + d];
 }
 
@@ -426,6 +327,50 @@ extern "C" void qsa_rope_kv_append_launch(
   hipLaunchKernelGGL(qsa_rope_kv_append, dim3(B * (QH + KVH)), dim3(D), 0,
                      stream, q, k, v, kc, vc, cos_t, sin_t, block_table,
                      seq_lens, B, QH, KVH, D, max_pages, qstride, kvstride);
+}
+
+// Un-fused single-step append (numerics tests / CPU-parity surface).
+__global__ void
+qsa_kv_append(const unsigned short* __restrict__ knew,  // [B, KVH, D]
+              const unsigned short* __restrict__ vnew,
+              unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
+              const int* __restrict__ block_table, const int* __restrict__ seq_lens,
+              int B, int KVH, int D, int max_pages, long long kvstride) {
+  const int b = blockIdx.x / KVH;
+  const int kvh = blockIdx.x % KVH;
+  const int d = threadIdx.x;
+  if (d >= D) return;
+  const int pos = seq_lens[b] - 1;
+  if (pos < 0) return;
+  const int page = block_table[(long long)b * max_pages + pos / QSA_PAGE];
+  const int pin = pos % QSA_PAGE;
+  const unsigned short kv = knew[(long long)b * kvstride + (long long)kvh * D + d];
+  const unsigned short vv = vnew[(long long)b * kvstride + (long long)kvh * D + d];
+  kc[((((long long)page * KVH + kvh) * (D / 8) + d / 8) * QSA_PAGE + pin) * 8 +
+     d % 8] = kv;
+  vc[(((long long)page * KVH + kvh) * D + d) * QSA_PAGE + pin] = vv;
+}
+
+// Prefill bulk scatter: T tokens' k/v by precomputed slot ids
+// (slot = page*64 + offset).  The V^T write is a 2-byte-per-thread scatter;
+// consecutive tokens of a page share cache lines per d, so L2 absorbs it.
+__global__ void
+qsa_kv_scatter(const unsigned short* __restrict__ knew,  // [T, KVH, D]
+               const unsigned short* __restrict__ vnew,
+               unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
+               const int* __restrict__ slots,  // [T]
+               int T, int KVH, int D) {
+  const long long t = blockIdx.x / KVH;
+  const int kvh = blockIdx.x % KVH;
+  const int d = threadIdx.x;
+  if (t >= T || d >= D) return;
+  const int slot = slots[t];
+  const long long page = slot / QSA_PAGE;
+  const int pin = slot % QSA_PAGE;
+  const unsigned short kv = knew[(t * KVH + kvh) * D + d];
+  const unsigned short vv = vnew[(t * KVH + kvh) * D + d];
+  kc[(((page * KVH + kvh) * (D / 8) + d / 8) * QSA_PAGE + pin) * 8 + d % 8] = kv;
+  vc[((page * KVH + kvh) * D + d) * QSA_PAGE + pin] = vv;
 }
 
 extern "C" void qsa_kv_append_launch(const unsigned short* knew,

@@ -78,7 +78,7 @@ def _make_paged_kv(B, KVH, D, seq_lens, npages_total):
     torch.manual_seed(7)
     kc = torch.randn(npages_total, KVH, D // 8, 64, 8, dtype=torch.bfloat16,
                      device=dev())
-    vc = torch.randn(npages_total, KVH, 64, D, dtype=torch.bfloat16, device=dev())
+    vc = torch.randn(npages_total, KVH, D, 64, dtype=torch.bfloat16, device=dev())
     max_pages = max((s + 63) // 64 for s in seq_lens)
     bt = torch.zeros(B, max_pages, dtype=torch.int32, device=dev())
     nxt = 0
@@ -115,7 +115,7 @@ def test_kv_append_then_attend(ext):
         pos = s - 1
         page = int(bt[b, pos // 64])
         pin = pos % 64
-        got_v = vc[page, :, pin, :].float().cpu()
+        got_v = vc[page, :, :, pin].float().cpu()
         torch.testing.assert_close(got_v, vnew[b].float().cpu(), atol=1e-3,
                                    rtol=1e-3)
         got_k = kc[page, :, :, pin, :].reshape(KVH, D).float().cpu()
@@ -127,14 +127,14 @@ def test_kv_scatter(ext):
     KVH, D = 2, 128
     T = 100
     kc = torch.zeros(4, KVH, D // 8, 64, 8, dtype=torch.bfloat16, device=dev())
-    vc = torch.zeros(4, KVH, 64, D, dtype=torch.bfloat16, device=dev())
+    vc = torch.zeros(4, KVH, D, 64, dtype=torch.bfloat16, device=dev())
     knew = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev())
     vnew = torch.randn(T, KVH, D, dtype=torch.bfloat16, device=dev())
     slots = torch.arange(T, dtype=torch.int32, device=dev()) + 28
     ext.kv_scatter(knew, vnew, kc, vc, slots)
     t = 40
     slot = int(slots[t])
-    got = vc[slot // 64, :, slot % 64, :].float().cpu()
+    got = vc[slot // 64, :, :, slot % 64].float().cpu()
     torch.testing.assert_close(got, vnew[t].float().cpu(), atol=1e-3, rtol=1e-3)
 
 
