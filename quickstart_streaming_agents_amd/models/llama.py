@@ -187,12 +187,14 @@ class LlamaModel:
 
     # ------------------------------------------------------------------
     def _split_qkv(self, qkv: torch.Tensor, n: int):
+        """Strided [n, H, D] views into the fused qkv buffer — the rope /
+        kv_scatter kernels take row strides, no copies."""
         c = self.cfg
         qd = self.n_q * c.d_head
         kd = self.n_kv * c.d_head
-        q = qkv[:, :qd].reshape(n, self.n_q, c.d_head).contiguous()
-        k = qkv[:, qd:qd + kd].reshape(n, self.n_kv, c.d_head).contiguous()
-        v = qkv[:, qd + kd:].reshape(n, self.n_kv, c.d_head).contiguous()
+        q = qkv[:, :qd].view(n, self.n_q, c.d_head)
+        k = qkv[:, qd:qd + kd].view(n, self.n_kv, c.d_head)
+        v = qkv[:, qd + kd:].view(n, self.n_kv, c.d_head)
         return q, k, v
 
     @torch.no_grad()

@@ -53,7 +53,7 @@ extern "C" void qsa_rope_kv_append_launch(unsigned short*,
 extern "C" void qsa_kv_scatter_launch(const unsigned short*,
                                       const unsigned short*, unsigned short*,
                                       unsigned short*, const int*, int, int,
-                                      int, hipStream_t);
+                                      int, long long, hipStream_t);
 extern "C" void qsa_skinny_gemm_launch(const unsigned short*,
                                        const unsigned short*, unsigned short*,
                                        int, int, long long, long long,
@@ -223,11 +223,14 @@ void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 
 void kv_scatter(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kc,
                 torch::Tensor vc, torch::Tensor slots) {
-  CHK_DEV(knew); CHK_CONT(knew); CHK_BF16(knew); CHK_BF16(vnew); CHK_I32(slots);
+  CHK_DEV(knew); CHK_BF16(knew); CHK_BF16(vnew); CHK_I32(slots);
   const int T = knew.size(0), KVH = knew.size(1), D = knew.size(2);
+  chk_hd_strided(knew, D, "knew"); chk_hd_strided(vnew, D, "vnew");
+  TORCH_CHECK(knew.stride(0) == vnew.stride(0), "k/v strides must match");
   if (T == 0) return;
   qsa_kv_scatter_launch(u16(knew), u16(vnew), u16m(kc), u16m(vc),
-                        slots.data_ptr<int>(), T, KVH, D, cur_stream());
+                        slots.data_ptr<int>(), T, KVH, D, knew.stride(0),
+                        cur_stream());
 }
 
 torch::Tensor pack_weight_frag(torch::Tensor w) {

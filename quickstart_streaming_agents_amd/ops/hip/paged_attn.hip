@@ -355,11 +355,11 @@ qsa_kv_append(const unsigned short* __restrict__ knew,  // [B, KVH, D]
 // (slot = page*64 + offset).  The V^T write is a 2-byte-per-thread scatter;
 // consecutive tokens of a page share cache lines per d, so L2 absorbs it.
 __global__ void
-qsa_kv_scatter(const unsigned short* __restrict__ knew,  // [T, KVH, D]
+qsa_kv_scatter(const unsigned short* __restrict__ knew,  // [T, KVH, D] (row stride kvstride)
                const unsigned short* __restrict__ vnew,
                unsigned short* __restrict__ kc, unsigned short* __restrict__ vc,
                const int* __restrict__ slots,  // [T]
-               int T, int KVH, int D) {
+               int T, int KVH, int D, long long kvstride) {
   const long long t = blockIdx.x / KVH;
   const int kvh = blockIdx.x % KVH;
   const int d = threadIdx.x;
@@ -367,8 +367,8 @@ qsa_kv_scatter(const unsigned short* __restrict__ knew,  // [T, KVH, D]
   const int slot = slots[t];
   const long long page = slot / QSA_PAGE;
   const int pin = slot % QSA_PAGE;
-  const unsigned short kv = knew[(t * KVH + kvh) * D + d];
-  const unsigned short vv = vnew[(t * KVH + kvh) * D + d];
+  const unsigned short kv = knew[t * kvstride + (long long)kvh * D + d];
+  const unsigned short vv = vnew[t * kvstride + (long long)kvh * D + d];
   kc[(((page * KVH + kvh) * (D / 8) + d / 8) * QSA_PAGE + pin) * 8 + d % 8] = kv;
   vc[((page * KVH + kvh) * D + d) * QSA_PAGE + pin] = vv;
 }
@@ -389,7 +389,7 @@ extern "C" void qsa_kv_scatter_launch(const unsigned short* knew,
                                       const unsigned short* vnew,
                                       unsigned short* kc, unsigned short* vc,
                                       const int* slots, int T, int KVH, int D,
-                                      hipStream_t stream) {
+                                      long long kvstride, hipStream_t stream) {
   hipLaunchKernelGGL(qsa_kv_scatter, dim3((long long)T * KVH), dim3(D), 0,
-                     stream, knew, vnew, kc, vc, slots, T, KVH, D);
+                     stream, knew, vnew, kc, vc, slots, T, KVH, D, kvstride);
 }
