@@ -281,6 +281,33 @@ class LlamaModel:
             kv.slot_ids(sid, s, n)
             for (_, sid, s), n in zip(items, lens)]).contiguous()
 
+        # ---- padded-batch attention prep (once per prefill call) --------
+        # Per-sequence attention loops were launch- and small-GEMM-bound;
+        # padding to n_max turns each layer's attention into ONE bmm pair +
+        # ONE masked-softmax launch across all sequences.  Per-row limits
+        # (start_i + qpos + 1, or 0 on pad rows) give exact causal + ragged
+        # masking; the softmax kernel zeroes beyond-limit columns.
+        R = self.n_q // self.n_kv
+        nb = len(items)
+        nmax = max(lens)
+        npmax = max((int(items[i][2]) + lens[i] + kv.PAGE - 1) // kv.PAGE
+                    for i in range(nb))
+        ctxp = npmax * kv.PAGE
+        dst_idx = torch.tensor(
+            [i * nmax + p for i, n in enumerate(lens) for p in range(n)],
+            dtype=torch.int64, device=dev)
+        page_idx = torch.zeros(nb, npmax, dtype=torch.int64)
+        for i, (_, sid, _) in enumerate(items):
+            pages = kv._seq_pages[sid]
+            page_idx[i, :len(pages)] = torch.tensor(pages,
+                                                    dtype=torch.int64)
+        flat_pages = page_idx.reshape(-1).to(dev)
+        limits = torch.zeros(nb, self.n_q, nmax, dtype=torch.int32)
+        for i, ((_, _, start), n) in enumerate(zip(items, lens)):
+            lim = torch.arange(start + 1, start + n + 1, dtype=torch.int32)
+            limits[i, :, :n] = lim.unsqueeze(0)
+        row_limits = limits.reshape(-1).to(dev)
+
         res = self.embed.index_select(0, tokens).contiguous()
         h = None
         mlp_out = None
@@ -294,32 +321,32 @@ class LlamaModel:
             q, k, v = self._split_qkv(qkv, T)
             D.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
             D.kv_scatter(k, v, kv.k[li], kv.v[li], slots)
-            attn = torch.empty(T, self.n_q * c.d_head, dtype=self.dtype,
-                               device=dev)
-            for (tt, sid, start), n, off in zip(items, lens, offs):
-                ctx = start + n
-                if start == 0:
-                    kseq = k[off:off + n].permute(1, 0, 2)
-                    vseq = v[off:off + n].permute(1, 0, 2)
-                else:
-                    kseq, vseq = self._gather_kv(kv.k[li], kv.v[li],
-                                                 kv._seq_pages[sid], ctx)
-                # GQA grouped: [KVH, R*n, D] x [KVH, ctx, D]; bf16 MFMA
-                # GEMMs (f32 bmm is 1/16 the MFMA rate on CDNA4)
-                R = self.n_q // self.n_kv
-                qf = q[off:off + n].permute(1, 0, 2).reshape(
-                    self.n_kv, R * n, c.d_head)
-                kf = kseq.contiguous()                         # [KVH, ctx, D]
-                vf = vseq.contiguous()
-                scores = (torch.bmm(qf, kf.transpose(1, 2)).float()
-                          * self.scale) \
-                    .reshape(self.n_q * n, ctx).contiguous()
-                D.softmax_rows_(scores, start, True, n, None)
-                probs = scores.reshape(self.n_kv, R * n, ctx) \
-                    .to(self.dtype)
-                a = torch.bmm(probs, vf)
-                attn[off:off + n] = a.reshape(self.n_q, n, c.d_head) \
-                    .permute(1, 0, 2).reshape(n, -1)
+            # padded q: [nb*nmax, n_q, D] -> [nb*KVH, R*nmax, D]
+            q_pad = torch.zeros(nb * nmax, self.n_q, c.d_head,
+                                dtype=self.dtype, device=dev)
+            q_pad.index_copy_(0, dst_idx, q)
+            qb = q_pad.view(nb, nmax, self.n_kv, R, c.d_head) \
+                .permute(0, 2, 3, 1, 4) \
+                .reshape(nb * self.n_kv, R * nmax, c.d_head)
+            # gather padded K/V from the paged cache (k/v just scattered)
+            ksel = kv.k[li].index_select(0, flat_pages) \
+                .view(nb, npmax, self.n_kv, c.d_head // 8, kv.PAGE, 8) \
+                .permute(0, 2, 1, 4, 3, 5) \
+                .reshape(nb * self.n_kv, ctxp, c.d_head)
+            vsel = kv.v[li].index_select(0, flat_pages) \
+                .view(nb, npmax, self.n_kv, c.d_head, kv.PAGE) \
+                .permute(0, 2, 1, 4, 3) \
+                .reshape(nb * self.n_kv, ctxp, c.d_head)
+            scores = (torch.bmm(qb, ksel.transpose(1, 2)).float()
+                      * self.scale).reshape(-1, ctxp).contiguous()
+            D.softmax_rows_(scores, 0, False, 0, row_limits)
+            probs = scores.reshape(nb * self.n_kv, R * nmax, ctxp) \
+                .to(self.dtype)
+            a = torch.bmm(probs, vsel)
+            attn = a.view(nb, self.n_kv, R, nmax, c.d_head) \
+                .permute(0, 3, 1, 2, 4) \
+                .reshape(nb * nmax, self.n_q * c.d_head) \
+                .index_select(0, dst_idx)
             o = self._tp_all_reduce(F.linear(attn, L["wo"]))
             h = D.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             mlp_out = self._ffn(L, h)
