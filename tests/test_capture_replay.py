@@ -57,3 +57,46 @@ def test_replay_purges_first(tmp_path):
                         "request_ts", window_ms=MIN5,
                         now_ms=1_800_000_000_000)
     assert b.topic("ride_requests").message_count() == n
+
+
+def test_publish_docs_and_sql_extract(tmp_path):
+    """publish_docs.py + sql_extractors.py parity: frontmatter chunking to
+    the documents topic; ```sql extraction with no-parse opt-out."""
+    from quickstart_streaming_agents_amd.labs.docs import (doc_records,
+                                                           publish_docs)
+    from quickstart_streaming_agents_amd.sql.extract import (
+        extract_sql_blocks, extract_statements)
+    md = tmp_path / "guide.md"
+    md.write_text(
+        "---\n"
+        "title: Flink Windows\n"
+        "pages: 3-5\n"
+        "fraud_categories: [dup, identity]\n"
+        "---\n"
+        "# Windows\n" + ("tumbling windows close on watermark. " * 30) +
+        "\n# Joins\nstate ttl evicts rows.\n")
+    recs = doc_records(str(md), max_chars=400)
+    assert len(recs) >= 2
+    assert recs[0]["title"] == "Flink Windows"
+    assert recs[0]["fraud_categories"] == ["dup", "identity"]
+    assert all(r["char_count"] == len(r["chunk"]) for r in recs)
+
+    b = Broker()
+    n = publish_docs(b, [str(md)], max_chars=400)
+    assert b.topic("documents").message_count() == n == len(recs)
+    # index ingests the published docs
+    from quickstart_streaming_agents_amd.labs.pipelines import lab2_build_index
+    from quickstart_streaming_agents_amd.vector.index import HashingEmbedder
+    idx = lab2_build_index(b, HashingEmbedder())
+    hits = idx.search(HashingEmbedder().embed("tumbling windows watermark"), 2)
+    assert hits and "window" in hits[0].chunk
+
+    walkthrough = (
+        "Intro\n```sql\nCREATE TABLE t1 (c STRING);\nSET 'a' = 'b';\n```\n"
+        "skip this:\n```sql no-parse\nDROP TABLE nope;\n```\n"
+        "```SQL\nCREATE TABLE t2 (d STRING);\n```\n")
+    blocks = extract_sql_blocks(walkthrough)
+    assert len(blocks) == 2
+    stmts = extract_statements(walkthrough)
+    assert len(stmts) == 3
+    assert "DROP" not in " ".join(stmts)
