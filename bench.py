@@ -38,8 +38,9 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
-    p.add_argument("--batch", type=int, default=24,
-                   help="decisions per step per GPU")
+    p.add_argument("--batch", type=int, default=128,
+                   help="decisions per step per GPU (concurrent agent "
+                        "episodes sharing batched decode)")
     p.add_argument("--model", default="llama3-8b")
     p.add_argument("--decode-tokens", type=int, default=64,
                    help="decode tokens per agent LLM turn")

@@ -197,6 +197,14 @@ class Engine:
 
     # ------------------------------------------------------------------
     def _admit(self) -> None:
+        # group like-sized prefills: padded-batch attention pads every
+        # admitted sequence to the batch max delta, so sort pending by
+        # (has-cached-prefix, delta desc) — fresh long prompts batch with
+        # each other, short continuation deltas (large ctx, tiny delta)
+        # batch together, minimizing both pad waste and KV-gather width
+        self.pending.sort(
+            key=lambda s: (s.cached_len > 0,
+                           -(len(s.prompt) - s.cached_len)))
         batch_items: list[tuple[torch.Tensor, int, int]] = []
         admitted: list[Sequence] = []
         new_tokens = 0
