@@ -59,7 +59,7 @@ _TIMING = _os.environ.get("QSA_TIMING", "") == "1"
 class Engine:
     def __init__(self, model: LlamaModel, kv_pages: int | None = None,
                  max_batch: int = 256, max_seq_len: int = 4096,
-                 eos_id: int | None = None, prefill_batch_tokens: int = 32768):
+                 eos_id: int | None = None, prefill_batch_tokens: int = 65536):
         self.model = model
         self.max_batch = max_batch
         self.max_seq_len = max_seq_len
