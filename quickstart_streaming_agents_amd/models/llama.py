@@ -337,11 +337,11 @@ class LlamaModel:
                 .view(nb, npmax, self.n_kv, c.d_head, kv.PAGE) \
                 .permute(0, 2, 1, 4, 3) \
                 .reshape(nb * self.n_kv, ctxp, c.d_head)
-            scores = (torch.bmm(qb, ksel.transpose(1, 2)).float()
-                      * self.scale).reshape(-1, ctxp).contiguous()
-            D.softmax_rows_(scores, 0, False, 0, row_limits)
-            probs = scores.reshape(nb * self.n_kv, R * nmax, ctxp) \
-                .to(self.dtype)
+            # bf16 scores + fused-scale masked softmax: no f32 round trip
+            scores = torch.bmm(qb, ksel.transpose(1, 2)) \
+                .reshape(-1, ctxp)
+            D.softmax_rows_bf16_(scores, self.scale, row_limits)
+            probs = scores.reshape(nb * self.n_kv, R * nmax, ctxp)
             a = torch.bmm(probs, vsel)
             attn = a.view(nb, self.n_kv, R, nmax, c.d_head) \
                 .permute(0, 3, 1, 2, 4) \

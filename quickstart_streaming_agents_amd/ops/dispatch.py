@@ -82,6 +82,22 @@ def softmax_rows_(scores: torch.Tensor, col_offset: int = 0,
     scores.copy_(torch.nan_to_num(out, nan=0.0))
 
 
+def softmax_rows_bf16_(scores: torch.Tensor, scale: float,
+                       row_limits: torch.Tensor) -> None:
+    """In-place bf16 masked softmax with folded scale; beyond-limit
+    columns zeroed (prefill padded-batch attention)."""
+    if _cuda(scores):
+        ext().softmax_rows_bf16_(scores, scale, row_limits)
+        return
+    rows, cols = scores.shape
+    s = scores.float() * scale
+    col = torch.arange(cols).unsqueeze(0)
+    masked = s.masked_fill(col >= row_limits.long().unsqueeze(1),
+                           float("-inf"))
+    out = torch.nan_to_num(torch.softmax(masked, dim=-1), nan=0.0)
+    scores.copy_(out.to(scores.dtype))
+
+
 # ---- paged KV -------------------------------------------------------------
 
 def kv_append(knew: torch.Tensor, vnew: torch.Tensor, kc: torch.Tensor,

@@ -33,6 +33,8 @@ extern "C" void qsa_rope_launch(unsigned short*, unsigned short*, const float*,
                                 long long, long long, hipStream_t);
 extern "C" void qsa_softmax_rows_launch(float*, int, int, int, int, int,
                                         const int*, hipStream_t);
+extern "C" void qsa_softmax_rows_bf16_launch(unsigned short*, long long, int,
+                                             float, const int*, hipStream_t);
 extern "C" void qsa_paged_attn_mfma_launch(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const int*, const int*, float*, float*, unsigned short*, float, int, int,
@@ -151,6 +153,17 @@ void softmax_rows_(torch::Tensor scores, long col_offset, bool causal,
   qsa_softmax_rows_launch(scores.data_ptr<float>(), rows, cols,
                           (int)col_offset, causal ? 1 : 0, (int)row_mod, rl,
                           cur_stream());
+}
+
+void softmax_rows_bf16_(torch::Tensor scores, double scale,
+                        torch::Tensor row_limits) {
+  CHK_DEV(scores); CHK_CONT(scores); CHK_BF16(scores); CHK_I32(row_limits);
+  const long long rows = scores.size(0);
+  const int cols = scores.size(1);
+  TORCH_CHECK(cols % 8 == 0, "cols % 8 == 0");
+  TORCH_CHECK(row_limits.numel() == rows, "one limit per row");
+  qsa_softmax_rows_bf16_launch(u16m(scores), rows, cols, (float)scale,
+                               row_limits.data_ptr<int>(), cur_stream());
 }
 
 torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor kc,
@@ -333,6 +346,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused residual-add + RMSNorm (bf16; residual updated in place)");
   m.def("swiglu", &swiglu, "fused silu(gate)*up (bf16)");
   m.def("rope_inplace", &rope_inplace, "rotary embedding in place (bf16)");
+  m.def("softmax_rows_bf16_", &softmax_rows_bf16_,
+        "bf16 masked row softmax with folded scale (in place)");
   m.def("softmax_rows_", &softmax_rows_, "row softmax in place (f32)",
         py::arg("scores"), py::arg("col_offset") = 0, py::arg("causal") = false,
         py::arg("row_mod") = 0, py::arg("row_limits") = py::none());
