@@ -217,3 +217,25 @@ def test_skinny_gemm_strided_rows():
     ref = a.float() @ w.float().T
     assert (out.float() - ref).abs().max().item() < 2e-2 * (
         ref.abs().max().item() + 1e-6)
+
+
+def test_softmax_rows_bf16_matches_ref():
+    from quickstart_streaming_agents_amd.ops import dispatch as D
+    torch.manual_seed(9)
+    rows, cols = 300, 256
+    scores = torch.randn(rows, cols, device="cuda:0",
+                         dtype=torch.bfloat16) * 4
+    limits = torch.randint(0, cols + 1, (rows,), dtype=torch.int32,
+                           device="cuda:0")
+    ref = scores.clone().cpu()
+    lim_cpu = limits.cpu()
+    D.softmax_rows_bf16_(ref, 0.125, lim_cpu)          # CPU reference path
+    D.softmax_rows_bf16_(scores, 0.125, limits)        # HIP kernel
+    got = scores.cpu().float()
+    torch.testing.assert_close(got, ref.float(), atol=2e-2, rtol=2e-2)
+    # beyond-limit strictly zero; rows sum to ~1 where limit > 0
+    for r in (0, 7, 123):
+        lim = int(lim_cpu[r])
+        assert (got[r, lim:] == 0).all()
+        if lim > 0:
+            assert abs(got[r, :lim].sum().item() - 1.0) < 2e-2
