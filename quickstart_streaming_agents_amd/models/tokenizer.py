@@ -11,6 +11,7 @@ from __future__ import annotations
 
 import hashlib
 import re
+from functools import lru_cache
 
 _SPLIT = re.compile(r"[A-Za-z0-9_$]+|[^\sA-Za-z0-9_]")
 
@@ -22,6 +23,7 @@ class HashTokenizer:
     def __init__(self, vocab_size: int = 128_256):
         self.vocab_size = vocab_size
 
+    @lru_cache(maxsize=65536)
     def _tok(self, piece: str) -> int:
         h = hashlib.blake2b(piece.encode(), digest_size=4).digest()
         return self.N_SPECIAL + int.from_bytes(h, "little") % (

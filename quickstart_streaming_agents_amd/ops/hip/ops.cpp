@@ -60,11 +60,6 @@ extern "C" void qsa_skinny_gemm_launch(const unsigned short*,
                                        const unsigned short*, unsigned short*,
                                        int, int, long long, long long,
                                        hipStream_t);
-extern "C" void qsa_skinny_gemm_m128_launch(const unsigned short*,
-                                            const unsigned short*,
-                                            unsigned short*, int, int,
-                                            long long, long long, int,
-                                            hipStream_t);
 extern "C" void qsa_skinny_gemm_probe_launch(const unsigned short*,
                                              const unsigned short*,
                                              unsigned short*, int, int,
@@ -266,27 +261,13 @@ torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor wf, long N, long K) {
   CHK_DEV(a); CHK_BF16(a); CHK_BF16(wf); CHK_CONT(wf);
   TORCH_CHECK(a.dim() == 2 && a.stride(1) == 1, "a rows must be contiguous");
   const int M = a.size(0);
-  TORCH_CHECK(M >= 1 && M <= 128, "skinny_gemm: M in [1,128]");
+  TORCH_CHECK(M >= 1 && M <= 32, "skinny_gemm: M in [1,32]");
   TORCH_CHECK(a.size(1) == K, "K mismatch");
   TORCH_CHECK(K % 256 == 0 && N % 16 == 0, "K%256==0, N%16==0");
   TORCH_CHECK(wf.numel() == (long long)N * K, "wf size");
   auto out = torch::empty({M, (long long)N}, a.options());
-  if (M <= 32)
-    qsa_skinny_gemm_launch(u16(a), u16(wf), u16m(out), M, (int)N, K,
-                           a.stride(0), cur_stream());
-  else
-    qsa_skinny_gemm_m128_launch(u16(a), u16(wf), u16m(out), M, (int)N, K,
-                                a.stride(0), 1, cur_stream());
-  return out;
-}
-
-torch::Tensor skinny_gemm_m128_probe(torch::Tensor a, torch::Tensor wf,
-                                     long N, long K, long nt_loads) {
-  CHK_DEV(a); CHK_BF16(a); CHK_BF16(wf); CHK_CONT(wf);
-  const int M = a.size(0);
-  auto out = torch::empty({M, (long long)N}, a.options());
-  qsa_skinny_gemm_m128_launch(u16(a), u16(wf), u16m(out), M, (int)N, K,
-                              a.stride(0), (int)nt_loads, cur_stream());
+  qsa_skinny_gemm_launch(u16(a), u16(wf), u16m(out), M, (int)N, K,
+                         a.stride(0), cur_stream());
   return out;
 }
 
@@ -380,8 +361,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "repack [N,K] bf16 into MFMA-fragment-major for skinny_gemm");
   m.def("skinny_gemm", &skinny_gemm,
         "decode-batch GEMM (M<=32) on the packed weight stream");
-  m.def("skinny_gemm_m128_probe", &skinny_gemm_m128_probe,
-        "M<=128 LDS-staged variant probe");
   m.def("skinny_gemm_probe", &skinny_gemm_probe,
         "ablation probe: waves/nt/variant sweep");
   m.def("topk_cosine", &topk_cosine, "exact cosine top-k over the HBM index");

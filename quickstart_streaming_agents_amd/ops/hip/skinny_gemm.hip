@@ -118,111 +118,6 @@ qsa_skinny_gemm_t(const unsigned short* __restrict__ A,   // [M,K] stride lda
   }
 }
 
-// ---------------------------------------------------------------------------
-// M <= 128 variant (decode at batch ~128): activations STAGE through LDS
-// once per k-chunk (cooperative coalesced fill, unconditional clamped
-// loads), so the 8 m-tiles' fragment re-reads never touch L2 — the direct
-// A-load design's L2 pressure (8x the W bytes at M=128) is what made it
-// lose.  8 waves each own one k-step of the chunk (8-way K split), W
-// streams nontemporal 1 KiB blocks, accumulators reduce through the SAME
-// LDS buffer after the last chunk (A image is dead by then).
-// ---------------------------------------------------------------------------
-template <bool NT>
-__global__ void __launch_bounds__(512)
-qsa_skinny_gemm_m128(const unsigned short* __restrict__ A,   // [M,K] lda
-                     const unsigned short* __restrict__ Wf,  // fragment-major
-                     unsigned short* __restrict__ Cbf,       // [M, N]
-                     int M, int N, long long K, long long lda) {
-  const int nt = blockIdx.x;              // 16-col n-tile
-  const int wave = threadIdx.x >> 6;
-  const int lane = threadIdx.x & 63;
-  const int tid = threadIdx.x;
-  const long long kchunks = K / QSA_KCH;
-
-  __shared__ unsigned short As[128][QSA_KCH + 8];   // ~66 KiB, reused below
-
-  f32x4 acc[8];
-#pragma unroll
-  for (int mt = 0; mt < 8; ++mt) acc[mt] = (f32x4){0.f, 0.f, 0.f, 0.f};
-
-  const int arow = lane & 15;
-  const int akoff = (lane >> 4) * 8;
-  const unsigned short* wbase =
-      Wf + (long long)nt * (K >> 5) * 512 +
-      (long long)((lane & 15) * 32 + (lane >> 4) * 8);
-
-  for (long long c = 0; c < kchunks; ++c) {
-    // stage A[0:128][c*256 : +256]: 2048 16-B pieces over 512 threads
-#pragma unroll
-    for (int p = 0; p < 4; ++p) {
-      const int piece = tid + p * 512;
-      const int row = piece >> 4;           // 16 pieces per 256-elem row
-      const int off16 = piece & 15;
-      const int rs = min(row, M - 1);       // clamp: no per-load branch
-      *reinterpret_cast<uint4*>(&As[row][off16 * 8]) =
-          *reinterpret_cast<const uint4*>(
-              A + (long long)rs * lda + c * QSA_KCH + off16 * 8);
-    }
-    __syncthreads();
-    const long long kk = c * 8 + wave;      // this wave's k-step
-    bf16x8 w;
-    if (NT)
-      w = __builtin_nontemporal_load(
-          reinterpret_cast<const bf16x8*>(wbase + kk * 512));
-    else
-      w = *reinterpret_cast<const bf16x8*>(wbase + kk * 512);
-    const int la = wave * 32 + akoff;
-#pragma unroll
-    for (int mt = 0; mt < 8; ++mt) {
-      const bf16x8 a =
-          *reinterpret_cast<const bf16x8*>(&As[mt * 16 + arow][la]);
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, w, acc[mt],
-                                                        0, 0, 0);
-    }
-    __syncthreads();
-  }
-
-  // ---- cross-wave reduction through the (dead) A buffer ----------------
-  float* red = reinterpret_cast<float*>(&As[0][0]);  // [8][64][32] f32
-#pragma unroll
-  for (int mt = 0; mt < 8; ++mt) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r)
-      red[((wave * 64 + lane) * 32) + mt * 4 + r] = acc[mt][r];
-  }
-  __syncthreads();
-  if (wave == 0) {
-    const int ncol = nt * 16 + (lane & 15);
-    const int mrow = (lane >> 4) * 4;
-#pragma unroll
-    for (int mt = 0; mt < 8; ++mt) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float v = 0.f;
-#pragma unroll
-        for (int wv = 0; wv < 8; ++wv)
-          v += red[((wv * 64 + lane) * 32) + mt * 4 + r];
-        const int m0 = mt * 16 + mrow + r;
-        if (m0 < M) Cbf[(long long)m0 * N + ncol] = f32_to_bf16(v);
-      }
-    }
-  }
-}
-
-extern "C" void qsa_skinny_gemm_m128_launch(const unsigned short* A,
-                                            const unsigned short* Wf,
-                                            unsigned short* Cbf, int M, int N,
-                                            long long K, long long lda,
-                                            int nt_loads,
-                                            hipStream_t stream) {
-  if (nt_loads)
-    hipLaunchKernelGGL((qsa_skinny_gemm_m128<true>), dim3(N / 16), dim3(512),
-                       0, stream, A, Wf, Cbf, M, N, K, lda);
-  else
-    hipLaunchKernelGGL((qsa_skinny_gemm_m128<false>), dim3(N / 16), dim3(512),
-                       0, stream, A, Wf, Cbf, M, N, K, lda);
-}
-
 extern "C" void qsa_skinny_gemm_launch(const unsigned short* A,
                                        const unsigned short* Wf,
                                        unsigned short* Cbf, int M, int N,
