@@ -214,3 +214,21 @@ def test_lab3_lab4_gpu_anomaly_matches_cpu_and_invariants():
     datagen.publish_lab4(b2, seed=42)
     rows4 = pipelines.lab4_anomalies(b2)
     assert [r["city"] for r in rows4] == ["Naples"]
+
+
+def test_sql_driven_lab1_on_gpu_engine():
+    """Catalog-driven lab1 runs end-to-end with the real decode engine on
+    the GPU (tiny preset): content invariants hold with actual LLM decode
+    in the loop."""
+    from quickstart_streaming_agents_amd.agents.mcp import StubMcpServer
+    from quickstart_streaming_agents_amd.labs.deploy import Deployment
+    dep = Deployment(labs=(1,), device="cuda:0", model="tiny")
+    dep.datagen(1)
+    srv = StubMcpServer().start()
+    try:
+        rows = dep.run(1, mcp_server=srv)
+    finally:
+        srv.stop()
+    assert len(rows) == 10
+    assert all(r["agent_status"] == "SUCCESS" for r in rows)
+    assert any(r["decision"] == "PRICE_MATCH" for r in rows)
