@@ -88,3 +88,27 @@ def test_deployment_destroy():
     dep.datagen(1)
     dep.destroy()
     assert not dep.catalog.tables and not dep.broker.topics
+
+
+def test_walkthrough_sql_roundtrip(tmp_path):
+    """The generated LAB*_SQL_COMMANDS.md artifacts re-parse into the same
+    catalog objects (reference pattern: tests run the SQL extracted from
+    the walkthrough markdown — testing/README.md:208-211)."""
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.extract import extract_statements
+    dep = Deployment(labs=(1, 3))
+    dep.write_summaries(str(tmp_path))
+    cat = Catalog()
+    for lab in (1, 3):
+        md = (tmp_path / f"LAB{lab}_SQL_COMMANDS.md").read_text()
+        stmts = extract_statements(md)
+        assert stmts, f"no SQL extracted for lab{lab}"
+        for s in stmts:
+            from quickstart_streaming_agents_amd.sql import parse as P
+            cat.apply(P.parse_statement(s))
+    spec = cat.agent_spec("price_match_agent")
+    assert spec.tools.allowed_tools == ("http_get", "send_email")
+    spec3 = cat.agent_spec("boat_dispatch_agent")
+    assert spec3.tools.allowed_tools == ("http_get", "http_post")
+    assert cat.ctas_info("anomalies_per_zone").anomaly[0][
+        "minTrainingSize"] == 286
