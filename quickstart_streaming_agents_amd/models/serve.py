@@ -59,8 +59,14 @@ _TIMING = _os.environ.get("QSA_TIMING", "") == "1"
 class Engine:
     def __init__(self, model: LlamaModel, kv_pages: int | None = None,
                  max_batch: int = 256, max_seq_len: int = 4096,
-                 eos_id: int | None = None, prefill_batch_tokens: int = 65536):
+                 eos_id: int | None = None, prefill_batch_tokens: int = 65536,
+                 temperature: float = 0.0):
         self.model = model
+        # temperature 0 = greedy (the deterministic benchmark contract);
+        # > 0 samples via the Gumbel-argmax trick, which stays a single
+        # argmax and is hipGraph-capture-safe (torch captures RNG state
+        # advancement, so replays draw fresh noise)
+        self.temperature = float(temperature)
         self.max_batch = max_batch
         self.max_seq_len = max_seq_len
         self.eos_id = eos_id  # None -> run to max_new_tokens (random weights)
@@ -80,6 +86,16 @@ class Engine:
             getattr(model, "tp_size", 1) == 1
         self._graph = None
         self._gbuf: dict = {}
+
+    def _sample(self, logits: torch.Tensor) -> torch.Tensor:
+        """Greedy at temperature 0; Gumbel-argmax sampling otherwise."""
+        if self.temperature <= 0.0:
+            return torch.argmax(logits, dim=-1)
+        u = torch.rand(logits.shape, device=logits.device,
+                       dtype=torch.float32).clamp_min_(1e-20)
+        gumbel = -torch.log(-torch.log(u).clamp_min_(1e-20))
+        return torch.argmax(logits.float() + self.temperature * gumbel,
+                            dim=-1)
 
     # ---- hipGraph decode -------------------------------------------------
     MAX_RUN = 512  # on-device token-history depth per graph run
@@ -107,7 +123,7 @@ class Engine:
             logits = self.model.forward_decode(
                 gb["tokens"], self.kv, gb["block_table"], gb["seq_lens"],
                 positions)
-            nxt = torch.argmax(logits, dim=-1)
+            nxt = self._sample(logits)
             gb["hist"].index_copy_(0, gb["ctr"], nxt.unsqueeze(0))
             gb["ctr"].add_(1)
             gb["tokens"].copy_(nxt)
@@ -244,7 +260,7 @@ class Engine:
         if _TIMING:
             torch.cuda.synchronize()
             self.stats.prefill_s += time.perf_counter() - _t0
-        first = torch.argmax(logits, dim=-1).tolist()
+        first = self._sample(logits).tolist()
         self.stats.prefill_batches += 1
         for seq, tok in zip(admitted, first):
             seq.out_tokens.append(int(tok))
@@ -291,7 +307,7 @@ class Engine:
             sl = self.kv.seq_lens_tensor(seq_ids)
             logits = self.model.forward_decode(tokens, self.kv, bt, sl,
                                                positions)
-            nxt = torch.argmax(logits, dim=-1).tolist()
+            nxt = self._sample(logits).tolist()
         for s, tok in zip(batch, nxt):
             s.out_tokens.append(int(tok))
             self._maybe_finish(s)

@@ -20,9 +20,17 @@ qsa_rmsnorm_kernel(const unsigned short* __restrict__ x,
   unsigned short* rr = res ? res + row * H : nullptr;
   __shared__ float scratch[8];
 
+  // register-cache the row between the two passes (H <= 8192 with 256
+  // threads = <=4 x uint4 per thread; compile-time indices so the cache
+  // stays in VGPRs, guide rule 20) — the normalize pass then re-reads
+  // nothing, which halves the dependent-latency chain of this small-grid
+  // decode kernel.
+  uint4 buf[4];
   float ss = 0.f;
-  // pass 1: sum of squares (and residual add, kept in f32 via re-read later)
-  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const int i = (threadIdx.x + c * blockDim.x) * 8;
+    if (i >= H) break;
     uint4 v = *reinterpret_cast<const uint4*>(xr + i);
     float acc = 0.f;
     unsigned int pk[4] = {v.x, v.y, v.z, v.w};
@@ -41,6 +49,7 @@ qsa_rmsnorm_kernel(const unsigned short* __restrict__ x,
         po[j] = f32x2_to_bf16x2(s0, s1);
       }
       *reinterpret_cast<uint4*>(rr + i) = outv;  // updated residual stream
+      buf[c] = outv;
     } else {
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
@@ -48,15 +57,51 @@ qsa_rmsnorm_kernel(const unsigned short* __restrict__ x,
         acc = fmaf(a.x, a.x, acc);
         acc = fmaf(a.y, a.y, acc);
       }
+      buf[c] = v;
     }
     ss += acc;
+  }
+  // rows wider than the register cache (H > 8192): uncached tail does the
+  // full pass-1 work (residual add + write) and pass 2 re-reads
+  const bool cached = H <= (int)blockDim.x * 8 * 4;
+  if (!cached) {
+    for (int i = (threadIdx.x + 4 * blockDim.x) * 8; i < H;
+         i += blockDim.x * 8) {
+      uint4 v = *reinterpret_cast<const uint4*>(xr + i);
+      unsigned int pk[4] = {v.x, v.y, v.z, v.w};
+      if (rr) {
+        uint4 rv = *reinterpret_cast<const uint4*>(rr + i);
+        unsigned int rpk[4] = {rv.x, rv.y, rv.z, rv.w};
+        uint4 outv;
+        unsigned int* po = &outv.x;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float2 a = bf16x2_to_f32x2(pk[j]);
+          float2 b = bf16x2_to_f32x2(rpk[j]);
+          float s0 = a.x + b.x, s1 = a.y + b.y;
+          ss = fmaf(s0, s0, ss);
+          ss = fmaf(s1, s1, ss);
+          po[j] = f32x2_to_bf16x2(s0, s1);
+        }
+        *reinterpret_cast<uint4*>(rr + i) = outv;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float2 a = bf16x2_to_f32x2(pk[j]);
+          ss = fmaf(a.x, a.x, ss);
+          ss = fmaf(a.y, a.y, ss);
+        }
+      }
+    }
   }
   ss = block_reduce_sum(ss, scratch);
   const float inv = rsqrtf(ss / (float)H + eps);
 
-  const unsigned short* src = rr ? rr : xr;
-  for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
-    uint4 v = *reinterpret_cast<const uint4*>(src + i);
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const int i = (threadIdx.x + c * blockDim.x) * 8;
+    if (i >= H) break;
+    uint4 v = buf[c];
     uint4 wv = *reinterpret_cast<const uint4*>(w + i);
     unsigned int pk[4] = {v.x, v.y, v.z, v.w};
     unsigned int wk[4] = {wv.x, wv.y, wv.z, wv.w};
@@ -69,6 +114,25 @@ qsa_rmsnorm_kernel(const unsigned short* __restrict__ x,
       po[j] = f32x2_to_bf16x2(a.x * inv * b.x, a.y * inv * b.y);
     }
     *reinterpret_cast<uint4*>(yr + i) = outv;
+  }
+  if (!cached) {
+    const unsigned short* src0 = rr ? rr : xr;
+    for (int i = (threadIdx.x + 4 * blockDim.x) * 8; i < H;
+         i += blockDim.x * 8) {
+      uint4 v = *reinterpret_cast<const uint4*>(src0 + i);
+      uint4 wv = *reinterpret_cast<const uint4*>(w + i);
+      unsigned int pk[4] = {v.x, v.y, v.z, v.w};
+      unsigned int wk[4] = {wv.x, wv.y, wv.z, wv.w};
+      uint4 outv;
+      unsigned int* po = &outv.x;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float2 a = bf16x2_to_f32x2(pk[j]);
+        float2 b = bf16x2_to_f32x2(wk[j]);
+        po[j] = f32x2_to_bf16x2(a.x * inv * b.x, a.y * inv * b.y);
+      }
+      *reinterpret_cast<uint4*>(yr + i) = outv;
+    }
   }
 }
 

@@ -239,3 +239,39 @@ def test_softmax_rows_bf16_matches_ref():
         assert (got[r, lim:] == 0).all()
         if lim > 0:
             assert abs(got[r, :lim].sum().item() - 1.0) < 2e-2
+
+
+def test_rope_kv_append_matches_composed(ext):
+    """Fused rope+append == rope_inplace followed by kv_append."""
+    import copy
+    torch.manual_seed(6)
+    B, QH, KVH, D = 5, 8, 2, 128
+    from quickstart_streaming_agents_amd.ops import cpu_ref
+    cos_t, sin_t = cpu_ref.rope_tables(256, D)
+    cos_t, sin_t = cos_t.to(dev()), sin_t.to(dev())
+    seq_lens = torch.tensor([3, 64, 65, 1, 120], dtype=torch.int32,
+                            device=dev())
+    npages = 3
+    bt = torch.arange(B * npages, dtype=torch.int32,
+                      device=dev()).reshape(B, npages)
+    q = torch.randn(B, QH, D, dtype=torch.bfloat16, device=dev())
+    k = torch.randn(B, KVH, D, dtype=torch.bfloat16, device=dev())
+    v = torch.randn(B, KVH, D, dtype=torch.bfloat16, device=dev())
+    kc1 = torch.zeros(B * npages, KVH, D // 8, 64, 8, dtype=torch.bfloat16,
+                      device=dev())
+    vc1 = torch.zeros(B * npages, KVH, D, 64, dtype=torch.bfloat16,
+                      device=dev())
+    kc2, vc2 = kc1.clone(), vc1.clone()
+
+    # composed reference
+    q2, k2 = q.clone(), k.clone()
+    pos = (seq_lens - 1).clamp(min=0).int()
+    ext.rope_inplace(q2, k2, cos_t, sin_t, pos)
+    ext.kv_append(k2, v, kc2, vc2, bt, seq_lens)
+    # fused
+    q1 = q.clone()
+    ext.rope_kv_append(q1, k, v, kc1, vc1, cos_t, sin_t, bt, seq_lens)
+
+    torch.testing.assert_close(q1, q2)
+    torch.testing.assert_close(kc1, kc2)
+    torch.testing.assert_close(vc1, vc2)

@@ -148,3 +148,29 @@ def test_interval_join_lab4_shape():
                         lambda c: c["city"], lambda a: a["city"],
                         lower_ms=-H6, upper_ms=0)
     assert [r["claim_id"] for r in out] == ["C1"]
+
+
+def test_engine_temperature_sampling_cpu():
+    """temperature > 0 samples (seeded-deterministic, differs from greedy);
+    temperature 0 stays the deterministic greedy contract."""
+    import torch
+
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    from quickstart_streaming_agents_amd.models.serve import Engine
+    model = LlamaModel(LlamaConfig.preset("tiny"), device="cpu",
+                       dtype=torch.float32, seed=3)
+    greedy = Engine(model, max_batch=2, max_seq_len=128)
+    g1 = greedy.generate_batch([[1, 5, 9]], [8])
+    g2 = Engine(model, max_batch=2, max_seq_len=128).generate_batch(
+        [[1, 5, 9]], [8])
+    assert g1 == g2
+
+    torch.manual_seed(0)
+    s1 = Engine(model, max_batch=2, max_seq_len=128,
+                temperature=5.0).generate_batch([[1, 5, 9]], [8])
+    torch.manual_seed(0)
+    s2 = Engine(model, max_batch=2, max_seq_len=128,
+                temperature=5.0).generate_batch([[1, 5, 9]], [8])
+    assert s1 == s2                      # seeded reproducibility
+    assert s1 != g1                      # high temperature diverges
