@@ -172,7 +172,11 @@ class Engine:
             self.stats.decode_s += time.perf_counter() - _t0
         hist = gb["hist"][:run, :n].t().tolist()  # one sync
         for s, toks in zip(batch, hist):
-            s.out_tokens.extend(int(t) for t in toks)
+            new = [int(t) for t in toks]
+            if self.eos_id is not None and self.eos_id in new:
+                # honor mid-run EOS: keep tokens up to and including it
+                new = new[: new.index(self.eos_id) + 1]
+            s.out_tokens.extend(new)
             self.kv._seq_len[s.seq_id] = len(s.prompt) + len(s.out_tokens)
             self._maybe_finish(s)
         self.stats.decode_tokens += n * run

@@ -174,3 +174,23 @@ def test_engine_temperature_sampling_cpu():
                 temperature=5.0).generate_batch([[1, 5, 9]], [8])
     assert s1 == s2                      # seeded reproducibility
     assert s1 != g1                      # high temperature diverges
+
+
+def test_engine_eos_mid_run():
+    """EOS inside a fixed-length decode run truncates the sequence there
+    (both the eager step path and the self-feeding graph path's token
+    history handling)."""
+    import torch
+
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    from quickstart_streaming_agents_amd.models.serve import Engine
+    model = LlamaModel(LlamaConfig.preset("tiny"), device="cpu",
+                       dtype=torch.float32, seed=3)
+    probe = Engine(model, max_batch=2, max_seq_len=128)
+    free = probe.generate_batch([[1, 5, 9]], [12])[0]
+    eos = free[3]  # token the model will emit at step 4
+    eng = Engine(model, max_batch=2, max_seq_len=128, eos_id=eos)
+    out = eng.generate_batch([[1, 5, 9]], [12])[0]
+    assert out == free[:4]
+    assert out[-1] == eos
