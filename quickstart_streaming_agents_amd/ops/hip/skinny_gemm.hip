@@ -148,18 +148,22 @@ qsa_skinny_gemm_m128(const unsigned short* __restrict__ A,   // [M,K] lda
 
   const int arow = lane & 15;
   const int akoff = (lane >> 4) * 8;
-  // stage geometry: 2048 16-B pieces over 512 threads = 4 per thread
-  const int srow[4] = {(tid + 0) >> 4, (tid + 512) >> 4,
-                       (tid + 1024) >> 4, (tid + 1536) >> 4};
-  const int soff[4] = {(tid + 0) & 15, (tid + 512) & 15,
-                       (tid + 1024) & 15, (tid + 1536) & 15};
+  // stage geometry: a 128 x 256-elem chunk = 4096 16-B pieces (32 per
+  // row) over 512 threads = 8 per thread
+  int srow[8], soff[8];
+#pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    const int piece = tid + p * 512;
+    srow[p] = piece >> 5;
+    soff[p] = piece & 31;
+  }
   const unsigned short* wbase =
       Wf + (long long)nt * (K >> 5) * 512 +
       (long long)((lane & 15) * 32 + (lane >> 4) * 8);
 
   auto load_chunk = [&](long long c, uint4* regs) {
 #pragma unroll
-    for (int p = 0; p < 4; ++p) {
+    for (int p = 0; p < 8; ++p) {
       const int rs = min(srow[p], M - 1);
       regs[p] = *reinterpret_cast<const uint4*>(
           A + (long long)rs * lda + c * QSA_KCH + soff[p] * 8);
@@ -167,7 +171,7 @@ qsa_skinny_gemm_m128(const unsigned short* __restrict__ A,   // [M,K] lda
   };
   auto store_chunk = [&](int buf, const uint4* regs) {
 #pragma unroll
-    for (int p = 0; p < 4; ++p)
+    for (int p = 0; p < 8; ++p)
       *reinterpret_cast<uint4*>(&As[buf][srow[p]][soff[p] * 8]) = regs[p];
   };
   auto load_w = [&](long long c) -> bf16x8 {
@@ -179,7 +183,7 @@ qsa_skinny_gemm_m128(const unsigned short* __restrict__ A,   // [M,K] lda
   };
 
   // prologue: chunk 0 -> LDS buf 0; chunk 1 -> regs; W(0) in flight
-  uint4 stage[4];
+  uint4 stage[8];
   load_chunk(0, stage);
   store_chunk(0, stage);
   bf16x8 w_cur = load_w(0);
