@@ -194,3 +194,23 @@ def test_engine_eos_mid_run():
     out = eng.generate_batch([[1, 5, 9]], [12])[0]
     assert out == free[:4]
     assert out[-1] == eos
+
+
+def test_engine_ragged_prefill_batches():
+    """Heavily ragged prompt lengths (1 vs ~90 tokens) through the padded
+    batch prefill: outputs must match running each prompt alone."""
+    import torch
+
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    from quickstart_streaming_agents_amd.models.serve import Engine
+    model = LlamaModel(LlamaConfig.preset("tiny"), device="cpu",
+                       dtype=torch.float32, seed=7)
+    prompts = [[1, 4], [2] + list(range(5, 95)), [3, 9, 9, 9],
+               list(range(30, 80))]
+    batched = Engine(model, max_batch=8,
+                     max_seq_len=256).generate_batch(prompts, [5] * 4)
+    for p, want in zip(prompts, batched):
+        solo = Engine(model, max_batch=1,
+                      max_seq_len=256).generate_batch([p], [5])[0]
+        assert solo == want
