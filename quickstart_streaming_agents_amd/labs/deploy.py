@@ -42,7 +42,11 @@ class Deployment:
         self.broker = Broker()
         self.catalog = Catalog()
         self.indexes: dict[int, VectorIndex] = {}
-        self.embedder = HashingEmbedder()
+        if device.startswith("cuda"):
+            from ..models.encoder import EmbeddingEncoder
+            self.embedder = EmbeddingEncoder(device=device)
+        else:
+            self.embedder = HashingEmbedder()
         self._llm = None
         self.catalog.execute(lab_sql("core"))
         for lab in self.labs:
@@ -87,6 +91,8 @@ class Deployment:
             else:
                 docs = datagen.lab4_policy_docs(seed=self.seed)
             idx.add_documents(docs, self.embedder)
+            if self.device.startswith("cuda"):
+                idx.to_torch(self.device)   # HBM-resident; GPU top-k path
             self.indexes[lab] = idx
         return self.indexes[lab]
 

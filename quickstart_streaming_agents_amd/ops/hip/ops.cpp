@@ -60,11 +60,6 @@ extern "C" void qsa_skinny_gemm_launch(const unsigned short*,
                                        const unsigned short*, unsigned short*,
                                        int, int, long long, long long,
                                        hipStream_t);
-extern "C" void qsa_skinny_gemm_m128_launch(const unsigned short*,
-                                            const unsigned short*,
-                                            unsigned short*, int, int,
-                                            long long, long long, int,
-                                            hipStream_t);
 extern "C" void qsa_skinny_gemm_probe_launch(const unsigned short*,
                                              const unsigned short*,
                                              unsigned short*, int, int,
@@ -276,19 +271,6 @@ torch::Tensor skinny_gemm(torch::Tensor a, torch::Tensor wf, long N, long K) {
   return out;
 }
 
-torch::Tensor skinny_gemm_m128(torch::Tensor a, torch::Tensor wf, long N,
-                               long K, long nt_loads) {
-  CHK_DEV(a); CHK_BF16(a); CHK_BF16(wf); CHK_CONT(wf);
-  TORCH_CHECK(a.dim() == 2 && a.stride(1) == 1, "a rows contiguous");
-  const int M = a.size(0);
-  TORCH_CHECK(M >= 1 && M <= 128, "M in [1,128]");
-  TORCH_CHECK(K % 256 == 0 && N % 16 == 0, "K%256==0, N%16==0");
-  auto out = torch::empty({M, (long long)N}, a.options());
-  qsa_skinny_gemm_m128_launch(u16(a), u16(wf), u16m(out), M, (int)N, K,
-                              a.stride(0), (int)nt_loads, cur_stream());
-  return out;
-}
-
 torch::Tensor skinny_gemm_probe(torch::Tensor a, torch::Tensor wf, long N,
                                 long K, long waves, long nt, long variant) {
   CHK_DEV(a); CHK_BF16(a); CHK_BF16(wf); CHK_CONT(wf);
@@ -379,8 +361,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "repack [N,K] bf16 into MFMA-fragment-major for skinny_gemm");
   m.def("skinny_gemm", &skinny_gemm,
         "decode-batch GEMM (M<=32) on the packed weight stream");
-  m.def("skinny_gemm_m128", &skinny_gemm_m128,
-        "pipelined LDS-staged M<=128 decode GEMM");
   m.def("skinny_gemm_probe", &skinny_gemm_probe,
         "ablation probe: waves/nt/variant sweep");
   m.def("topk_cosine", &topk_cosine, "exact cosine top-k over the HBM index");

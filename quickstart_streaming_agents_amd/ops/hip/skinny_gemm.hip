@@ -118,140 +118,6 @@ qsa_skinny_gemm_t(const unsigned short* __restrict__ A,   // [M,K] stride lda
   }
 }
 
-// ---------------------------------------------------------------------------
-// M <= 128 variant (decode at batch ~128): 8 m-tiles per 16-col n-tile.
-// Direct per-wave A loads would re-read 8x the W bytes through L2; instead
-// A stages through LDS, SOFTWARE-PIPELINED (guide T14): while the 8 waves
-// compute chunk c from LDS buffer c&1, every thread already holds chunk
-// c+1 in registers (loads issued before the compute phase) and the next
-// W block is in flight — the write pass lands after the barrier into the
-// other buffer.  8 waves split the chunk's 8 k-steps; accumulators reduce
-// through LDS buffer 0 at the end (A image dead by then).
-// ---------------------------------------------------------------------------
-template <bool NT>
-__global__ void __launch_bounds__(512)
-qsa_skinny_gemm_m128(const unsigned short* __restrict__ A,   // [M,K] lda
-                     const unsigned short* __restrict__ Wf,  // fragment-major
-                     unsigned short* __restrict__ Cbf,       // [M, N]
-                     int M, int N, long long K, long long lda) {
-  const int nt = blockIdx.x;
-  const int wave = threadIdx.x >> 6;
-  const int lane = threadIdx.x & 63;
-  const int tid = threadIdx.x;
-  const long long kchunks = K / QSA_KCH;
-
-  __shared__ unsigned short As[2][128][QSA_KCH + 8];   // 2 x ~66 KiB
-
-  f32x4 acc[8];
-#pragma unroll
-  for (int mt = 0; mt < 8; ++mt) acc[mt] = (f32x4){0.f, 0.f, 0.f, 0.f};
-
-  const int arow = lane & 15;
-  const int akoff = (lane >> 4) * 8;
-  // stage geometry: a 128 x 256-elem chunk = 4096 16-B pieces (32 per
-  // row) over 512 threads = 8 per thread
-  int srow[8], soff[8];
-#pragma unroll
-  for (int p = 0; p < 8; ++p) {
-    const int piece = tid + p * 512;
-    srow[p] = piece >> 5;
-    soff[p] = piece & 31;
-  }
-  const unsigned short* wbase =
-      Wf + (long long)nt * (K >> 5) * 512 +
-      (long long)((lane & 15) * 32 + (lane >> 4) * 8);
-
-  auto load_chunk = [&](long long c, uint4* regs) {
-#pragma unroll
-    for (int p = 0; p < 8; ++p) {
-      const int rs = min(srow[p], M - 1);
-      regs[p] = *reinterpret_cast<const uint4*>(
-          A + (long long)rs * lda + c * QSA_KCH + soff[p] * 8);
-    }
-  };
-  auto store_chunk = [&](int buf, const uint4* regs) {
-#pragma unroll
-    for (int p = 0; p < 8; ++p)
-      *reinterpret_cast<uint4*>(&As[buf][srow[p]][soff[p] * 8]) = regs[p];
-  };
-  auto load_w = [&](long long c) -> bf16x8 {
-    const long long kk = c * 8 + wave;
-    if (NT)
-      return __builtin_nontemporal_load(
-          reinterpret_cast<const bf16x8*>(wbase + kk * 512));
-    return *reinterpret_cast<const bf16x8*>(wbase + kk * 512);
-  };
-
-  // prologue: chunk 0 -> LDS buf 0; chunk 1 -> regs; W(0) in flight
-  uint4 stage[8];
-  load_chunk(0, stage);
-  store_chunk(0, stage);
-  bf16x8 w_cur = load_w(0);
-  if (kchunks > 1) load_chunk(1, stage);
-  __syncthreads();
-
-  for (long long c = 0; c < kchunks; ++c) {
-    const int buf = (int)(c & 1);
-    bf16x8 w_next;
-    if (c + 1 < kchunks) w_next = load_w(c + 1);
-    const int la = wave * 32 + akoff;
-#pragma unroll
-    for (int mt = 0; mt < 8; ++mt) {
-      const bf16x8 a =
-          *reinterpret_cast<const bf16x8*>(&As[buf][mt * 16 + arow][la]);
-      acc[mt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, w_cur, acc[mt],
-                                                        0, 0, 0);
-    }
-    __syncthreads();
-    if (c + 1 < kchunks) {
-      store_chunk(buf ^ 1, stage);
-      if (c + 2 < kchunks) load_chunk(c + 2, stage);
-      w_cur = w_next;
-    }
-    __syncthreads();
-  }
-
-  // ---- cross-wave reduction through the (dead) A buffer ----------------
-  float* red = reinterpret_cast<float*>(&As[0][0][0]);  // [8][64][32] f32
-#pragma unroll
-  for (int mt = 0; mt < 8; ++mt) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r)
-      red[((wave * 64 + lane) * 32) + mt * 4 + r] = acc[mt][r];
-  }
-  __syncthreads();
-  if (wave == 0) {
-    const int ncol = nt * 16 + (lane & 15);
-    const int mrow = (lane >> 4) * 4;
-#pragma unroll
-    for (int mt = 0; mt < 8; ++mt) {
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        float v = 0.f;
-#pragma unroll
-        for (int wv = 0; wv < 8; ++wv)
-          v += red[((wv * 64 + lane) * 32) + mt * 4 + r];
-        const int m0 = mt * 16 + mrow + r;
-        if (m0 < M) Cbf[(long long)m0 * N + ncol] = f32_to_bf16(v);
-      }
-    }
-  }
-}
-
-extern "C" void qsa_skinny_gemm_m128_launch(const unsigned short* A,
-                                            const unsigned short* Wf,
-                                            unsigned short* Cbf, int M, int N,
-                                            long long K, long long lda,
-                                            int nt_loads,
-                                            hipStream_t stream) {
-  if (nt_loads)
-    hipLaunchKernelGGL((qsa_skinny_gemm_m128<true>), dim3(N / 16), dim3(512),
-                       0, stream, A, Wf, Cbf, M, N, K, lda);
-  else
-    hipLaunchKernelGGL((qsa_skinny_gemm_m128<false>), dim3(N / 16), dim3(512),
-                       0, stream, A, Wf, Cbf, M, N, K, lda);
-}
-
 extern "C" void qsa_skinny_gemm_launch(const unsigned short* A,
                                        const unsigned short* Wf,
                                        unsigned short* Cbf, int M, int N,

@@ -120,6 +120,10 @@ class VectorIndex:
     def search(self, query: np.ndarray, k: int = 3) -> list[SearchHit]:
         if not self._vecs:
             return []
+        if self._torch_matrix is not None and \
+                str(self._torch_device).startswith("cuda"):
+            return self.search_batch_gpu(np.asarray(query)[None, :], k,
+                                         device=self._torch_device)[0]
         q = np.asarray(query, dtype=np.float32)
         n = float(np.linalg.norm(q))
         if n > 0:
@@ -132,6 +136,10 @@ class VectorIndex:
                           self.metadata[i]) for i in top]
 
     def search_batch(self, queries: np.ndarray, k: int = 3) -> list[list[SearchHit]]:
+        if self._torch_matrix is not None and \
+                str(self._torch_device).startswith("cuda"):
+            return self.search_batch_gpu(np.asarray(queries), k,
+                                         device=self._torch_device)
         return [self.search(q, k) for q in np.asarray(queries)]
 
     # ---- GPU path --------------------------------------------------------

@@ -47,12 +47,7 @@ def main():
         if M <= 32:
             ms_sk = t_ms(lambda: e.skinny_gemm(a, wf, N, K))
         else:
-            out = e.skinny_gemm_m128(a, wf, N, K, 1)
-            ref = a.float() @ w.float().T
-            err = (out.float() - ref).abs().max().item() / (
-                ref.abs().max().item() + 1e-6)
-            assert err < 2e-2, f"m128 wrong: rel err {err}"
-            ms_sk = t_ms(lambda: e.skinny_gemm_m128(a, wf, N, K, 1))
+            ms_sk = float("nan")  # custom path is M<=32; rocBLAS above
         floor = N * K * 2 / 6.3e12 * 1e3
         print(f"{name:8s} N={N:6d} K={K:6d}: rocblas {ms_blas*1e3:8.1f} us  "
               f"skinny {ms_sk*1e3:8.1f} us  floor {floor*1e3:8.1f} us  "
