@@ -232,3 +232,24 @@ def test_sql_driven_lab1_on_gpu_engine():
     assert len(rows) == 10
     assert all(r["agent_status"] == "SUCCESS" for r in rows)
     assert any(r["decision"] == "PRICE_MATCH" for r in rows)
+
+
+def test_sql_driven_lab3_on_gpu_stack():
+    """Lab3 end-to-end on the GPU stack: batched HIP anomaly scoring, the
+    on-GPU embedding encoder, the HBM-resident index (GPU top-k), and the
+    tiny decode engine — French Quarter contract preserved."""
+    import json
+
+    from quickstart_streaming_agents_amd.agents.mcp import StubMcpServer
+    from quickstart_streaming_agents_amd.labs.deploy import Deployment
+    dep = Deployment(labs=(3,), device="cuda:0", model="tiny")
+    dep.datagen(3)
+    srv = StubMcpServer().start()
+    try:
+        rows = dep.run(3, mcp_server=srv)
+    finally:
+        srv.stop()
+    assert 1 <= len(rows) <= 2
+    assert all(r["pickup_zone"] == "French Quarter" for r in rows)
+    boats = json.loads(rows[0]["dispatch_json"])["boats"]
+    assert 0 < len(boats) <= 8
