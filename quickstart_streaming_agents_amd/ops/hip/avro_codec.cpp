@@ -295,8 +295,9 @@ static py::object decode(const NodeP& s, Reader& r) {
         for (long long i = 0; i < n; ++i) {
           long long klen = r.vlong();
           const unsigned char* q = r.take((size_t)klen);
-          std::string k(reinterpret_cast<const char*>(q), (size_t)klen);
-          out[k.c_str()] = decode(s->values, r);
+          // py::str keeps embedded NULs (c_str() would truncate)
+          out[py::str(std::string(reinterpret_cast<const char*>(q),
+                                  (size_t)klen))] = decode(s->values, r);
         }
       }
       return out;

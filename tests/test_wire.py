@@ -142,3 +142,51 @@ def test_native_codec_byte_identical():
                 assert abs(back[k] - v) < 1e-9
             else:
                 assert back[k] == v
+
+
+def test_codec_fuzz_python_vs_native():
+    """Property test: random records serialize byte-identically in the
+    Python and C++ codecs and round-trip through both."""
+    import pytest
+    from hypothesis import given, settings, strategies as st
+
+    from quickstart_streaming_agents_amd.ops import ext, have_ext
+    from quickstart_streaming_agents_amd.wire.avro import (Schema,
+                                                           deserialize,
+                                                           serialize)
+    if not have_ext():
+        pytest.skip("extension not built")
+    defn = {
+        "type": "record", "name": "fuzz", "fields": [
+            {"name": "s", "type": "string"},
+            {"name": "n", "type": "long"},
+            {"name": "d", "type": "double"},
+            {"name": "b", "type": "boolean"},
+            {"name": "opt", "type": ["null", "string"], "default": None},
+            {"name": "arr", "type": {"type": "array", "items": "long"}},
+            {"name": "m", "type": {"type": "map", "values": "string"}},
+        ]}
+    py = Schema(defn)
+    codec = ext().AvroCodec(defn)
+
+    @settings(max_examples=200, deadline=None)
+    @given(st.text(max_size=80),
+           st.integers(min_value=-(2 ** 62), max_value=2 ** 62),
+           st.floats(allow_nan=False, allow_infinity=False),
+           st.booleans(),
+           st.one_of(st.none(), st.text(max_size=20)),
+           st.lists(st.integers(min_value=-10 ** 12, max_value=10 ** 12),
+                    max_size=8),
+           st.dictionaries(st.text(min_size=1, max_size=10),
+                           st.text(max_size=10), max_size=5))
+    def roundtrip(s, n, d, b, opt, arr, m):
+        rec = {"s": s, "n": n, "d": d, "b": b, "opt": opt, "arr": arr,
+               "m": m}
+        raw_py = serialize(py, 3, rec)
+        raw_cc = bytes(codec.serialize(3, rec))
+        assert raw_py == raw_cc
+        _, back_py = deserialize(py, raw_cc)
+        _, back_cc = codec.deserialize(raw_py)
+        assert back_py == back_cc == rec
+
+    roundtrip()
