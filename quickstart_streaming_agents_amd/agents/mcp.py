@@ -19,6 +19,8 @@ from __future__ import annotations
 
 import json
 import threading
+import time
+import urllib.error
 import urllib.request
 from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
 from typing import Any, Callable
@@ -47,8 +49,23 @@ class McpClient:
             self.endpoint, data=json.dumps(payload).encode(),
             headers={"Content-Type": "application/json",
                      "Accept": "application/json, text/event-stream"})
-        with urllib.request.urlopen(req, timeout=self.timeout_s) as resp:
-            body = json.loads(resp.read().decode())
+        # transport-level retry with backoff (the reference wraps flaky
+        # externals the same way, deploy.py:83-91): under hundreds of
+        # concurrent episodes a connection can be refused/reset; only
+        # transport errors retry — JSON-RPC errors surface immediately
+        body = None
+        delay = 0.05
+        for attempt in range(4):
+            try:
+                with urllib.request.urlopen(req,
+                                            timeout=self.timeout_s) as resp:
+                    body = json.loads(resp.read().decode())
+                break
+            except (urllib.error.URLError, ConnectionError, OSError):
+                if attempt == 3:
+                    raise
+                time.sleep(delay)
+                delay *= 2
         if "error" in body:
             raise McpError(str(body["error"]))
         return body.get("result")
@@ -236,7 +253,11 @@ class _Handler(BaseHTTPRequestHandler):
 class StubMcpServer:
     """In-process stub: MCP endpoint + competitor site + vessel API."""
 
+    request_queue_size = 256   # default backlog of 5 drops connections
+                               # under hundreds of concurrent episodes
+
     def __init__(self, host: str = "127.0.0.1", port: int = 0):
+        ThreadingHTTPServer.request_queue_size = self.request_queue_size
         self.httpd = ThreadingHTTPServer((host, port), _Handler)
         self.httpd.ctx = {"emails": [], "dispatches": []}
         self._thread: threading.Thread | None = None
