@@ -393,3 +393,210 @@ extern "C" void qsa_kv_scatter_launch(const unsigned short* knew,
   hipLaunchKernelGGL(qsa_kv_scatter, dim3((long long)T * KVH), dim3(D), 0,
                      stream, knew, vnew, kc, vc, slots, T, KVH, D, kvstride);
 }
+
+// ---------------------------------------------------------------------------
+// Varlen flash PREFILL attention over the paged cache (K4 prefill path).
+//
+// One WAVE per (16-row q-block, q head): streams the sequence's K/V pages
+// directly from the paged cache (K [P,KVH,D/8,64,8], V^T [P,KVH,D,64] —
+// the same coalesced MFMA fragment reads as the decode kernel), online
+// softmax per q-row column, causal masking against the row's absolute
+// position (cached prefix start + row).  No K/V gather, no padding, no
+// materialized score matrix: each KV byte is read once per wave and the
+// bmm+softmax+bmm chain collapses into one kernel per layer.
+//
+// Host passes per-q-block maps (item id, first q position) plus per-item
+// row offset / cached start / new length and the padded page table.
+// ---------------------------------------------------------------------------
+template <int D>
+__global__ void __launch_bounds__(256)
+qsa_paged_attn_prefill(const unsigned short* __restrict__ q,   // [T, QH, D] strided
+                       const unsigned short* __restrict__ kc,
+                       const unsigned short* __restrict__ vc,
+                       const int* __restrict__ block_table,    // [nb, npmax]
+                       const int* __restrict__ qb_item,        // [QB]
+                       const int* __restrict__ qb_pos0,        // [QB]
+                       const int* __restrict__ item_off,       // [nb] row offset
+                       const int* __restrict__ item_start,     // [nb] cached
+                       const int* __restrict__ item_len,       // [nb] new rows
+                       unsigned short* __restrict__ out,       // [T, QH*D]
+                       float scale, int QB, int QH, int KVH, int npmax,
+                       long long qstride) {
+  constexpr int KSTEPS = D / 32;
+  constexpr int DTILES = D / 16;
+  const int lane = threadIdx.x & 63;
+  const int gw = blockIdx.x * 4 + (threadIdx.x >> 6);  // global wave id
+  if (gw >= QB * QH) return;
+  const int qb = gw / QH;
+  const int qh = gw % QH;
+  const int R = QH / KVH;
+  const int kvh = qh / R;
+  const int item = qb_item[qb];
+  const int pos0 = qb_pos0[qb];          // first new-row index in the item
+  const int n_i = item_len[item];
+  const int start = item_start[item];
+  const int off = item_off[item];
+  const int col = lane & 15;             // q-row column in the score tiles
+  const int hi = lane >> 4;
+
+  // ---- Q B-fragments: lane holds Q[row pos0+col][k = ks*32 + hi*8+e] --
+  const int qrow_l = min(pos0 + col, n_i - 1);   // clamp pad rows
+  const unsigned short* qrow =
+      q + (long long)(off + qrow_l) * qstride + (long long)qh * D;
+  bf16x8_a qf[KSTEPS];
+#pragma unroll
+  for (int ks = 0; ks < KSTEPS; ++ks)
+    qf[ks] = *reinterpret_cast<const bf16x8_a*>(qrow + ks * 32 + hi * 8);
+
+  float m = -3.0e38f, lsum = 0.f;
+  f32x4_a o_acc[DTILES];
+#pragma unroll
+  for (int dt = 0; dt < DTILES; ++dt)
+    o_acc[dt] = (f32x4_a){0.f, 0.f, 0.f, 0.f};
+
+  // causal horizon: the block's LAST row attends to start+pos0+15 (abs)
+  const int max_abs = start + min(pos0 + 15, n_i - 1);
+  const int npages = (max_abs + QSA_PAGE) / QSA_PAGE;  // ceil(max_abs+1/64)
+  const int* btab = block_table + (long long)item * npmax;
+
+  for (int pi = 0; pi < npages; ++pi) {
+    const int page = btab[pi];
+    const unsigned short* kbase =
+        kc + ((((long long)page * KVH + kvh) * (D / 8)) * QSA_PAGE) * 8;
+    const unsigned short* vbase =
+        vc + (((long long)page * KVH + kvh) * D) * QSA_PAGE;
+    const int nvalid = min(max_abs + 1 - pi * QSA_PAGE, QSA_PAGE);
+
+    f32x4_a sc[4];
+#pragma unroll
+    for (int pt = 0; pt < 4; ++pt) {
+      f32x4_a acc = {0.f, 0.f, 0.f, 0.f};
+      const int pos = pt * 16 + col;
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        const bf16x8_a kf = *reinterpret_cast<const bf16x8_a*>(
+            kbase + (((long long)(ks * 4 + hi) * QSA_PAGE) + pos) * 8);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, qf[ks], acc,
+                                                      0, 0, 0);
+      }
+      sc[pt] = acc;
+    }
+    // mask: kpos (abs) must be <= start + qrow (abs) and < nvalid bound;
+    // C layout row kpos = pt*16 + hi*4 + r, col = q-row
+    const int qabs = start + pos0 + col;          // this column's row
+    float pagemax = -3.0e38f;
+#pragma unroll
+    for (int pt = 0; pt < 4; ++pt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int kpos = pt * 16 + hi * 4 + r;
+        const int kabs = pi * QSA_PAGE + kpos;
+        const bool ok = (kpos < nvalid) && (kabs <= qabs) &&
+                        (pos0 + col < n_i);
+        float v = ok ? sc[pt][r] * scale : -3.0e38f;
+        sc[pt][r] = v;
+        pagemax = fmaxf(pagemax, v);
+      }
+    }
+    pagemax = fmaxf(pagemax, __shfl_xor(pagemax, 16, QSA_WAVE));
+    pagemax = fmaxf(pagemax, __shfl_xor(pagemax, 32, QSA_WAVE));
+    // skip only when EVERY column is masked: the skip must be
+    // wave-uniform — MFMA reads all 64 lanes' source registers
+    // regardless of EXEC, so divergence around the PV MFMAs corrupts
+    // active lanes' products
+    if (__all(pagemax <= -3.0e38f)) continue;
+    const float m_new = fmaxf(m, pagemax);
+    float alpha = __expf(m - m_new);
+    if (m <= -3.0e38f) alpha = 0.f;
+    m = m_new;
+#pragma unroll
+    for (int dt = 0; dt < DTILES; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha;
+    }
+    unsigned int ppk[8];
+    float psum = 0.f;
+#pragma unroll
+    for (int pt = 0; pt < 4; ++pt) {
+      float p0f = 0.f, p1f = 0.f, p2f = 0.f, p3f = 0.f;
+      if (sc[pt][0] > -1.0e38f) p0f = __expf(sc[pt][0] - m_new);
+      if (sc[pt][1] > -1.0e38f) p1f = __expf(sc[pt][1] - m_new);
+      if (sc[pt][2] > -1.0e38f) p2f = __expf(sc[pt][2] - m_new);
+      if (sc[pt][3] > -1.0e38f) p3f = __expf(sc[pt][3] - m_new);
+      psum += p0f + p1f + p2f + p3f;
+      ppk[pt * 2] = f32x2_to_bf16x2(p0f, p1f);
+      ppk[pt * 2 + 1] = f32x2_to_bf16x2(p2f, p3f);
+    }
+    psum += __shfl_xor(psum, 16, QSA_WAVE);
+    psum += __shfl_xor(psum, 32, QSA_WAVE);
+    lsum = lsum * alpha + psum;
+
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int src_a = col + (((2 * hi) & 3) << 4);
+      const int src_b = col + (((2 * hi + 1) & 3) << 4);
+      const int tA = half * 2;
+      const int tB = half * 2 + 1;
+      const unsigned int a0A = __shfl(ppk[tA * 2], src_a, QSA_WAVE);
+      const unsigned int a1A = __shfl(ppk[tA * 2 + 1], src_a, QSA_WAVE);
+      const unsigned int b0A = __shfl(ppk[tA * 2], src_b, QSA_WAVE);
+      const unsigned int b1A = __shfl(ppk[tA * 2 + 1], src_b, QSA_WAVE);
+      const unsigned int a0B = __shfl(ppk[tB * 2], src_a, QSA_WAVE);
+      const unsigned int a1B = __shfl(ppk[tB * 2 + 1], src_a, QSA_WAVE);
+      const unsigned int b0B = __shfl(ppk[tB * 2], src_b, QSA_WAVE);
+      const unsigned int b1B = __shfl(ppk[tB * 2 + 1], src_b, QSA_WAVE);
+      const bool lo = hi < 2;
+      bf16x8_a pa;
+      unsigned int* pa_u = reinterpret_cast<unsigned int*>(&pa);
+      pa_u[0] = lo ? a0A : a0B;
+      pa_u[1] = lo ? a1A : a1B;
+      pa_u[2] = lo ? b0A : b0B;
+      pa_u[3] = lo ? b1A : b1B;
+#pragma unroll
+      for (int dt = 0; dt < DTILES; ++dt) {
+        const bf16x8_a vf = *reinterpret_cast<const bf16x8_a*>(
+            vbase + ((long long)(dt * 16 + col) * QSA_PAGE) + half * 32 +
+            hi * 8);
+        o_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vf,
+                                                            o_acc[dt],
+                                                            0, 0, 0);
+      }
+    }
+  }
+
+  // ---- epilogue: O C-layout row = q-row = hi*4 + r, col = dim ----------
+  // per-row 1/l lives in the lanes of its COLUMN; fetch via shfl
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qr = hi * 4 + r;               // q-row this lane holds
+    const float lr = __shfl(lsum, qr, 16);   // lanes 0-15 hold cols 0-15
+    const float inv = (lr > 0.f) ? 1.f / lr : 0.f;
+    if (pos0 + qr < n_i) {
+      unsigned short* orow =
+          out + (long long)(off + pos0 + qr) * (QH * D) + (long long)qh * D;
+#pragma unroll
+      for (int dt = 0; dt < DTILES; ++dt)
+        orow[dt * 16 + col] = f32_to_bf16(o_acc[dt][r] * inv);
+    }
+  }
+}
+
+extern "C" void qsa_paged_attn_prefill_launch(
+    const unsigned short* q, const unsigned short* kc,
+    const unsigned short* vc, const int* block_table, const int* qb_item,
+    const int* qb_pos0, const int* item_off, const int* item_start,
+    const int* item_len, unsigned short* out, float scale, int QB, int QH,
+    int KVH, int npmax, int D, long long qstride, hipStream_t stream) {
+  const long long waves = (long long)QB * QH;
+  dim3 grid((unsigned)((waves + 3) / 4));
+  if (D == 128)
+    hipLaunchKernelGGL((qsa_paged_attn_prefill<128>), grid, dim3(256), 0,
+                       stream, q, kc, vc, block_table, qb_item, qb_pos0,
+                       item_off, item_start, item_len, out, scale, QB, QH,
+                       KVH, npmax, qstride);
+  else if (D == 64)
+    hipLaunchKernelGGL((qsa_paged_attn_prefill<64>), grid, dim3(256), 0,
+                       stream, q, kc, vc, block_table, qb_item, qb_pos0,
+                       item_off, item_start, item_len, out, scale, QB, QH,
+                       KVH, npmax, qstride);
+}
