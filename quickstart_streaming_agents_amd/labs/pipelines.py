@@ -37,6 +37,7 @@ from ..agents.parse import (parse_lab1_sections, parse_lab3_sections,
 from ..agents.runner import AgentSpec, Continue, Finish, ToolCall, ToolSet, episode
 from ..agents.schedule import run_episodes
 from ..runtime.anomaly import AnomalyDetector
+from ..runtime.trace import Tracer
 from ..runtime.joins import TTLTable, enrich_join, interval_join
 from ..runtime.windows import TumblingWindows, aggregate
 from ..vector.index import VectorIndex, vector_search_agg
@@ -191,11 +192,16 @@ def lab1_user_prompt(order: dict, competitor_url: str, email_recipient: str) -> 
 def lab1_run(broker: Broker, llm_batch, tool_fn, competitor_url: str,
              email_recipient: str = "customer@example.com",
              max_new_tokens: int = 64, agent: AgentSpec | None = None,
-             state_ttl_ms: int = 3_600_000) -> list[dict]:
+             state_ttl_ms: int = 3_600_000,
+             tracer: Tracer | None = None) -> list[dict]:
     """enriched_orders -> AI_RUN_AGENT(price_match_agent) ->
     price_match_results rows (and topic).  `agent` (e.g. from
     sql.catalog.Catalog.agent_spec) overrides the built-in spec."""
-    enriched = lab1_enriched_orders(broker, state_ttl_ms=state_ttl_ms)
+    tracer = tracer or Tracer("lab1", enabled=False)
+    with tracer.stage("enrich_join") as sp:
+        enriched = lab1_enriched_orders(broker, state_ttl_ms=state_ttl_ms)
+        if sp:
+            sp.records_out = len(enriched)
     tools = ToolSet("lab1_remote_mcp", allowed_tools=("http_get", "send_email"),
                     request_timeout_s=30.0)
     agent = agent or AgentSpec("price_match_agent", "remote_mcp_model",
@@ -207,7 +213,11 @@ def lab1_run(broker: Broker, llm_batch, tool_fn, competitor_url: str,
                 max_new_tokens=max_new_tokens)
         for o in enriched
     ]
-    results = run_episodes(episodes, llm_batch, tool_fn)
+    with tracer.stage("ai_run_agent", records_in=len(episodes)) as sp:
+        results = run_episodes(episodes, llm_batch, tool_fn)
+        if sp:
+            sp.records_out = sum(r.status == "SUCCESS" for r in results)
+    tracer.count("decisions", len(results))
     out_topic = broker.create_topic("price_match_results")
     rows = []
     for o, r in zip(enriched, results):

@@ -281,32 +281,53 @@ class LlamaModel:
             kv.slot_ids(sid, s, n)
             for (_, sid, s), n in zip(items, lens)]).contiguous()
 
-        # ---- padded-batch attention prep (once per prefill call) --------
-        # Per-sequence attention loops were launch- and small-GEMM-bound;
-        # padding to n_max turns each layer's attention into ONE bmm pair +
-        # ONE masked-softmax launch across all sequences.  Per-row limits
-        # (start_i + qpos + 1, or 0 on pad rows) give exact causal + ragged
-        # masking; the softmax kernel zeroes beyond-limit columns.
+        # ---- attention prep (once per prefill call) ---------------------
         R = self.n_q // self.n_kv
         nb = len(items)
         nmax = max(lens)
         npmax = max((int(items[i][2]) + lens[i] + kv.PAGE - 1) // kv.PAGE
                     for i in range(nb))
         ctxp = npmax * kv.PAGE
-        dst_idx = torch.tensor(
-            [i * nmax + p for i, n in enumerate(lens) for p in range(n)],
-            dtype=torch.int64, device=dev)
-        page_idx = torch.zeros(nb, npmax, dtype=torch.int64)
-        for i, (_, sid, _) in enumerate(items):
-            pages = kv._seq_pages[sid]
-            page_idx[i, :len(pages)] = torch.tensor(pages,
-                                                    dtype=torch.int64)
-        flat_pages = page_idx.reshape(-1).to(dev)
-        limits = torch.zeros(nb, self.n_q, nmax, dtype=torch.int32)
-        for i, ((_, _, start), n) in enumerate(zip(items, lens)):
-            lim = torch.arange(start + 1, start + n + 1, dtype=torch.int32)
-            limits[i, :, :n] = lim.unsqueeze(0)
-        row_limits = limits.reshape(-1).to(dev)
+        use_hip_prefill = str(dev).startswith("cuda") and \
+            self.cfg.d_head in (64, 128)
+        if use_hip_prefill:
+            # varlen flash prefill over the paged cache: per-q-block maps
+            # (one WAVE per 16 q rows x head in the kernel)
+            qb_item, qb_pos0 = [], []
+            for i, n in enumerate(lens):
+                for p0 in range(0, n, 16):
+                    qb_item.append(i)
+                    qb_pos0.append(p0)
+            qb_item_t = torch.tensor(qb_item, dtype=torch.int32, device=dev)
+            qb_pos0_t = torch.tensor(qb_pos0, dtype=torch.int32, device=dev)
+            off_t = torch.tensor(offs[:-1], dtype=torch.int32, device=dev)
+            start_t = torch.tensor([int(s) for _, _, s in items],
+                                   dtype=torch.int32, device=dev)
+            len_t = torch.tensor(lens, dtype=torch.int32, device=dev)
+            bt = torch.zeros(nb, npmax, dtype=torch.int32)
+            for i, (_, sid, _) in enumerate(items):
+                pages = kv._seq_pages[sid]
+                bt[i, :len(pages)] = torch.tensor(pages, dtype=torch.int32)
+            bt_t = bt.to(dev)
+        else:
+            # CPU reference: padded-batch bmm pair + masked softmax.
+            # Per-row limits (start_i + qpos + 1, or 0 on pad rows) give
+            # exact causal + ragged masking.
+            dst_idx = torch.tensor(
+                [i * nmax + p for i, n in enumerate(lens) for p in range(n)],
+                dtype=torch.int64, device=dev)
+            page_idx = torch.zeros(nb, npmax, dtype=torch.int64)
+            for i, (_, sid, _) in enumerate(items):
+                pages = kv._seq_pages[sid]
+                page_idx[i, :len(pages)] = torch.tensor(pages,
+                                                        dtype=torch.int64)
+            flat_pages = page_idx.reshape(-1).to(dev)
+            limits = torch.zeros(nb, self.n_q, nmax, dtype=torch.int32)
+            for i, ((_, _, start), n) in enumerate(zip(items, lens)):
+                lim = torch.arange(start + 1, start + n + 1,
+                                   dtype=torch.int32)
+                limits[i, :, :n] = lim.unsqueeze(0)
+            row_limits = limits.reshape(-1).to(dev)
 
         res = self.embed.index_select(0, tokens).contiguous()
         h = None
@@ -321,32 +342,39 @@ class LlamaModel:
             q, k, v = self._split_qkv(qkv, T)
             D.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
             D.kv_scatter(k, v, kv.k[li], kv.v[li], slots)
-            # padded q: [nb*nmax, n_q, D] -> [nb*KVH, R*nmax, D]
-            q_pad = torch.zeros(nb * nmax, self.n_q, c.d_head,
-                                dtype=self.dtype, device=dev)
-            q_pad.index_copy_(0, dst_idx, q)
-            qb = q_pad.view(nb, nmax, self.n_kv, R, c.d_head) \
-                .permute(0, 2, 3, 1, 4) \
-                .reshape(nb * self.n_kv, R * nmax, c.d_head)
-            # gather padded K/V from the paged cache (k/v just scattered)
-            ksel = kv.k[li].index_select(0, flat_pages) \
-                .view(nb, npmax, self.n_kv, c.d_head // 8, kv.PAGE, 8) \
-                .permute(0, 2, 1, 4, 3, 5) \
-                .reshape(nb * self.n_kv, ctxp, c.d_head)
-            vsel = kv.v[li].index_select(0, flat_pages) \
-                .view(nb, npmax, self.n_kv, c.d_head, kv.PAGE) \
-                .permute(0, 2, 1, 4, 3) \
-                .reshape(nb * self.n_kv, ctxp, c.d_head)
-            # bf16 scores + fused-scale masked softmax: no f32 round trip
-            scores = torch.bmm(qb, ksel.transpose(1, 2)) \
-                .reshape(-1, ctxp)
-            D.softmax_rows_bf16_(scores, self.scale, row_limits)
-            probs = scores.reshape(nb * self.n_kv, R * nmax, ctxp)
-            a = torch.bmm(probs, vsel)
-            attn = a.view(nb, self.n_kv, R, nmax, c.d_head) \
-                .permute(0, 3, 1, 2, 4) \
-                .reshape(nb * nmax, self.n_q * c.d_head) \
-                .index_select(0, dst_idx)
+            if use_hip_prefill:
+                # ONE varlen flash kernel: streams K/V pages directly, no
+                # gather/pad/score materialization
+                attn = D.ext().paged_attn_prefill(
+                    q, kv.k[li], kv.v[li], bt_t, qb_item_t, qb_pos0_t,
+                    off_t, start_t, len_t, self.scale)
+            else:
+                # padded q: [nb*nmax, n_q, D] -> [nb*KVH, R*nmax, D]
+                q_pad = torch.zeros(nb * nmax, self.n_q, c.d_head,
+                                    dtype=self.dtype, device=dev)
+                q_pad.index_copy_(0, dst_idx, q)
+                qb = q_pad.view(nb, nmax, self.n_kv, R, c.d_head) \
+                    .permute(0, 2, 3, 1, 4) \
+                    .reshape(nb * self.n_kv, R * nmax, c.d_head)
+                # gather padded K/V from the paged cache (just scattered)
+                ksel = kv.k[li].index_select(0, flat_pages) \
+                    .view(nb, npmax, self.n_kv, c.d_head // 8, kv.PAGE, 8) \
+                    .permute(0, 2, 1, 4, 3, 5) \
+                    .reshape(nb * self.n_kv, ctxp, c.d_head)
+                vsel = kv.v[li].index_select(0, flat_pages) \
+                    .view(nb, npmax, self.n_kv, c.d_head, kv.PAGE) \
+                    .permute(0, 2, 1, 4, 3) \
+                    .reshape(nb * self.n_kv, ctxp, c.d_head)
+                # bf16 scores + fused-scale masked softmax
+                scores = torch.bmm(qb, ksel.transpose(1, 2)) \
+                    .reshape(-1, ctxp)
+                D.softmax_rows_bf16_(scores, self.scale, row_limits)
+                probs = scores.reshape(nb * self.n_kv, R * nmax, ctxp)
+                a = torch.bmm(probs, vsel)
+                attn = a.view(nb, self.n_kv, R, nmax, c.d_head) \
+                    .permute(0, 3, 1, 2, 4) \
+                    .reshape(nb * nmax, self.n_q * c.d_head) \
+                    .index_select(0, dst_idx)
             o = self._tp_all_reduce(F.linear(attn, L["wo"]))
             h = D.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             mlp_out = self._ffn(L, h)

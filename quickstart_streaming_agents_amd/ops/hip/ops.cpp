@@ -44,6 +44,11 @@ extern "C" void qsa_kv_append_launch(const unsigned short*,
                                      unsigned short*, const int*, const int*,
                                      int, int, int, int, long long,
                                      hipStream_t);
+extern "C" void qsa_paged_attn_prefill_launch(
+    const unsigned short*, const unsigned short*, const unsigned short*,
+    const int*, const int*, const int*, const int*, const int*, const int*,
+    unsigned short*, float, int, int, int, int, int, long long,
+    hipStream_t);
 extern "C" void qsa_rope_kv_append_launch(unsigned short*,
                                           const unsigned short*,
                                           const unsigned short*,
@@ -214,6 +219,34 @@ void kv_append(torch::Tensor knew, torch::Tensor vnew, torch::Tensor kc,
                        cur_stream());
 }
 
+torch::Tensor paged_attn_prefill(torch::Tensor q, torch::Tensor kc,
+                                 torch::Tensor vc, torch::Tensor block_table,
+                                 torch::Tensor qb_item, torch::Tensor qb_pos0,
+                                 torch::Tensor item_off,
+                                 torch::Tensor item_start,
+                                 torch::Tensor item_len, double scale) {
+  CHK_DEV(q); CHK_BF16(q); CHK_BF16(kc); CHK_BF16(vc);
+  CHK_CONT(kc); CHK_CONT(vc); CHK_I32(block_table); CHK_CONT(block_table);
+  CHK_I32(qb_item); CHK_I32(qb_pos0); CHK_I32(item_off);
+  CHK_I32(item_start); CHK_I32(item_len);
+  const int T = q.size(0), QH = q.size(1), D = q.size(2);
+  chk_hd_strided(q, D, "q");
+  const int KVH = kc.size(1);
+  TORCH_CHECK(D == 128 || D == 64, "D must be 64/128");
+  TORCH_CHECK(QH % KVH == 0 && QH / KVH <= 16, "GQA ratio <= 16");
+  TORCH_CHECK(vc.size(2) == D && vc.size(3) == 64, "V layout [P,KVH,D,64]");
+  const int QB = qb_item.numel();
+  const int npmax = block_table.size(1);
+  auto out = torch::empty({(long long)T, (long long)QH * D}, q.options());
+  qsa_paged_attn_prefill_launch(
+      u16(q), u16(kc), u16(vc), block_table.data_ptr<int>(),
+      qb_item.data_ptr<int>(), qb_pos0.data_ptr<int>(),
+      item_off.data_ptr<int>(), item_start.data_ptr<int>(),
+      item_len.data_ptr<int>(), u16m(out), (float)scale, QB, QH, KVH, npmax,
+      D, q.stride(0), cur_stream());
+  return out;
+}
+
 void rope_kv_append(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                     torch::Tensor kc, torch::Tensor vc, torch::Tensor cos_t,
                     torch::Tensor sin_t, torch::Tensor block_table,
@@ -351,6 +384,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_rows_", &softmax_rows_, "row softmax in place (f32)",
         py::arg("scores"), py::arg("col_offset") = 0, py::arg("causal") = false,
         py::arg("row_mod") = 0, py::arg("row_limits") = py::none());
+  m.def("paged_attn_prefill", &paged_attn_prefill,
+        "varlen flash prefill attention over the paged cache");
   m.def("paged_attn_decode", &paged_attn_decode,
         "paged-attention decode (bf16, GQA, page=64)");
   m.def("kv_append", &kv_append, "append one step's k/v to the paged cache");

@@ -140,3 +140,27 @@ def test_retry_and_poll():
     with pytest.raises(ValueError):
         retry_with_backoff(lambda: (_ for _ in ()).throw(ValueError("x")),
                            attempts=2, sleep=lambda s: None)
+
+
+def test_lab1_pipeline_tracing():
+    """Per-stage spans + counters come out of a real pipeline run."""
+    from quickstart_streaming_agents_amd.agents.mcp import (McpClient,
+                                                            StubMcpServer)
+    from quickstart_streaming_agents_amd.labs import datagen, pipelines
+    from quickstart_streaming_agents_amd.wire import Broker
+    b = Broker()
+    datagen.publish_lab1(b, seed=42)
+    srv = StubMcpServer().start()
+    try:
+        tr = Tracer("lab1")
+        rows = pipelines.lab1_run(
+            b, pipelines.StubLLM(), pipelines.mcp_tool_fn(
+                McpClient(srv.mcp_endpoint)),
+            competitor_url=f"{srv.base_url}/competitor", tracer=tr)
+    finally:
+        srv.stop()
+    s = tr.summary()
+    assert s["stages"]["enrich_join"]["records_out"] == len(rows) == 10
+    assert s["stages"]["ai_run_agent"]["records_out"] == 10
+    assert s["counters"]["decisions"] == 10
+    assert s["stages"]["ai_run_agent"]["total_s"] > 0
