@@ -288,8 +288,10 @@ class LlamaModel:
         npmax = max((int(items[i][2]) + lens[i] + kv.PAGE - 1) // kv.PAGE
                     for i in range(nb))
         ctxp = npmax * kv.PAGE
+        import os as _os
         use_hip_prefill = str(dev).startswith("cuda") and \
-            self.cfg.d_head in (64, 128)
+            self.cfg.d_head in (64, 128) and \
+            _os.environ.get("QSA_NO_HIP_PREFILL") != "1"
         if use_hip_prefill:
             # varlen flash prefill over the paged cache: per-q-block maps
             # (one WAVE per 16 q rows x head in the kernel)
