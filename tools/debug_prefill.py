@@ -11,7 +11,9 @@ from quickstart_streaming_agents_amd.ops import ext
 torch.manual_seed(0)
 dev = "cuda:0"
 QH, KVH, D = 4, 2, 64
-n, start = 75, 0
+import sys as _s
+n = int(_s.argv[1]) if len(_s.argv) > 1 else 75
+start = int(_s.argv[2]) if len(_s.argv) > 2 else 0
 T = n
 npages = (start + n + 63) // 64
 P = npages + 2
@@ -55,5 +57,8 @@ print("per-head max:", per_head.tolist())
 if bad:
     r = bad[0]
     print("row", r, "per-head err:", err[r].amax(dim=1).tolist())
-    print("got[., :6]:", got[r, 0, :6].tolist())
-    print("ref[., :6]:", ref[r, 0, :6].tolist())
+    h = int(err[r].amax(dim=1).argmax())
+    grid = (err[r, h] > 0.05).int().reshape(D // 16, 16)
+    print(f"head {h} bad-dim grid [dt, col]:")
+    for dt in range(D // 16):
+        print("  dt", dt, "".join(str(int(x)) for x in grid[dt]))
