@@ -131,10 +131,16 @@ qsa_paged_attn_mfma(const unsigned short* __restrict__ q,   // [B, QH, D]
     float alpha = __expf(m - m_new);
     if (m <= -3.0e38f) alpha = 0.f;
     m = m_new;
+    // o_acc element (dt, r) belongs to OUTPUT ROW hi*4 + r (the PV
+    // C-layout), while alpha lives per score COLUMN (this lane's l&15):
+    // fetch each row's own alpha from the lane holding that column —
+    // using the lane's own alpha zeroes/mis-scales other rows' history
+    // whenever per-row maxima diverge across pages
 #pragma unroll
-    for (int dt = 0; dt < DTILES; ++dt) {
+    for (int r = 0; r < 4; ++r) {
+      const float alpha_r = __shfl(alpha, hi * 4 + r, 16);
 #pragma unroll
-      for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha;
+      for (int dt = 0; dt < DTILES; ++dt) o_acc[dt][r] *= alpha_r;
     }
     // P = exp(sc - m) (invalid -> 0), packed bf16x2 per tile quad
     unsigned int ppk[8];   // tile pt -> 2 uints (4 bf16 = quads r0..3)
@@ -509,10 +515,16 @@ qsa_paged_attn_prefill(const unsigned short* __restrict__ q,   // [T, QH, D] str
     float alpha = __expf(m - m_new);
     if (m <= -3.0e38f) alpha = 0.f;
     m = m_new;
+    // o_acc element (dt, r) belongs to OUTPUT ROW hi*4 + r (the PV
+    // C-layout), while alpha lives per score COLUMN (this lane's l&15):
+    // fetch each row's own alpha from the lane holding that column —
+    // using the lane's own alpha zeroes/mis-scales other rows' history
+    // whenever per-row maxima diverge across pages
 #pragma unroll
-    for (int dt = 0; dt < DTILES; ++dt) {
+    for (int r = 0; r < 4; ++r) {
+      const float alpha_r = __shfl(alpha, hi * 4 + r, 16);
 #pragma unroll
-      for (int r = 0; r < 4; ++r) o_acc[dt][r] *= alpha;
+      for (int dt = 0; dt < DTILES; ++dt) o_acc[dt][r] *= alpha_r;
     }
     unsigned int ppk[8];
     float psum = 0.f;

@@ -275,3 +275,35 @@ def test_rope_kv_append_matches_composed(ext):
     torch.testing.assert_close(q1, q2)
     torch.testing.assert_close(kc1, kc2)
     torch.testing.assert_close(vc1, vc2)
+
+
+def test_paged_attention_forced_rescale(ext):
+    """Force the online-softmax rescale with per-head divergent maxima at
+    a LATE page (guide rule: a rare data-dependent branch needs its own
+    test) — this catches per-row/column alpha mix-ups the random-data
+    test cannot."""
+    torch.manual_seed(8)
+    B, QH, KVH, D = 2, 8, 2, 128
+    seqlen = 3 * 64  # 3 pages
+    kc = torch.randn(8, KVH, D // 8, 64, 8, dtype=torch.bfloat16,
+                     device=dev()) * 0.3
+    vc = torch.randn(8, KVH, D, 64, dtype=torch.bfloat16, device=dev())
+    q = torch.randn(B, QH, D, dtype=torch.bfloat16, device=dev())
+    # spike K at page 2 position 7 so scores jump there; make q for SOME
+    # heads align with the spike direction and others anti-align
+    spike = torch.randn(D, device=dev(), dtype=torch.bfloat16) * 3
+    kc[2 + 3, :, :, 7, :] = spike.reshape(D // 8, 8).unsqueeze(0) * 2
+    kc[2, :, :, 7, :] = spike.reshape(D // 8, 8).unsqueeze(0) * 2
+    for h in range(QH):
+        sgn = 1.0 if h % 2 == 0 else -1.0
+        q[:, h] = sgn * spike + torch.randn(D, device=dev(),
+                                            dtype=torch.bfloat16) * 0.2
+    bt = torch.tensor([[0, 1, 2], [3, 4, 5]], dtype=torch.int32,
+                      device=dev())
+    sl = torch.tensor([seqlen, seqlen - 10], dtype=torch.int32,
+                      device=dev())
+    out = ext.paged_attn_decode(q, kc, vc, bt, sl, 0.088)
+    ref = cpu_ref.paged_attn_ref(q.cpu(), kc.cpu(), vc.cpu(), bt.cpu(),
+                                 sl.cpu(), 0.088)
+    torch.testing.assert_close(out.float().cpu(), ref.float(), atol=4e-2,
+                               rtol=4e-2)
