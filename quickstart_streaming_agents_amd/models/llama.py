@@ -290,7 +290,7 @@ class LlamaModel:
         ctxp = npmax * kv.PAGE
         import os as _os
         use_hip_prefill = str(dev).startswith("cuda") and \
-            self.cfg.d_head in (64, 128) and \
+            self.cfg.d_head in (64, 128) and self.n_q % 4 == 0 and \
             _os.environ.get("QSA_NO_HIP_PREFILL") != "1"
         if use_hip_prefill:
             # varlen flash prefill over the paged cache: per-q-block maps

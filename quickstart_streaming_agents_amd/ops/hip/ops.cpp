@@ -234,6 +234,8 @@ torch::Tensor paged_attn_prefill(torch::Tensor q, torch::Tensor kc,
   const int KVH = kc.size(1);
   TORCH_CHECK(D == 128 || D == 64, "D must be 64/128");
   TORCH_CHECK(QH % KVH == 0 && QH / KVH <= 16, "GQA ratio <= 16");
+  TORCH_CHECK(QH % 4 == 0, "QH % 4 == 0 (4 head-waves per workgroup "
+              "lockstep on a shared page stream)");
   TORCH_CHECK(vc.size(2) == D && vc.size(3) == 64, "V layout [P,KVH,D,64]");
   const int QB = qb_item.numel();
   const int npmax = block_table.size(1);

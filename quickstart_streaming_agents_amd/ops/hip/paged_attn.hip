@@ -432,9 +432,14 @@ qsa_paged_attn_prefill(const unsigned short* __restrict__ q,   // [T, QH, D] str
   constexpr int DTILES = D / 16;
   const int lane = threadIdx.x & 63;
   const int gw = blockIdx.x * 4 + (threadIdx.x >> 6);  // global wave id
-  if (gw >= QB * QH) return;
-  const int qb = gw / QH;
-  const int qh = gw % QH;
+  // the 4 waves of a workgroup are 4 consecutive heads of the SAME
+  // q-block (and, at R>=4, the same kv head): they stream identical
+  // K/V pages, so keeping them in lockstep (per-page barrier below)
+  // turns 3 of the 4 streams into L1 hits
+  const bool live = gw < QB * QH;
+  const int gws = live ? gw : QB * QH - 1;   // dead waves shadow the last
+  const int qb = gws / QH;
+  const int qh = gws % QH;
   const int R = QH / KVH;
   const int kvh = qh / R;
   const int item = qb_item[qb];
@@ -466,6 +471,7 @@ qsa_paged_attn_prefill(const unsigned short* __restrict__ q,   // [T, QH, D] str
   const int* btab = block_table + (long long)item * npmax;
 
   for (int pi = 0; pi < npages; ++pi) {
+    __syncthreads();   // lockstep the 4 head-waves on the shared page
     const int page = btab[pi];
     const unsigned short* kbase =
         kc + ((((long long)page * KVH + kvh) * (D / 8)) * QSA_PAGE) * 8;
