@@ -8,10 +8,15 @@ projections.  Weights are random-init bf16 (the north-star benchmark runs
 synthetic data / random weights; the architecture is the named config).
 
 Decode path (the hot path): fully fused per layer —
-  rmsnorm_residual -> merged qkv GEMM -> rope_inplace -> kv_append ->
-  paged_attn_decode -> o GEMM -> rmsnorm_residual -> merged gate_up GEMM ->
-  swiglu -> down GEMM
-Prefill: per-sequence chunk with GEMM scores + the causal softmax kernel.
+  rmsnorm_residual -> qkv (weight-streaming skinny GEMM or rocBLAS by
+  measured winner per shape) -> fused rope+paged-KV-append ->
+  MFMA flash-decode paged attention -> o GEMM -> rmsnorm_residual ->
+  gate_up GEMM -> swiglu -> down GEMM; the whole step is hipGraph-captured
+  and self-feeding (models/serve.py).
+Prefill: fused per-token projections over the varlen concatenation, then
+ONE varlen flash-attention kernel per layer streaming K/V straight from
+the paged cache (ops/hip/paged_attn.hip); the CPU test path keeps a
+padded-batch bmm + masked-softmax reference of the same semantics.
 
 Tensor parallelism (Megatron-style, RCCL over xGMI): attention heads and
 FFN columns shard across the TP group; wo / wdown are row-parallel with ONE
