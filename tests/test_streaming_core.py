@@ -214,3 +214,18 @@ def test_engine_ragged_prefill_batches():
         solo = Engine(model, max_batch=1,
                       max_seq_len=256).generate_batch([p], [5])[0]
         assert solo == want
+
+
+def test_engine_kv_exhaustion_raises():
+    """Prompts that can never fit the KV pool fail loudly, not silently."""
+    import pytest
+    import torch
+
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    from quickstart_streaming_agents_amd.models.serve import Engine
+    model = LlamaModel(LlamaConfig.preset("tiny"), device="cpu",
+                       dtype=torch.float32, seed=3)
+    eng = Engine(model, kv_pages=2, max_batch=2, max_seq_len=256)
+    with pytest.raises(MemoryError):
+        eng.generate_batch([list(range(1, 200))], [100])
