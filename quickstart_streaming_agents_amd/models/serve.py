@@ -348,7 +348,13 @@ class Engine:
                 self._retire()
         else:
             while self.pending or self.running:
-                self.step()
+                n_pending = len(self.pending)
+                remaining = self.step()
+                if remaining and not self.running and \
+                        len(self.pending) == n_pending:
+                    raise MemoryError(
+                        "decode stalled: pending prompts cannot be "
+                        "admitted (KV pages exhausted?)")
         if torch.cuda.is_available():
             torch.cuda.synchronize()
         self.stats.wall_s += time.perf_counter() - t0
