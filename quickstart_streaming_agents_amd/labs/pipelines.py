@@ -338,17 +338,81 @@ def _detect_series(det: AnomalyDetector, rows: list[dict],
     per_key = det.batch_results_gpu(series)
     return [per_key[k][i] for k, i in idx]
 
+
+
+def _window_rows_gpu(broker: Broker, topic: str, schema, ts_field: str,
+                     key_field: str, win_ms: int,
+                     value_field: str | None = None) -> list[dict]:
+    """Columnar native decode (C++ AvroCodec.decode_columns) + the HIP
+    segmented (key, window) aggregation kernel -> closed-window rows in
+    the same (window_start, key) order as the CPU TumblingWindows path
+    (GPU test asserts equality on the lab datagen)."""
+    import torch
+
+    from ..ops import ext
+    recs = broker.topic(topic).read_all()
+    payloads = [bytes(r.value) for r in recs]
+    codec = ext().AvroCodec(schema)
+    cols = codec.decode_columns(
+        payloads, [ts_field, key_field] +
+        ([value_field] if value_field else []))
+    ts = np.asarray(cols[ts_field], dtype=np.int64)
+    keys = cols[key_field]
+    names = sorted(set(keys))
+    kid = {k: i for i, k in enumerate(names)}
+    key_ids = np.fromiter((kid[k] for k in keys), dtype=np.int32,
+                          count=len(keys))
+    t0 = int(ts.min() // win_ms) * win_ms
+    nwin = int((ts.max() - t0) // win_ms) + 1
+    dev = "cuda"
+    vals_t = None
+    if value_field:
+        vals = np.asarray([float(x) for x in cols[value_field]],
+                          dtype=np.float32)
+        vals_t = torch.from_numpy(vals).to(dev)
+    counts, sums = ext().window_agg(
+        torch.from_numpy(ts).to(dev),
+        torch.from_numpy(key_ids).to(dev), vals_t, t0, win_ms, nwin,
+        len(names))
+    counts = counts.cpu().numpy()
+    sums = sums.cpu().numpy()
+    rows = []
+    for w in range(nwin):
+        for ki, name in enumerate(names):
+            cnt = int(counts[ki, w])
+            if cnt == 0:
+                continue
+            start = t0 + w * win_ms
+            row = {"key": name, "window_start": start,
+                   "window_end": start + win_ms,
+                   "window_time": start + win_ms - 1,
+                   "request_count": cnt}
+            if value_field:
+                row["total_" + value_field] = float(sums[ki, w])
+                row["claim_count"] = cnt
+            rows.append(row)
+    return rows
+
 def lab3_anomalies(broker: Broker, params: dict | None = None) -> list[dict]:
     """5-min TUMBLE per pickup_zone + ML_DETECT_ANOMALIES; keep
     is_anomaly AND request_count > upper_bound (LAB3:99-198)."""
     params = params or {"minTrainingSize": 286, "maxTrainingSize": 7000,
                         "confidencePercentage": 99.9, "enableStl": False}
-    rides = [r for _, r in AvroConsumer(broker, "ride_requests",
-                                        schemas.RIDE_REQUESTS).poll()]
-    tw = TumblingWindows(MIN5_MS, lambda r: r["pickup_zone"],
-                         lambda r: r["request_ts"], watermark_delay_ms=5000)
-    rows = aggregate(tw.feed(rides) + tw.flush(), {"request_count": len})
-    rows.sort(key=lambda r: (r["window_start"], r["key"]))
+    import torch
+    if torch.cuda.is_available():
+        # native columnar decode + HIP windowed aggregation (K7/K10)
+        rows = _window_rows_gpu(broker, "ride_requests",
+                                schemas.RIDE_REQUESTS, "request_ts",
+                                "pickup_zone", MIN5_MS)
+    else:
+        rides = [r for _, r in AvroConsumer(broker, "ride_requests",
+                                            schemas.RIDE_REQUESTS).poll()]
+        tw = TumblingWindows(MIN5_MS, lambda r: r["pickup_zone"],
+                             lambda r: r["request_ts"],
+                             watermark_delay_ms=5000)
+        rows = aggregate(tw.feed(rides) + tw.flush(),
+                         {"request_count": len})
+        rows.sort(key=lambda r: (r["window_start"], r["key"]))
     det = AnomalyDetector.from_json_params(params)
     out = []
     apz_topic = broker.create_topic("anomalies_per_zone")
@@ -530,14 +594,25 @@ class Lab4FraudPolicy:
 def lab4_anomalies(broker: Broker, params: dict | None = None) -> list[dict]:
     params = params or {"minTrainingSize": 8, "maxTrainingSize": 50,
                         "confidencePercentage": 95.0, "enableStl": False}
-    claims = [c for _, c in AvroConsumer(broker, "claims", schemas.CLAIMS).poll()]
-    tw = TumblingWindows(H6_MS, lambda r: r["city"],
-                         lambda r: r["claim_timestamp"], watermark_delay_ms=5000)
-    rows = aggregate(tw.feed(claims) + tw.flush(), {
-        "total_claim_amount": lambda rs: sum(float(r["claim_amount"]) for r in rs),
-        "claim_count": len,
-    })
-    rows.sort(key=lambda r: (r["window_start"], r["key"]))
+    import torch
+    if torch.cuda.is_available():
+        rows = _window_rows_gpu(broker, "claims", schemas.CLAIMS,
+                                "claim_timestamp", "city", H6_MS,
+                                value_field="claim_amount")
+        for r in rows:
+            r["total_claim_amount"] = r.pop("total_claim_amount")
+    else:
+        claims = [c for _, c in AvroConsumer(broker, "claims",
+                                             schemas.CLAIMS).poll()]
+        tw = TumblingWindows(H6_MS, lambda r: r["city"],
+                             lambda r: r["claim_timestamp"],
+                             watermark_delay_ms=5000)
+        rows = aggregate(tw.feed(claims) + tw.flush(), {
+            "total_claim_amount": lambda rs: sum(float(r["claim_amount"])
+                                                 for r in rs),
+            "claim_count": len,
+        })
+        rows.sort(key=lambda r: (r["window_start"], r["key"]))
     det = AnomalyDetector.from_json_params(params)
     out = []
     topic = broker.create_topic("claims_anomalies_by_city")

@@ -9,6 +9,7 @@
 // record, array, map, union, enum, and the Confluent framing
 // magic 0x00 + 4-byte big-endian schema id (publish_lab3_data.py:96-122).
 #include <torch/extension.h>
+#include <pybind11/numpy.h>
 
 #include <cstring>
 #include <memory>
@@ -360,6 +361,108 @@ class Codec {
     return out;
   }
 
+  // Columnar batch decode (K10: record batches feeding columnar GPU
+  // buffers): long/int/boolean columns land in int64 numpy arrays,
+  // float/double in float64, strings in a py::list — one pass over the
+  // wire bytes, no per-record Python dicts.
+  py::dict decode_columns(py::sequence payloads,
+                          std::vector<std::string> columns) const {
+    if (root_->type != T::Record)
+      throw std::runtime_error("decode_columns needs a record schema");
+    const size_t n = payloads.size();
+    // per-column storage
+    std::vector<int> kind(columns.size(), -1);  // 0=int64 1=f64 2=obj
+    std::vector<std::vector<long long>> icols(columns.size());
+    std::vector<std::vector<double>> fcols(columns.size());
+    std::vector<py::list> ocols(columns.size());
+    auto col_of = [&](const std::string& name) -> int {
+      for (size_t c = 0; c < columns.size(); ++c)
+        if (columns[c] == name) return (int)c;
+      return -1;
+    };
+    for (size_t i = 0; i < n; ++i) {
+      std::string raw = py::cast<py::bytes>(payloads[i]);
+      if (raw.size() < 5 || raw[0] != 0)
+        throw std::runtime_error("not Confluent wire format");
+      const unsigned char* p =
+          reinterpret_cast<const unsigned char*>(raw.data());
+      Reader r{p + 5, p + raw.size()};
+      for (const auto& f : root_->fields) {
+        const int c = col_of(f.name);
+        const NodeP& fs = f.schema;
+        // resolve through a [null, X] union for extraction
+        if (c < 0) {
+          decode(fs, r);  // skip (still must advance the reader)
+          continue;
+        }
+        NodeP eff = fs;
+        bool is_null = false;
+        if (fs->type == T::Union) {
+          long long idx = r.vlong();
+          eff = fs->branches.at((size_t)idx);
+          is_null = eff->type == T::Null;
+        }
+        switch (is_null ? T::Null : eff->type) {
+          case T::Int:
+          case T::Long: {
+            long long v = r.vlong();
+            if (kind[c] < 0) kind[c] = 0;
+            if (kind[c] == 0) icols[c].push_back(v);
+            else if (kind[c] == 1) fcols[c].push_back((double)v);
+            else ocols[c].append(py::int_(v));
+            break;
+          }
+          case T::Boolean: {
+            bool v = *r.take(1) != 0;
+            if (kind[c] < 0) kind[c] = 0;
+            if (kind[c] == 0) icols[c].push_back(v ? 1 : 0);
+            else ocols[c].append(py::bool_(v));
+            break;
+          }
+          case T::Float: {
+            float f32v;
+            std::memcpy(&f32v, r.take(4), 4);
+            if (kind[c] < 0) kind[c] = 1;
+            fcols[c].push_back((double)f32v);
+            break;
+          }
+          case T::Double: {
+            double dv;
+            std::memcpy(&dv, r.take(8), 8);
+            if (kind[c] < 0) kind[c] = 1;
+            fcols[c].push_back(dv);
+            break;
+          }
+          default: {
+            // strings / complex / null -> object column
+            py::object v = is_null ? py::object(py::none())
+                                   : decode(eff, r);
+            if (kind[c] < 0) kind[c] = 2;
+            ocols[c].append(v);
+            break;
+          }
+        }
+      }
+    }
+    py::dict out;
+    for (size_t c = 0; c < columns.size(); ++c) {
+      if (kind[c] == 0) {
+        py::array_t<long long> arr((py::ssize_t)icols[c].size());
+        std::memcpy(arr.mutable_data(), icols[c].data(),
+                    icols[c].size() * sizeof(long long));
+        out[columns[c].c_str()] = arr;
+      } else if (kind[c] == 1) {
+        py::array_t<double> arr((py::ssize_t)fcols[c].size());
+        std::memcpy(arr.mutable_data(), fcols[c].data(),
+                    fcols[c].size() * sizeof(double));
+        out[columns[c].c_str()] = arr;
+      } else {
+        out[columns[c].c_str()] = ocols[c];
+      }
+    }
+    return out;
+  }
+
  private:
   NodeP root_;
   py::object defn_;  // keeps default-value objects alive
@@ -373,5 +476,6 @@ void register_avro(py::module_& m) {
       .def("serialize", &qsa_avro::Codec::serialize)
       .def("deserialize", &qsa_avro::Codec::deserialize)
       .def("serialize_batch", &qsa_avro::Codec::serialize_batch)
+      .def("decode_columns", &qsa_avro::Codec::decode_columns)
       .def("deserialize_batch", &qsa_avro::Codec::deserialize_batch);
 }

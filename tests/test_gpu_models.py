@@ -253,3 +253,29 @@ def test_sql_driven_lab3_on_gpu_stack():
     assert all(r["pickup_zone"] == "French Quarter" for r in rows)
     boats = json.loads(rows[0]["dispatch_json"])["boats"]
     assert 0 < len(boats) <= 8
+
+
+def test_gpu_window_rows_match_cpu_assigner():
+    """Columnar decode + HIP window_agg produces exactly the CPU
+    TumblingWindows rows on the lab3 stream."""
+    from quickstart_streaming_agents_amd.labs import datagen, schemas
+    from quickstart_streaming_agents_amd.labs.pipelines import (
+        MIN5_MS, _window_rows_gpu)
+    from quickstart_streaming_agents_amd.runtime.windows import (
+        TumblingWindows, aggregate)
+    from quickstart_streaming_agents_amd.wire import AvroConsumer, Broker
+    b = Broker()
+    datagen.publish_lab3(b, seed=42)
+    gpu_rows = _window_rows_gpu(b, "ride_requests", schemas.RIDE_REQUESTS,
+                                "request_ts", "pickup_zone", MIN5_MS)
+    rides = [r for _, r in AvroConsumer(b, "ride_requests",
+                                        schemas.RIDE_REQUESTS).poll()]
+    tw = TumblingWindows(MIN5_MS, lambda r: r["pickup_zone"],
+                         lambda r: r["request_ts"], watermark_delay_ms=5000)
+    cpu_rows = aggregate(tw.feed(rides) + tw.flush(),
+                         {"request_count": len})
+    cpu_rows.sort(key=lambda r: (r["window_start"], r["key"]))
+    assert [(r["key"], r["window_start"], r["request_count"])
+            for r in gpu_rows] == \
+        [(r["key"], r["window_start"], r["request_count"])
+         for r in cpu_rows]
