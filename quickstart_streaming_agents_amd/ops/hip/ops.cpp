@@ -193,6 +193,7 @@ torch::Tensor paged_attn_decode(torch::Tensor q, torch::Tensor kc,
   // (b, kvh, split); 4 splits share a workgroup)
   int ns = (int)std::min<long long>(32, std::max<long long>(
       1, (2048 + (long long)B * KVH - 1) / ((long long)B * KVH)));
+  if (ns > 1) ns = std::max(ns, 4);   // fill all 4 waves per workgroup
   ns = std::min(ns, std::max(1, max_pages));
   auto opts_f = q.options().dtype(at::kFloat);
   auto part_o = torch::empty({B, QH, ns, D}, opts_f);
