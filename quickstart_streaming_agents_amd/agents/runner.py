@@ -81,16 +81,45 @@ class Continue:
 Action = "ToolCall | Finish | Continue"
 
 
+def extract_tool_call_json(text: str) -> str | None:
+    """The JSON object after ``TOOL_CALL`` with BALANCED braces (a
+    non-greedy regex truncates nested ``arguments`` objects at the first
+    closing brace)."""
+    m = re.search(r"TOOL_CALL\s*:?\s*\{", text or "", re.IGNORECASE)
+    if not m:
+        return None
+    start = m.end() - 1
+    depth = 0
+    in_str = False
+    esc = False
+    for i in range(start, len(text)):
+        ch = text[i]
+        if in_str:
+            if esc:
+                esc = False
+            elif ch == "\\":
+                esc = True
+            elif ch == '"':
+                in_str = False
+        elif ch == '"':
+            in_str = True
+        elif ch == "{":
+            depth += 1
+        elif ch == "}":
+            depth -= 1
+            if depth == 0:
+                return text[start:i + 1]
+    return None
+
+
 class ToolCallPolicy:
     """Parse ``TOOL_CALL {json}`` syntax from model output; finish otherwise."""
 
-    TOOL_RE = re.compile(r"TOOL_CALL\s*:?\s*(\{[\s\S]*?\})", re.IGNORECASE)
-
     def __call__(self, text: str, iteration: int, ctx: dict) -> Any:
-        m = self.TOOL_RE.search(text or "")
-        if m:
+        raw = extract_tool_call_json(text or "")
+        if raw is not None:
             try:
-                call = json.loads(m.group(1))
+                call = json.loads(raw)
                 return ToolCall(call.get("name", ""),
                                 call.get("arguments", {}) or {})
             except json.JSONDecodeError:
@@ -214,10 +243,10 @@ def ai_tool_invoke(model_llm: Callable[[str, int], str],
                            for name, desc in tool_descriptions.items())
     text = model_llm(f"{prompt}\n\nAvailable tools:\n{tool_block}", 128)
     out: dict[str, str] = {"response": text}
-    m = ToolCallPolicy.TOOL_RE.search(text or "")
-    if m:
+    raw = extract_tool_call_json(text or "")
+    if raw is not None:
         try:
-            call = json.loads(m.group(1))
+            call = json.loads(raw)
             name = call.get("name", "")
             if name in tool_descriptions:
                 try:
