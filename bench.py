@@ -38,7 +38,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
-    p.add_argument("--batch", type=int, default=128,
+    p.add_argument("--batch", type=int, default=192,
                    help="decisions per step per GPU (concurrent agent "
                         "episodes sharing batched decode)")
     p.add_argument("--model", default="llama3-8b")
