@@ -34,7 +34,10 @@ class Deployment:
     """One in-process deployment: broker + catalog + per-lab runtime."""
 
     def __init__(self, labs=ALL_LABS, device: str = "cpu",
-                 model: str = "tiny", seed: int = 42):
+                 model: str | None = "tiny", seed: int = 42):
+        """model=None resolves the engine preset from the catalog's
+        CREATE MODEL options ('local.model', core.sql) — the SQL surface
+        is the source of truth; an explicit name is the CLI override."""
         self.labs = tuple(labs)
         self.device = device
         self.model = model
@@ -57,12 +60,19 @@ class Deployment:
                 self.broker.create_topic(name)
 
     # ---- model backends --------------------------------------------------
+    def resolved_model(self) -> str:
+        if self.model:
+            return self.model
+        md = self.catalog.models.get("llm_textgen_model")
+        return (md.options.get("local.model", "llama3-8b")
+                if md else "llama3-8b")
+
     def llm(self):
         if self._llm is None:
             if self.device.startswith("cuda"):
                 from ..models import build_model
                 from ..models.serve import Engine, EngineLLM
-                m = build_model(self.model, device=self.device,
+                m = build_model(self.resolved_model(), device=self.device,
                                 seed=self.seed)
                 self._llm = EngineLLM(Engine(m, max_batch=32,
                                              max_seq_len=2048))

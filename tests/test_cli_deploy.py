@@ -112,3 +112,12 @@ def test_walkthrough_sql_roundtrip(tmp_path):
     assert spec3.tools.allowed_tools == ("http_get", "http_post")
     assert cat.ctas_info("anomalies_per_zone").anomaly[0][
         "minTrainingSize"] == 286
+
+
+def test_model_resolves_from_sql():
+    """CREATE MODEL WITH('local.model'=...) drives engine selection when
+    no CLI override is given."""
+    dep = Deployment(labs=(1,), model=None)
+    assert dep.resolved_model() == "llama3-8b"
+    dep2 = Deployment(labs=(1,), model="tiny")
+    assert dep2.resolved_model() == "tiny"
