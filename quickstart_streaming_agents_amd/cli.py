@@ -128,14 +128,19 @@ def main(argv=None) -> int:
     d = sub.add_parser("deploy", help="build catalog + topics, write summaries")
     d.add_argument("--labs", default="1,2,3,4")
     d.add_argument("--device", default="cpu")
-    d.add_argument("--model", default="tiny")
+    d.add_argument("--model", default=None,
+                   help="engine preset; default resolves from the lab "
+                        "SQL CREATE MODEL options")
     d.add_argument("--dir", default="deploy_out")
     d.set_defaults(fn=cmd_deploy)
 
     r = sub.add_parser("run", help="datagen + run one lab end-to-end")
     r.add_argument("--lab", type=int, required=True)
     r.add_argument("--device", default="cpu")
-    r.add_argument("--model", default="tiny")
+    r.add_argument("--model", default=None,
+                   help="engine preset; default resolves from the lab "
+                        "SQL CREATE MODEL options (CPU runs use the "
+                        "stub LLM regardless)")
     r.add_argument("--max-print", type=int, default=3)
     r.set_defaults(fn=cmd_run)
 
