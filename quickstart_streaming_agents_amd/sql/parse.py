@@ -217,13 +217,16 @@ def _parse_with_options(seg: str) -> dict:
 def _find_matching_paren(s: str, open_idx: int) -> int:
     depth = 0
     in_str = False
-    for i in range(open_idx, len(s)):
+    i = open_idx
+    while i < len(s):
         ch = s[i]
         if in_str:
-            if ch == "'" and not (i + 1 < len(s) and s[i + 1] == "'"):
-                in_str = False
-            continue
-        if ch == "'":
+            if ch == "'":
+                if i + 1 < len(s) and s[i + 1] == "'":
+                    i += 1          # '' escape: consume both quotes
+                else:
+                    in_str = False
+        elif ch == "'":
             in_str = True
         elif ch == "(":
             depth += 1
@@ -231,6 +234,7 @@ def _find_matching_paren(s: str, open_idx: int) -> int:
             depth -= 1
             if depth == 0:
                 return i
+        i += 1
     raise ValueError("unbalanced parens")
 
 
@@ -255,18 +259,22 @@ def _extract_with(stmt: str) -> tuple[str, dict]:
 def _depth_at(s: str, idx: int) -> int:
     depth = 0
     in_str = False
-    for i in range(idx):
+    i = 0
+    while i < idx:
         ch = s[i]
         if in_str:
-            if ch == "'" and not (i + 1 < len(s) and s[i + 1] == "'"):
-                in_str = False
-            continue
-        if ch == "'":
+            if ch == "'":
+                if i + 1 < len(s) and s[i + 1] == "'":
+                    i += 1          # '' escape: consume both quotes
+                else:
+                    in_str = False
+        elif ch == "'":
             in_str = True
         elif ch == "(":
             depth += 1
         elif ch == ")":
             depth -= 1
+        i += 1
     return depth
 
 

@@ -173,3 +173,35 @@ def test_statement_splitting_respects_strings():
     assert len(stmts) == 2
     st = P.parse_statement(stmts[0])
     assert st.value == "x;y"
+
+
+def test_parser_edge_cases():
+    from quickstart_streaming_agents_amd.sql import parse as P
+    # WITH before AS (lab3 anomalies_enriched pattern)
+    t = P.parse_statement(
+        "CREATE TABLE x WITH ('changelog.mode' = 'append') AS "
+        "SELECT a FROM b")
+    assert t.as_select.startswith("SELECT")
+    assert t.options["changelog.mode"] == "append"
+    # quoted value containing parens/semicolons/escaped quote
+    c = P.parse_statement(
+        "CREATE CONNECTION `c.x` WITH ('endpoint' = 'http://h/p?(a;b)', "
+        "'note' = 'it''s fine')")
+    assert c.name == "c.x"
+    assert c.options["endpoint"] == "http://h/p?(a;b)"
+    assert c.options["note"] == "it's fine"
+    # nested generic types in columns
+    t2 = P.parse_statement(
+        "CREATE TABLE t (m MAP<STRING, ARRAY<FLOAT>>, "
+        "r ROW<a INT, b STRING>, c DECIMAL(10, 2))")
+    assert [col.name for col in t2.columns] == ["m", "r", "c"]
+    assert t2.columns[0].type == "MAP<STRING, ARRAY<FLOAT>>"
+    # agent with multiple tools
+    a = P.parse_statement(
+        "CREATE AGENT ag USING MODEL m USING PROMPT 'p' "
+        "USING TOOLS t1, t2 WITH ('max_iterations' = '3')")
+    assert a.tools == ["t1", "t2"]
+    # unsupported statement raises
+    import pytest
+    with pytest.raises(ValueError):
+        P.parse_statement("ALTER TABLE x ADD COLUMN y STRING")
