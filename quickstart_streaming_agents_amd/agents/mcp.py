@@ -212,15 +212,20 @@ class _Handler(BaseHTTPRequestHandler):
                                      "required": ["to", "subject", "body"]}},
                 ]}
             elif method == "tools/call":
-                result = self._tool_call(params.get("name", ""),
-                                         params.get("arguments", {}) or {})
+                # tool-execution failures map to an isError *result* per
+                # MCP; protocol-level failures below are JSON-RPC errors
+                try:
+                    result = self._tool_call(params.get("name", ""),
+                                             params.get("arguments", {}) or {})
+                except Exception as e:
+                    result = {"isError": True,
+                              "content": [{"type": "text", "text": str(e)}]}
             else:
                 raise McpError(f"unknown method {method}")
             body = {"jsonrpc": "2.0", "id": rid, "result": result}
-        except Exception as e:  # tool errors -> isError result per MCP
+        except McpError as e:  # method not found -> JSON-RPC error object
             body = {"jsonrpc": "2.0", "id": rid,
-                    "result": {"isError": True,
-                               "content": [{"type": "text", "text": str(e)}]}}
+                    "error": {"code": -32601, "message": str(e)}}
         self._send(200, json.dumps(body).encode())
 
     def _tool_call(self, name: str, args: dict) -> dict:
@@ -250,15 +255,16 @@ class _Handler(BaseHTTPRequestHandler):
         raise McpError(f"unknown tool {name}")
 
 
-class StubMcpServer:
-    """In-process stub: MCP endpoint + competitor site + vessel API."""
-
+class _StubHttpServer(ThreadingHTTPServer):
     request_queue_size = 256   # default backlog of 5 drops connections
                                # under hundreds of concurrent episodes
 
+
+class StubMcpServer:
+    """In-process stub: MCP endpoint + competitor site + vessel API."""
+
     def __init__(self, host: str = "127.0.0.1", port: int = 0):
-        ThreadingHTTPServer.request_queue_size = self.request_queue_size
-        self.httpd = ThreadingHTTPServer((host, port), _Handler)
+        self.httpd = _StubHttpServer((host, port), _Handler)
         self.httpd.ctx = {"emails": [], "dispatches": []}
         self._thread: threading.Thread | None = None
 
