@@ -190,3 +190,56 @@ def test_codec_fuzz_python_vs_native():
         assert back_py == back_cc == rec
 
     roundtrip()
+
+
+def test_registry_backward_compatibility_enforced():
+    """Confluent SR default BACKWARD mode: a new subject version must be
+    able to read data written with the previous one."""
+    import pytest
+
+    from quickstart_streaming_agents_amd.wire.registry import (
+        IncompatibleSchemaError, SchemaRegistry, schema_incompatibilities)
+    base = {"type": "record", "name": "o", "fields": [
+        {"name": "id", "type": "string"},
+        {"name": "price", "type": "double"}]}
+    reg = SchemaRegistry()
+    reg.register("orders-value", base)
+
+    # add field WITH default -> compatible new version
+    ok = {"type": "record", "name": "o", "fields": base["fields"] + [
+        {"name": "region", "type": "string", "default": "us"}]}
+    assert reg.check_compatible("orders-value", ok) == []
+    reg.register("orders-value", ok)
+
+    # add field WITHOUT default -> rejected
+    bad = {"type": "record", "name": "o", "fields": base["fields"] + [
+        {"name": "must", "type": "string"}]}
+    assert "without a default" in reg.check_compatible(
+        "orders-value", bad)[0]
+    with pytest.raises(IncompatibleSchemaError):
+        reg.register("orders-value", bad)
+
+    # removing a writer field is fine backward (reader ignores it)
+    narrower = {"type": "record", "name": "o", "fields": [
+        {"name": "id", "type": "string"}]}
+    reg.register("orders-value", narrower)
+
+    # NONE mode disables the check
+    reg2 = SchemaRegistry()
+    reg2.set_mode("s-value", "NONE")
+    reg2.register("s-value", base)
+    reg2.register("s-value", bad)
+
+    # promotions + unions + enum symbol removal
+    from quickstart_streaming_agents_amd.wire.avro import Schema
+    assert schema_incompatibilities(Schema("double"), Schema("int")) == []
+    assert schema_incompatibilities(Schema("int"), Schema("double"))
+    assert schema_incompatibilities(
+        Schema(["null", "string"]), Schema("string")) == []
+    assert schema_incompatibilities(
+        Schema("string"), Schema(["null", "string"]))
+    old_enum = {"type": "enum", "name": "v", "symbols": ["A", "B"]}
+    new_enum = {"type": "enum", "name": "v", "symbols": ["A"]}
+    assert "symbols removed" in schema_incompatibilities(
+        Schema(new_enum), Schema(old_enum))[0]
+    assert schema_incompatibilities(Schema(old_enum), Schema(new_enum)) == []
