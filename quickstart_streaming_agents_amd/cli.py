@@ -103,6 +103,13 @@ def cmd_destroy(args) -> int:
     return 0
 
 
+def cmd_serve(args) -> int:
+    from .serve_api import main as serve_main
+    return serve_main(["--host", args.host, "--port", str(args.port),
+                       "--device", args.device] +
+                      (["--model", args.model] if args.model else []))
+
+
 def cmd_tests(args) -> int:
     import os
     import subprocess
@@ -154,6 +161,14 @@ def main(argv=None) -> int:
     x = sub.add_parser("destroy", help="remove deployment artifacts")
     x.add_argument("--dir", default="deploy_out")
     x.set_defaults(fn=cmd_destroy)
+
+    s = sub.add_parser("serve", help="HTTP serving API "
+                       "(completions/embeddings/search/agents)")
+    s.add_argument("--host", default="127.0.0.1")
+    s.add_argument("--port", type=int, default=8080)
+    s.add_argument("--device", default="cpu")
+    s.add_argument("--model", default=None)
+    s.set_defaults(fn=cmd_serve)
 
     t = sub.add_parser("tests", help="run the test suite "
                        "(scripts/run_tests.py parity)")

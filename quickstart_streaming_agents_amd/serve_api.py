@@ -1,0 +1,194 @@
+"""HTTP serving surface for the engine (production-serving face of the
+framework).
+
+The reference exposes its capabilities only through Flink SQL statements
+against managed endpoints; a local MI355X deployment needs
+a network-facing entry point for the same four capabilities, so this
+module serves them over HTTP (FastAPI):
+
+    POST /v1/completions   -> K4 ML_PREDICT('llm_textgen_model', ...)
+    POST /v1/embeddings    -> K1 ML_PREDICT('llm_embedding_model', ...)
+    POST /v1/search        -> K2 VECTOR_SEARCH_AGG(..., k)
+    POST /v1/agents/{name} -> K6 AI_RUN_AGENT(agent, prompt[, key])
+    GET  /healthz, /v1/status
+
+Everything is dependency-injected: `create_app` takes the same
+`llm_batch` callable the lab pipelines use (labs/pipelines.py StubLLM on
+CPU, models/serve.py EngineLLM on a GPU), a HashingEmbedder-compatible
+embedder, a VectorIndex, and a dict of agent episode callables — so the
+app is unit-testable on CPU and serves the hipGraph engine unchanged on
+an MI355X.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Any, Callable, Optional, Union
+
+from pydantic import BaseModel, Field
+
+
+class CompletionRequest(BaseModel):
+    prompt: Union[str, list[str]]
+    max_tokens: int = Field(default=64, ge=1, le=4096)
+
+
+class EmbeddingRequest(BaseModel):
+    input: Union[str, list[str]]
+
+
+class SearchRequest(BaseModel):
+    query: Union[str, list[float]]
+    k: int = Field(default=3, ge=1, le=100)
+
+
+class AgentRequest(BaseModel):
+    prompt: str
+    record_key: Optional[str] = None
+
+
+def create_app(llm: Callable[[list[str], list[int]], list[str]],
+               embedder=None,
+               index=None,
+               agents: dict[str, Callable[..., Any]] | None = None):
+    """Build the FastAPI app around injected engine components.
+
+    `agents` maps agent name -> callable(prompt, record_key) returning an
+    object with .status/.response/.iterations/.tool_calls (the
+    agents/runner.py episode result shape).
+    """
+    from fastapi import FastAPI, HTTPException
+
+    app = FastAPI(title="quickstart-streaming-agents-amd", version="0.1")
+    app.state.started = time.time()
+    app.state.requests = 0
+    agents = agents or {}
+
+    @app.get("/healthz")
+    def healthz():
+        return {"status": "ok"}
+
+    @app.get("/v1/status")
+    def status():
+        return {
+            "uptime_s": round(time.time() - app.state.started, 3),
+            "requests": app.state.requests,
+            "capabilities": {
+                "completions": True,
+                "embeddings": embedder is not None,
+                "search": index is not None and embedder is not None,
+                "agents": sorted(agents),
+            },
+        }
+
+    @app.post("/v1/completions")
+    def completions(req: CompletionRequest):
+        app.state.requests += 1
+        prompts = [req.prompt] if isinstance(req.prompt, str) else req.prompt
+        if not prompts:
+            raise HTTPException(400, "empty prompt list")
+        texts = llm(prompts, [req.max_tokens] * len(prompts))
+        return {"object": "text_completion",
+                "choices": [{"index": i, "text": t}
+                            for i, t in enumerate(texts)]}
+
+    @app.post("/v1/embeddings")
+    def embeddings(req: EmbeddingRequest):
+        app.state.requests += 1
+        if embedder is None:
+            raise HTTPException(503, "no embedder configured")
+        texts = [req.input] if isinstance(req.input, str) else req.input
+        vecs = [embedder.embed(t) for t in texts]
+        return {"object": "list",
+                "data": [{"index": i, "embedding": v.tolist()}
+                         for i, v in enumerate(vecs)],
+                "dims": len(vecs[0]) if vecs else 0}
+
+    @app.post("/v1/search")
+    def search(req: SearchRequest):
+        app.state.requests += 1
+        if index is None:
+            raise HTTPException(503, "no vector index configured")
+        if isinstance(req.query, str):
+            if embedder is None:
+                raise HTTPException(503, "no embedder for text queries")
+            q = embedder.embed(req.query)
+        else:
+            import numpy as np
+            q = np.asarray(req.query, dtype="float32")
+            if q.shape != (index.dim,):
+                raise HTTPException(
+                    400, f"query dims {q.shape} != index dim {index.dim}")
+        hits = index.search(q, req.k)
+        return {"hits": [{"document_id": h.document_id, "chunk": h.chunk,
+                          "score": h.score, "metadata": h.metadata}
+                         for h in hits]}
+
+    @app.post("/v1/agents/{name}")
+    def run_agent(name: str, req: AgentRequest):
+        app.state.requests += 1
+        fn = agents.get(name)
+        if fn is None:
+            raise HTTPException(404, f"unknown agent {name!r}")
+        res = fn(req.prompt, req.record_key)
+        return {"agent": name,
+                "status": res.status,
+                "response": res.response,
+                "iterations": getattr(res, "iterations", None),
+                "tool_calls": getattr(res, "tool_calls", None)}
+
+    return app
+
+
+def build_lab_app(device: str = "cpu", model: str | None = None,
+                  labs: tuple[int, ...] = (1, 2)):
+    """Wire the app from a lab Deployment: the deployment's LLM (stub on
+    CPU, hipGraph engine on 'cuda'), the lab2 document index, and every
+    CREATE AGENT in the lab catalogs exposed at /v1/agents/{name} with
+    AI_RUN_AGENT semantics (model-driven TOOL_CALL loop against the stub
+    MCP server's tool set)."""
+    from .agents.mcp import McpClient, StubMcpServer
+    from .agents.runner import drive_episode, episode
+    from .labs.deploy import Deployment
+    from .labs.pipelines import mcp_tool_fn
+
+    dep = Deployment(labs=labs, device=device, model=model)
+    if 2 in labs:
+        dep.datagen(2)
+    server = StubMcpServer().start()
+    tool_fn = mcp_tool_fn(McpClient(server.mcp_endpoint))
+    llm_batch = dep.llm()
+
+    def make_agent_fn(spec):
+        def run(prompt: str, record_key=None):
+            return drive_episode(
+                episode(spec, prompt),
+                lambda p, t: llm_batch([p], [t])[0], tool_fn)
+        return run
+
+    agents = {name: make_agent_fn(dep.catalog.agent_spec(name))
+              for name in dep.catalog.agents}
+    app = create_app(llm_batch, embedder=dep.embedder,
+                     index=dep._index(2) if 2 in labs else None,
+                     agents=agents)
+    app.state.mcp_server = server          # kept alive with the app
+    return app
+
+
+def main(argv=None) -> int:
+    import argparse
+
+    import uvicorn
+    p = argparse.ArgumentParser(prog="qsa-serve")
+    p.add_argument("--host", default="127.0.0.1")
+    p.add_argument("--port", type=int, default=8080)
+    p.add_argument("--device", default="cpu")
+    p.add_argument("--model", default=None)
+    args = p.parse_args(argv)
+    app = build_lab_app(device=args.device, model=args.model)
+    uvicorn.run(app, host=args.host, port=args.port)
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())
