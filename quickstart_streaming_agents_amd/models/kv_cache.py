@@ -1,4 +1,5 @@
-"""Paged KV cache for the decode engine (page = 64 positions).
+"""Paged KV cache for the decode engine (page = 64 positions);
+backs the K4 agent-LLM decode of SURVEY.md 2.4.
 
 Layouts match the HIP decode kernel (ops/hip/paged_attn.hip):
   K: [pages, KVH, D/8, 64, 8]  (d-major x8 -> coalesced MFMA A-frag reads)

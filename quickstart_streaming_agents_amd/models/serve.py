@@ -1,4 +1,6 @@
-"""Continuous-batching serving engine for the agent LLM.
+"""Continuous-batching serving engine for the agent LLM — the local
+replacement for the managed ML_PREDICT('llm_textgen_model') endpoint
+(reference terraform/core/main.tf:461,495 models).
 
 Sequences are admitted into the running decode batch as KV pages free up
 (vLLM-style): each step() admits pending prompts via ONE batched prefill

@@ -1,4 +1,6 @@
-"""Deterministic air-gapped tokenizer.
+"""Deterministic air-gapped tokenizer (BASELINE.json: synthetic data /
+random-init weights contract — token identity only has to be
+deterministic, not linguistic).
 
 No network means no real BPE vocab files; the north-star benchmark runs
 random-init weights, so token IDENTITY only needs to be deterministic and

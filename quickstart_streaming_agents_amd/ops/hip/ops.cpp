@@ -1,5 +1,8 @@
 // Torch bindings for the qsa MI355X (gfx950) kernels.  HIP-native: no CUDA
 // naming, no compatibility shims — this extension only builds for ROCm.
+// Operator inventory these kernels implement: SURVEY.md 2.4 K1-K10
+// (embedding, cosine top-k, anomaly scoring, paged-attention LLM decode,
+// window aggregation, Avro wire codec).
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>

@@ -1,5 +1,6 @@
 // Skinny-batch decode GEMM for CDNA4 (gfx950): C[M,N] = A[M,K] @ W[N,K]^T,
 // M <= 32 (the decode batch), bf16 in / bf16 out.
+// Serves the K4 agent-LLM decode projections.
 //
 // Decode projections are HBM-bandwidth-bound on the WEIGHT stream (the
 // activations are KB-sized): speed-of-light is W bytes / 6.3 TB/s.  rocBLAS

@@ -1,4 +1,6 @@
-"""Distributed context for the 8x MI355X node.
+"""Distributed context for the 8x MI355X node (the
+reference's implicit parallelism — Kafka partitions / managed Flink —
+mapped to explicit DP/TP/EP groups).
 
 One process per GPU over torch.distributed: backend "nccl" IS RCCL on
 ROCm, collectives ride xGMI (7 p2p links x ~153 GB/s per GPU).  CPU-only

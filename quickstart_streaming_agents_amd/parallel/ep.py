@@ -1,4 +1,5 @@
-"""Expert-parallel token exchange (MoE all-to-all over xGMI).
+"""Expert-parallel token exchange (MoE all-to-all over xGMI;
+BASELINE.json config 5: Mixtral-8x7B TP+EP across 8 MI355X).
 
 Mixtral's routed FFN shards experts across the EP group (8 experts over up
 to 8 GPUs: each rank holds n_experts/ep_size experts' full weights —
