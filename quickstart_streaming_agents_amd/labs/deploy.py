@@ -24,6 +24,13 @@ SQL_DIR = os.path.join(os.path.dirname(os.path.abspath(__file__)), "sql")
 ALL_LABS = (1, 2, 3, 4)
 
 
+def _maybe(res, name):
+    try:
+        return res(name)
+    except Exception:
+        return None
+
+
 def lab_sql(lab: int | str) -> str:
     name = lab if isinstance(lab, str) else f"lab{lab}"
     with open(os.path.join(SQL_DIR, f"{name}.sql")) as fh:
@@ -145,6 +152,122 @@ class Deployment:
                 self.broker, llm, self.embedder, self._index(4),
                 params=params, agent=agent, **kw)
         raise ValueError(f"unknown lab {lab}")
+
+    # ---- generic SQL execution (sql/exec.py) -----------------------------
+    FINAL_TABLE = {1: "price_match_results", 2: "search_results_response",
+                   3: "completed_actions", 4: "claims_reviewed"}
+
+    def sql_executor(self, lab: int, mcp_server=None):
+        """Build a generic SqlExecutor for one lab's catalog: the same
+        statements labs/sql/*.sql declare, executed by sql/exec.py instead
+        of the hand-fused pipelines.  Prompt expressions the reference
+        inlines as giant CONCATs are supplied as named bindings; scripted
+        episode policies (the deterministic CPU stand-ins for a real
+        tool-calling LLM) come from agent_policies."""
+        from ..sql.exec import SqlExecutor
+        from . import schemas
+
+        topic_schemas = {
+            "orders": schemas.ORDERS, "customers": schemas.CUSTOMERS,
+            "products": schemas.PRODUCTS,
+            "ride_requests": schemas.RIDE_REQUESTS,
+            "claims": schemas.CLAIMS, "queries": schemas.QUERIES,
+            "documents": schemas.DOCUMENTS,
+        }
+        bindings: dict = {}
+        policies: dict = {}
+        indexes: dict = {}
+        tool_fn = None
+        if mcp_server is not None:
+            from ..agents.mcp import McpClient
+            tool_fn = pipelines.mcp_tool_fn(
+                McpClient(mcp_server.mcp_endpoint))
+        if lab == 1:
+            url = f"{mcp_server.base_url}/competitor" if mcp_server else ""
+            email = "customer@example.com"
+
+            def _order(res):
+                vals = {k: _maybe(res, k) for k in
+                        ("order_id", "product_name", "price", "order_price",
+                         "customer_email", "list_price")}
+                return {k: v for k, v in vals.items() if v is not None}
+
+            bindings["user_prompt"] = lambda res: pipelines.lab1_user_prompt(
+                _order(res), url, email)
+            policies["price_match_agent"] = \
+                lambda res: pipelines.Lab1PriceMatchPolicy(_order(res), url,
+                                                           email)
+        elif lab == 3:
+            indexes["documents_vectordb_lab3"] = self._index(3)
+            base = mcp_server.base_url if mcp_server else ""
+
+            def _surge(res):
+                return pipelines.lab3_surge_query(
+                    res("pickup_zone"), int(res("window_time")),
+                    int(res("request_count")), float(res("forecast_value")))
+
+            def _summarize(res):
+                ctx = "\n\n".join(
+                    f"[doc {i} | score {float(res(f'score{i}') or 0):.3f}] "
+                    f"{res(f'chunk{i}')}"
+                    for i in (1, 2, 3) if _maybe(res, f"chunk{i}"))
+                return (f"Answer the question using only the context "
+                        f"below.\n\nContext:\n{ctx}\n\n"
+                        f"Question: {_surge(res)}\nAnswer:")
+
+            bindings["surge_query"] = _surge
+            bindings["summarize_prompt"] = _summarize
+            policies["boat_dispatch_agent"] = \
+                lambda res: pipelines.Lab3DispatchPolicy(
+                    res("pickup_zone"), f"{base}/api/vessel_catalog",
+                    f"{base}/api/dispatch")
+        elif lab == 4:
+            indexes["fema_policies_vectordb"] = self._index(4)
+
+            def _claim(res):
+                return {k: _maybe(res, k) for k in
+                        ("claim_id", "applicant_name", "city", "claim_amount",
+                         "claim_narrative", "is_primary_residence",
+                         "damage_assessed", "has_insurance",
+                         "insurance_amount", "shared_account",
+                         "shared_phone")}
+
+            def _investigation(res):
+                c = _claim(res)
+                chunks = "\n".join(str(_maybe(res, f"chunk_{i}") or "")
+                                   for i in (1, 2, 3))
+                return (f"CLAIM {c['claim_id']} ({c['city']}): "
+                        f"${c['claim_amount']}\n"
+                        f"Narrative: {c['claim_narrative']}\n\n"
+                        f"Policy excerpts:\n{chunks}\n\nVerdict?")
+
+            bindings["investigation_prompt"] = _investigation
+            policies["claims_fraud_investigation_agent"] = \
+                lambda res: pipelines.Lab4FraudPolicy(_claim(res))
+        elif lab == 2:
+            indexes["documents_vectordb_lab2"] = self._index(2)
+
+            def _rag(res):
+                ctx = "\n\n".join(
+                    f"[doc {i} | score {float(res(f'score_{i}') or 0):.3f}] "
+                    f"{res(f'chunk_{i}')}"
+                    for i in (1, 2, 3) if _maybe(res, f"chunk_{i}"))
+                return (f"Answer the question using only the context "
+                        f"below.\n\nContext:\n{ctx}\n\n"
+                        f"Question: {res('query')}\nAnswer:")
+
+            bindings["rag_prompt"] = _rag
+        return SqlExecutor(self.catalog, self.broker, schemas=topic_schemas,
+                           embedder=self.embedder, indexes=indexes,
+                           llm_batch=self.llm(), tool_fn=tool_fn,
+                           bindings=bindings, agent_policies=policies)
+
+    def run_sql(self, lab: int, mcp_server=None) -> list[dict]:
+        """Run one lab end-to-end through the GENERIC SQL executor (the
+        hand-fused pipelines are `run()`)."""
+        ex = self.sql_executor(lab, mcp_server=mcp_server)
+        ex.run_inserts()
+        return ex.run_table(self.FINAL_TABLE[lab])
 
     # ---- destroy / summary ----------------------------------------------
     def destroy(self) -> None:

@@ -27,8 +27,12 @@ INSERT INTO queries_embed
 SELECT query, embedding
 FROM queries, LATERAL TABLE(ML_PREDICT('llm_embedding_model', query));
 
+-- unpack the top-3 hits into numbered columns (lab2 main.tf:292 shape)
 CREATE TABLE search_results AS
-SELECT qe.query, r.document_id, r.chunk, r.score
+SELECT qe.query,
+       r.chunk1 AS chunk_1, r.score1 AS score_1,
+       r.chunk2 AS chunk_2, r.score2 AS score_2,
+       r.chunk3 AS chunk_3, r.score3 AS score_3
 FROM queries_embed qe
 CROSS JOIN LATERAL TABLE(
   VECTOR_SEARCH_AGG(documents_vectordb_lab2, DESCRIPTOR(embedding),

@@ -42,14 +42,16 @@ CREATE TABLE documents_vectordb_lab3 (
 );
 
 CREATE TABLE anomalies_per_zone AS
-SELECT pickup_zone, window_time, request_count,
+SELECT pickup_zone, window_time, COUNT(*) AS request_count,
   ML_DETECT_ANOMALIES(CAST(request_count AS DOUBLE), window_time,
     JSON_OBJECT('minTrainingSize' VALUE 286, 'maxTrainingSize' VALUE 7000,
                 'confidencePercentage' VALUE 99.9, 'enableStl' VALUE FALSE))
     OVER (PARTITION BY pickup_zone ORDER BY window_time
           RANGE UNBOUNDED PRECEDING) AS anomaly
 FROM TABLE(TUMBLE(TABLE ride_requests, DESCRIPTOR(request_ts),
-                  INTERVAL '5' MINUTE));
+                  INTERVAL '5' MINUTE))
+GROUP BY pickup_zone, window_start, window_end, window_time
+HAVING anomaly.is_anomaly AND request_count > anomaly.upper_bound;
 
 CREATE TABLE anomalies_enriched WITH ('changelog.mode' = 'append') AS
 SELECT rad.pickup_zone, rad.window_time, rad.request_count,

@@ -4,14 +4,24 @@
 
 SET 'sql.state-ttl' = '14 d';
 
+-- 17 columns like the reference claims table (lab4 main.tf:55-76)
 CREATE TABLE claims (
   claim_id STRING,
   applicant_name STRING,
   city STRING,
-  state STRING,
+  is_primary_residence STRING,
+  damage_assessed STRING,
   claim_amount STRING,
-  damage_type STRING,
+  has_insurance STRING,
+  insurance_amount STRING,
   claim_narrative STRING,
+  assessment_date STRING,
+  disaster_date STRING,
+  previous_claims_count STRING,
+  last_claim_date STRING,
+  assessment_source STRING,
+  shared_account STRING,
+  shared_phone STRING,
   claim_timestamp TIMESTAMP(3),
   WATERMARK FOR claim_timestamp AS claim_timestamp - INTERVAL '5' SECOND
 );
@@ -34,18 +44,25 @@ CREATE TABLE fema_policies_vectordb (
 );
 
 CREATE TABLE claims_anomalies_by_city AS
-SELECT city, window_time, total_claim_amount,
+SELECT city, window_time,
+  SUM(CAST(claim_amount AS DOUBLE)) AS total_claim_amount,
+  COUNT(*) AS claim_count,
   ML_DETECT_ANOMALIES(total_claim_amount, window_time,
     JSON_OBJECT('minTrainingSize' VALUE 8, 'maxTrainingSize' VALUE 50,
                 'confidencePercentage' VALUE 95.0, 'enableStl' VALUE FALSE))
     OVER (PARTITION BY city ORDER BY window_time
           RANGE UNBOUNDED PRECEDING) AS anomaly
 FROM TABLE(TUMBLE(TABLE claims, DESCRIPTOR(claim_timestamp),
-                  INTERVAL '6' HOUR));
+                  INTERVAL '6' HOUR))
+GROUP BY city, window_start, window_end, window_time
+HAVING anomaly.is_anomaly AND total_claim_amount > anomaly.upper_bound;
 
 CREATE TABLE claims_to_investigate AS
 SELECT c.claim_id, c.applicant_name, c.city, c.claim_amount,
-       c.damage_type, c.claim_narrative, c.claim_timestamp
+       c.claim_narrative, c.claim_timestamp,
+       c.is_primary_residence, c.damage_assessed, c.has_insurance,
+       c.insurance_amount, c.shared_account, c.shared_phone,
+       c.previous_claims_count
 FROM claims c
 JOIN claims_anomalies_by_city a
   ON c.city = a.city
