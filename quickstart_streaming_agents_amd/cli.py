@@ -36,7 +36,10 @@ def cmd_run(args) -> int:
     dep.datagen(args.lab)
     server = StubMcpServer().start()
     try:
-        rows = dep.run(args.lab, mcp_server=server)
+        if args.sql:
+            rows = dep.run_sql(args.lab, mcp_server=server)
+        else:
+            rows = dep.run(args.lab, mcp_server=server)
     finally:
         server.stop()
     for r in rows[: args.max_print]:
@@ -149,6 +152,9 @@ def main(argv=None) -> int:
                         "SQL CREATE MODEL options (CPU runs use the "
                         "stub LLM regardless)")
     r.add_argument("--max-print", type=int, default=3)
+    r.add_argument("--sql", action="store_true",
+                   help="run through the generic SQL CTAS executor "
+                        "(sql/exec.py) instead of the fused pipelines")
     r.set_defaults(fn=cmd_run)
 
     g = sub.add_parser("datagen", help="publish one lab's synthetic stream")
