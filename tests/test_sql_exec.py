@@ -115,3 +115,52 @@ def test_executor_plain_projection_and_where():
     rows = SqlExecutor(cat, broker).run_table("big")
     assert rows == [{"name": "b", "q": 3.0, "tag": "b-x"},
                     {"name": "c", "q": 5.0, "tag": "c-x"}]
+
+
+def test_evaluator_expressions():
+    from quickstart_streaming_agents_amd.sql.exec import Evaluator, _Row
+    ev = Evaluator(bindings={"greet": lambda res: f"hi {res('name')}"})
+    row = _Row({"t": {"name": "ada", "n": 3, "ts": 10_000_000,
+                      "text": "Price: $42.50 end"}})
+    e = lambda x: ev.eval(x, row)
+    assert e("'it''s'") == "it's"
+    assert e("42") == 42 and e("4.5") == 4.5
+    assert e("TRUE") is True
+    assert e("t.name") == "ada" and e("name") == "ada"
+    assert e("CONCAT('a-', t.name, '-', t.n)") == "a-ada-3"
+    assert e("CAST(t.n AS DOUBLE)") == 3.0
+    assert e("CAST('7.9' AS INT)") == 7
+    assert e("UPPER(t.name)") == "ADA"
+    assert e("REGEXP_EXTRACT(t.text, '\\$(\\d+\\.\\d+)', 1)") == "42.50"
+    assert e("greet") == "hi ada"
+    assert e("t.ts - INTERVAL '1' HOUR") == 10_000_000 - 3_600_000
+    assert e("t.ts + INTERVAL '5' SECOND") == 10_005_000
+
+
+def test_evaluator_coalesce_and_errors():
+    import pytest
+
+    from quickstart_streaming_agents_amd.sql.exec import (Evaluator,
+                                                          SqlExecError, _Row)
+    ev = Evaluator()
+    row = _Row({"t": {"a": None, "b": "x"}})
+    assert ev.eval("COALESCE(t.a, t.b)", row) == "x"
+    with pytest.raises(SqlExecError, match="unbound"):
+        ev.eval("nope", row)
+
+
+def test_predicates_and_between():
+    from quickstart_streaming_agents_amd.sql.exec import Evaluator, _Row
+    ev = Evaluator()
+    row = _Row({"c": {"city": "Naples", "amt": 50.0, "ts": 500,
+                      "narr": ""}, "a": {"wt": 600, "flag": True}})
+    p = lambda c: ev.pred(c, row)
+    assert p("c.city = 'Naples' AND c.amt > 10")
+    assert not p("c.city <> 'Naples'")
+    assert p("c.narr <> 'x'")
+    assert not p("c.narr <> ''")
+    assert p("c.ts BETWEEN a.wt - INTERVAL '1' SECOND AND a.wt")
+    assert not p("c.ts BETWEEN 501 AND 600")
+    assert p("a.flag AND c.amt >= 50")
+    # string with AND inside stays intact
+    assert p("c.city <> 'rock AND roll'")
