@@ -451,8 +451,10 @@ class SqlExecutor:
         self._cache.pop(name, None)
         return self.table_rows(name)
 
-    def run_inserts(self) -> None:
-        """Apply catalog INSERT statements (VALUES and INSERT..SELECT)."""
+    def run_inserts(self, values_only: bool = False) -> None:
+        """Apply catalog INSERT statements (VALUES and INSERT..SELECT).
+        values_only=True applies just the VALUES inserts — streaming
+        pipelines (sql/stream.py) own the INSERT..SELECT statements."""
         for ins in self.catalog.inserts:
             topic = self.broker.create_topic(ins.table)
             if ins.values:
@@ -460,7 +462,7 @@ class SqlExecutor:
                         self.catalog.tables[ins.table].columns]
                 for vals in ins.values:
                     topic.append(dict(zip(cols, vals)), partition=0)
-            elif ins.select:
+            elif ins.select and not values_only:
                 for row in self.run_select(ins.select, sink=ins.table):
                     topic.append(row, partition=0)
             self._cache.pop(ins.table, None)

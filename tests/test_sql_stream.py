@@ -1,0 +1,123 @@
+"""Streaming (incremental) SQL execution tests (sql/stream.py): chunked
+arrival produces the same final results as batch execution, windows close
+on watermark advance (not at end), and a snapshot/restore mid-stream
+resumes exactly (the reference's continuous-Flink + replay-from-offset
+behavior)."""
+
+import json
+
+import pytest
+
+from quickstart_streaming_agents_amd.agents.mcp import StubMcpServer
+from quickstart_streaming_agents_amd.labs.deploy import Deployment
+from quickstart_streaming_agents_amd.sql.stream import StreamingPipeline
+
+
+@pytest.fixture(scope="module")
+def mcp():
+    with StubMcpServer() as srv:
+        yield srv
+
+
+def _chunked_records(dep, topic_name, n_chunks):
+    """Drain a populated topic and return its raw records in n chunks."""
+    topic = dep.broker.topics[topic_name]
+    recs = topic.read_all()
+    topic.purge()
+    size = max(1, len(recs) // n_chunks)
+    return [recs[i:i + size] for i in range(0, len(recs), size)]
+
+
+def _replay(dep, topic_name, recs):
+    t = dep.broker.topics[topic_name]
+    for r in recs:
+        t.append(r.value, key=r.key, timestamp_ms=r.timestamp_ms,
+                 partition=0)
+
+
+def test_lab3_streaming_matches_batch(mcp):
+    # batch reference
+    ref = Deployment(labs=(3,), device="cpu")
+    ref.datagen(3)
+    batch_rows = ref.run_sql(3, mcp_server=mcp)
+
+    # streaming run: same data arriving in 5 chunks
+    dep = Deployment(labs=(3,), device="cpu")
+    dep.datagen(3)
+    chunks = _chunked_records(dep, "ride_requests", 5)
+    pipe = StreamingPipeline(dep.sql_executor(3, mcp_server=mcp))
+    incr: list = []
+    for ch in chunks:
+        _replay(dep, "ride_requests", ch)
+        out = pipe.advance()
+        incr += out["completed_actions"]
+    final = pipe.finish()
+    incr += final["completed_actions"]
+
+    assert len(incr) == len(batch_rows) >= 1
+    assert [r["pickup_zone"] for r in incr] == \
+        [r["pickup_zone"] for r in batch_rows]
+    assert all(r["pickup_zone"] == "French Quarter" for r in incr)
+    # at least the training-period windows closed before the final flush
+    # (incremental watermark closure, not a single end-of-input batch)
+    assert pipe.by_sink["anomalies_per_zone"].windows.wm.current > 0
+
+
+def test_lab1_streaming_join_both_arrival_orders(mcp):
+    ref = Deployment(labs=(1,), device="cpu")
+    ref.datagen(1)
+    want = sorted(r["order_id"] for r in ref.run_sql(1, mcp_server=mcp))
+
+    # orders arrive BEFORE the dim tables: the two-sided join must emit
+    # when the late dimension side shows up
+    dep = Deployment(labs=(1,), device="cpu")
+    dep.datagen(1)
+    orders = _chunked_records(dep, "orders", 1)[0]
+    customers = _chunked_records(dep, "customers", 1)[0]
+    products = _chunked_records(dep, "products", 1)[0]
+    pipe = StreamingPipeline(dep.sql_executor(1, mcp_server=mcp))
+    got: list = []
+    _replay(dep, "orders", orders)
+    got += pipe.advance()["price_match_results"]
+    assert got == []                      # dims not there yet
+    _replay(dep, "customers", customers)
+    got += pipe.advance()["price_match_results"]
+    _replay(dep, "products", products)
+    got += pipe.advance()["price_match_results"]
+    assert sorted(r["order_id"] for r in got) == want
+    assert all(r["agent_status"] == "SUCCESS" for r in got)
+
+
+def test_lab3_snapshot_restore_resumes_exactly(mcp):
+    # uninterrupted reference
+    ref = Deployment(labs=(3,), device="cpu")
+    ref.datagen(3)
+    chunks_ref = _chunked_records(ref, "ride_requests", 4)
+    pipe_ref = StreamingPipeline(ref.sql_executor(3, mcp_server=mcp))
+    out_ref: list = []
+    for ch in chunks_ref:
+        _replay(ref, "ride_requests", ch)
+        out_ref += pipe_ref.advance()["completed_actions"]
+    out_ref += pipe_ref.finish()["completed_actions"]
+
+    # interrupted: advance 2 chunks, snapshot, restore into a NEW pipeline
+    dep = Deployment(labs=(3,), device="cpu")
+    dep.datagen(3)
+    chunks = _chunked_records(dep, "ride_requests", 4)
+    pipe1 = StreamingPipeline(dep.sql_executor(3, mcp_server=mcp))
+    out: list = []
+    for ch in chunks[:2]:
+        _replay(dep, "ride_requests", ch)
+        out += pipe1.advance()["completed_actions"]
+    snap = json.loads(json.dumps(pipe1.snapshot()))   # must be JSON-safe
+
+    pipe2 = StreamingPipeline(dep.sql_executor(3, mcp_server=mcp))
+    pipe2.restore(snap)
+    for ch in chunks[2:]:
+        _replay(dep, "ride_requests", ch)
+        out += pipe2.advance()["completed_actions"]
+    out += pipe2.finish()["completed_actions"]
+
+    assert [r["pickup_zone"] for r in out] == \
+        [r["pickup_zone"] for r in out_ref]
+    assert len(out) == len(out_ref) >= 1
