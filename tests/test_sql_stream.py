@@ -121,3 +121,25 @@ def test_lab3_snapshot_restore_resumes_exactly(mcp):
     assert [r["pickup_zone"] for r in out] == \
         [r["pickup_zone"] for r in out_ref]
     assert len(out) == len(out_ref) >= 1
+
+
+def test_lab2_streaming_query_arrives_late(mcp):
+    """queries -> queries_embed (INSERT..SELECT) -> search_results ->
+    search_results_response reacts to a query published AFTER the
+    pipeline started (continuous-statement behavior)."""
+    dep = Deployment(labs=(2,), device="cpu")
+    dep.datagen(2)
+    ex = dep.sql_executor(2)
+    ex.run_inserts(values_only=True)       # the terraform sample INSERT
+    pipe = StreamingPipeline(ex)
+    out1 = pipe.advance()
+    n1 = len(out1["search_results_response"])
+    assert n1 >= 1                         # the sample query answered
+    # nothing new -> nothing emitted
+    assert pipe.advance()["search_results_response"] == []
+    # late query flows through the whole chain in one advance
+    dep.broker.topics["queries"].append(
+        {"query": "What is a watermark in Flink?"}, partition=0)
+    out2 = pipe.advance()
+    assert len(out2["search_results_response"]) == 1
+    assert out2["search_results_response"][0]["response"]
