@@ -279,3 +279,22 @@ def test_gpu_window_rows_match_cpu_assigner():
             for r in gpu_rows] == \
         [(r["key"], r["window_start"], r["request_count"])
          for r in cpu_rows]
+
+
+def test_generic_sql_executor_on_gpu_engine():
+    """The generic CTAS executor (sql/exec.py) drives the REAL GPU stack:
+    EngineLLM decode, on-GPU embedding encoder, HBM-resident index —
+    lab3's French Quarter contract holds end-to-end through generic SQL
+    execution (CPU equivalence is tests/test_sql_exec.py)."""
+    from quickstart_streaming_agents_amd.agents.mcp import StubMcpServer
+    from quickstart_streaming_agents_amd.labs.deploy import Deployment
+    dep = Deployment(labs=(3,), device="cuda:0", model="tiny")
+    dep.datagen(3)
+    srv = StubMcpServer().start()
+    try:
+        rows = dep.run_sql(3, mcp_server=srv)
+    finally:
+        srv.stop()
+    assert 1 <= len(rows) <= 2
+    assert all(r["pickup_zone"] == "French Quarter" for r in rows)
+    assert all(r["api_response"] for r in rows)
