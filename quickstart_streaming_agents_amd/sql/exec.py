@@ -115,6 +115,8 @@ class Evaluator:
             return up == "TRUE"
         if up.startswith("MAP["):
             return None                      # debug maps: accepted, ignored
+        if up.startswith("CASE") and up.endswith("END"):
+            return self._case(expr, row)
         m = re.match(r"(\w+)\s*\(", expr)
         if m and expr.endswith(")"):
             close = P._find_matching_paren(expr, m.end() - 1)
@@ -139,6 +141,32 @@ class Evaluator:
                 raise SqlExecError(f"unbound identifier {expr!r} "
                                    f"(add a binding?)")
         raise SqlExecError(f"unsupported expression {expr!r}")
+
+    def _case(self, expr: str, row: _Row):
+        """CASE WHEN cond THEN v [WHEN ...] [ELSE v] END (searched form)."""
+        body = expr.strip()[4:].strip()
+        if body.upper().endswith("END"):
+            body = body[:-3].strip()
+        # split on top-level WHEN/ELSE keywords
+        tokens = re.split(r"\b(WHEN|THEN|ELSE)\b", body,
+                          flags=re.IGNORECASE)
+        # tokens like ['', 'WHEN', cond, 'THEN', val, 'WHEN', ...]
+        i = 0
+        default = None
+        while i < len(tokens):
+            kw = tokens[i].strip().upper()
+            if kw == "WHEN":
+                cond, val = tokens[i + 1], tokens[i + 3]
+                assert tokens[i + 2].strip().upper() == "THEN", expr
+                if self.pred(cond, row):
+                    return self.eval(val, row)
+                i += 4
+            elif kw == "ELSE":
+                default = tokens[i + 1]
+                i += 2
+            else:
+                i += 1
+        return self.eval(default, row) if default is not None else None
 
     def _call(self, fn: str, args: list[str], row: _Row, expr: str):
         if fn == "CONCAT":

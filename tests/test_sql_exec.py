@@ -164,3 +164,24 @@ def test_predicates_and_between():
     assert p("a.flag AND c.amt >= 50")
     # string with AND inside stays intact
     assert p("c.city <> 'rock AND roll'")
+
+
+def test_case_when_expression():
+    from quickstart_streaming_agents_amd.sql.exec import Evaluator, _Row
+    ev = Evaluator()
+    # the lab3 reference builds time-of-day buckets with CASE
+    # (LAB3-Walkthrough.md:271-339 surge query CONCAT)
+    expr = ("CASE WHEN t.h >= 5 AND t.h < 12 THEN 'morning' "
+            "WHEN t.h >= 12 AND t.h < 17 THEN 'afternoon' "
+            "WHEN t.h >= 17 AND t.h < 21 THEN 'evening' "
+            "ELSE 'late night' END")
+    for h, want in ((6, "morning"), (13, "afternoon"), (19, "evening"),
+                    (2, "late night"), (23, "late night")):
+        assert ev.eval(expr, _Row({"t": {"h": h}})) == want
+    # CASE nested inside CONCAT
+    got = ev.eval("CONCAT('bucket=', CASE WHEN t.h > 0 THEN 'day' "
+                  "ELSE 'night' END)", _Row({"t": {"h": 1}}))
+    assert got == "bucket=day"
+    # no ELSE and nothing matches -> NULL
+    assert ev.eval("CASE WHEN t.h > 99 THEN 'x' END",
+                   _Row({"t": {"h": 1}})) is None
