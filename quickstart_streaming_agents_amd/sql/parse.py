@@ -169,9 +169,14 @@ def _ident(s: str) -> str:
 
 
 def _split_top(s: str, sep: str = ",") -> list[str]:
-    """Split on sep at paren/angle/string depth 0."""
+    """Split on sep at paren/angle/string depth 0.
+
+    '<' only opens generic-type depth when immediately followed by an
+    identifier character (MAP<STRING, ...>); a comparison like
+    ``a >= 90`` or ``a < 12`` must not unbalance the scan."""
     parts, cur = [], []
     depth = 0
+    angle = 0
     in_str = False
     i, n = 0, len(s)
     while i < n:
@@ -187,13 +192,20 @@ def _split_top(s: str, sep: str = ",") -> list[str]:
         elif ch == "'":
             in_str = True
             cur.append(ch)
-        elif ch in "(<[":
+        elif ch in "([":
             depth += 1
             cur.append(ch)
-        elif ch in ")>]":
+        elif ch in ")]":
             depth -= 1
             cur.append(ch)
-        elif ch == sep and depth == 0:
+        elif ch == "<" and i + 1 < n and (s[i + 1].isalnum()
+                                          or s[i + 1] in "_<"):
+            angle += 1
+            cur.append(ch)
+        elif ch == ">" and angle > 0:
+            angle -= 1
+            cur.append(ch)
+        elif ch == sep and depth == 0 and angle == 0:
             parts.append("".join(cur))
             cur = []
         else:

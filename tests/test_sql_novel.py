@@ -1,0 +1,79 @@
+"""A NOVEL user statement (not one of the four labs) through the generic
+executor — the 'bring your own SQL' capability the reference gives users
+via Confluent Flink (docs/SQL.md walks through this example)."""
+
+from quickstart_streaming_agents_amd.sql.catalog import Catalog
+from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+from quickstart_streaming_agents_amd.wire import Broker
+
+DDL = """
+CREATE TABLE sensor_readings (
+  sensor_id STRING,
+  site STRING,
+  temp_c DOUBLE,
+  reading_ts TIMESTAMP_LTZ(3),
+  WATERMARK FOR reading_ts AS reading_ts - INTERVAL '5' SECOND
+);
+
+CREATE TABLE site_meta (site STRING, region STRING, alert_email STRING);
+
+CREATE TABLE hot_windows AS
+SELECT site, window_time, COUNT(*) AS n_readings,
+  AVG(CAST(temp_c AS DOUBLE)) AS avg_temp,
+  ML_DETECT_ANOMALIES(avg_temp, window_time,
+    JSON_OBJECT('minTrainingSize' VALUE 4, 'maxTrainingSize' VALUE 50,
+                'confidencePercentage' VALUE 95.0, 'enableStl' VALUE FALSE))
+    OVER (PARTITION BY site ORDER BY window_time
+          RANGE UNBOUNDED PRECEDING) AS anomaly
+FROM TABLE(TUMBLE(TABLE sensor_readings, DESCRIPTOR(reading_ts),
+                  INTERVAL '10' MINUTE))
+GROUP BY site, window_start, window_end, window_time
+HAVING anomaly.is_anomaly AND avg_temp > anomaly.upper_bound;
+
+CREATE TABLE hot_alerts AS
+SELECT h.site, m.region, m.alert_email, h.avg_temp,
+  CASE WHEN h.avg_temp >= 90 THEN 'CRITICAL'
+       WHEN h.avg_temp >= 70 THEN 'WARNING'
+       ELSE 'INFO' END AS severity,
+  CONCAT('Heat anomaly at ', h.site, ' (', m.region, '): avg ',
+         CAST(h.avg_temp AS INT), 'C') AS message
+FROM hot_windows h
+JOIN site_meta m ON h.site = m.site
+WHERE m.region <> 'decommissioned';
+"""
+
+
+def test_novel_anomaly_alert_statement():
+    cat = Catalog()
+    cat.execute(DDL)
+    broker = Broker()
+    meta = broker.create_topic("site_meta")
+    meta.append({"site": "plant-a", "region": "gulf",
+                 "alert_email": "ops@a"}, partition=0)
+    meta.append({"site": "plant-b", "region": "decommissioned",
+                 "alert_email": "ops@b"}, partition=0)
+    readings = broker.create_topic("sensor_readings")
+    MIN10 = 600_000
+    ts = 0
+    for w in range(12):                       # 11 calm windows then a spike
+        for s, base in (("plant-a", 20.0), ("plant-b", 20.0)):
+            for i in range(3):
+                t = w * MIN10 + i * 60_000
+                temp = base + (80.0 if w == 11 else 0.0) + 0.1 * i
+                readings.append({"sensor_id": f"{s}-{i}", "site": s,
+                                 "temp_c": temp, "reading_ts": t},
+                                partition=0)
+        ts = w * MIN10
+
+    ex = SqlExecutor(cat, broker)
+    alerts = ex.run_table("hot_alerts")
+    # plant-b spiked too but its site row is decommissioned -> filtered
+    assert len(alerts) == 1
+    a = alerts[0]
+    assert a["site"] == "plant-a" and a["region"] == "gulf"
+    assert a["severity"] == "CRITICAL"
+    assert a["message"].startswith("Heat anomaly at plant-a (gulf)")
+    assert a["avg_temp"] > 90
+    # sinks materialized as topics for downstream consumers
+    assert broker.topics["hot_windows"].message_count() == 2
+    assert broker.topics["hot_alerts"].message_count() == 1
