@@ -106,6 +106,42 @@ def cmd_destroy(args) -> int:
     return 0
 
 
+def cmd_sql(args) -> int:
+    """Execute a .sql file (the lab grammar, docs/SQL.md) against JSONL
+    topic data: --data topic=path.jsonl (repeatable), then print the rows
+    of --table (or every CTAS)."""
+    from .sql.catalog import Catalog
+    from .sql.exec import SqlExecutor
+    from .wire import Broker
+    cat = Catalog()
+    with open(args.file) as fh:
+        cat.execute(fh.read())
+    broker = Broker()
+    for spec in args.data or []:
+        topic_name, _, path = spec.partition("=")
+        t = broker.create_topic(topic_name)
+        with open(path) as fh:
+            for line in fh:
+                line = line.strip()
+                if line:
+                    t.append(json.loads(line), partition=0)
+    ex = SqlExecutor(cat, broker)
+    ex.run_inserts()
+    targets = [args.table] if args.table else \
+        [n for n, t in cat.tables.items() if t.as_select]
+    for name in targets:
+        if args.explain:
+            print(f"-- plan for {name}:")
+            for step in ex.explain(name):
+                print(f"   {step}")
+            continue
+        rows = ex.run_table(name)
+        print(f"-- {name}: {len(rows)} rows")
+        for r in rows[: args.max_print]:
+            print(json.dumps(r, default=str)[:400])
+    return 0
+
+
 def cmd_serve(args) -> int:
     from .serve_api import main as serve_main
     return serve_main(["--host", args.host, "--port", str(args.port),
@@ -167,6 +203,18 @@ def main(argv=None) -> int:
     x = sub.add_parser("destroy", help="remove deployment artifacts")
     x.add_argument("--dir", default="deploy_out")
     x.set_defaults(fn=cmd_destroy)
+
+    q = sub.add_parser("sql", help="execute a .sql file against JSONL "
+                       "topic data (docs/SQL.md grammar)")
+    q.add_argument("--file", required=True)
+    q.add_argument("--data", action="append",
+                   help="topic=path.jsonl (repeatable)")
+    q.add_argument("--table", default=None,
+                   help="CTAS to materialize (default: all)")
+    q.add_argument("--explain", action="store_true",
+                   help="print the stage plan instead of executing")
+    q.add_argument("--max-print", type=int, default=5)
+    q.set_defaults(fn=cmd_sql)
 
     s = sub.add_parser("serve", help="HTTP serving API "
                        "(completions/embeddings/search/agents)")

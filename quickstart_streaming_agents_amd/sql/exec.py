@@ -495,6 +495,37 @@ class SqlExecutor:
                     topic.append(row, partition=0)
             self._cache.pop(ins.table, None)
 
+    def explain(self, table: str) -> list[str]:
+        """Human-readable stage plan for a CTAS (what run_table will do)."""
+        t = self.catalog.tables[table]
+        if not t.as_select:
+            return [f"scan topic {table}"]
+        info = analyze_select(t.as_select)
+        clauses = _split_clauses(t.as_select)
+        from_clause, laterals = _extract_laterals(clauses["from"])
+        plan = []
+        if info.tumble:
+            plan.append(
+                f"tumble {info.tumble['table']} every "
+                f"{info.tumble['window_ms']} ms on {info.tumble['ts_col']}")
+            if info.anomaly:
+                plan.append(f"anomaly-detect {info.anomaly[0]}")
+        else:
+            tables = _parse_joins(from_clause)
+            plan.append(f"scan {tables[0][0]}")
+            for name, alias, cond in tables[1:]:
+                plan.append(f"hash-join {name} [{alias}] on {cond}")
+        for lat in laterals:
+            plan.append(f"lateral {lat['call'].split('(')[0].strip()}"
+                        f"{' AS ' + lat['alias'] if lat['alias'] else ''}")
+        for key in ("where", "having"):
+            if clauses[key]:
+                plan.append(f"{key} {clauses[key]}")
+        if clauses["limit"]:
+            plan.append(f"limit {clauses['limit']}")
+        plan.append(f"project -> {table}")
+        return plan
+
     # -- core ---------------------------------------------------------------
     def run_select(self, select_sql: str, sink: str | None = None
                    ) -> list[dict]:

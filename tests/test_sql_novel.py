@@ -77,3 +77,45 @@ def test_novel_anomaly_alert_statement():
     # sinks materialized as topics for downstream consumers
     assert broker.topics["hot_windows"].message_count() == 2
     assert broker.topics["hot_alerts"].message_count() == 1
+
+
+def test_cli_sql_runner(tmp_path):
+    """The `sql` CLI subcommand executes a user .sql file against JSONL
+    topic data and prints materialized rows / plans."""
+    import json as J
+
+    from quickstart_streaming_agents_amd.cli import main
+    sql_file = tmp_path / "alerts.sql"
+    sql_file.write_text(DDL)
+    readings = tmp_path / "readings.jsonl"
+    MIN10 = 600_000
+    with open(readings, "w") as fh:
+        for w in range(12):
+            for i in range(3):
+                t = w * MIN10 + i * 60_000
+                temp = 20.0 + (80.0 if w == 11 else 0.0) + 0.1 * i
+                fh.write(J.dumps({"sensor_id": f"s{i}", "site": "plant-a",
+                                  "temp_c": temp, "reading_ts": t}) + "\n")
+    meta = tmp_path / "meta.jsonl"
+    meta.write_text(J.dumps({"site": "plant-a", "region": "gulf",
+                             "alert_email": "ops@a"}) + "\n")
+    rc = main(["sql", "--file", str(sql_file),
+               "--data", f"sensor_readings={readings}",
+               "--data", f"site_meta={meta}", "--table", "hot_alerts"])
+    assert rc == 0
+    rc2 = main(["sql", "--file", str(sql_file), "--explain"])
+    assert rc2 == 0
+
+
+def test_explain_plans():
+    cat = Catalog()
+    cat.execute(DDL)
+    ex = SqlExecutor(cat, Broker())
+    hw = ex.explain("hot_windows")
+    assert any("tumble sensor_readings" in s for s in hw)
+    assert any("anomaly-detect" in s for s in hw)
+    assert any("having" in s for s in hw)
+    ha = ex.explain("hot_alerts")
+    assert any(s.startswith("scan hot_windows") for s in ha)
+    assert any("hash-join site_meta" in s for s in ha)
+    assert ha[-1] == "project -> hot_alerts"
