@@ -434,7 +434,7 @@ class SqlExecutor:
                  llm_batch=None, tool_fn=None,
                  bindings: dict[str, Callable] | None = None,
                  agent_policies: dict[str, Callable] | None = None,
-                 max_new_tokens: int = 96):
+                 max_new_tokens: int = 96, tracer=None):
         self.catalog = catalog
         self.broker = broker
         self.schemas = schemas or {}
@@ -445,6 +445,10 @@ class SqlExecutor:
         self.ev = Evaluator(bindings)
         self.agent_policies = agent_policies or {}
         self.max_new_tokens = max_new_tokens
+        if tracer is None:
+            from ..runtime.trace import Tracer
+            tracer = Tracer("sql", enabled=False)
+        self.tracer = tracer
         self._cache: dict[str, list[dict]] = {}
 
     # -- row sources --------------------------------------------------------
@@ -534,13 +538,22 @@ class SqlExecutor:
         from_clause, laterals = _extract_laterals(clauses["from"])
         items = _parse_select_items(clauses["select"])
 
-        if info.tumble:
-            rows = self._tumble_rows(info, items, select_sql)
-        else:
-            rows = self._join_rows(from_clause)
+        stage_name = "tumble" if info.tumble else "scan_join"
+        with self.tracer.stage(f"{sink or 'select'}:{stage_name}") as sp:
+            if info.tumble:
+                rows = self._tumble_rows(info, items, select_sql)
+            else:
+                rows = self._join_rows(from_clause)
+            if sp:
+                sp.records_out = len(rows)
 
         for lat in laterals:
-            rows = self._apply_lateral(lat, rows, items, select_sql)
+            op = lat["call"].split("(")[0].strip().lower()
+            with self.tracer.stage(f"{sink or 'select'}:{op}",
+                                   records_in=len(rows)) as sp:
+                rows = self._apply_lateral(lat, rows, items, select_sql)
+                if sp:
+                    sp.records_out = len(rows)
 
         for cond_key in ("where", "having"):
             cond = clauses[cond_key]

@@ -119,3 +119,26 @@ def test_explain_plans():
     assert any(s.startswith("scan hot_windows") for s in ha)
     assert any("hash-join site_meta" in s for s in ha)
     assert ha[-1] == "project -> hot_alerts"
+
+
+def test_executor_tracing():
+    from quickstart_streaming_agents_amd.runtime.trace import Tracer
+    cat = Catalog()
+    cat.execute(DDL)
+    broker = Broker()
+    broker.create_topic("site_meta").append(
+        {"site": "plant-a", "region": "gulf", "alert_email": "x"},
+        partition=0)
+    t = broker.create_topic("sensor_readings")
+    for w in range(12):
+        for i in range(3):
+            t.append({"sensor_id": f"s{i}", "site": "plant-a",
+                      "temp_c": 20.0 + (80.0 if w == 11 else 0.0),
+                      "reading_ts": w * 600_000 + i * 60_000}, partition=0)
+    tracer = Tracer("novel", enabled=True)
+    ex = SqlExecutor(cat, broker, tracer=tracer)
+    ex.run_table("hot_alerts")
+    stages = tracer.summary()["stages"]
+    assert "hot_windows:tumble" in stages
+    assert "hot_alerts:scan_join" in stages
+    assert stages["hot_windows:tumble"]["records_out"] >= 1
