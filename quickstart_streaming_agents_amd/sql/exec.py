@@ -301,23 +301,56 @@ _CLAUSE_RE = re.compile(
     re.IGNORECASE | re.DOTALL)
 
 
+_CLAUSE_KWS = ("WHERE", "GROUP BY", "HAVING", "ORDER BY", "LIMIT")
+
+
+def _kw_top(s: str, kws=_CLAUSE_KWS) -> tuple[int, int, str] | None:
+    """First top-level (outside parens/strings) clause keyword:
+    (start, end, keyword) or None."""
+    depth, in_str = 0, False
+    i, n = 0, len(s)
+    while i < n:
+        ch = s[i]
+        if in_str:
+            if ch == "'":
+                if i + 1 < n and s[i + 1] == "'":
+                    i += 1
+                else:
+                    in_str = False
+        elif ch == "'":
+            in_str = True
+        elif ch == "(":
+            depth += 1
+        elif ch == ")":
+            depth -= 1
+        elif depth == 0 and (i == 0 or not s[i - 1].isalnum()):
+            for kw in kws:
+                if s[i:i + len(kw)].upper() == kw and \
+                        (i + len(kw) >= n or not s[i + len(kw)].isalnum()):
+                    return i, i + len(kw), kw
+        i += 1
+    return None
+
+
 def _split_clauses(select_sql: str) -> dict:
     m = _CLAUSE_RE.search(select_sql)
     if not m:
         raise SqlExecError("no SELECT ... FROM found")
     rest = m.group("rest")
     out = {"select": m.group("select").strip(), "where": None,
-           "having": None, "group_by": None, "limit": None}
-    for kw, key in (("WHERE", "where"), ("GROUP BY", "group_by"),
-                    ("HAVING", "having"), ("LIMIT", "limit")):
-        mm = re.search(rf"\b{kw}\b", rest, re.IGNORECASE)
-        if mm:
-            tail = rest[mm.end():]
-            stop = re.search(r"\b(WHERE|GROUP BY|HAVING|LIMIT)\b", tail,
-                             re.IGNORECASE)
-            out[key] = (tail[:stop.start()] if stop else tail).strip()
-            rest = rest[:mm.start()] + (tail[stop.start():] if stop else "")
-    out["from"] = rest.strip().rstrip(";").strip()
+           "having": None, "group_by": None, "order_by": None,
+           "limit": None}
+    first = _kw_top(rest)
+    out["from"] = (rest[:first[0]] if first else rest) \
+        .strip().rstrip(";").strip()
+    while first:
+        start, end, kw = first
+        tail = rest[end:]
+        nxt = _kw_top(tail)
+        body = (tail[:nxt[0]] if nxt else tail).strip().rstrip(";").strip()
+        out[kw.lower().replace(" ", "_")] = body
+        rest = tail
+        first = nxt
     return out
 
 
@@ -560,6 +593,8 @@ class SqlExecutor:
             if cond:
                 rows = [r for r in rows if self.ev.pred(cond, r)]
 
+        if clauses["order_by"]:
+            rows = self._order(rows, clauses["order_by"])
         if clauses["limit"]:
             rows = rows[: int(clauses["limit"].split()[0])]
 
@@ -569,6 +604,20 @@ class SqlExecutor:
             for row in out:
                 topic.append(row, partition=0)
         return out
+
+    def _order(self, rows: list[_Row], order_by: str) -> list[_Row]:
+        """Stable multi-key ORDER BY col [ASC|DESC], applied last-first."""
+        specs = []
+        for item in P._split_top(order_by):
+            toks = item.strip().split()
+            desc = toks[-1].upper() == "DESC"
+            expr = (" ".join(toks[:-1])
+                    if toks[-1].upper() in ("ASC", "DESC") else item.strip())
+            specs.append((expr, desc))
+        for expr, desc in reversed(specs):
+            rows = sorted(rows, key=lambda r: self.ev.eval(expr, r),
+                          reverse=desc)
+        return rows
 
     # -- FROM stage ---------------------------------------------------------
     def _join_rows(self, from_clause: str) -> list[_Row]:

@@ -185,3 +185,24 @@ def test_case_when_expression():
     # no ELSE and nothing matches -> NULL
     assert ev.eval("CASE WHEN t.h > 99 THEN 'x' END",
                    _Row({"t": {"h": 1}})) is None
+
+
+def test_order_by_clause():
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+    cat = Catalog()
+    cat.execute("""
+    CREATE TABLE src (name STRING, grp STRING, qty INT);
+    CREATE TABLE top2 AS
+    SELECT s.name, s.grp, s.qty FROM src s
+    ORDER BY s.grp ASC, s.qty DESC LIMIT 3;
+    """)
+    broker = Broker()
+    t = broker.create_topic("src")
+    for name, grp, qty in (("a", "y", 5), ("b", "x", 1), ("c", "x", 9),
+                           ("d", "y", 7), ("e", "x", 4)):
+        t.append({"name": name, "grp": grp, "qty": qty}, partition=0)
+    rows = SqlExecutor(cat, broker).run_table("top2")
+    assert [(r["grp"], r["qty"]) for r in rows] == \
+        [("x", 9), ("x", 4), ("x", 1)]
