@@ -83,6 +83,10 @@ class Evaluator:
 
     def __init__(self, bindings: dict[str, Callable] | None = None):
         self.bindings = bindings or {}
+        self.fns: dict[str, Callable] = {}   # extension functions
+                                             # (executor registers
+                                             # ML_PREDICT / AI_TOOL_INVOKE
+                                             # scalar forms)
 
     # -- helpers -----------------------------------------------------------
     @staticmethod
@@ -113,8 +117,13 @@ class Evaluator:
         up = expr.upper()
         if up in ("TRUE", "FALSE"):
             return up == "TRUE"
-        if up.startswith("MAP["):
-            return None                      # debug maps: accepted, ignored
+        if up.startswith("MAP[") and expr.endswith("]"):
+            # MAP['k','v',...] literal -> dict (MAP[] -> {})
+            inner = expr[4:-1].strip()
+            if not inner:
+                return {}
+            vals = [self.eval(a, row) for a in P._split_top(inner)]
+            return dict(zip(vals[0::2], vals[1::2]))
         if up.startswith("CASE") and up.endswith("END"):
             return self._case(expr, row)
         m = re.match(r"(\w+)\s*\(", expr)
@@ -195,6 +204,8 @@ class Evaluator:
             pattern = self._unquote(args[1].strip())
             group = int(args[2]) if len(args) > 2 else 1
             return regexp_extract(str(subject or ""), pattern, group)
+        if fn in self.fns:
+            return self.fns[fn](args, row)
         raise SqlExecError(f"unsupported function in {expr!r}")
 
     # -- boolean predicates -------------------------------------------------
@@ -335,7 +346,13 @@ def _kw_top(s: str, kws=_CLAUSE_KWS) -> tuple[int, int, str] | None:
 def _split_clauses(select_sql: str) -> dict:
     m = _CLAUSE_RE.search(select_sql)
     if not m:
-        raise SqlExecError("no SELECT ... FROM found")
+        ms = re.match(r"\s*SELECT\b(.*)$", select_sql,
+                      re.IGNORECASE | re.DOTALL)
+        if not ms:
+            raise SqlExecError("no SELECT found")
+        return {"select": ms.group(1).strip().rstrip(";").strip(),
+                "from": None, "where": None, "having": None,
+                "group_by": None, "order_by": None, "limit": None}
     rest = m.group("rest")
     out = {"select": m.group("select").strip(), "where": None,
            "having": None, "group_by": None, "order_by": None,
@@ -482,6 +499,8 @@ class SqlExecutor:
             from ..runtime.trace import Tracer
             tracer = Tracer("sql", enabled=False)
         self.tracer = tracer
+        self.ev.fns["ML_PREDICT"] = self._fn_ml_predict
+        self.ev.fns["AI_TOOL_INVOKE"] = self._fn_tool_invoke
         self._cache: dict[str, list[dict]] = {}
 
     # -- row sources --------------------------------------------------------
@@ -532,6 +551,31 @@ class SqlExecutor:
                     topic.append(row, partition=0)
             self._cache.pop(ins.table, None)
 
+    # -- scalar operator forms (walkthrough smoke statements:
+    # LAB1-Walkthrough.md:66-92 SELECT ML_PREDICT / AI_TOOL_INVOKE) ------
+    def _fn_ml_predict(self, args: list[str], row: _Row):
+        model_name = args[0].strip().strip("'")
+        model = self.catalog.models.get(model_name)
+        text = str(self.ev.eval(args[1], row))
+        if model and model.outputs and \
+                "ARRAY" in model.outputs[0].type.upper():
+            if self.embedder is None:
+                raise SqlExecError("no embedder configured")
+            return self.embedder.embed(text).tolist()
+        if self.llm_batch is None:
+            raise SqlExecError("no LLM configured")
+        return self.llm_batch([text], [self.max_new_tokens])[0]
+
+    def _fn_tool_invoke(self, args: list[str], row: _Row):
+        from ..agents.runner import ai_tool_invoke
+        if self.llm_batch is None:
+            raise SqlExecError("no LLM configured")
+        prompt = str(self.ev.eval(args[1], row))
+        tools = self.ev.eval(args[3], row) if len(args) > 3 else {}
+        return ai_tool_invoke(
+            lambda p, t: self.llm_batch([p], [t])[0],
+            self.tool_fn or (lambda n, a: ""), prompt, tools or {})
+
     def explain(self, table: str) -> list[str]:
         """Human-readable stage plan for a CTAS (what run_table will do)."""
         t = self.catalog.tables[table]
@@ -568,12 +612,14 @@ class SqlExecutor:
                    ) -> list[dict]:
         info = analyze_select(select_sql)
         clauses = _split_clauses(select_sql)
-        from_clause, laterals = _extract_laterals(clauses["from"])
+        from_clause, laterals = _extract_laterals(clauses["from"] or "")
         items = _parse_select_items(clauses["select"])
 
         stage_name = "tumble" if info.tumble else "scan_join"
         with self.tracer.stage(f"{sink or 'select'}:{stage_name}") as sp:
-            if info.tumble:
+            if clauses["from"] is None:
+                rows = [_Row({})]            # scalar SELECT (smoke tests)
+            elif info.tumble:
                 rows = self._tumble_rows(info, items, select_sql)
             else:
                 rows = self._join_rows(from_clause)

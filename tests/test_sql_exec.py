@@ -206,3 +206,50 @@ def test_order_by_clause():
     rows = SqlExecutor(cat, broker).run_table("top2")
     assert [(r["grp"], r["qty"]) for r in rows] == \
         [("x", 9), ("x", 4), ("x", 1)]
+
+
+def test_walkthrough_smoke_statements():
+    """The reference walkthrough smoke tests (LAB1-Walkthrough.md:66-92):
+    scalar SELECT ML_PREDICT / AI_TOOL_INVOKE without FROM."""
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.vector.index import HashingEmbedder
+    from quickstart_streaming_agents_amd.wire import Broker
+
+    cat = Catalog()
+    cat.execute("""
+    CREATE MODEL llm_textgen_model INPUT (prompt STRING)
+      OUTPUT (response STRING) WITH ('provider' = 'local');
+    CREATE MODEL llm_embedding_model INPUT (text STRING)
+      OUTPUT (embedding ARRAY<FLOAT>) WITH ('provider' = 'local');
+    """)
+    calls = []
+
+    def llm(prompts, toks):
+        calls.append(prompts)
+        return ['TOOL_CALL {"name": "send_email", "arguments": '
+                '{"to": "a@b"}}'] * len(prompts)
+
+    ex = SqlExecutor(cat, Broker(), embedder=HashingEmbedder(),
+                     llm_batch=llm,
+                     tool_fn=lambda n, a: f"sent to {a['to']}")
+    # textgen smoke
+    [row] = ex.run_select(
+        "SELECT ML_PREDICT('llm_textgen_model', 'What is Flink?') "
+        "AS answer")
+    assert row["answer"].startswith("TOOL_CALL")
+    # embedding smoke: 1536 dims
+    [erow] = ex.run_select(
+        "SELECT ML_PREDICT('llm_embedding_model', 'hello') AS embedding")
+    assert len(erow["embedding"]) == 1536
+    # tool-invoke smoke: tool called, dict flattened into the row
+    [trow] = ex.run_select(
+        "SELECT AI_TOOL_INVOKE('remote_mcp_model', 'send the email', "
+        "MAP[], MAP['send_email', 'Send an email']) AS agent_output")
+    assert trow["send_email"] == "sent to a@b"
+    assert "response" in trow
+    # MAP literal evaluates to a dict
+    from quickstart_streaming_agents_amd.sql.exec import _Row
+    assert ex.ev.eval("MAP['a','1','b','2']", _Row({})) == \
+        {"a": "1", "b": "2"}
+    assert ex.ev.eval("MAP[]", _Row({})) == {}
