@@ -127,6 +127,15 @@ def cmd_sql(args) -> int:
                     t.append(json.loads(line), partition=0)
     ex = SqlExecutor(cat, broker)
     ex.run_inserts()
+    # read-only statements in the file print their results
+    from .sql import parse as P
+    for st in P.parse_script(open(args.file).read()):
+        if isinstance(st, P.ShowStmt):
+            print(f"-- SHOW {st.kind}: {', '.join(cat.show(st.kind))}")
+        elif isinstance(st, P.DescribeStmt):
+            print(f"-- DESCRIBE {st.name}:")
+            for col, ty in cat.describe(st.name):
+                print(f"   {col:32s} {ty}")
     targets = [args.table] if args.table else \
         [n for n, t in cat.tables.items() if t.as_select]
     for name in targets:

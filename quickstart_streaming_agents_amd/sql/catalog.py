@@ -122,7 +122,23 @@ class Catalog:
             self.apply(st)
         return parsed
 
+    def show(self, kind: str) -> list[str]:
+        store = {"TABLES": self.tables, "MODELS": self.models,
+                 "CONNECTIONS": self.connections, "TOOLS": self.tools,
+                 "AGENTS": self.agents}[kind.upper()]
+        return sorted(store)
+
+    def describe(self, name: str) -> list[tuple[str, str]]:
+        """(column, type) rows for a table (Flink DESCRIBE shape)."""
+        t = self.tables[name]
+        out = [(col.name, col.type) for col in t.columns]
+        if t.as_select and not out:
+            out = [("(CTAS)", t.as_select[:80] + "...")]
+        return out
+
     def apply(self, st) -> None:
+        if isinstance(st, (P.ShowStmt, P.DescribeStmt)):
+            return                       # read-only statements
         if isinstance(st, P.SetStmt):
             self.session[st.key] = st.value
         elif isinstance(st, P.CreateTable):

@@ -298,9 +298,30 @@ _SET_RE = re.compile(r"^SET\s+'([^']+)'\s*=\s*'([^']*)'\s*$",
                      re.IGNORECASE | re.DOTALL)
 
 
+@dataclass
+class ShowStmt:
+    kind: str                    # TABLES | MODELS | CONNECTIONS | ...
+
+
+@dataclass
+class DescribeStmt:
+    name: str
+
+
 def parse_statement(stmt: str):
     s = stmt.strip()
     up = s.upper()
+    if up.startswith("SHOW"):
+        m = re.match(r"SHOW\s+(TABLES|MODELS|CONNECTIONS|TOOLS|AGENTS)\s*$",
+                     s, re.IGNORECASE)
+        if not m:
+            raise ValueError(f"bad SHOW: {s!r}")
+        return ShowStmt(m.group(1).upper())
+    if up.startswith("DESCRIBE") or up.startswith("DESC "):
+        m = re.match(r"DESC(?:RIBE)?\s+(\S+)\s*$", s, re.IGNORECASE)
+        if not m:
+            raise ValueError(f"bad DESCRIBE: {s!r}")
+        return DescribeStmt(_ident(m.group(1)))
     if up.startswith("SET"):
         m = _SET_RE.match(s)
         if not m:

@@ -205,3 +205,21 @@ def test_parser_edge_cases():
     import pytest
     with pytest.raises(ValueError):
         P.parse_statement("ALTER TABLE x ADD COLUMN y STRING")
+
+
+def test_show_and_describe():
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    cat = Catalog()
+    cat.execute(LAB1_DDL)
+    assert cat.show("TABLES") == ["orders", "ride_requests"]
+    assert cat.show("AGENTS") == ["price_match_agent"]
+    assert "remote-mcp-connection" in cat.show("CONNECTIONS")
+    cols = cat.describe("orders")
+    assert cols[0] == ("order_id", "STRING")
+    assert ("price", "DOUBLE") in cols
+    # parses as statements too (read-only; apply() is a no-op)
+    st = P.parse_statement("SHOW TABLES")
+    assert st.kind == "TABLES"
+    d = P.parse_statement("DESCRIBE `orders`")
+    assert d.name == "orders"
+    cat.execute("SHOW MODELS; DESCRIBE orders;")   # no-ops, no error
