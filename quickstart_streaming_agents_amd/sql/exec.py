@@ -24,8 +24,11 @@ Supported surface (the grammar of labs/sql/*.sql):
   LATERAL TABLE(AI_RUN_AGENT('agent', prompt[, key][, MAP[...]]))
     AS alias(col, ...)
   WHERE / HAVING conjunctions of comparisons, <>, BETWEEN, bare booleans
-  LIMIT n; expressions: literals, refs, CONCAT, CAST, COALESCE,
-  REGEXP_EXTRACT, +/- INTERVAL 'n' U
+  ORDER BY col [ASC|DESC][, ...]; LIMIT n
+  expressions: literals, refs, CONCAT, CAST, COALESCE, UPPER/LOWER,
+  REGEXP_EXTRACT, CASE WHEN, MAP[...] literals, +/- INTERVAL 'n' U
+  scalar smoke forms: SELECT ML_PREDICT('m', 'q') / AI_TOOL_INVOKE(...)
+  without FROM (LAB1-Walkthrough.md:66-92)
 
 Unbound identifiers (the reference inlines giant CONCAT prompt
 expressions; we hoist them into named *bindings*) resolve through the
