@@ -94,7 +94,6 @@ def cmd_validate(args) -> int:
 
 def cmd_destroy(args) -> int:
     import os
-    import shutil
     removed = []
     for f in ("DEPLOYED_RESOURCES.md",) + tuple(
             f"LAB{i}_SQL_COMMANDS.md" for i in (1, 2, 3, 4)):

@@ -15,7 +15,6 @@ import json
 import time
 
 from ..wire import AvroConsumer, AvroProducer, Broker
-from ..wire.avro import Schema
 
 
 def capture_topic(broker: Broker, topic: str, schema, path: str) -> int:

@@ -20,7 +20,6 @@ Reproduces the reference's operating points and determinism contracts
 from __future__ import annotations
 
 import random
-from dataclasses import dataclass
 
 from . import schemas
 from ..wire.topics import AvroProducer, Broker

@@ -25,7 +25,6 @@ from dataclasses import dataclass, field
 
 import torch
 
-from .kv_cache import PagedKVCache
 from .llama import LlamaModel
 from .tokenizer import HashTokenizer
 

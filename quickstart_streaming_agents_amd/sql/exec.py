@@ -40,7 +40,7 @@ without widening the SQL grammar.
 from __future__ import annotations
 
 import re
-from typing import Any, Callable
+from typing import Callable
 
 from . import parse as P
 from .catalog import Catalog, analyze_select

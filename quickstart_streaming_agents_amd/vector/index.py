@@ -16,10 +16,8 @@ merges local (score, id) heaps via all-gather (parallel/shard_index.py).
 from __future__ import annotations
 
 import hashlib
-import math
 import re
 from dataclasses import dataclass, field
-from typing import Any
 
 import numpy as np
 

@@ -12,7 +12,6 @@ the TopicNameStrategy the reference publishers use (``<topic>-value`` /
 from __future__ import annotations
 
 import threading
-from typing import Any
 
 from .avro import Schema
 
