@@ -68,6 +68,30 @@ def create_app(llm: Callable[[list[str], list[int]], list[str]],
     def healthz():
         return {"status": "ok"}
 
+    # per-app Prometheus registry (a fresh one per create_app so tests
+    # can build many apps without duplicate-timeseries errors)
+    from prometheus_client import (CONTENT_TYPE_LATEST, CollectorRegistry,
+                                   Counter, Histogram, generate_latest)
+    registry = CollectorRegistry()
+    req_count = Counter("qsa_requests_total", "requests per endpoint",
+                        ["endpoint"], registry=registry)
+    req_latency = Histogram("qsa_request_seconds", "request latency",
+                            ["endpoint"], registry=registry)
+    app.state.prom_registry = registry
+
+    def _observe(endpoint: str, t0: float) -> None:
+        req_count.labels(endpoint).inc()
+        req_latency.labels(endpoint).observe(time.time() - t0)
+
+    @app.get("/metrics")
+    def metrics():
+        """Prometheus exposition (request counts + latency histograms —
+        the observability layer the reference delegates to Confluent's
+        managed UI)."""
+        from fastapi import Response
+        return Response(generate_latest(registry),
+                        media_type=CONTENT_TYPE_LATEST)
+
     @app.get("/v1/status")
     def status():
         return {
@@ -83,11 +107,13 @@ def create_app(llm: Callable[[list[str], list[int]], list[str]],
 
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):
+        t0 = time.time()
         app.state.requests += 1
         prompts = [req.prompt] if isinstance(req.prompt, str) else req.prompt
         if not prompts:
             raise HTTPException(400, "empty prompt list")
         texts = llm(prompts, [req.max_tokens] * len(prompts))
+        _observe("completions", t0)
         return {"object": "text_completion",
                 "choices": [{"index": i, "text": t}
                             for i, t in enumerate(texts)]}
@@ -97,8 +123,10 @@ def create_app(llm: Callable[[list[str], list[int]], list[str]],
         app.state.requests += 1
         if embedder is None:
             raise HTTPException(503, "no embedder configured")
+        t0 = time.time()
         texts = [req.input] if isinstance(req.input, str) else req.input
         vecs = [embedder.embed(t) for t in texts]
+        _observe("embeddings", t0)
         return {"object": "list",
                 "data": [{"index": i, "embedding": v.tolist()}
                          for i, v in enumerate(vecs)],
@@ -119,7 +147,9 @@ def create_app(llm: Callable[[list[str], list[int]], list[str]],
             if q.shape != (index.dim,):
                 raise HTTPException(
                     400, f"query dims {q.shape} != index dim {index.dim}")
+        t0 = time.time()
         hits = index.search(q, req.k)
+        _observe("search", t0)
         return {"hits": [{"document_id": h.document_id, "chunk": h.chunk,
                           "score": h.score, "metadata": h.metadata}
                          for h in hits]}
@@ -130,7 +160,9 @@ def create_app(llm: Callable[[list[str], list[int]], list[str]],
         fn = agents.get(name)
         if fn is None:
             raise HTTPException(404, f"unknown agent {name!r}")
+        t0 = time.time()
         res = fn(req.prompt, req.record_key)
+        _observe("agents", t0)
         return {"agent": name,
                 "status": res.status,
                 "response": res.response,

@@ -85,3 +85,11 @@ def test_validation_errors():
                       json={"input": "x"}).status_code == 503
         assert c.post("/v1/search",
                       json={"query": "x"}).status_code == 503
+
+
+def test_metrics_endpoint(client):
+    client.post("/v1/completions", json={"prompt": "count me"})
+    body = client.get("/metrics").text
+    assert "qsa_requests_total" in body
+    assert 'endpoint="completions"' in body
+    assert "qsa_request_seconds" in body
