@@ -298,3 +298,32 @@ def test_generic_sql_executor_on_gpu_engine():
     assert 1 <= len(rows) <= 2
     assert all(r["pickup_zone"] == "French Quarter" for r in rows)
     assert all(r["api_response"] for r in rows)
+
+
+def test_serve_api_on_gpu_engine():
+    """The HTTP serving surface (serve_api.py) wired to the REAL GPU
+    stack: EngineLLM completions, GPU embedder + HBM index search, and
+    an agent episode — all through the FastAPI app."""
+    import pytest as _pytest
+    _pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    from quickstart_streaming_agents_amd.serve_api import build_lab_app
+    app = build_lab_app(device="cuda:0", model="tiny", labs=(1, 2))
+    try:
+        with TestClient(app) as c:
+            out = c.post("/v1/completions",
+                         json={"prompt": "hello", "max_tokens": 8}).json()
+            assert out["choices"][0]["text"]
+            emb = c.post("/v1/embeddings", json={"input": "x"}).json()
+            assert emb["dims"] == 1536
+            hits = c.post("/v1/search",
+                          json={"query": "How do I create a Flink table?",
+                                "k": 2}).json()["hits"]
+            assert len(hits) == 2 and hits[0]["score"] >= hits[1]["score"]
+            ag = c.post("/v1/agents/price_match_agent",
+                        json={"prompt": "check AirPods"}).json()
+            assert ag["status"] in ("SUCCESS", "FAILED")
+            assert "qsa_requests_total" in c.get("/metrics").text
+    finally:
+        app.state.mcp_server.stop()
