@@ -377,3 +377,21 @@ class StreamingPipeline:
         for sink, s in snap.items():
             if sink in self.by_sink:
                 self.by_sink[sink].restore(s)
+
+    # -- durable form (runtime/checkpoint.py: torn-write-safe files) -----
+    def checkpoint(self, root: str, pipeline: str = "sql",
+                   shard: int = 0) -> int:
+        """Persist the pipeline state to disk; returns the checkpoint id."""
+        from ..runtime.checkpoint import CheckpointStore, PipelineState
+        store = CheckpointStore(root, pipeline, shard)
+        return store.save(PipelineState(operator=self.snapshot()))
+
+    def resume(self, root: str, pipeline: str = "sql",
+               shard: int = 0) -> bool:
+        """Restore the latest on-disk checkpoint; False if none exists."""
+        from ..runtime.checkpoint import CheckpointStore
+        state = CheckpointStore(root, pipeline, shard).load()
+        if state is None:
+            return False
+        self.restore(state.operator)
+        return True

@@ -143,3 +143,40 @@ def test_lab2_streaming_query_arrives_late(mcp):
     out2 = pipe.advance()
     assert len(out2["search_results_response"]) == 1
     assert out2["search_results_response"][0]["response"]
+
+
+def test_durable_checkpoint_resume(mcp, tmp_path):
+    """Pipeline state survives a process 'crash' via the torn-write-safe
+    CheckpointStore files (runtime/checkpoint.py)."""
+    ref = Deployment(labs=(3,), device="cpu")
+    ref.datagen(3)
+    chunks_ref = _chunked_records(ref, "ride_requests", 3)
+    pipe_ref = StreamingPipeline(ref.sql_executor(3, mcp_server=mcp))
+    out_ref: list = []
+    for ch in chunks_ref:
+        _replay(ref, "ride_requests", ch)
+        out_ref += pipe_ref.advance()["completed_actions"]
+    out_ref += pipe_ref.finish()["completed_actions"]
+
+    dep = Deployment(labs=(3,), device="cpu")
+    dep.datagen(3)
+    chunks = _chunked_records(dep, "ride_requests", 3)
+    pipe1 = StreamingPipeline(dep.sql_executor(3, mcp_server=mcp))
+    out: list = []
+    _replay(dep, "ride_requests", chunks[0])
+    out += pipe1.advance()["completed_actions"]
+    cp = pipe1.checkpoint(str(tmp_path), "lab3")
+    assert cp == 1
+    del pipe1                                    # "crash"
+
+    pipe2 = StreamingPipeline(dep.sql_executor(3, mcp_server=mcp))
+    assert pipe2.resume(str(tmp_path), "lab3")
+    for ch in chunks[1:]:
+        _replay(dep, "ride_requests", ch)
+        out += pipe2.advance()["completed_actions"]
+    out += pipe2.finish()["completed_actions"]
+    assert [r["pickup_zone"] for r in out] == \
+        [r["pickup_zone"] for r in out_ref]
+    # fresh pipeline with no checkpoint
+    assert not StreamingPipeline(
+        dep.sql_executor(3, mcp_server=mcp)).resume(str(tmp_path), "other")
