@@ -108,13 +108,13 @@ def cmd_destroy(args) -> int:
 def cmd_sql(args) -> int:
     """Execute a .sql file (the lab grammar, docs/SQL.md) against JSONL
     topic data: --data topic=path.jsonl (repeatable), then print the rows
-    of --table (or every CTAS)."""
+    of --table (or every CTAS).  --interactive reads further statements
+    from stdin after the file (if any) is applied."""
+    from .sql import parse as P
     from .sql.catalog import Catalog
-    from .sql.exec import SqlExecutor
+    from .sql.exec import SqlExecError, SqlExecutor
     from .wire import Broker
     cat = Catalog()
-    with open(args.file) as fh:
-        cat.execute(fh.read())
     broker = Broker()
     for spec in args.data or []:
         topic_name, _, path = spec.partition("=")
@@ -125,28 +125,54 @@ def cmd_sql(args) -> int:
                 if line:
                     t.append(json.loads(line), partition=0)
     ex = SqlExecutor(cat, broker)
-    ex.run_inserts()
-    # read-only statements in the file print their results
-    from .sql import parse as P
-    for st in P.parse_script(open(args.file).read()):
-        if isinstance(st, P.ShowStmt):
-            print(f"-- SHOW {st.kind}: {', '.join(cat.show(st.kind))}")
-        elif isinstance(st, P.DescribeStmt):
-            print(f"-- DESCRIBE {st.name}:")
-            for col, ty in cat.describe(st.name):
-                print(f"   {col:32s} {ty}")
-    targets = [args.table] if args.table else \
-        [n for n, t in cat.tables.items() if t.as_select]
-    for name in targets:
-        if args.explain:
-            print(f"-- plan for {name}:")
-            for step in ex.explain(name):
-                print(f"   {step}")
-            continue
-        rows = ex.run_table(name)
-        print(f"-- {name}: {len(rows)} rows")
-        for r in rows[: args.max_print]:
-            print(json.dumps(r, default=str)[:400])
+
+    def run_script(text: str, materialize: bool) -> None:
+        for st in P.parse_script(text):
+            cat.apply(st)
+            if isinstance(st, P.ShowStmt):
+                print(f"-- SHOW {st.kind}: {', '.join(cat.show(st.kind))}")
+            elif isinstance(st, P.DescribeStmt):
+                print(f"-- DESCRIBE {st.name}:")
+                for col, ty in cat.describe(st.name):
+                    print(f"   {col:32s} {ty}")
+            elif isinstance(st, P.InsertInto) and materialize:
+                ex.run_inserts()
+                cat.inserts.clear()
+            elif isinstance(st, P.CreateTable) and st.as_select and \
+                    materialize:
+                if args.explain:
+                    print(f"-- plan for {st.name}:")
+                    for step in ex.explain(st.name):
+                        print(f"   {step}")
+                    continue
+                rows = ex.run_table(st.name)
+                print(f"-- {st.name}: {len(rows)} rows")
+                for r in rows[: args.max_print]:
+                    print(json.dumps(r, default=str)[:400])
+
+    if args.file:
+        with open(args.file) as fh:
+            text = fh.read()
+        if args.table:
+            cat.execute(text)
+            ex.run_inserts()
+            rows = ex.run_table(args.table)
+            print(f"-- {args.table}: {len(rows)} rows")
+            for r in rows[: args.max_print]:
+                print(json.dumps(r, default=str)[:400])
+        else:
+            run_script(text, materialize=True)
+    if args.interactive:
+        print("-- interactive: end statements with ';' (EOF to quit)")
+        buf = []
+        for line in sys.stdin:
+            buf.append(line)
+            if line.rstrip().endswith(";"):
+                try:
+                    run_script("".join(buf), materialize=True)
+                except (ValueError, SqlExecError, KeyError) as e:
+                    print(f"-- error: {e}")
+                buf = []
     return 0
 
 
@@ -214,7 +240,9 @@ def main(argv=None) -> int:
 
     q = sub.add_parser("sql", help="execute a .sql file against JSONL "
                        "topic data (docs/SQL.md grammar)")
-    q.add_argument("--file", required=True)
+    q.add_argument("--file", default=None)
+    q.add_argument("--interactive", action="store_true",
+                   help="read further statements from stdin")
     q.add_argument("--data", action="append",
                    help="topic=path.jsonl (repeatable)")
     q.add_argument("--table", default=None,

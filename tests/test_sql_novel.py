@@ -142,3 +142,28 @@ def test_executor_tracing():
     assert "hot_windows:tumble" in stages
     assert "hot_alerts:scan_join" in stages
     assert stages["hot_windows:tumble"]["records_out"] >= 1
+
+
+def test_cli_sql_interactive(tmp_path, monkeypatch, capsys):
+    """--interactive reads statements from stdin: DDL, data via the file
+    path, SHOW, incremental CTAS materialization, graceful errors."""
+    import io
+
+    from quickstart_streaming_agents_amd.cli import main
+    meta = tmp_path / "m.jsonl"
+    meta.write_text('{"site": "a", "region": "gulf"}\n')
+    stdin = io.StringIO(
+        "CREATE TABLE site_meta (site STRING, region STRING);\n"
+        "SHOW TABLES;\n"
+        "DESCRIBE site_meta;\n"
+        "CREATE TABLE gulf AS SELECT m.site FROM site_meta m "
+        "WHERE m.region = 'gulf';\n"
+        "NOT REAL SQL;\n")
+    monkeypatch.setattr("sys.stdin", stdin)
+    rc = main(["sql", "--interactive", "--data", f"site_meta={meta}"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "SHOW TABLES: site_meta" in out
+    assert "site " in out             # DESCRIBE column listing
+    assert "gulf: 1 rows" in out
+    assert "-- error:" in out         # bad statement didn't crash the REPL
