@@ -105,6 +105,23 @@ def cmd_destroy(args) -> int:
     return 0
 
 
+def cmd_capture(args) -> int:
+    """Capture a lab topic to JSONL (scripts/capture_lab{1,3}_data.py
+    parity): datagen then dump base64 wire payloads for replay."""
+    from .labs import schemas
+    from .labs.capture import capture_topic
+    from .labs.deploy import Deployment
+    dep = Deployment(labs=(args.lab,), device="cpu")
+    dep.datagen(args.lab)
+    schema = {1: schemas.ORDERS, 2: schemas.QUERIES,
+              3: schemas.RIDE_REQUESTS, 4: schemas.CLAIMS}[args.lab]
+    topic = args.topic or {1: "orders", 2: "queries",
+                           3: "ride_requests", 4: "claims"}[args.lab]
+    n = capture_topic(dep.broker, topic, schema, args.out)
+    print(f"[capture] {topic}: {n} records -> {args.out}")
+    return 0 if n else 1
+
+
 def cmd_sql(args) -> int:
     """Execute a .sql file (the lab grammar, docs/SQL.md) against JSONL
     topic data: --data topic=path.jsonl (repeatable), then print the rows
@@ -237,6 +254,13 @@ def main(argv=None) -> int:
     x = sub.add_parser("destroy", help="remove deployment artifacts")
     x.add_argument("--dir", default="deploy_out")
     x.set_defaults(fn=cmd_destroy)
+
+    c = sub.add_parser("capture", help="capture a lab topic to JSONL "
+                       "(wire payloads, base64) for replay")
+    c.add_argument("--lab", type=int, required=True)
+    c.add_argument("--topic", default=None)
+    c.add_argument("--out", required=True)
+    c.set_defaults(fn=cmd_capture)
 
     q = sub.add_parser("sql", help="execute a .sql file against JSONL "
                        "topic data (docs/SQL.md grammar)")

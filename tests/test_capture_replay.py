@@ -100,3 +100,12 @@ def test_publish_docs_and_sql_extract(tmp_path):
     stmts = extract_statements(walkthrough)
     assert len(stmts) == 3
     assert "DROP" not in " ".join(stmts)
+
+
+def test_cli_capture_subcommand(tmp_path):
+    from quickstart_streaming_agents_amd.cli import main
+    out = tmp_path / "rides.jsonl"
+    rc = main(["capture", "--lab", "3", "--out", str(out)])
+    assert rc == 0
+    lines = out.read_text().strip().splitlines()
+    assert len(lines) >= 28_000          # the lab3 volume contract
