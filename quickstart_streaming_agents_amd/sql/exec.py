@@ -668,6 +668,18 @@ class SqlExecutor:
                           reverse=desc)
         return rows
 
+    def watermark_delay_ms(self, table: str) -> int:
+        """Delay from the table's WATERMARK clause (ts - INTERVAL 'n' U);
+        5 s when the table declares none (the lab default)."""
+        t = self.catalog.tables.get(table)
+        if t is not None and t.watermark:
+            m = re.search(r"INTERVAL\s+'(\d+)'\s+(\w+)", t.watermark[1],
+                          re.IGNORECASE)
+            if m:
+                from .catalog import _interval_ms
+                return _interval_ms(m.group(1), m.group(2))
+        return 5000
+
     # -- FROM stage ---------------------------------------------------------
     def _join_rows(self, from_clause: str) -> list[_Row]:
         tables = _parse_joins(from_clause)
@@ -737,10 +749,10 @@ class SqlExecutor:
                 fn, arg = m.group(1).upper(), m.group(2).strip()
                 aggs[alias] = (fn, arg)
 
-        tw = TumblingWindows(tum["window_ms"],
-                             key_fn=lambda r: r[key_col],
-                             ts_fn=lambda r: r[tum["ts_col"]],
-                             watermark_delay_ms=5000)
+        tw = TumblingWindows(
+            tum["window_ms"], key_fn=lambda r: r[key_col],
+            ts_fn=lambda r: r[tum["ts_col"]],
+            watermark_delay_ms=self.watermark_delay_ms(tum["table"]))
         panes = tw.feed(src) + tw.flush()
         panes.sort(key=lambda p: (p.window_start, str(p.key)))
         rows = []

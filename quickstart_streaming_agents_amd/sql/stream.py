@@ -140,7 +140,8 @@ class StreamingQuery:
                                         m.group(2).strip())
             self.windows = TumblingWindows(
                 tum["window_ms"], key_fn=lambda r: r[self.key_col],
-                ts_fn=lambda r: r[tum["ts_col"]], watermark_delay_ms=5000)
+                ts_fn=lambda r: r[tum["ts_col"]],
+                watermark_delay_ms=ex.watermark_delay_ms(tum["table"]))
             self.detector = None
             self.anom_alias = "anomaly"
             self.anom_value_expr = None

@@ -253,3 +253,19 @@ def test_walkthrough_smoke_statements():
     assert ex.ev.eval("MAP['a','1','b','2']", _Row({})) == \
         {"a": "1", "b": "2"}
     assert ex.ev.eval("MAP[]", _Row({})) == {}
+
+
+def test_watermark_delay_from_ddl():
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+    cat = Catalog()
+    cat.execute("""
+    CREATE TABLE ev (k STRING, ts TIMESTAMP_LTZ(3),
+      WATERMARK FOR ts AS ts - INTERVAL '30' SECOND);
+    CREATE TABLE nv (k STRING, ts TIMESTAMP_LTZ(3));
+    """)
+    ex = SqlExecutor(cat, Broker())
+    assert ex.watermark_delay_ms("ev") == 30_000
+    assert ex.watermark_delay_ms("nv") == 5_000     # default
+    assert ex.watermark_delay_ms("missing") == 5_000
