@@ -916,10 +916,18 @@ class SqlExecutor:
                 if isinstance(v, dict):
                     out.update(v)
                 continue
+            if expr.strip() == "*" or expr.strip().endswith(".*"):
+                prefix = expr.strip()[:-2]
+                for ns_alias, d in row.ns.items():
+                    if prefix and ns_alias != prefix:
+                        continue
+                    for k, val in d.items():
+                        out.setdefault(k, val)
+                continue
             v = self.ev.eval(expr, row)
             if isinstance(v, dict):
                 out.update(v)
-            elif hasattr(v, "tolist") and alias in ("embedding",):
+            elif hasattr(v, "tolist"):       # numpy arrays/scalars -> JSON
                 out[alias] = v.tolist()
             else:
                 out[alias] = v

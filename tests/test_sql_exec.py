@@ -269,3 +269,22 @@ def test_watermark_delay_from_ddl():
     assert ex.watermark_delay_ms("ev") == 30_000
     assert ex.watermark_delay_ms("nv") == 5_000     # default
     assert ex.watermark_delay_ms("missing") == 5_000
+
+
+def test_select_star_and_qualified_star():
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+    cat = Catalog()
+    cat.execute("""
+    CREATE TABLE a (x STRING, y INT);
+    CREATE TABLE b (x STRING, z INT);
+    CREATE TABLE all_cols AS SELECT * FROM a s JOIN b t ON s.x = t.x;
+    CREATE TABLE left_only AS SELECT s.* FROM a s JOIN b t ON s.x = t.x;
+    """)
+    broker = Broker()
+    broker.create_topic("a").append({"x": "k", "y": 1}, partition=0)
+    broker.create_topic("b").append({"x": "k", "z": 9}, partition=0)
+    ex = SqlExecutor(cat, broker)
+    assert ex.run_table("all_cols") == [{"x": "k", "y": 1, "z": 9}]
+    assert ex.run_table("left_only") == [{"x": "k", "y": 1}]
