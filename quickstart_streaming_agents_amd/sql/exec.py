@@ -689,8 +689,14 @@ class SqlExecutor:
         rows = [_Row({alias0: r}) for r in self.table_rows(name0)]
         for name, alias, cond in tables[1:]:
             right = self.table_rows(name)
-            if cond is None:               # comma join (cross) — rare
-                rows = [r.child() for r in rows for _ in right]
+            if cond is None:               # comma join = cross join
+                crossed = []
+                for r in rows:
+                    for rr in right:
+                        c = r.child()
+                        c.ns[alias] = rr
+                        crossed.append(c)
+                rows = crossed
                 continue
             terms = _split_bool(cond)
             eq_pairs, residual = [], []

@@ -288,3 +288,26 @@ def test_select_star_and_qualified_star():
     ex = SqlExecutor(cat, broker)
     assert ex.run_table("all_cols") == [{"x": "k", "y": 1, "z": 9}]
     assert ex.run_table("left_only") == [{"x": "k", "y": 1}]
+
+
+def test_comma_cross_join():
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+    cat = Catalog()
+    cat.execute("""
+    CREATE TABLE l (a STRING);
+    CREATE TABLE r (b STRING);
+    CREATE TABLE x AS SELECT lt.a, rt.b FROM l lt, r rt;
+    """)
+    broker = Broker()
+    tl = broker.create_topic("l")
+    tr = broker.create_topic("r")
+    for a in ("1", "2"):
+        tl.append({"a": a}, partition=0)
+    for b in ("x", "y", "z"):
+        tr.append({"b": b}, partition=0)
+    rows = SqlExecutor(cat, broker).run_table("x")
+    assert len(rows) == 6
+    assert {(r["a"], r["b"]) for r in rows} == \
+        {(a, b) for a in "12" for b in "xyz"}
