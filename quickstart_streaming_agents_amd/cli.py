@@ -36,7 +36,9 @@ def cmd_run(args) -> int:
     dep.datagen(args.lab)
     server = StubMcpServer().start()
     try:
-        if args.sql:
+        if args.stream:
+            rows = dep.run_stream(args.lab, mcp_server=server)
+        elif args.sql:
             rows = dep.run_sql(args.lab, mcp_server=server)
         else:
             rows = dep.run(args.lab, mcp_server=server)
@@ -242,6 +244,9 @@ def main(argv=None) -> int:
     r.add_argument("--sql", action="store_true",
                    help="run through the generic SQL CTAS executor "
                         "(sql/exec.py) instead of the fused pipelines")
+    r.add_argument("--stream", action="store_true",
+                   help="run through the incremental streaming executor "
+                        "(sql/stream.py)")
     r.set_defaults(fn=cmd_run)
 
     g = sub.add_parser("datagen", help="publish one lab's synthetic stream")

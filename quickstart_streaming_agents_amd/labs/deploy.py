@@ -269,6 +269,21 @@ class Deployment:
         ex.run_inserts()
         return ex.run_table(self.FINAL_TABLE[lab])
 
+    def run_stream(self, lab: int, mcp_server=None) -> list[dict]:
+        """Run one lab through the STREAMING executor (sql/stream.py):
+        incremental consumption + watermark-driven window closure, then a
+        bounded-input flush."""
+        from ..sql.stream import StreamingPipeline
+        ex = self.sql_executor(lab, mcp_server=mcp_server)
+        ex.run_inserts(values_only=True)
+        pipe = StreamingPipeline(ex)
+        final = self.FINAL_TABLE[lab]
+        rows = pipe.advance().get(final, [])
+        for sink, extra in pipe.finish().items():
+            if sink == final:
+                rows += extra
+        return rows
+
     # ---- destroy / summary ----------------------------------------------
     def destroy(self) -> None:
         """Drop lab objects in reverse order (destroy.py:140-146)."""

@@ -180,3 +180,12 @@ def test_durable_checkpoint_resume(mcp, tmp_path):
     # fresh pipeline with no checkpoint
     assert not StreamingPipeline(
         dep.sql_executor(3, mcp_server=mcp)).resume(str(tmp_path), "other")
+
+
+def test_run_stream_deployment_wrapper(mcp):
+    dep = Deployment(labs=(4,), device="cpu")
+    dep.datagen(4)
+    rows = dep.run_stream(4)
+    assert rows and len(rows) <= 10
+    from quickstart_streaming_agents_amd.agents.parse import LAB4_VERDICTS
+    assert all(r["verdict"] in LAB4_VERDICTS for r in rows)
