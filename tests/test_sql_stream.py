@@ -189,3 +189,15 @@ def test_run_stream_deployment_wrapper(mcp):
     assert rows and len(rows) <= 10
     from quickstart_streaming_agents_amd.agents.parse import LAB4_VERDICTS
     assert all(r["verdict"] in LAB4_VERDICTS for r in rows)
+
+
+def test_run_stream_lab1_and_lab2(mcp):
+    dep1 = Deployment(labs=(1,), device="cpu")
+    dep1.datagen(1)
+    rows1 = dep1.run_stream(1, mcp_server=mcp)
+    assert rows1 and all(r["agent_status"] == "SUCCESS" for r in rows1)
+
+    dep2 = Deployment(labs=(2,), device="cpu")
+    dep2.datagen(2)
+    rows2 = dep2.run_stream(2)
+    assert rows2 and all(r["response"] for r in rows2)
