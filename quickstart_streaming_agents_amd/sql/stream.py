@@ -14,6 +14,13 @@ offsets, window panes, detector state and join buffers through the
 JSON-serializable forms of runtime/checkpoint.py, so a pipeline can
 resume mid-stream after a crash (the reference's recovery story is
 replay-from-offset on Confluent Flink; SURVEY.md 2.5 elasticity row).
+
+Known bound: join buffers grow with the stream (Flink bounds them with
+`SET 'sql.state-ttl'`, which needs a per-table event-time column to
+apply generically; the hand-fused pipelines honor TTL via
+runtime/joins.TTLTable on the columns the labs declare —
+labs/pipelines.py lab1_enriched_orders).  The lab4 interval join is
+bounded semantically by its BETWEEN predicate regardless.
 """
 
 from __future__ import annotations
