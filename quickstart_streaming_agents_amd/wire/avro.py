@@ -241,3 +241,34 @@ def peek_schema_id(payload: bytes) -> int | None:
     if len(payload) >= 5 and payload[0] == MAGIC_BYTE:
         return struct.unpack(">I", payload[1:5])[0]
     return None
+
+
+def project_to_reader(reader: Schema, value):
+    """Avro schema-resolution projection: reshape a value decoded with the
+    WRITER schema into the READER schema's record shape (reader-only
+    fields take their declared defaults; writer-only fields are dropped).
+    Registry BACKWARD compatibility (registry.py) guarantees the defaults
+    exist.  Recurses through records/arrays/maps; other types pass
+    through unchanged."""
+    t = reader.type
+    if t == "record" and isinstance(value, dict):
+        out = {}
+        for name, fs, fdef in reader.fields:
+            if name in value:
+                out[name] = project_to_reader(fs, value[name])
+            elif "default" in fdef:
+                out[name] = fdef["default"]
+            else:
+                raise KeyError(
+                    f"field {name!r} missing and has no default")
+        return out
+    if t == "array" and isinstance(value, list):
+        return [project_to_reader(reader.items, v) for v in value]
+    if t == "map" and isinstance(value, dict):
+        return {k: project_to_reader(reader.values, v)
+                for k, v in value.items()}
+    if t == "union":
+        for b in reader.branches:
+            if (value is None) == (b.type == "null"):
+                return project_to_reader(b, value)
+    return value

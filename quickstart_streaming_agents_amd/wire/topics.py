@@ -22,7 +22,8 @@ import threading
 from dataclasses import dataclass
 from typing import Any
 
-from .avro import Schema, deserialize, serialize
+from .avro import (Schema, deserialize, peek_schema_id,
+                   project_to_reader, serialize)
 from .registry import SchemaRegistry
 
 
@@ -176,6 +177,30 @@ class AvroConsumer:
         self.value_schema = value_schema if isinstance(value_schema, Schema) else Schema(value_schema)
         self._codec = _native_codec(self.value_schema)
         self._offsets: dict[int, int] = {}
+        self._reader_canonical = self.value_schema.canonical()
+        self._writers: dict[int, tuple[Schema, Any] | None] = {}
+
+    def _writer_for(self, raw: bytes):
+        """Confluent semantics: decode with the WRITER schema named by the
+        wire schema id, then project to this consumer's reader schema
+        (None = same schema or unknown id -> decode with the reader)."""
+        sid = peek_schema_id(raw)
+        if sid is None:
+            return None
+        hit = self._writers.get(sid, False)
+        if hit is not False:
+            return hit
+        try:
+            writer = self.broker.registry.by_id(sid)
+        except KeyError:
+            self._writers[sid] = None
+            return None
+        if writer.canonical() == self._reader_canonical:
+            self._writers[sid] = None
+            return None
+        entry = (writer, _native_codec(writer))
+        self._writers[sid] = entry
+        return entry
 
     def poll(self, max_count: int | None = None) -> list[tuple[Record, Any]]:
         topic = self.broker.topic(self.topic_name)
@@ -186,7 +211,15 @@ class AvroConsumer:
             for r in recs:
                 v = r.value
                 if isinstance(v, (bytes, bytearray)):
-                    if self._codec is not None:
+                    writer = self._writer_for(bytes(v))
+                    if writer is not None:
+                        wschema, wcodec = writer
+                        if wcodec is not None:
+                            _, v = wcodec.deserialize(bytes(v))
+                        else:
+                            _, v = deserialize(wschema, bytes(v))
+                        v = project_to_reader(self.value_schema, v)
+                    elif self._codec is not None:
                         _, v = self._codec.deserialize(bytes(v))
                     else:
                         _, v = deserialize(self.value_schema, bytes(v))

@@ -243,3 +243,32 @@ def test_registry_backward_compatibility_enforced():
     assert "symbols removed" in schema_incompatibilities(
         Schema(new_enum), Schema(old_enum))[0]
     assert schema_incompatibilities(Schema(old_enum), Schema(new_enum)) == []
+
+
+def test_consumer_resolves_writer_schema_by_id():
+    """Confluent decode semantics: a consumer built with the NEW (reader)
+    schema reads records written with the OLD schema — the wire schema id
+    selects the writer schema from the registry and reader-only fields
+    take their defaults."""
+    from quickstart_streaming_agents_amd.wire import Broker
+    from quickstart_streaming_agents_amd.wire.topics import (AvroConsumer,
+                                                             AvroProducer)
+    v1 = {"type": "record", "name": "o", "fields": [
+        {"name": "id", "type": "string"},
+        {"name": "price", "type": "double"}]}
+    v2 = {"type": "record", "name": "o", "fields": [
+        {"name": "id", "type": "string"},
+        {"name": "price", "type": "double"},
+        {"name": "region", "type": "string", "default": "us"}]}
+    broker = Broker()
+    AvroProducer(broker, "orders", v1).produce({"id": "a", "price": 1.5})
+    AvroProducer(broker, "orders", v2).produce(
+        {"id": "b", "price": 2.5, "region": "eu"})
+
+    rows = [v for _, v in AvroConsumer(broker, "orders", v2).poll()]
+    assert rows == [{"id": "a", "price": 1.5, "region": "us"},
+                    {"id": "b", "price": 2.5, "region": "eu"}]
+    # an OLD-schema consumer still reads new records (extra field dropped)
+    old_rows = [v for _, v in AvroConsumer(broker, "orders", v1).poll()]
+    assert old_rows == [{"id": "a", "price": 1.5},
+                        {"id": "b", "price": 2.5}]
