@@ -130,11 +130,24 @@ def cmd_sql(args) -> int:
     of --table (or every CTAS).  --interactive reads further statements
     from stdin after the file (if any) is applied."""
     from .sql import parse as P
-    from .sql.catalog import Catalog
     from .sql.exec import SqlExecError, SqlExecutor
-    from .wire import Broker
-    cat = Catalog()
-    broker = Broker()
+    server = None
+    if args.lab:
+        # lab context: catalog + datagen topics + stub-LLM engine wiring,
+        # so ML_PREDICT / VECTOR_SEARCH_AGG / AI_RUN_AGENT statements run
+        from .agents.mcp import StubMcpServer
+        from .labs.deploy import Deployment
+        dep = Deployment(labs=(args.lab,), device=args.device)
+        dep.datagen(args.lab)
+        server = StubMcpServer().start()
+        ex = dep.sql_executor(args.lab, mcp_server=server)
+        cat, broker = dep.catalog, dep.broker
+    else:
+        from .sql.catalog import Catalog
+        from .wire import Broker
+        cat = Catalog()
+        broker = Broker()
+        ex = SqlExecutor(cat, broker)
     for spec in args.data or []:
         topic_name, _, path = spec.partition("=")
         t = broker.create_topic(topic_name)
@@ -143,7 +156,6 @@ def cmd_sql(args) -> int:
                 line = line.strip()
                 if line:
                     t.append(json.loads(line), partition=0)
-    ex = SqlExecutor(cat, broker)
 
     def run_script(text: str, materialize: bool) -> None:
         for st in P.parse_script(text):
@@ -192,6 +204,8 @@ def cmd_sql(args) -> int:
                 except (ValueError, SqlExecError, KeyError) as e:
                     print(f"-- error: {e}")
                 buf = []
+    if server is not None:
+        server.stop()
     return 0
 
 
@@ -270,6 +284,10 @@ def main(argv=None) -> int:
     q = sub.add_parser("sql", help="execute a .sql file against JSONL "
                        "topic data (docs/SQL.md grammar)")
     q.add_argument("--file", default=None)
+    q.add_argument("--lab", type=int, default=None,
+                   help="load a lab catalog + datagen topics + stub "
+                        "engine so AI statements execute")
+    q.add_argument("--device", default="cpu")
     q.add_argument("--interactive", action="store_true",
                    help="read further statements from stdin")
     q.add_argument("--data", action="append",

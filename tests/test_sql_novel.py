@@ -167,3 +167,24 @@ def test_cli_sql_interactive(tmp_path, monkeypatch, capsys):
     assert "site " in out             # DESCRIBE column listing
     assert "gulf: 1 rows" in out
     assert "-- error:" in out         # bad statement didn't crash the REPL
+
+
+def test_cli_sql_lab_context_runs_ai_statements(monkeypatch, capsys):
+    """`sql --lab 2 --interactive`: ML functions execute against the lab
+    deployment (stub LLM + index) straight from the REPL."""
+    import io
+
+    from quickstart_streaming_agents_amd.cli import main
+    stdin = io.StringIO(
+        "CREATE TABLE smoke AS SELECT "
+        "ML_PREDICT('llm_textgen_model', 'What is Flink?') AS answer;\n"
+        "CREATE TABLE hits AS SELECT qe.query, r.chunk1 AS c "
+        "FROM queries_embed qe CROSS JOIN LATERAL TABLE("
+        "VECTOR_SEARCH_AGG(documents_vectordb_lab2, DESCRIPTOR(embedding),"
+        " qe.embedding, 1)) AS r;\n")
+    monkeypatch.setattr("sys.stdin", stdin)
+    # populate queries_embed first via the lab INSERT..SELECT
+    rc = main(["sql", "--lab", "2", "--interactive"])
+    assert rc == 0
+    out = capsys.readouterr().out
+    assert "smoke: 1 rows" in out
