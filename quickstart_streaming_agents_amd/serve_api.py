@@ -94,7 +94,7 @@ def create_app(llm: Callable[[list[str], list[int]], list[str]],
 
     @app.get("/v1/status")
     def status():
-        return {
+        out = {
             "uptime_s": round(time.time() - app.state.started, 3),
             "requests": app.state.requests,
             "capabilities": {
@@ -104,6 +104,19 @@ def create_app(llm: Callable[[list[str], list[int]], list[str]],
                 "agents": sorted(agents),
             },
         }
+        engine = getattr(llm, "engine", None)
+        if engine is not None and hasattr(engine, "stats"):
+            st = engine.stats
+            out["engine"] = {
+                "prefill_tokens": st.prefill_tokens,
+                "cached_prefix_tokens": st.cached_prefix_tokens,
+                "decode_tokens": st.decode_tokens,
+                "decode_steps": st.decode_steps,
+                "prefill_batches": st.prefill_batches,
+            }
+        if index is not None:
+            out["index_docs"] = len(index)
+        return out
 
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):

@@ -93,3 +93,10 @@ def test_metrics_endpoint(client):
     assert "qsa_requests_total" in body
     assert 'endpoint="completions"' in body
     assert "qsa_request_seconds" in body
+
+
+def test_status_reports_index_size(client):
+    st = client.get("/v1/status").json()
+    assert st["index_docs"] > 0            # the lab2 document index
+    # CPU stub LLM has no engine stats block
+    assert "engine" not in st
