@@ -77,7 +77,15 @@ class Topic:
     def append(self, value: Any, key: Any = None, timestamp_ms: int = 0,
                partition: int | None = None) -> Record:
         if partition is None:
-            partition = (hash(key) % len(self.partitions)) if key is not None else 0
+            if key is not None:
+                # Kafka's default partitioner (murmur2 on the key bytes) —
+                # deterministic across processes, unlike Python's salted
+                # hash(); keyed DP sharding depends on this
+                from ..parallel.stream_shard import partition_for_key
+                kb = key if isinstance(key, (str, bytes)) else str(key)
+                partition = partition_for_key(kb, len(self.partitions))
+            else:
+                partition = 0
         rec = Record(self.name, partition, -1, timestamp_ms, key, value)
         self.partitions[partition].append(rec)
         return rec

@@ -272,3 +272,19 @@ def test_consumer_resolves_writer_schema_by_id():
     old_rows = [v for _, v in AvroConsumer(broker, "orders", v1).poll()]
     assert old_rows == [{"id": "a", "price": 1.5},
                         {"id": "b", "price": 2.5}]
+
+
+def test_keyed_partitioning_is_murmur2():
+    """Default keyed partitioning must be Kafka's murmur2 (deterministic
+    across processes) — Python's salted hash() would scatter keys
+    differently every run."""
+    from quickstart_streaming_agents_amd.parallel.stream_shard import \
+        partition_for_key
+    from quickstart_streaming_agents_amd.wire.topics import Topic
+    t = Topic("t", num_partitions=8)
+    keys = [f"customer-{i}@example.com" for i in range(50)]
+    for k in keys:
+        rec = t.append({"k": k}, key=k)
+        assert rec.partition == partition_for_key(k, 8)
+    # keyless appends pin partition 0 (the labs' watermark determinism)
+    assert t.append({"k": None}).partition == 0
