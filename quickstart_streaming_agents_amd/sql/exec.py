@@ -72,9 +72,13 @@ class _Row:
         if "." in name:
             alias, col = name.split(".", 1)
             d = self.ns.get(alias)
-            if d is not None and col in d:
-                return d[col]
-            raise KeyError(name)
+            if d is not None:
+                if col in d:
+                    return d[col]
+                raise KeyError(name)
+            # alias gone (e.g. after GROUP BY collapsed namespaces):
+            # fall through to a bare-column lookup
+            name = col
         for d in self.ns.values():
             if name in d:
                 return d[name]
@@ -626,6 +630,9 @@ class SqlExecutor:
                 rows = self._tumble_rows(info, items, select_sql)
             else:
                 rows = self._join_rows(from_clause)
+                if clauses["group_by"]:      # plain (non-windowed) GROUP BY
+                    rows = self._group_rows(rows, clauses["group_by"],
+                                            items)
             if sp:
                 sp.records_out = len(rows)
 
@@ -652,6 +659,35 @@ class SqlExecutor:
             topic = self.broker.create_topic(sink)
             for row in out:
                 topic.append(row, partition=0)
+        return out
+
+    def _group_rows(self, rows: list[_Row], group_by: str,
+                    items) -> list[_Row]:
+        """Plain GROUP BY over a bounded row set: key exprs + the
+        COUNT/SUM/AVG select aggregates, one output row per group."""
+        key_exprs = [e.strip() for e in P._split_top(group_by)]
+        aggs = {}
+        for expr, alias in items:
+            m = re.match(r"(COUNT|SUM|AVG)\s*\(\s*(.*)\s*\)$", expr.strip(),
+                         re.IGNORECASE | re.DOTALL)
+            if m:
+                aggs[alias] = (m.group(1).upper(), m.group(2).strip())
+        groups: dict[tuple, list[_Row]] = {}
+        for r in rows:
+            k = tuple(self.ev.eval(e, r) for e in key_exprs)
+            groups.setdefault(k, []).append(r)
+        out = []
+        for k, members in groups.items():
+            cols = {e.strip("`").split(".")[-1]: v
+                    for e, v in zip(key_exprs, k)}
+            for alias, (fn, arg) in aggs.items():
+                if fn == "COUNT":
+                    cols[alias] = len(members)
+                else:
+                    vals = [float(self.ev.eval(arg, m)) for m in members]
+                    cols[alias] = (sum(vals) if fn == "SUM"
+                                   else sum(vals) / max(len(vals), 1))
+            out.append(_Row({"_g": cols}))
         return out
 
     def _order(self, rows: list[_Row], order_by: str) -> list[_Row]:

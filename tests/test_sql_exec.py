@@ -311,3 +311,28 @@ def test_comma_cross_join():
     assert len(rows) == 6
     assert {(r["a"], r["b"]) for r in rows} == \
         {(a, b) for a in "12" for b in "xyz"}
+
+
+def test_plain_group_by():
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+    cat = Catalog()
+    cat.execute("""
+    CREATE TABLE sales (region STRING, amt DOUBLE);
+    CREATE TABLE by_region AS
+    SELECT s.region, COUNT(*) AS n, SUM(s.amt) AS total,
+           AVG(s.amt) AS mean
+    FROM sales s GROUP BY s.region
+    ORDER BY total DESC;
+    """)
+    broker = Broker()
+    t = broker.create_topic("sales")
+    for region, amt in (("e", 10.0), ("w", 5.0), ("e", 30.0), ("w", 1.0),
+                        ("n", 7.0)):
+        t.append({"region": region, "amt": amt}, partition=0)
+    rows = SqlExecutor(cat, broker).run_table("by_region")
+    assert rows == [
+        {"region": "e", "n": 2, "total": 40.0, "mean": 20.0},
+        {"region": "n", "n": 1, "total": 7.0, "mean": 7.0},
+        {"region": "w", "n": 2, "total": 6.0, "mean": 3.0}]
