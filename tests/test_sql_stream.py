@@ -201,3 +201,17 @@ def test_run_stream_lab1_and_lab2(mcp):
     dep2.datagen(2)
     rows2 = dep2.run_stream(2)
     assert rows2 and all(r["response"] for r in rows2)
+
+
+def test_ctas_cycle_detected():
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import (SqlExecError,
+                                                          SqlExecutor)
+    from quickstart_streaming_agents_amd.wire import Broker
+    cat = Catalog()
+    cat.execute("""
+    CREATE TABLE a AS SELECT x.v FROM b x;
+    CREATE TABLE b AS SELECT y.v FROM a y;
+    """)
+    with pytest.raises(SqlExecError, match="cycle"):
+        StreamingPipeline(SqlExecutor(cat, Broker()))
