@@ -47,10 +47,16 @@ class AgentRequest(BaseModel):
     record_key: Optional[str] = None
 
 
+class SqlRequest(BaseModel):
+    statement: str
+    max_rows: int = Field(default=20, ge=1, le=1000)
+
+
 def create_app(llm: Callable[[list[str], list[int]], list[str]],
                embedder=None,
                index=None,
-               agents: dict[str, Callable[..., Any]] | None = None):
+               agents: dict[str, Callable[..., Any]] | None = None,
+               sql_executor=None):
     """Build the FastAPI app around injected engine components.
 
     `agents` maps agent name -> callable(prompt, record_key) returning an
@@ -116,7 +122,48 @@ def create_app(llm: Callable[[list[str], list[int]], list[str]],
             }
         if index is not None:
             out["index_docs"] = len(index)
+        out["capabilities"]["sql"] = sql_executor is not None
         return out
+
+    @app.post("/v1/sql")
+    def run_sql(req: SqlRequest):
+        """Execute one statement (docs/SQL.md grammar) against the
+        deployment's catalog/topics: DDL applies, CTAS materializes and
+        returns rows, scalar SELECTs evaluate."""
+        app.state.requests += 1
+        if sql_executor is None:
+            raise HTTPException(503, "no SQL executor configured")
+        from .sql import parse as P
+        from .sql.exec import SqlExecError
+        t0 = time.time()
+        cat = sql_executor.catalog
+        try:
+            result: dict = {"ok": True}
+            if req.statement.strip().rstrip(";").upper().startswith(
+                    "SELECT"):
+                rows = sql_executor.run_select(
+                    req.statement.strip().rstrip(";"))
+                result["row_count"] = len(rows)
+                result["rows"] = rows[: req.max_rows]
+            else:
+                for st in P.parse_script(req.statement):
+                    cat.apply(st)
+                    if isinstance(st, P.ShowStmt):
+                        result["show"] = cat.show(st.kind)
+                    elif isinstance(st, P.DescribeStmt):
+                        result["describe"] = cat.describe(st.name)
+                    elif isinstance(st, P.InsertInto):
+                        sql_executor.run_inserts()
+                        cat.inserts.clear()
+                    elif isinstance(st, P.CreateTable) and st.as_select:
+                        rows = sql_executor.run_table(st.name)
+                        result["table"] = st.name
+                        result["row_count"] = len(rows)
+                        result["rows"] = rows[: req.max_rows]
+        except (ValueError, SqlExecError, KeyError) as e:
+            raise HTTPException(400, f"{type(e).__name__}: {e}")
+        _observe("sql", t0)
+        return result
 
     @app.post("/v1/completions")
     def completions(req: CompletionRequest):
@@ -215,7 +262,9 @@ def build_lab_app(device: str = "cpu", model: str | None = None,
               for name in dep.catalog.agents}
     app = create_app(llm_batch, embedder=dep.embedder,
                      index=dep._index(2) if 2 in labs else None,
-                     agents=agents)
+                     agents=agents,
+                     sql_executor=dep.sql_executor(labs[0],
+                                                   mcp_server=server))
     app.state.mcp_server = server          # kept alive with the app
     return app
 

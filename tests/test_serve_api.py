@@ -100,3 +100,15 @@ def test_status_reports_index_size(client):
     assert st["index_docs"] > 0            # the lab2 document index
     # CPU stub LLM has no engine stats block
     assert "engine" not in st
+
+
+def test_sql_endpoint(client):
+    st = client.post("/v1/sql", json={
+        "statement": "SELECT ML_PREDICT('llm_textgen_model', 'hi') "
+                     "AS answer"}).json()
+    assert st["ok"] and st["rows"][0]["answer"]
+    ddl = client.post("/v1/sql", json={
+        "statement": "CREATE TABLE web_t (a STRING); SHOW TABLES;"}).json()
+    assert "web_t" in ddl["show"]
+    bad = client.post("/v1/sql", json={"statement": "FROB x"})
+    assert bad.status_code == 400
