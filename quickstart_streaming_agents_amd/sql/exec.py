@@ -603,12 +603,16 @@ class SqlExecutor:
             plan.append(f"scan {tables[0][0]}")
             for name, alias, cond in tables[1:]:
                 plan.append(f"hash-join {name} [{alias}] on {cond}")
+            if clauses["group_by"]:
+                plan.append(f"group-by {clauses['group_by']}")
         for lat in laterals:
             plan.append(f"lateral {lat['call'].split('(')[0].strip()}"
                         f"{' AS ' + lat['alias'] if lat['alias'] else ''}")
         for key in ("where", "having"):
             if clauses[key]:
                 plan.append(f"{key} {clauses[key]}")
+        if clauses["order_by"]:
+            plan.append(f"order-by {clauses['order_by']}")
         if clauses["limit"]:
             plan.append(f"limit {clauses['limit']}")
         plan.append(f"project -> {table}")
