@@ -923,12 +923,18 @@ class SqlExecutor:
         agent_name = args[0].strip().strip("'").strip("`")
         spec = self.catalog.agent_spec(agent_name)
         policy_fn = self.agent_policies.get(agent_name)
+        # MAP['debug','true'] anywhere in the call -> per-episode traces
+        # (LAB1-Walkthrough.md:253 debug semantics)
+        debug = any(
+            str((self.ev.eval(a, rows[0]) or {}).get("debug", "")
+                if rows else "").lower() == "true"
+            for a in args[2:] if a.strip().upper().startswith("MAP["))
         eps = []
         for r in rows:
             prompt = str(self.ev.eval(args[1], r))
             policy = policy_fn(lambda n, _r=r: _r.resolve(n)) \
                 if policy_fn else None
-            eps.append(episode(spec, prompt, policy=policy,
+            eps.append(episode(spec, prompt, policy=policy, debug=debug,
                                max_new_tokens=self.max_new_tokens))
         if self.llm_batch is None:
             raise SqlExecError("no LLM configured for AI_RUN_AGENT")
@@ -938,6 +944,8 @@ class SqlExecutor:
         cols = lat["cols"] or ["status", "response"]
         for r, res in zip(rows, results):
             ns = {"status": res.status, "response": res.response}
+            if debug:
+                ns["debug_trace"] = res.trace
             r.ns[alias] = {c: ns.get(c) for c in cols} if cols else ns
         return rows
 

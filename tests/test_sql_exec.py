@@ -336,3 +336,30 @@ def test_plain_group_by():
         {"region": "e", "n": 2, "total": 40.0, "mean": 20.0},
         {"region": "n", "n": 1, "total": 7.0, "mean": 7.0},
         {"region": "w", "n": 2, "total": 6.0, "mean": 3.0}]
+
+
+def test_run_agent_debug_map_collects_traces():
+    """MAP['debug','true'] on AI_RUN_AGENT surfaces per-episode traces
+    (the reference's debug map semantics, LAB1-Walkthrough.md:253)."""
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+    cat = Catalog()
+    cat.execute("""
+    CREATE MODEL m INPUT (p STRING) OUTPUT (r STRING)
+      WITH ('provider' = 'local');
+    CREATE AGENT ag USING MODEL m USING PROMPT 'sys'
+      WITH ('max_iterations' = '3');
+    CREATE TABLE src (q STRING);
+    CREATE TABLE out_dbg AS
+    SELECT s.q, agent_result.status, agent_result.debug_trace
+    FROM src s,
+    LATERAL TABLE(AI_RUN_AGENT('ag', s.q, MAP['debug','true']))
+      AS agent_result(status, response, debug_trace);
+    """)
+    broker = Broker()
+    broker.create_topic("src").append({"q": "hello"}, partition=0)
+    ex = SqlExecutor(cat, broker, llm_batch=lambda ps, ts: ["done"] * len(ps))
+    rows = ex.run_table("out_dbg")
+    assert rows[0]["status"] == "SUCCESS"
+    assert any("model_output" in t for t in rows[0]["debug_trace"])
