@@ -28,9 +28,20 @@ sources = [
     os.path.join(HIP_DIR, "streaming.hip"),
 ]
 
+from setuptools import find_packages  # noqa: E402
+
 setup(
-    name="qsa-hip",
+    name="quickstart-streaming-agents-amd",
     version="0.1.0",
+    description=("MI355X-native streaming-agent engine: Kafka-wire "
+                 "ingest, Flink-SQL-subset CREATE AGENT surface, "
+                 "hand-written CDNA4 HIP kernels, RCCL over xGMI"),
+    packages=find_packages(include=["quickstart_streaming_agents_amd*"]),
+    package_data={"quickstart_streaming_agents_amd": [
+        "labs/sql/*.sql", "data/*.csv"]},
+    python_requires=">=3.10",
+    entry_points={"console_scripts": [
+        "qsa=quickstart_streaming_agents_amd.cli:main"]},
     ext_modules=[
         CUDAExtension(
             name="quickstart_streaming_agents_amd.qsa_hip",
