@@ -88,3 +88,36 @@ def test_request_ids_increment_and_404(server):
     assert cl._id >= 2
     with pytest.raises(urllib.error.HTTPError):
         urllib.request.urlopen(server.base_url + "/nope", timeout=10)
+
+
+def test_concurrent_tool_call_burst(server):
+    """Hundreds of concurrent episodes hit the stub MCP server at once
+    (the bench's burst shape): every call must succeed — backlog 256 +
+    client retry absorb connection churn (no dropped connections)."""
+    import threading
+
+    from quickstart_streaming_agents_amd.agents.mcp import McpClient
+    errors = []
+    results = []
+    lock = threading.Lock()
+
+    def worker(i):
+        try:
+            cl = McpClient(server.mcp_endpoint)
+            out = cl.tools_call("send_email", {
+                "to": f"u{i}@example.com", "subject": "s", "body": "b"})
+            with lock:
+                results.append(out)
+        except Exception as e:        # noqa: BLE001
+            with lock:
+                errors.append(repr(e))
+
+    threads = [threading.Thread(target=worker, args=(i,))
+               for i in range(120)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=60)
+    assert not errors, errors[:3]
+    assert len(results) == 120
+    assert len(server.emails) >= 120
