@@ -83,6 +83,12 @@ class TumblingWindows:
         ready.sort(key=lambda k: (self._panes[k].window_start, str(k[0])))
         return [self._panes.pop(k) for k in ready]
 
+    @property
+    def late_dropped(self) -> int:
+        """Rows dropped for arriving beyond the watermark (the metric
+        Flink exposes as numLateRecordsDropped)."""
+        return self._late_dropped
+
     def flush(self) -> list[WindowResult]:
         """Close every remaining pane (end of bounded input)."""
         self.wm.observe(1 << 62)

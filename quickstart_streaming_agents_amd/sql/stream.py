@@ -268,6 +268,20 @@ class StreamingQuery:
             rows = stage.on_left(rows) + from_right
         return self._finish_rows(rows)
 
+    def stats(self) -> dict:
+        """Operator-level counters for monitoring (rows emitted, join
+        buffer sizes, late-dropped window rows)."""
+        out = {"sink": self.sink, "emitted": self.emitted}
+        if self.info.tumble:
+            out["late_dropped"] = self.windows.late_dropped
+            out["open_panes"] = len(self.windows._panes)
+        if self.joins:
+            out["join_buffered"] = sum(
+                sum(len(v) for v in st.left_buf.values()) +
+                sum(len(v) for v in st.right_buf.values())
+                for _, st in self.joins)
+        return out
+
     def flush(self) -> list[dict]:
         """Bounded-input end: close every remaining window pane."""
         if not self.info.tumble:
@@ -377,6 +391,9 @@ class StreamingPipeline:
             if not moved:
                 break
         return out
+
+    def stats(self) -> list[dict]:
+        return [q.stats() for q in self.queries]
 
     def snapshot(self) -> dict:
         return {q.sink: q.snapshot() for q in self.queries}

@@ -215,3 +215,16 @@ def test_ctas_cycle_detected():
     """)
     with pytest.raises(SqlExecError, match="cycle"):
         StreamingPipeline(SqlExecutor(cat, Broker()))
+
+
+def test_pipeline_stats_counters(mcp):
+    dep = Deployment(labs=(3,), device="cpu")
+    dep.datagen(3)
+    pipe = StreamingPipeline(dep.sql_executor(3, mcp_server=mcp))
+    pipe.advance()
+    pipe.finish()
+    stats = {s["sink"]: s for s in pipe.stats()}
+    apz = stats["anomalies_per_zone"]
+    assert apz["emitted"] >= 1
+    assert apz["late_dropped"] >= 0 and apz["open_panes"] == 0
+    assert stats["completed_actions"]["emitted"] == apz["emitted"]
