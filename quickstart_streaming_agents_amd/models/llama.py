@@ -55,6 +55,7 @@ class LlamaConfig:
     rope_theta: float = 500_000.0
     max_pos: int = 8192
     norm_eps: float = 1e-5
+    attn_bias: bool = False     # Qwen2-family: biases on the q/k/v proj
 
     @classmethod
     def preset(cls, name: str) -> "LlamaConfig":
@@ -67,6 +68,18 @@ class LlamaConfig:
             return cls(name=name, vocab_size=2048, hidden=256, n_layers=2,
                        n_q_heads=4, n_kv_heads=2, d_head=64, ffn=512,
                        max_pos=2048)
+        if name in ("qwen2-7b", "qwen2.5-7b"):
+            # Qwen2-7B geometry: GQA 28/4, wide FFN, qkv biases,
+            # rope theta 1e6 (the other mainstream dense-7B family)
+            return cls(name=name, vocab_size=152_064, hidden=3584,
+                       n_layers=28, n_q_heads=28, n_kv_heads=4, d_head=128,
+                       ffn=18_944, rope_theta=1_000_000.0, max_pos=8192,
+                       attn_bias=True)
+        if name == "tiny-qwen":
+            return cls(name=name, vocab_size=2048, hidden=256, n_layers=2,
+                       n_q_heads=4, n_kv_heads=2, d_head=64, ffn=512,
+                       max_pos=2048, rope_theta=1_000_000.0,
+                       attn_bias=True)
         raise ValueError(f"unknown preset {name}")
 
 
@@ -128,6 +141,13 @@ class LlamaModel:
                 "mlp_norm": torch.ones(c.hidden, device=device, dtype=dtype),
                 "wqkv": wqkv, "wo": wo,
             }
+            if c.attn_bias:
+                bq = w(c.n_q_heads * c.d_head)
+                bk = w(c.n_kv_heads * c.d_head)
+                bv = w(c.n_kv_heads * c.d_head)
+                layer["bqkv"] = torch.cat(
+                    [rows(bq, tp_size), rows(bk, tp_size),
+                     rows(bv, tp_size)]).contiguous()
             layer.update(self._ffn_weights(w, rows, cols))
             self.layers.append(layer)
         # rope tables
@@ -220,6 +240,8 @@ class LlamaModel:
                 h = D.rmsnorm_residual(mlp_out, res, L["attn_norm"],
                                        c.norm_eps)
             qkv = self._linear(h, L["wqkv"], L.get("wqkv_f"))
+            if "bqkv" in L:
+                qkv = qkv + L["bqkv"]
             # strided [B, H, D] views straight into the fused qkv buffer —
             # the kernels take row strides, no contiguous() copies
             qd = self.n_q * c.d_head
@@ -345,7 +367,7 @@ class LlamaModel:
             else:
                 h = D.rmsnorm_residual(mlp_out, res, L["attn_norm"],
                                        c.norm_eps)
-            qkv = F.linear(h, L["wqkv"])
+            qkv = F.linear(h, L["wqkv"], L.get("bqkv"))
             q, k, v = self._split_qkv(qkv, T)
             D.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
             D.kv_scatter(k, v, kv.k[li], kv.v[li], slots)

@@ -228,3 +228,54 @@ def test_dp_sharded_sql_anomalies_match_unsharded():
     assert res["total_kept"] == res["n_records"]      # exact cover
     assert res["merged"] == res["ref"]
     assert len(res["ref"]) >= 1                       # the FQ anomaly
+
+
+def _tp_qwen_worker(rank, world):
+    import torch.distributed as dist
+
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    from quickstart_streaming_agents_amd.models.serve import Engine
+    dist.init_process_group("gloo")
+    cfg = LlamaConfig.preset("tiny-qwen")
+    model = LlamaModel(cfg, device="cpu", dtype=torch.float32, seed=11,
+                       tp_rank=rank, tp_size=world, tp_group=None)
+    eng = Engine(model, max_batch=4, max_seq_len=256)
+    outs = eng.generate_batch([[3, 7, 11], [2, 9, 4, 6]], [5, 5])
+    dist.destroy_process_group()
+    return outs
+
+
+@pytest.mark.timeout(300)
+def test_tp2_qwen_bias_matches_tp1():
+    """Qwen2-family (qkv biases): TP=2 == TP=1 — the bias shards with the
+    column-parallel rows."""
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    from quickstart_streaming_agents_amd.models.serve import Engine
+    ref = Engine(LlamaModel(LlamaConfig.preset("tiny-qwen"), device="cpu",
+                            dtype=torch.float32, seed=11),
+                 max_batch=4, max_seq_len=256)
+    want = ref.generate_batch([[3, 7, 11], [2, 9, 4, 6]], [5, 5])
+    got = spawn_world(_tp_qwen_worker, world=2)
+    assert got[0] == want and got[1] == want
+
+
+def test_qwen_preset_geometry_and_bias_changes_output():
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    from quickstart_streaming_agents_amd.models.serve import Engine
+    cfg = LlamaConfig.preset("qwen2-7b")
+    assert (cfg.n_q_heads, cfg.n_kv_heads, cfg.hidden, cfg.ffn) == \
+        (28, 4, 3584, 18_944)
+    assert cfg.attn_bias and cfg.rope_theta == 1_000_000.0
+    # bias actually participates: zeroing it changes the decode output
+    m = LlamaModel(LlamaConfig.preset("tiny-qwen"), device="cpu",
+                   dtype=torch.float32, seed=5)
+    out1 = Engine(m, max_batch=2, max_seq_len=128).generate_batch(
+        [[1, 2, 3]], [4])
+    for L in m.layers:
+        L["bqkv"].zero_()
+    out2 = Engine(m, max_batch=2, max_seq_len=128).generate_batch(
+        [[1, 2, 3]], [4])
+    assert out1 != out2
