@@ -327,3 +327,20 @@ def test_serve_api_on_gpu_engine():
             assert "qsa_requests_total" in c.get("/metrics").text
     finally:
         app.state.mcp_server.stop()
+
+
+def test_qwen_bias_engine_on_gpu():
+    """Qwen2-family decode (qkv biases) through the full GPU engine:
+    hipGraph capture with the bias add in the captured region, greedy
+    decode deterministic across two engines."""
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    from quickstart_streaming_agents_amd.models.serve import Engine
+    cfg = LlamaConfig.preset("tiny-qwen")
+    outs = []
+    for _ in range(2):
+        eng = Engine(LlamaModel(cfg, device="cuda:0", seed=13),
+                     max_batch=4, max_seq_len=256)
+        outs.append(eng.generate_batch([[3, 7, 11], [2, 9, 4, 6]], [6, 6]))
+    assert outs[0] == outs[1]
+    assert all(len(o) == 6 for o in outs[0])
