@@ -47,6 +47,15 @@ class MixtralConfig(LlamaConfig):
             return cls(name=name, vocab_size=2048, hidden=256, n_layers=2,
                        n_q_heads=4, n_kv_heads=2, d_head=64, ffn=512,
                        max_pos=2048, n_experts=4, top_k=2)
+        if name == "mixtral-8x22b":
+            # ~141B params: ~282 GB bf16 — does NOT fit one 288 GB GPU
+            # with KV cache; serve with EP>=2 (parallel/ep.py shards the
+            # 8 experts across ranks, each rank ~46 GB of experts +
+            # ~16 GB shared)
+            return cls(name=name, vocab_size=32_768, hidden=6144,
+                       n_layers=56, n_q_heads=48, n_kv_heads=8,
+                       d_head=128, ffn=16_384, rope_theta=1_000_000.0,
+                       max_pos=8192, n_experts=8, top_k=2)
         raise ValueError(f"unknown preset {name}")
 
 

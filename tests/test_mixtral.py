@@ -57,3 +57,13 @@ def test_mixtral_ep2_matches_ep1():
     results = spawn_world(_ep2_worker, world=2)
     assert results[0] == ref, f"EP2 {results[0]} != EP1 {ref}"
     assert results[1] == ref
+
+
+def test_8x22b_preset_geometry():
+    cfg = MixtralConfig.preset("mixtral-8x22b")
+    assert (cfg.hidden, cfg.n_layers, cfg.n_experts, cfg.top_k) == \
+        (6144, 56, 8, 2)
+    # bf16 expert bytes per rank at EP=2 stay under one GPU's HBM
+    per_expert = 3 * cfg.ffn * cfg.hidden * 2          # w1/w2/w3 bf16
+    per_rank_experts = cfg.n_layers * (cfg.n_experts // 2) * per_expert
+    assert per_rank_experts < 288e9
