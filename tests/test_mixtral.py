@@ -60,6 +60,7 @@ def test_mixtral_ep2_matches_ep1():
 
 
 def test_8x22b_preset_geometry():
+    from quickstart_streaming_agents_amd.models.mixtral import MixtralConfig
     cfg = MixtralConfig.preset("mixtral-8x22b")
     assert (cfg.hidden, cfg.n_layers, cfg.n_experts, cfg.top_k) == \
         (6144, 56, 8, 2)
