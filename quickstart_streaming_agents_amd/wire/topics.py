@@ -179,10 +179,14 @@ class AvroProducer:
 class AvroConsumer:
     """Consume wire-format records back into dicts (Avro-or-passthrough tolerant)."""
 
-    def __init__(self, broker: Broker, topic: str, value_schema: Schema | str | dict):
+    def __init__(self, broker: Broker, topic: str, value_schema: Schema | str | dict,
+                 key_schema: Schema | str | dict | None = None):
         self.broker = broker
         self.topic_name = topic
         self.value_schema = value_schema if isinstance(value_schema, Schema) else Schema(value_schema)
+        self.key_schema = (key_schema if isinstance(key_schema, Schema)
+                           else Schema(key_schema)) \
+            if key_schema is not None else None
         self._codec = _native_codec(self.value_schema)
         self._offsets: dict[int, int] = {}
         self._reader_canonical = self.value_schema.canonical()
@@ -231,6 +235,11 @@ class AvroConsumer:
                         _, v = self._codec.deserialize(bytes(v))
                     else:
                         _, v = deserialize(self.value_schema, bytes(v))
+                if self.key_schema is not None and \
+                        isinstance(r.key, (bytes, bytearray)):
+                    _, dk = deserialize(self.key_schema, bytes(r.key))
+                    r = Record(r.topic, r.partition, r.offset,
+                               r.timestamp_ms, dk, r.value)
                 out.append((r, v))
             self._offsets[pi] = start + len(recs)
         return out

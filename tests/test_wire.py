@@ -288,3 +288,25 @@ def test_keyed_partitioning_is_murmur2():
         assert rec.partition == partition_for_key(k, 8)
     # keyless appends pin partition 0 (the labs' watermark determinism)
     assert t.append({"k": None}).partition == 0
+
+
+def test_keyed_produce_consume_roundtrip():
+    """Key schemas round-trip like value schemas (TopicNameStrategy
+    <topic>-key subject): the consumer decodes wire-format keys back."""
+    from quickstart_streaming_agents_amd.wire import Broker
+    from quickstart_streaming_agents_amd.wire.topics import (AvroConsumer,
+                                                             AvroProducer)
+    vs = {"type": "record", "name": "v", "fields": [
+        {"name": "n", "type": "int"}]}
+    ks = {"type": "record", "name": "k", "fields": [
+        {"name": "id", "type": "string"}]}
+    broker = Broker()
+    p = AvroProducer(broker, "t", vs, key_schema=ks)
+    p.produce({"n": 1}, key={"id": "a"})
+    p.produce({"n": 2}, key={"id": "b"})
+    assert "t-key" in broker.registry.subjects()
+    out = AvroConsumer(broker, "t", vs, key_schema=ks).poll()
+    assert [(r.key["id"], v["n"]) for r, v in out] == [("a", 1), ("b", 2)]
+    # without a key schema the raw wire bytes come back untouched
+    raw = AvroConsumer(broker, "t", vs).poll()
+    assert isinstance(raw[0][0].key, (bytes, bytearray))
