@@ -86,6 +86,18 @@ def cmd_validate(args) -> int:
         ok &= "http_get" in names
     finally:
         srv.stop()
+    from .sql.catalog import Catalog
+    from .sql.exec import SqlExecutor
+    from .wire import Broker
+    cat = Catalog()
+    cat.execute("CREATE TABLE t (a STRING); "
+                "CREATE TABLE u AS SELECT s.a FROM t s WHERE s.a <> '';")
+    b = Broker()
+    b.create_topic("t").append({"a": "x"}, partition=0)
+    rows = SqlExecutor(cat, b).run_table("u")
+    print(f"[validate] SQL executor round trip: "
+          f"{'OK' if rows == [{'a': 'x'}] else 'FAIL'}")
+    ok &= rows == [{"a": "x"}]
     from .ops import have_ext
     print(f"[validate] qsa_hip extension importable: {have_ext()}")
     import torch
