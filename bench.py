@@ -160,6 +160,8 @@ def main():
     total_decisions = args.batch * args.steps * world
     value = total_decisions / elapsed
     p50_ms = statistics.median(step_times) * 1000.0
+    p95_ms = (sorted(step_times)[max(0, int(len(step_times) * 0.95) - 1)]
+              * 1000.0 if step_times else 0.0)
 
     if rank == 0:
         out = {
@@ -171,6 +173,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1000.0, 2),
             "p50_e2e_ms": round(p50_ms, 2),
+            "p95_e2e_ms": round(p95_ms, 2),
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": round(value / BASELINE_DECISIONS_PER_SEC, 1),
