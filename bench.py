@@ -160,7 +160,10 @@ def main():
     total_decisions = args.batch * args.steps * world
     value = total_decisions / elapsed
     p50_ms = statistics.median(step_times) * 1000.0
-    p95_ms = (sorted(step_times)[max(0, int(len(step_times) * 0.95) - 1)]
+    import math
+    p95_ms = (sorted(step_times)[
+        min(len(step_times) - 1,
+            max(0, math.ceil(len(step_times) * 0.95) - 1))]
               * 1000.0 if step_times else 0.0)
 
     if rank == 0:
