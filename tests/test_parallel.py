@@ -37,11 +37,17 @@ def _run_worker(fn, rank, world, port, q, args):
         q.put((rank, "err", f"{e}\n{traceback.format_exc()}"))
 
 
+_PORT_SEQ = [0]
+
+
 def spawn_world(fn, world=2, args=(), timeout=240):
     """Run fn(rank, world, *args) in `world` processes; returns rank->result."""
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29600 + (os.getpid() % 500)
+    # fresh port per call: a previous rendezvous socket in TIME_WAIT on
+    # the same port can fail the TCP store bind
+    _PORT_SEQ[0] += 1
+    port = 29600 + (os.getpid() + _PORT_SEQ[0] * 7) % 500
     procs = [ctx.Process(target=_run_worker, args=(fn, r, world, port, q, args))
              for r in range(world)]
     for p in procs:
