@@ -178,6 +178,10 @@ def cmd_sql(args) -> int:
                 print(f"-- DESCRIBE {st.name}:")
                 for col, ty in cat.describe(st.name):
                     print(f"   {col:32s} {ty}")
+            elif isinstance(st, P.ExplainStmt):
+                print(f"-- plan for {st.name}:")
+                for step in ex.explain(st.name):
+                    print(f"   {step}")
             elif isinstance(st, P.InsertInto) and materialize:
                 ex.run_inserts()
                 cat.inserts.clear()

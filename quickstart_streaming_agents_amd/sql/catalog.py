@@ -137,7 +137,8 @@ class Catalog:
         return out
 
     def apply(self, st) -> None:
-        if isinstance(st, (P.ShowStmt, P.DescribeStmt)):
+        if isinstance(st, (P.ShowStmt, P.DescribeStmt,
+                           P.ExplainStmt)):
             return                       # read-only statements
         if isinstance(st, P.SetStmt):
             self.session[st.key] = st.value

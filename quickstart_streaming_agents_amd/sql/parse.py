@@ -308,6 +308,11 @@ class DescribeStmt:
     name: str
 
 
+@dataclass
+class ExplainStmt:
+    name: str                    # CTAS table to plan
+
+
 def parse_statement(stmt: str):
     s = stmt.strip()
     up = s.upper()
@@ -317,6 +322,11 @@ def parse_statement(stmt: str):
         if not m:
             raise ValueError(f"bad SHOW: {s!r}")
         return ShowStmt(m.group(1).upper())
+    if up.startswith("EXPLAIN"):
+        m = re.match(r"EXPLAIN\s+(\S+)\s*$", s, re.IGNORECASE)
+        if not m:
+            raise ValueError(f"bad EXPLAIN: {s!r}")
+        return ExplainStmt(_ident(m.group(1)))
     if up.startswith("DESCRIBE") or up.startswith("DESC "):
         m = re.match(r"DESC(?:RIBE)?\s+(\S+)\s*$", s, re.IGNORECASE)
         if not m:

@@ -223,3 +223,10 @@ def test_show_and_describe():
     d = P.parse_statement("DESCRIBE `orders`")
     assert d.name == "orders"
     cat.execute("SHOW MODELS; DESCRIBE orders;")   # no-ops, no error
+
+
+def test_explain_statement_parses():
+    st = P.parse_statement("EXPLAIN anomalies_per_zone")
+    assert st.name == "anomalies_per_zone"
+    cat = Catalog()
+    cat.execute(LAB1_DDL + "; EXPLAIN orders;")   # read-only no-op
