@@ -71,8 +71,12 @@ JOIN claims_anomalies_by_city a
 WHERE c.claim_narrative <> ''
 LIMIT 10;
 
+-- carry the structured claim through: the fraud agent's checklist needs
+-- amounts/residence/insurance/shared-identity fields (LAB4:249-310)
 CREATE TABLE claims_to_investigate_with_policies AS
-SELECT ci.claim_id, ci.claim_narrative,
+SELECT ci.claim_id, ci.claim_narrative, ci.city, ci.claim_amount,
+       ci.is_primary_residence, ci.damage_assessed, ci.has_insurance,
+       ci.insurance_amount, ci.shared_account, ci.shared_phone,
        search_results.chunk1, search_results.chunk2, search_results.chunk3
 FROM claims_to_investigate ci,
 LATERAL TABLE(ML_PREDICT('llm_embedding_model', ci.claim_narrative)),

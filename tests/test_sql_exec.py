@@ -67,6 +67,9 @@ def test_lab3_generic_contract(mcp):
 
 
 def test_lab4_generic_contract(mcp):
+    ref_dep = Deployment(labs=(4,), device="cpu")
+    ref_dep.datagen(4)
+    hand = ref_dep.run(4)
     dep = Deployment(labs=(4,), device="cpu")
     dep.datagen(4)
     rows = dep.run_sql(4)
@@ -75,6 +78,10 @@ def test_lab4_generic_contract(mcp):
     for r in rows:
         assert r["verdict"] in LAB4_VERDICTS
         assert r["summary"]
+    # verdict-for-verdict parity with the hand-fused pipeline: the
+    # structured claim survives the CTAS chain into the fraud checklist
+    hand_verdicts = {h["claim_id"]: h["verdict"] for h in hand}
+    assert {r["claim_id"]: r["verdict"] for r in rows} == hand_verdicts
     # anomaly city contract: exactly Naples
     anoms = dep.broker.topics["claims_anomalies_by_city"].read_all()
     cities = {m.value["city"] for m in anoms}
