@@ -39,7 +39,9 @@ def murmur2(data: bytes) -> int:
     if length >= 1:
         h ^= data[i]
         h = (h * m) & 0xFFFFFFFF
-        h ^= h >> 13
+    # finalization runs unconditionally (Java Utils.murmur2 applies
+    # h ^= h >>> 13 after the tail switch, even for length % 4 == 0)
+    h ^= h >> 13
     h = (h * m) & 0xFFFFFFFF
     h ^= h >> 15
     return h & 0x7FFFFFFF

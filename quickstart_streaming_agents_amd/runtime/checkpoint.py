@@ -76,7 +76,6 @@ class CheckpointStore:
             if os.path.exists(tmp):
                 os.unlink(tmp)
         # prune old checkpoints
-        old = sorted(x for x in (self.latest_id() or 0,) )
         ids = sorted(int(f[5:-5]) for f in os.listdir(self.dir)
                      if f.startswith("ckpt-") and f.endswith(".json"))
         for i in ids[:-keep]:

@@ -15,12 +15,14 @@ JSON-serializable forms of runtime/checkpoint.py, so a pipeline can
 resume mid-stream after a crash (the reference's recovery story is
 replay-from-offset on Confluent Flink; SURVEY.md 2.5 elasticity row).
 
-Known bound: join buffers grow with the stream (Flink bounds them with
-`SET 'sql.state-ttl'`, which needs a per-table event-time column to
-apply generically; the hand-fused pipelines honor TTL via
-runtime/joins.TTLTable on the columns the labs declare —
-labs/pipelines.py lab1_enriched_orders).  The lab4 interval join is
-bounded semantically by its BETWEEN predicate regardless.
+Join state is bounded by ``SET 'sql.state-ttl'`` (reference
+LAB1-Walkthrough.md:119-120, LAB4:124): every buffered join row is
+stamped with the statement's stream time (max event-time observed so
+far, advanced by any source column the tables declare as event time —
+their WATERMARK column or first TIMESTAMP column), and entries whose
+stamp falls behind stream_time - ttl are evicted — the same
+"state not updated for TTL" contract Flink's keyed-state TTL gives.
+Statements with no `sql.state-ttl` keep state forever, as Flink does.
 """
 
 from __future__ import annotations
@@ -37,7 +39,8 @@ from .exec import (SqlExecError, SqlExecutor, _extract_laterals,
 class _TwoSidedJoin:
     """One streaming equi-join stage (+ residual predicates): buffers both
     sides, emits each matched pair exactly once (when the later side
-    arrives)."""
+    arrives).  Buffer entries are (stream_time_stamp, row); `evict()`
+    drops entries whose stamp falls behind the TTL cutoff."""
 
     def __init__(self, ex: SqlExecutor, alias: str, cond: str):
         self.ex = ex
@@ -56,8 +59,9 @@ class _TwoSidedJoin:
                     self.eq_right.append(r.split(".", 1)[1])
                     continue
             self.residual.append(t)
-        self.left_buf: dict[tuple, list[_Row]] = {}
-        self.right_buf: dict[tuple, list[dict]] = {}
+        self.left_buf: dict[tuple, list[tuple[int, _Row]]] = {}
+        self.right_buf: dict[tuple, list[tuple[int, dict]]] = {}
+        self.evicted = 0
 
     def _lkey(self, row: _Row) -> tuple:
         return tuple(self.ex.ev.eval(e, row) for e in self.eq_left)
@@ -72,40 +76,65 @@ class _TwoSidedJoin:
             return cand
         return None
 
-    def on_left(self, rows: list[_Row]) -> list[_Row]:
+    def on_left(self, rows: list[_Row], stamp: int = 0) -> list[_Row]:
         out = []
         for row in rows:
             k = self._lkey(row)
-            self.left_buf.setdefault(k, []).append(row)
-            for d in self.right_buf.get(k, ()):
+            self.left_buf.setdefault(k, []).append((stamp, row))
+            for _, d in self.right_buf.get(k, ()):
                 c = self._pair(row, d)
                 if c is not None:
                     out.append(c)
         return out
 
-    def on_right(self, dicts: list[dict]) -> list[_Row]:
+    def on_right(self, dicts: list[dict], stamp: int = 0) -> list[_Row]:
         out = []
         for d in dicts:
             k = self._rkey(d)
-            self.right_buf.setdefault(k, []).append(d)
-            for row in self.left_buf.get(k, ()):
+            self.right_buf.setdefault(k, []).append((stamp, d))
+            for _, row in self.left_buf.get(k, ()):
                 c = self._pair(row, d)
                 if c is not None:
                     out.append(c)
         return out
+
+    def evict(self, cutoff: int) -> int:
+        """Drop buffered entries stamped before `cutoff`; returns count."""
+        n = 0
+        for buf in (self.left_buf, self.right_buf):
+            dead_keys = []
+            for k, entries in buf.items():
+                kept = [e for e in entries if e[0] >= cutoff]
+                if len(kept) != len(entries):
+                    n += len(entries) - len(kept)
+                    if kept:
+                        buf[k] = kept
+                    else:
+                        dead_keys.append(k)
+            for k in dead_keys:
+                del buf[k]
+        self.evicted += n
+        return n
+
+    def size(self) -> int:
+        return (sum(len(v) for v in self.left_buf.values())
+                + sum(len(v) for v in self.right_buf.values()))
 
     def snapshot(self) -> dict:
         """JSON-serializable buffers (joins precede LATERAL stages in the
         grammar, so buffered rows are plain column dicts)."""
-        return {"left": [[list(k), [r.ns for r in rows]]
+        return {"left": [[list(k), [[ts, r.ns] for ts, r in rows]]
                          for k, rows in self.left_buf.items()],
-                "right": [[list(k), ds]
-                          for k, ds in self.right_buf.items()]}
+                "right": [[list(k), [[ts, d] for ts, d in ds]]
+                          for k, ds in self.right_buf.items()],
+                "evicted": self.evicted}
 
     def restore(self, snap: dict) -> None:
-        self.left_buf = {tuple(k): [_Row(ns) for ns in rows]
+        self.left_buf = {tuple(k): [(ts, _Row(ns)) for ts, ns in rows]
                          for k, rows in snap.get("left", [])}
-        self.right_buf = {tuple(k): ds for k, ds in snap.get("right", [])}
+        self.right_buf = {tuple(k): [(ts, d) for ts, d in ds]
+                          for k, ds in snap.get("right", [])}
+        self.evicted = snap.get("evicted", 0)
 
 
 class StreamingQuery:
@@ -123,7 +152,7 @@ class StreamingQuery:
         self.limit = (int(self.clauses["limit"].split()[0])
                       if self.clauses["limit"] else None)
         self.emitted = 0
-        self._offsets: dict[str, int] = {}
+        self._offsets: dict[str, dict[int, int]] = {}  # table -> {part -> off}
         self._consumers: dict[str, Any] = {}
 
         if self.info.tumble:
@@ -177,6 +206,35 @@ class StreamingQuery:
                 (name, _TwoSidedJoin(ex, alias, cond))
                 for name, alias, cond in tables[1:]]
 
+        # ---- state TTL (SET 'sql.state-ttl'; LAB1-Walkthrough.md:119-120)
+        self.ttl_ms = ex.catalog.state_ttl_ms()
+        self._stream_time = -(1 << 62)   # max event ts observed so far
+        self._ts_cols = {t: self._event_time_col(t)
+                         for t in [self.stream_table]
+                         + [name for name, _ in self.joins]}
+
+    def _event_time_col(self, table: str) -> str | None:
+        """The column that advances stream time for TTL: the table's
+        WATERMARK column if declared, else its first TIMESTAMP column."""
+        t = self.ex.catalog.tables.get(table)
+        if t is None:
+            return None
+        if t.watermark:
+            return t.watermark[0]
+        for c in t.columns:
+            if c.type.upper().startswith("TIMESTAMP"):
+                return c.name
+        return None
+
+    def _observe(self, table: str, dicts: list[dict]) -> None:
+        col = self._ts_cols.get(table)
+        if col is None:
+            return
+        for d in dicts:
+            v = d.get(col)
+            if isinstance(v, (int, float)) and v > self._stream_time:
+                self._stream_time = int(v)
+
     # -- incremental sources ------------------------------------------------
     def _new_source_rows(self, table: str) -> list[dict]:
         t = self.ex.catalog.tables.get(table)
@@ -191,13 +249,18 @@ class StreamingQuery:
         topic = self.ex.broker.topics.get(table)
         if topic is None:
             return []
-        recs = topic.read_all()
-        off = self._offsets.get(table, 0)
-        self._offsets[table] = len(recs)
+        # per-partition offsets (like AvroConsumer.poll): a late-timestamp
+        # append must be delivered exactly once, which a slice of the
+        # globally timestamp-sorted read_all() cannot guarantee
+        offs = self._offsets.setdefault(table, {})
         out = []
-        for rec in recs[off:]:
-            v = rec.value
-            out.append(v if isinstance(v, dict) else {"value": v})
+        for pi, p in enumerate(topic.partitions):
+            start = offs.get(pi, 0)
+            recs = p.read(start)
+            offs[pi] = start + len(recs)
+            for rec in recs:
+                v = rec.value
+                out.append(v if isinstance(v, dict) else {"value": v})
         return out
 
     # -- stages -------------------------------------------------------------
@@ -258,15 +321,30 @@ class StreamingQuery:
             panes = self.windows.feed(new)
             panes.sort(key=lambda p: (p.window_start, str(p.key)))
             return self._finish_rows(self._window_close(panes))
-        rows = [_Row({self.stream_alias: d})
-                for d in self._new_source_rows(self.stream_table)]
+        new_left = self._new_source_rows(self.stream_table)
+        self._observe(self.stream_table, new_left)
+        new_right = {name: self._new_source_rows(name)
+                     for name, _ in self.joins}
+        for name, ds in new_right.items():
+            self._observe(name, ds)
+        stamp = self._stream_time
+        rows = [_Row({self.stream_alias: d}) for d in new_left]
+        finished: list[_Row] = []
         for i, (name, stage) in enumerate(self.joins):
-            # new right-side rows first: they match already-buffered lefts
-            from_right = stage.on_right(self._new_source_rows(name))
+            # new right-side rows first: they match already-buffered lefts.
+            # The cascade through later stages fully joins them, so they go
+            # straight to `finished` — pushing them back through `rows`
+            # would buffer and emit each matched pair twice.
+            from_right = stage.on_right(new_right[name], stamp)
             for _, later in self.joins[i + 1:]:
-                from_right = later.on_left(from_right)
-            rows = stage.on_left(rows) + from_right
-        return self._finish_rows(rows)
+                from_right = later.on_left(from_right, stamp)
+            finished.extend(from_right)
+            rows = stage.on_left(rows, stamp)
+        if self.ttl_ms is not None and self.joins:
+            cutoff = self._stream_time - self.ttl_ms
+            for _, stage in self.joins:
+                stage.evict(cutoff)
+        return self._finish_rows(rows + finished)
 
     def stats(self) -> dict:
         """Operator-level counters for monitoring (rows emitted, join
@@ -276,10 +354,8 @@ class StreamingQuery:
             out["late_dropped"] = self.windows.late_dropped
             out["open_panes"] = len(self.windows._panes)
         if self.joins:
-            out["join_buffered"] = sum(
-                sum(len(v) for v in st.left_buf.values()) +
-                sum(len(v) for v in st.right_buf.values())
-                for _, st in self.joins)
+            out["join_buffered"] = sum(st.size() for _, st in self.joins)
+            out["join_evicted"] = sum(st.evicted for _, st in self.joins)
         return out
 
     def flush(self) -> list[dict]:
@@ -293,7 +369,9 @@ class StreamingQuery:
     # -- checkpoint ---------------------------------------------------------
     def snapshot(self) -> dict:
         from ..runtime.checkpoint import snapshot_anomaly, snapshot_windows
-        snap: dict = {"offsets": dict(self._offsets),
+        snap: dict = {"offsets": {t: {str(p): o for p, o in offs.items()}
+                                  for t, offs in self._offsets.items()},
+                      "stream_time": self._stream_time,
                       "consumer_offsets": {
                           t: c.offsets() if hasattr(c, "offsets")
                           else getattr(c, "_offsets", None)
@@ -309,7 +387,13 @@ class StreamingQuery:
 
     def restore(self, snap: dict) -> None:
         from ..runtime.checkpoint import restore_anomaly, restore_windows
-        self._offsets = dict(snap.get("offsets", {}))
+        self._offsets = {}
+        for t, offs in snap.get("offsets", {}).items():
+            if isinstance(offs, dict):
+                self._offsets[t] = {int(p): o for p, o in offs.items()}
+            else:   # legacy count form (single-partition topics)
+                self._offsets[t] = {0: int(offs)}
+        self._stream_time = snap.get("stream_time", -(1 << 62))
         self.emitted = snap.get("emitted", 0)
         for t, offs in (snap.get("consumer_offsets") or {}).items():
             if offs is None:

@@ -285,3 +285,19 @@ def test_qwen_preset_geometry_and_bias_changes_output():
     out2 = Engine(m, max_batch=2, max_seq_len=128).generate_batch(
         [[1, 2, 3]], [4])
     assert out1 != out2
+
+
+def test_murmur2_matches_java_kafka_client():
+    """Regression: finalization (h ^= h >>> 13) must
+    run unconditionally, including keys whose length % 4 == 0.  Vectors
+    are org.apache.kafka.common.utils.UtilsTest#testMurmur2, in
+    Utils.toPositive (& 0x7fffffff) form as the partitioner applies."""
+    from quickstart_streaming_agents_amd.parallel.stream_shard import murmur2
+    java = {  # key -> signed int32 from the Java client
+        b"21": -973932308,
+        b"foobar": -790332482,
+        b"a-little-bit-long-string": -985981536,  # len % 4 == 0
+        b"abc": 479470107,
+    }
+    for k, signed in java.items():
+        assert murmur2(k) == (signed & 0x7FFFFFFF), k

@@ -228,3 +228,120 @@ def test_pipeline_stats_counters(mcp):
     assert apz["emitted"] >= 1
     assert apz["late_dropped"] >= 0 and apz["open_panes"] == 0
     assert stats["completed_actions"]["emitted"] == apz["emitted"]
+
+
+def test_lab1_dim_resend_emits_each_pair_once(mcp):
+    """Regression: the multi-join cascade used to push
+    cascaded right-side results back through later stages, double-buffering
+    them — re-sending a dimension row then emitted duplicates."""
+    dep = Deployment(labs=(1,), device="cpu")
+    dep.datagen(1)
+    orders = _chunked_records(dep, "orders", 1)[0]
+    products = _chunked_records(dep, "products", 1)[0]
+    customers = _chunked_records(dep, "customers", 1)[0]
+    pipe = StreamingPipeline(dep.sql_executor(1, mcp_server=mcp))
+    got: list = []
+    # the advisor's arrival order: orders -> products -> customers
+    _replay(dep, "orders", orders)
+    got += pipe.advance()["price_match_results"]
+    _replay(dep, "products", products)
+    got += pipe.advance()["price_match_results"]
+    _replay(dep, "customers", customers)
+    got += pipe.advance()["price_match_results"]
+    ids = [r["order_id"] for r in got]
+    assert len(ids) == len(set(ids)) >= 1   # each order exactly once
+
+    # re-send ONE product record: each buffered order of that product
+    # must pair with the new copy exactly once (streaming-join semantics)
+    from quickstart_streaming_agents_amd.labs.schemas import ORDERS, PRODUCTS
+    from quickstart_streaming_agents_amd.wire.topics import AvroConsumer
+    first = AvroConsumer(dep.broker, "products", PRODUCTS).poll()[0][0]
+    decoded = [v for _, v in AvroConsumer(
+        dep.broker, "orders", ORDERS).poll()]
+    pid_rec = [v for _, v in AvroConsumer(
+        dep.broker, "products", PRODUCTS).poll()][0]
+    expect = sum(1 for o in decoded if o["product_id"] == pid_rec["product_id"])
+    dep.broker.topics["products"].append(
+        first.value, key=first.key, timestamp_ms=first.timestamp_ms,
+        partition=0)
+    new = pipe.advance()["price_match_results"]
+    assert len(new) == expect >= 1
+    new_ids = [r["order_id"] for r in new]
+    assert len(new_ids) == len(set(new_ids))
+
+
+def test_raw_topic_late_timestamp_append_delivered_once():
+    """Regression: the schema-less incremental source
+    sliced the timestamp-sorted read_all() by count, so a late-timestamp
+    append re-delivered one record and dropped the new one."""
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+    cat = Catalog()
+    cat.execute("CREATE TABLE sink AS SELECT r.id AS id FROM raw r;")
+    broker = Broker()
+    t = broker.create_topic("raw")
+    pipe = StreamingPipeline(SqlExecutor(cat, broker))
+    t.append({"id": 1}, timestamp_ms=100, partition=0)
+    t.append({"id": 2}, timestamp_ms=200, partition=0)
+    out = pipe.advance()["sink"]
+    assert [r["id"] for r in out] == [1, 2]
+    # late event-time record appended after processing ts=100,200
+    t.append({"id": 3}, timestamp_ms=50, partition=0)
+    out = pipe.advance()["sink"]
+    assert [r["id"] for r in out] == [3]
+    assert pipe.advance()["sink"] == []
+
+
+def test_state_ttl_bounds_join_buffers():
+    """`SET 'sql.state-ttl'` evicts join state older than the TTL behind
+    the statement's stream time (LAB1-Walkthrough.md:119-120): buffers
+    stay bounded on a long stream, and rows inside the TTL still join."""
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+
+    def build(ttl_stmt):
+        cat = Catalog()
+        cat.execute(f"""
+        {ttl_stmt}
+        CREATE TABLE ev (k STRING, ev_ts TIMESTAMP_LTZ(3));
+        CREATE TABLE dim (k STRING, payload STRING, up_ts TIMESTAMP_LTZ(3));
+        CREATE TABLE joined AS
+          SELECT e.k AS k, d.payload AS payload
+          FROM ev e JOIN dim d ON e.k = d.k;
+        """)
+        broker = Broker()
+        broker.create_topic("ev")
+        broker.create_topic("dim")
+        return broker, StreamingPipeline(SqlExecutor(cat, broker))
+
+    HOUR = 3_600_000
+    broker, pipe = build("SET 'sql.state-ttl' = '1 HOURS';")
+    # 50 keyed events spaced 10 min apart -> stream time spans >8 h
+    for i in range(50):
+        broker.topics["ev"].append(
+            {"k": f"k{i}", "ev_ts": i * 10 * 60_000}, partition=0)
+        pipe.advance()
+    stats = {s["sink"]: s for s in pipe.stats()}["joined"]
+    # only events within the last hour of stream time remain buffered
+    assert stats["join_buffered"] <= 7
+    assert stats["join_evicted"] >= 40
+    # a dim row arriving now joins live events, not evicted ones
+    broker.topics["dim"].append(
+        {"k": "k49", "payload": "live", "up_ts": 49 * 10 * 60_000},
+        partition=0)
+    broker.topics["dim"].append(
+        {"k": "k0", "payload": "stale", "up_ts": 49 * 10 * 60_000},
+        partition=0)
+    out = pipe.advance()["joined"]
+    assert [r["k"] for r in out] == ["k49"]
+
+    # without the SET, nothing is evicted (Flink's default: state forever)
+    broker2, pipe2 = build("")
+    for i in range(50):
+        broker2.topics["ev"].append(
+            {"k": f"k{i}", "ev_ts": i * 10 * 60_000}, partition=0)
+        pipe2.advance()
+    stats2 = {s["sink"]: s for s in pipe2.stats()}["joined"]
+    assert stats2["join_buffered"] == 50 and stats2["join_evicted"] == 0
