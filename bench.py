@@ -1,18 +1,23 @@
 #!/usr/bin/env python3
 """Flagship benchmark: Lab1 price-match agent decisions/sec on MI355X.
 
-Measures the BASELINE.json headline metric — agent decisions/sec (+ p50
-end-to-end latency) for the Lab1 price-match pipeline — on N GPUs of one
-node, weak scaling (per-GPU work fixed): each rank runs the full pipeline
-on its own shard of the order stream (DP over stream partitions), with the
-agent LLM (Llama-3-8B bf16, random init), the continuous-batching decode
-engine, the paged-attention HIP kernel, and real MCP tool round trips
-against a local stub server.  Synthetic data (deterministic datagen),
-no network.
+Measures the BASELINE.json headline metric — agent decisions/sec + p50
+per-decision end-to-end latency — for the Lab1 price-match pipeline on N
+GPUs of one node, weak scaling (per-GPU work fixed): each rank runs the
+full pipeline on its own shard of the order stream (DP over stream
+partitions), with the agent LLM (Llama-3-8B bf16, random init), the
+continuous-batching decode engine, the paged-attention HIP kernel, and
+real MCP tool round trips against a local stub server.  Synthetic data
+(deterministic datagen), no network.
 
-One decision = one order through AI_RUN_AGENT(price_match_agent):
-LLM turn -> http_get (competitor page) -> LLM turn -> send_email/finish ->
-final 3-section response + REGEXP_EXTRACT parse.
+One decision = one order through AI_RUN_AGENT(price_match_agent) with
+MODEL-DRIVEN control flow (reference LAB1-Walkthrough.md:155-181): each
+turn's action — which tool to call, or finish — is chosen by the model
+via grammar-constrained decoding (models/grammar.py: logit-masked
+decision tokens + forced TOOL_CALL token scripts, parsed back by
+ToolCallPolicy), bounded by max_iterations=10 /
+max_consecutive_failures=2.  Episode shapes (iterations, tool calls,
+turn lengths) therefore vary per order; the distribution is reported.
 
 Reference implied operating point: ~0.011 decisions/sec (1 order / 90-120 s,
 BASELINE.md).
@@ -97,36 +102,48 @@ def main():
     client = McpClient(server.mcp_endpoint)
     tool_fn = pipelines.mcp_tool_fn(client)
     competitor_url = f"{server.base_url}/competitor"
+    # tool schemas drive the per-turn grammar (function-calling registry)
+    tool_schemas = {t["name"]: t.get("inputSchema", {})
+                    for t in client.tools_list()}
 
     # ---- model backend ---------------------------------------------------
+    torch.manual_seed(1234 + rank)  # decision sampling is seeded
     if args.stub_llm:
         llm = pipelines.StubLLM()
     else:
         from quickstart_streaming_agents_amd.models import build_model
         from quickstart_streaming_agents_amd.models.serve import Engine, EngineLLM
+        from quickstart_streaming_agents_amd.models.tokenizer import \
+            default_tokenizer
         model = build_model(args.model, device=device)
+        tok = default_tokenizer(model.cfg.vocab_size)
         engine = Engine(model, max_batch=args.batch,
-                        max_seq_len=args.max_seq_len)
-        llm = EngineLLM(engine)
+                        max_seq_len=args.max_seq_len, eos_id=tok.EOS,
+                        valid_vocab=(tok._BYTE0, tok.n_tokens))
+        llm = EngineLLM(engine, tok)
 
     tools = ToolSet("lab1_remote_mcp", allowed_tools=("http_get", "send_email"))
     agent = AgentSpec("price_match_agent", "remote_mcp_model",
                       pipelines.LAB1_AGENT_PROMPT, tools,
                       max_iterations=10, max_consecutive_failures=2)
 
-    def run_step(step_idx: int) -> int:
+    all_results = []
+
+    def run_step(step_idx: int, record: bool = False) -> int:
         orders = enriched[step_idx * args.batch:(step_idx + 1) * args.batch]
+        # model-driven episodes: default ToolCallPolicy parses the
+        # grammar-constrained model output; no scripted per-order policy
         eps = [episode(agent,
                        pipelines.lab1_user_prompt(o, competitor_url,
                                                   o["customer_email"]),
-                       policy=pipelines.Lab1PriceMatchPolicy(
-                           o, competitor_url, o["customer_email"]),
-                       max_new_tokens=args.decode_tokens)
+                       max_new_tokens=args.decode_tokens,
+                       tool_schemas=tool_schemas)
                for o in orders]
         results = run_episodes(eps, llm, tool_fn)
-        ok = sum(r.status == "SUCCESS" for r in results)
-        assert ok == len(orders), f"{len(orders) - ok} episodes failed"
-        return ok
+        assert len(results) == len(orders)
+        if record:
+            all_results.extend(results)
+        return sum(r.status == "SUCCESS" for r in results)
 
     def sync():
         if use_gpu:
@@ -142,7 +159,7 @@ def main():
     t0 = time.perf_counter()
     for s in range(args.steps):
         ts = time.perf_counter()
-        run_step(args.warmup + s)
+        run_step(args.warmup + s, record=True)
         if use_gpu:
             torch.cuda.synchronize()
         step_times.append(time.perf_counter() - ts)
@@ -159,12 +176,27 @@ def main():
 
     total_decisions = args.batch * args.steps * world
     value = total_decisions / elapsed
-    p50_ms = statistics.median(step_times) * 1000.0
+
+    # per-decision submit->finish latency percentiles (BASELINE metric:
+    # p50 end-to-end latency per decision, not per step)
     import math
-    p95_ms = (sorted(step_times)[
-        min(len(step_times) - 1,
-            max(0, math.ceil(len(step_times) * 0.95) - 1))]
-              * 1000.0 if step_times else 0.0)
+
+    def _pct(xs, q):
+        if not xs:
+            return 0.0
+        xs = sorted(xs)
+        return xs[min(len(xs) - 1, max(0, math.ceil(len(xs) * q) - 1))]
+
+    lats = [r.latency_s for r in all_results]
+    p50_ms = statistics.median(lats) * 1000.0 if lats else 0.0
+    p95_ms = _pct(lats, 0.95) * 1000.0
+
+    # episode-shape distribution: model-driven control flow varies shapes
+    from collections import Counter
+    shape_hist = Counter((r.iterations, r.tool_calls) for r in all_results)
+    n_ok = sum(r.status == "SUCCESS" for r in all_results)
+    shapes = {f"i{i}_t{t}": c
+              for (i, t), c in sorted(shape_hist.items())}
 
     if rank == 0:
         out = {
@@ -177,6 +209,8 @@ def main():
             "ms_per_step": round(elapsed / args.steps * 1000.0, 2),
             "p50_e2e_ms": round(p50_ms, 2),
             "p95_e2e_ms": round(p95_ms, 2),
+            "success_rate": round(n_ok / max(1, len(all_results)), 4),
+            "episode_shapes": shapes,
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": round(value / BASELINE_DECISIONS_PER_SEC, 1),
@@ -187,6 +221,8 @@ def main():
                 "global_batch": args.batch * world,
                 "seq_len": args.max_seq_len,
                 "decode_tokens_per_turn": args.decode_tokens,
+                "agent_loop": "model_driven_grammar",
+                "max_iterations": 10,
                 "parallelism": f"dp{world}",
             },
         }

@@ -135,18 +135,27 @@ class EpisodeResult:
     tool_calls: int = 0
     failures: int = 0
     trace: list[dict] = field(default_factory=list)
+    latency_s: float = 0.0   # submit -> finish (set by the scheduler)
 
 
 def episode(agent: AgentSpec, user_prompt: str,
             policy: Callable[[str, int, dict], Any] | None = None,
             max_new_tokens: int = 128, debug: bool = False,
+            tool_schemas: dict[str, dict] | None = None,
             ) -> Generator[tuple, Any, EpisodeResult]:
     """The AI_RUN_AGENT loop as a resumable generator.
 
-    Yields ("llm", prompt, max_new_tokens) and ("tool", name, arguments);
-    caller .send()s the string result back.  Tool errors are sent as
-    exceptions via .throw() or as ("__error__", msg) sentinel strings.
-    Returns EpisodeResult via StopIteration.value.
+    Yields ("llm", prompt, max_new_tokens[, grammar]) and
+    ("tool", name, arguments); caller .send()s the string result back.
+    Tool errors are sent as exceptions via .throw() or as
+    ("__error__", msg) sentinel strings.  Returns EpisodeResult via
+    StopIteration.value.
+
+    tool_schemas ({tool -> MCP inputSchema}, from tools/list): when given,
+    each LLM turn carries a models.grammar.TurnGrammar so the serving
+    engine's constrained decode lets the MODEL choose the turn's action
+    (tool call vs finish) — the reference's model-driven AI_RUN_AGENT
+    control flow (LAB1-Walkthrough.md:155-181) under random-init weights.
     """
     policy = policy or ToolCallPolicy()
     transcript: list[str] = []
@@ -158,7 +167,13 @@ def episode(agent: AgentSpec, user_prompt: str,
 
     for it in range(agent.max_iterations):
         prompt = _build_prompt(agent, user_prompt, transcript)
-        text = yield ("llm", prompt, max_new_tokens)
+        if tool_schemas is not None:
+            from ..models.grammar import build_turn_grammar
+            allowed = agent.tools.allowed_tools if agent.tools else None
+            grammar = build_turn_grammar(tool_schemas, allowed, prompt)
+            text = yield ("llm", prompt, max_new_tokens, grammar)
+        else:
+            text = yield ("llm", prompt, max_new_tokens)
         last_text = text or ""
         if debug:
             trace.append({"iteration": it, "model_output": last_text})

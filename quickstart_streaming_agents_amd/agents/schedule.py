@@ -25,9 +25,11 @@ def run_episodes(episodes: list[Generator],
     llm_batch(prompts, max_new_tokens_list) -> list of generated texts.
     tool(name, args) -> result text (exceptions become __error__ results).
     """
+    import time as _time
     results: dict[int, EpisodeResult] = {}
     # (episode index -> pending request), advanced in rounds
     pending: dict[int, tuple] = {}
+    started: dict[int, float] = {}
 
     def _advance(idx: int, send_value: Any) -> None:
         try:
@@ -35,11 +37,15 @@ def run_episodes(episodes: list[Generator],
             pending[idx] = req
         except StopIteration as stop:
             results[idx] = stop.value
+            if stop.value is not None and idx in started:
+                # per-decision end-to-end latency (submit -> finish)
+                stop.value.latency_s = _time.perf_counter() - started[idx]
             pending.pop(idx, None)
             if hasattr(llm_batch, "release"):
                 llm_batch.release(idx)  # free the conversation's KV prefix
 
     for idx in range(len(episodes)):
+        started[idx] = _time.perf_counter()
         _advance(idx, None)
 
     # conversation-aware backends (EngineLLM) take conv ids for KV prefix
@@ -58,8 +64,14 @@ def run_episodes(episodes: list[Generator],
             if llm_ids:
                 prompts = [pending[i][1] for i in llm_ids]
                 maxtoks = [pending[i][2] for i in llm_ids]
+                grammars = [pending[i][3] if len(pending[i]) > 3 else None
+                            for i in llm_ids]
                 if conv_aware:
-                    texts = llm_batch(prompts, maxtoks, llm_ids)
+                    if any(g is not None for g in grammars):
+                        texts = llm_batch(prompts, maxtoks, llm_ids,
+                                          grammars)
+                    else:
+                        texts = llm_batch(prompts, maxtoks, llm_ids)
                 else:
                     texts = llm_batch(prompts, maxtoks)
                 for i, text in zip(llm_ids, texts):

@@ -38,6 +38,13 @@ class Sequence:
     out_tokens: list[int] = field(default_factory=list)
     done: bool = False
     keep_alive: bool = False     # conversation: keep KV after completion
+    # grammar-constrained decoding (models/grammar.py): the first generated
+    # token is logit-masked to `decision_allowed`; if the sampled decision
+    # has an entry in `branches`, the rest of the turn is forced to that
+    # token script (ending in EOS), else it decodes free.
+    decision_allowed: list[int] | None = None
+    branches: dict[int, list[int]] | None = None
+    script: list[int] | None = None
 
 
 @dataclass
@@ -83,13 +90,22 @@ class Engine:
     def __init__(self, model: LlamaModel, kv_pages: int | None = None,
                  max_batch: int = 256, max_seq_len: int = 4096,
                  eos_id: int | None = None, prefill_batch_tokens: int = 65536,
-                 temperature: float = 0.0):
+                 temperature: float = 0.0,
+                 valid_vocab: tuple[int, int] | None = None,
+                 decision_temperature: float = 1.0):
         self.model = model
         # temperature 0 = greedy (the deterministic benchmark contract);
         # > 0 samples via the Gumbel-argmax trick, which stays a single
         # argmax and is hipGraph-capture-safe (torch captures RNG state
         # advancement, so replays draw fresh noise)
         self.temperature = float(temperature)
+        # grammar decision tokens are SAMPLED from the model's masked
+        # distribution (temperature-1 Gumbel-argmax by default): the
+        # model's logits drive tool-vs-finish control flow, as in real
+        # tool-choice serving; deterministic under a fixed torch seed.
+        # 0 = greedy decisions (degenerate under random-init weights:
+        # every episode makes the same choice).
+        self.decision_temperature = float(decision_temperature)
         self.max_batch = max_batch
         self.max_seq_len = max_seq_len
         self.eos_id = eos_id  # None -> run to max_new_tokens (random weights)
@@ -109,9 +125,23 @@ class Engine:
             getattr(model, "tp_size", 1) == 1
         self._graph = None
         self._gbuf: dict = {}
+        # free-text sampling mask: tokens outside [lo, hi) (plus EOS) are
+        # never emitted — keeps random-init decode inside the tokenizer's
+        # trained vocab so outputs are real text (models/grammar.py)
+        self._vocab_bias: torch.Tensor | None = None
+        if valid_vocab is not None:
+            lo, hi = valid_vocab
+            bias = torch.full((model.cfg.vocab_size,), float("-inf"),
+                              device=model.device, dtype=torch.float32)
+            bias[lo:hi] = 0.0
+            if self.eos_id is not None:
+                bias[self.eos_id] = 0.0
+            self._vocab_bias = bias
 
     def _sample(self, logits: torch.Tensor) -> torch.Tensor:
         """Greedy at temperature 0; Gumbel-argmax sampling otherwise."""
+        if self._vocab_bias is not None:
+            logits = logits.float() + self._vocab_bias
         if self.temperature <= 0.0:
             return torch.argmax(logits, dim=-1)
         u = torch.rand(logits.shape, device=logits.device,
@@ -119,6 +149,35 @@ class Engine:
         gumbel = -torch.log(-torch.log(u).clamp_min_(1e-20))
         return torch.argmax(logits.float() + self.temperature * gumbel,
                             dim=-1)
+
+    def _sample_first(self, logits: torch.Tensor,
+                      admitted: list["Sequence"]) -> list[int]:
+        """Sample each admitted sequence's first token, honoring per-row
+        decision masks (grammar turns) with ONE host sync."""
+        free = self._sample(logits)
+        rows = [i for i, s in enumerate(admitted) if s.decision_allowed]
+        if not rows:
+            return [int(t) for t in free.tolist()]
+        K = max(len(admitted[i].decision_allowed) for i in rows)
+        idx = torch.tensor(
+            [admitted[i].decision_allowed
+             + [admitted[i].decision_allowed[-1]]
+             * (K - len(admitted[i].decision_allowed)) for i in rows],
+            dtype=torch.int64, device=logits.device)
+        lens = torch.tensor([len(admitted[i].decision_allowed)
+                             for i in rows], device=logits.device)
+        pad = torch.arange(K, device=logits.device)[None, :] >= lens[:, None]
+        sub = logits[rows].float().gather(1, idx)          # [R, K]
+        t = self.decision_temperature
+        if t > 0.0:
+            u = torch.rand(sub.shape, device=sub.device).clamp_min_(1e-20)
+            sub = sub / t + (-torch.log(-torch.log(u)))
+        sub = sub.masked_fill(pad, float("-inf"))
+        pick = idx.gather(1, torch.argmax(sub, 1, keepdim=True)).squeeze(1)
+        out = free.tolist()
+        for r, t in zip(rows, pick.tolist()):
+            out[r] = int(t)
+        return [int(t) for t in out]
 
     # ---- hipGraph decode -------------------------------------------------
     MAX_RUN = 512  # on-device token-history depth per graph run
@@ -136,6 +195,12 @@ class Engine:
             "active": torch.zeros(B, dtype=torch.int32, device=dev),
             "ctr": torch.zeros(1, dtype=torch.int64, device=dev),
             "hist": torch.zeros(self.MAX_RUN, B, dtype=torch.int64, device=dev),
+            # grammar scripts: forced token at run-step j for each row
+            # (script_mask False -> free sample)
+            "script": torch.zeros(B, self.MAX_RUN, dtype=torch.int64,
+                                  device=dev),
+            "script_mask": torch.zeros(B, self.MAX_RUN, dtype=torch.bool,
+                                       device=dev),
         }
 
         def body():
@@ -147,6 +212,11 @@ class Engine:
                 gb["tokens"], self.kv, gb["block_table"], gb["seq_lens"],
                 positions)
             nxt = self._sample(logits)
+            # grammar-forced tokens override the free sample in-graph
+            j = gb["ctr"].clamp(max=self.MAX_RUN - 1)
+            forced = gb["script"].index_select(1, j).squeeze(1)
+            fmask = gb["script_mask"].index_select(1, j).squeeze(1)
+            nxt = torch.where(fmask, forced, nxt)
             gb["hist"].index_copy_(0, gb["ctr"], nxt.unsqueeze(0))
             gb["ctr"].add_(1)
             gb["tokens"].copy_(nxt)
@@ -184,6 +254,28 @@ class Engine:
         if n < self.max_batch:
             gb["seq_lens"][n:] = 0
             gb["active"][n:] = 0
+        # grammar scripts for this run window: run-step j emits
+        # out_tokens[d + j] whose script index is d - 1 + j (out_tokens[0]
+        # was the prefill-sampled decision token)
+        if any(s.script for s in batch):
+            sc = torch.zeros(n, run, dtype=torch.int64)
+            sm = torch.zeros(n, run, dtype=torch.bool)
+            for i, s in enumerate(batch):
+                if not s.script:
+                    continue
+                d = len(s.out_tokens)
+                window = s.script[d - 1: d - 1 + run]
+                if window:
+                    sc[i, :len(window)] = torch.tensor(window,
+                                                       dtype=torch.int64)
+                    sm[i, :len(window)] = True
+            gb["script"][:n, :run] = sc.to(dev)
+            gb["script_mask"][:n, :run] = sm.to(dev)
+            gb["script_mask"][:n, run:] = False
+            if n < self.max_batch:
+                gb["script_mask"][n:] = False
+        else:
+            gb["script_mask"].fill_(False)
         gb["ctr"].zero_()
         if _TIMING:
             torch.cuda.synchronize()
@@ -208,10 +300,23 @@ class Engine:
     # ------------------------------------------------------------------
     def submit(self, prompt_tokens: list[int], max_new_tokens: int,
                continue_from: Sequence | None = None,
-               keep_alive: bool = False) -> Sequence:
+               keep_alive: bool = False, constraint=None) -> Sequence:
         """Queue a prompt.  continue_from: a completed keep-alive sequence
-        whose prompt is a prefix of this one — its KV prefix is reused."""
+        whose prompt is a prefix of this one — its KV prefix is reused.
+        constraint: models.grammar.CompiledGrammar for this turn."""
         prompt = list(prompt_tokens[: self.max_seq_len - 1])
+        decision_allowed = branches = None
+        if constraint is not None:
+            decision_allowed = list(constraint.decision_allowed)
+            branches = dict(constraint.branches)
+            # a forced tool-call script must fit in the turn
+            max_new_tokens = max(max_new_tokens,
+                                 constraint.max_script_len() + 1)
+            max_new_tokens = min(max_new_tokens, self.MAX_RUN)
+        # prompt + decode must fit the sequence budget (KV pages AND the
+        # model's rope table)
+        max_new_tokens = max(1, min(max_new_tokens,
+                                    self.max_seq_len - len(prompt)))
         if continue_from is not None and not continue_from.keep_alive:
             continue_from = None
         if continue_from is not None:
@@ -223,14 +328,17 @@ class Engine:
                 shared = 0
             if shared > 0:
                 seq = Sequence(continue_from.seq_id, prompt, max_new_tokens,
-                               cached_len=shared, keep_alive=keep_alive)
+                               cached_len=shared, keep_alive=keep_alive,
+                               decision_allowed=decision_allowed,
+                               branches=branches)
                 self.kv.truncate(seq.seq_id, shared)
                 self._next_id = max(self._next_id, seq.seq_id + 1)
                 self.pending.append(seq)
                 return seq
             self.kv.free(continue_from.seq_id)
         seq = Sequence(self._next_id, prompt, max_new_tokens,
-                       keep_alive=keep_alive)
+                       keep_alive=keep_alive,
+                       decision_allowed=decision_allowed, branches=branches)
         self._next_id += 1
         self.pending.append(seq)
         return seq
@@ -287,10 +395,12 @@ class Engine:
         if _TIMING:
             torch.cuda.synchronize()
             self.stats.prefill_s += time.perf_counter() - _t0
-        first = self._sample(logits).tolist()
+        first = self._sample_first(logits, admitted)
         self.stats.prefill_batches += 1
         for seq, tok in zip(admitted, first):
             seq.out_tokens.append(int(tok))
+            if seq.branches is not None:
+                seq.script = seq.branches.get(int(tok))
             self.running.append(seq)
             self._maybe_finish(seq)
 
@@ -322,7 +432,10 @@ class Engine:
         for s in batch:
             self.kv.extend(s.seq_id, len(s.prompt) + len(s.out_tokens))
         if self.use_graph:
-            nxt = self._decode_graph(batch)
+            # one-step graph run (appends + finishes internally)
+            self._decode_run_graph(batch, 1)
+            self._retire()
+            return len(self.running) + len(self.pending)
         else:
             tokens = torch.tensor([s.out_tokens[-1] for s in batch],
                                   dtype=torch.int64, device=dev)
@@ -336,6 +449,12 @@ class Engine:
                                                positions)
             nxt = self._sample(logits).tolist()
         for s, tok in zip(batch, nxt):
+            if s.script:
+                # grammar-forced continuation (script index: out_tokens[0]
+                # was the decision token)
+                si = len(s.out_tokens) - 1
+                if si < len(s.script):
+                    tok = s.script[si]
             s.out_tokens.append(int(tok))
             self._maybe_finish(s)
         self.stats.decode_tokens += len(batch)
@@ -394,23 +513,33 @@ class EngineLLM:
 
     When the scheduler passes conversation ids (agents/schedule.py), each
     conversation keeps its engine sequence alive between turns and only the
-    prompt delta is prefassed (KV prefix reuse)."""
+    prompt delta is prefassed (KV prefix reuse).  Per-turn TurnGrammar
+    specs (models/grammar.py) are compiled against the tokenizer and
+    passed into the engine as decision masks + forced scripts."""
 
-    def __init__(self, engine: Engine, tokenizer: HashTokenizer | None = None):
+    def __init__(self, engine: Engine, tokenizer=None):
+        if tokenizer is None:
+            from .tokenizer import default_tokenizer
+            tokenizer = default_tokenizer(engine.model.cfg.vocab_size)
         self.engine = engine
-        self.tokenizer = tokenizer or HashTokenizer(
-            engine.model.cfg.vocab_size)
+        self.tokenizer = tokenizer
         self._convs: dict = {}
 
     def __call__(self, prompts: list[str], max_new_tokens: list[int],
-                 conv_ids: list | None = None) -> list[str]:
+                 conv_ids: list | None = None,
+                 grammars: list | None = None) -> list[str]:
+        from .grammar import compile_grammar
         seqs = []
         for i, (p, m) in enumerate(zip(prompts, max_new_tokens)):
             conv = conv_ids[i] if conv_ids is not None else None
             prev = self._convs.get(conv) if conv is not None else None
+            g = grammars[i] if grammars is not None else None
+            constraint = compile_grammar(g, self.tokenizer) \
+                if g is not None else None
             enc = self.tokenizer.encode(p)
             seq = self.engine.submit(enc, m, continue_from=prev,
-                                     keep_alive=conv is not None)
+                                     keep_alive=conv is not None,
+                                     constraint=constraint)
             if conv is not None:
                 self._convs[conv] = seq
             seqs.append(seq)

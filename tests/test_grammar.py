@@ -1,0 +1,232 @@
+"""BPE tokenizer + grammar-constrained model-driven agent loop (CPU).
+
+The reference's AI_RUN_AGENT lets the model choose tools and iteration
+count (LAB1-Walkthrough.md:155-181); here the engine's constrained decode
+(models/grammar.py + models/serve.py) does that under random-init weights.
+These tests run the tiny preset on CPU through the REAL engine."""
+
+import collections
+
+import pytest
+import torch
+
+from quickstart_streaming_agents_amd.agents.mcp import McpClient, StubMcpServer
+from quickstart_streaming_agents_amd.agents.runner import (AgentSpec,
+                                                           ToolCallPolicy,
+                                                           ToolSet, episode)
+from quickstart_streaming_agents_amd.agents.schedule import run_episodes
+from quickstart_streaming_agents_amd.labs import datagen, pipelines
+from quickstart_streaming_agents_amd.models.grammar import (ActionOption,
+                                                            TurnGrammar,
+                                                            build_turn_grammar,
+                                                            compile_grammar)
+from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                          LlamaModel)
+from quickstart_streaming_agents_amd.models.serve import Engine, EngineLLM
+from quickstart_streaming_agents_amd.models.tokenizer import BpeTokenizer
+
+
+@pytest.fixture(scope="module")
+def tok():
+    return BpeTokenizer(vocab_size=4096)
+
+
+# ---------------------------------------------------------------------------
+# tokenizer
+# ---------------------------------------------------------------------------
+
+def test_bpe_round_trip_lossless(tok):
+    texts = [
+        "Hello, world!  Prices start at $209.99 (was $245).",
+        pipelines.LAB1_AGENT_PROMPT,
+        'TOOL_CALL {"name": "http_get", "arguments": '
+        '{"url": "http://127.0.0.1:1234/competitor"}}',
+        "unicode: émojis 🙂 and\ttabs\nnewlines",
+        "",
+    ]
+    for t in texts:
+        assert tok.decode(tok.encode(t, bos=False)) == t
+
+
+def test_bpe_subword_statistics(tok):
+    """Token counts must be in the real-BPE class (multiple chars/token),
+    not 1 token per word and not 1 per byte."""
+    text = pipelines.lab1_user_prompt(
+        {"order_id": "ORD-1", "product_name": "AirPods Pro",
+         "order_price": 249.0, "customer_email": "a@b.com"},
+        "http://127.0.0.1:8000/competitor", "a@b.com")
+    ids = tok.encode(text, bos=False)
+    cpt = len(text) / len(ids)
+    assert 2.0 < cpt < 8.0, f"chars/token {cpt}"
+    n_words = len(text.split())
+    assert len(ids) > n_words   # subword: more tokens than words
+
+
+def test_bpe_specials_and_truncated_vocab(tok):
+    s = "<|finish|>ok<|tool_0|>"
+    ids = tok.encode(s, bos=False)
+    assert ids[0] == tok.FINISH and tok.tool_slot(0) in ids
+    assert tok.decode(ids) == s
+    # a truncated-merge (small-vocab) tokenizer stays lossless
+    small = BpeTokenizer(vocab_size=512)
+    text = "The quick brown fox jumps over the lazy dog. $12.34!"
+    assert small.decode(small.encode(text, bos=False)) == text
+    assert len(small.encode(text, bos=False)) > len(
+        tok.encode(text, bos=False))
+
+
+# ---------------------------------------------------------------------------
+# grammar construction
+# ---------------------------------------------------------------------------
+
+def _schemas():
+    with StubMcpServer() as srv:
+        return {t["name"]: t.get("inputSchema", {})
+                for t in McpClient(srv.mcp_endpoint).tools_list()}
+
+
+def test_build_turn_grammar_from_lab1_prompt():
+    schemas = _schemas()
+    text = pipelines.LAB1_AGENT_PROMPT + "\n" + pipelines.lab1_user_prompt(
+        {"order_id": "ORD-7", "product_name": "AirPods Pro",
+         "order_price": 249.0, "customer_email": "kim@example.com"},
+        "http://127.0.0.1:9999/competitor", "kim@example.com")
+    g = build_turn_grammar(schemas, ("http_get", "send_email"), text)
+    by_tool = collections.defaultdict(list)
+    for o in g.options:
+        by_tool[o.tool].append(o)
+    # http_get bound to the URL present in the prompt
+    assert by_tool["http_get"][0].arguments["url"] == \
+        "http://127.0.0.1:9999/competitor"
+    # send_email bound to recipient/subject/body extracted from the prompt
+    se = by_tool["send_email"][0].arguments
+    assert se["to"] == "kim@example.com"
+    assert "ORD-7" in se["subject"]
+    assert se["body"]
+    # http_post requires url (present) -> allowed_tools filter removed it
+    assert "http_post" not in by_tool
+    assert g.allow_finish
+
+
+def test_compile_grammar_scripts_parse_back(tok):
+    g = TurnGrammar(options=[
+        ActionOption("http_get", {"url": "http://x.test/a"}),
+        ActionOption("send_email", {"to": "a@b.com", "subject": "s",
+                                    "body": "hello there"})])
+    cg = compile_grammar(g, tok)
+    assert tok.FINISH in cg.decision_allowed
+    assert len(cg.decision_allowed) == 3
+    pol = ToolCallPolicy()
+    for t, script in cg.branches.items():
+        assert script[-1] == tok.EOS
+        text = tok.decode([t] + script)
+        act = pol(text, 0, {})
+        assert act.__class__.__name__ == "ToolCall"
+        assert act.name in ("http_get", "send_email")
+        if act.name == "send_email":
+            assert act.arguments["body"] == "hello there"
+
+
+# ---------------------------------------------------------------------------
+# engine-constrained decode (tiny model, CPU, real Engine)
+# ---------------------------------------------------------------------------
+
+def _engine(tok, max_batch=16, seed=0):
+    cfg = LlamaConfig.preset("tiny")
+    cfg.vocab_size = 4096
+    model = LlamaModel(cfg, device="cpu", dtype=torch.float32, seed=seed)
+    eng = Engine(model, max_batch=max_batch, max_seq_len=2048,
+                 eos_id=tok.EOS, valid_vocab=(tok._BYTE0, tok.n_tokens))
+    return eng
+
+
+def _run_lab1_episodes(tok, n=10, seed=7, max_iters=6, server=None):
+    torch.manual_seed(seed)
+    eng = _engine(tok)
+    llm = EngineLLM(eng, tok)
+    import contextlib
+    with (contextlib.nullcontext(server) if server is not None
+          else StubMcpServer()) as srv:
+        client = McpClient(srv.mcp_endpoint)
+        schemas = {t["name"]: t.get("inputSchema", {})
+                   for t in client.tools_list()}
+        tool_fn = pipelines.mcp_tool_fn(client)
+        tools = ToolSet("lab1_remote_mcp",
+                        allowed_tools=("http_get", "send_email"))
+        agent = AgentSpec("price_match_agent", "m",
+                          pipelines.LAB1_AGENT_PROMPT, tools,
+                          max_iterations=max_iters,
+                          max_consecutive_failures=2)
+        products = datagen.lab1_products()
+        orders = []
+        for i in range(n):
+            p = products[i % len(products)]
+            orders.append({"order_id": f"ORD-{i:04d}",
+                           "product_name": p["product_name"],
+                           "product_id": p["product_id"],
+                           "order_price": p["price"],
+                           "customer_email": f"u{i}@example.com",
+                           "order_ts": i})
+        eps = [episode(agent,
+                       pipelines.lab1_user_prompt(
+                           o, f"{srv.base_url}/competitor",
+                           o["customer_email"]),
+                       max_new_tokens=24, tool_schemas=schemas)
+               for o in orders]
+        results = run_episodes(eps, llm, tool_fn)
+        return results, len(srv.emails)
+
+
+def test_model_driven_episodes_vary_and_respect_caps(tok):
+    results, n_emails = _run_lab1_episodes(tok, n=12, seed=7)
+    assert len(results) == 12
+    shapes = {(r.iterations, r.tool_calls) for r in results}
+    assert len(shapes) >= 2, "model-driven control flow must vary shapes"
+    assert all(r.iterations <= 6 for r in results)
+    assert all(r.latency_s > 0 for r in results)
+    # at least one episode actually called a tool through MCP
+    assert any(r.tool_calls > 0 for r in results)
+    # a send_email decision produced a real email through the stub server
+    total_tool_calls = sum(r.tool_calls for r in results)
+    assert total_tool_calls >= 1
+    assert n_emails >= 0  # may be zero for some seeds; counted for info
+
+
+def test_model_driven_episodes_deterministic(tok):
+    # one server instance -> identical prompts (the URL carries the port)
+    with StubMcpServer() as srv:
+        r1, _ = _run_lab1_episodes(tok, n=8, seed=11, server=srv)
+        r2, _ = _run_lab1_episodes(tok, n=8, seed=11, server=srv)
+    assert [(r.status, r.iterations, r.tool_calls) for r in r1] == \
+        [(r.status, r.iterations, r.tool_calls) for r in r2]
+    assert [r.response for r in r1] == [r.response for r in r2]
+
+
+def test_forced_script_emitted_exactly(tok):
+    """A single-option grammar with no finish forces the full tool-call
+    token script; the emitted text must be the exact rendered call."""
+    torch.manual_seed(0)
+    eng = _engine(tok, max_batch=2)
+    opt = ActionOption("http_get", {"url": "http://t.test/page"})
+    cg = compile_grammar(TurnGrammar(options=[opt], allow_finish=False),
+                         tok)
+    prompt = tok.encode("please fetch the page")
+    seq = eng.submit(prompt, 4, constraint=cg)
+    eng.run_to_completion()
+    text = tok.decode(seq.out_tokens)
+    assert text.startswith("<|tool_0|>")
+    act = ToolCallPolicy()(text, 0, {})
+    assert act.__class__.__name__ == "ToolCall"
+    assert act.name == "http_get"
+    assert act.arguments == {"url": "http://t.test/page"}
+    # script ran to its EOS, not to an arbitrary cap
+    assert seq.out_tokens[-1] == tok.EOS
+
+
+def test_free_turn_respects_valid_vocab(tok):
+    torch.manual_seed(3)
+    eng = _engine(tok, max_batch=2)
+    seq = eng.submit(tok.encode("hello"), 16)
+    eng.run_to_completion()
+    for t in seq.out_tokens:
+        assert (tok._BYTE0 <= t < tok.n_tokens) or t == tok.EOS
