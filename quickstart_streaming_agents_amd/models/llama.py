@@ -158,22 +158,37 @@ class LlamaModel:
         self.rope_cos = ang.cos().float().to(device)
         self.rope_sin = ang.sin().float().to(device)
         self.scale = 1.0 / (c.d_head ** 0.5)
-        # decode-path weight packing: fragment-major copies for the
-        # weight-streaming skinny GEMM (288 GB HBM affords the duplicate)
+        # decode-path weight packing (288 GB HBM affords the duplicates):
+        # fp8 (e4m3 per-channel, W8A16) stream copies by default — the
+        # decode projections are weight-bandwidth-bound, so fp8 halves the
+        # stream (ops/hip/skinny_gemm_fp8.hip); QSA_FP8=0 falls back to
+        # the bf16 fragment copies for the small shapes.
+        import os as _os
+        self.use_fp8 = _os.environ.get("QSA_FP8", "1") == "1"
         self.lm_head_f = None
+        self.lm_head_f8 = None
         if str(device).startswith("cuda") and dtype == torch.bfloat16:
-            def want(shape):
-                # only shapes where skinny beats rocBLAS (kernel_bench):
-                # duplicating the big-N weights buys nothing
-                return D.can_pack_weight(*shape) and \
-                    shape[0] * shape[1] <= 6144 * 4096
-            for L in self.layers:
-                for key in ("wqkv", "wo", "wgu", "wdown"):
-                    wt = L.get(key)
-                    if wt is not None and want(wt.shape):
-                        L[key + "_f"] = D.pack_weight_frag(wt)
-            if want(self.lm_head.shape):
-                self.lm_head_f = D.pack_weight_frag(self.lm_head)
+            if self.use_fp8:
+                for L in self.layers:
+                    for key in ("wqkv", "wo", "wgu", "wdown"):
+                        wt = L.get(key)
+                        if wt is not None and D.can_pack_weight(*wt.shape):
+                            L[key + "_f8"] = D.pack_weight_fp8(wt)
+                if D.can_pack_weight(*self.lm_head.shape):
+                    self.lm_head_f8 = D.pack_weight_fp8(self.lm_head)
+            else:
+                def want(shape):
+                    # only shapes where bf16 skinny beats rocBLAS
+                    # (kernel_bench): duplicating big-N weights buys nothing
+                    return D.can_pack_weight(*shape) and \
+                        shape[0] * shape[1] <= 6144 * 4096
+                for L in self.layers:
+                    for key in ("wqkv", "wo", "wgu", "wdown"):
+                        wt = L.get(key)
+                        if wt is not None and want(wt.shape):
+                            L[key + "_f"] = D.pack_weight_frag(wt)
+                if want(self.lm_head.shape):
+                    self.lm_head_f = D.pack_weight_frag(self.lm_head)
 
     def _ffn_weights(self, w, rows, cols) -> dict:
         """Per-layer FFN weights; the generator order is part of the
@@ -193,15 +208,19 @@ class LlamaModel:
                             n_pages, self.device, self.dtype)
 
     def _linear(self, x: torch.Tensor, w: torch.Tensor,
-                wf: torch.Tensor | None = None) -> torch.Tensor:
+                wf: torch.Tensor | None = None,
+                wf8: tuple | None = None) -> torch.Tensor:
         """Projection: decode-sized cuda batches go through the
-        weight-streaming skinny-GEMM kernel, everything else rocBLAS."""
+        weight-streaming skinny GEMM — fp8 (W8A16, half the weight
+        stream) when packed, else bf16; everything else rocBLAS."""
         n, k = w.shape
-        if wf is not None and x.is_cuda and x.shape[0] <= 32 and \
-                n * k <= 6144 * 4096:
-            # skinny wins the small decode shapes (qkv/wo); rocBLAS is at
-            # the bandwidth floor for the big-N ones (kernel_bench sweep)
-            return D.skinny_linear(x, wf, n, k)
+        if x.is_cuda and x.shape[0] <= 32:
+            if wf8 is not None:
+                return D.skinny_linear_fp8(x, wf8[0], wf8[1], n, k)
+            if wf is not None and n * k <= 6144 * 4096:
+                # bf16 skinny wins only the small decode shapes (qkv/wo);
+                # rocBLAS is at the bandwidth floor for the big-N ones
+                return D.skinny_linear(x, wf, n, k)
         return F.linear(x, w)
 
     def _tp_all_reduce(self, t: torch.Tensor) -> torch.Tensor:
@@ -239,7 +258,8 @@ class LlamaModel:
             else:
                 h = D.rmsnorm_residual(mlp_out, res, L["attn_norm"],
                                        c.norm_eps)
-            qkv = self._linear(h, L["wqkv"], L.get("wqkv_f"))
+            qkv = self._linear(h, L["wqkv"], L.get("wqkv_f"),
+                               L.get("wqkv_f8"))
             if "bqkv" in L:
                 qkv = qkv + L["bqkv"]
             # strided [B, H, D] views straight into the fused qkv buffer —
@@ -254,19 +274,23 @@ class LlamaModel:
             attn = D.paged_attn_decode(q, kv.k[li], kv.v[li], block_table,
                                        seq_lens, self.scale)
             o = self._tp_all_reduce(
-                self._linear(attn.view(B, -1), L["wo"], L.get("wo_f")))
+                self._linear(attn.view(B, -1), L["wo"], L.get("wo_f"),
+                             L.get("wo_f8")))
             h = D.rmsnorm_residual(o, res, L["mlp_norm"], c.norm_eps)
             mlp_out = self._ffn(L, h)
         final_h = D.rmsnorm_residual(mlp_out, res, self.final_norm, c.norm_eps)
-        return self._linear(final_h, self.lm_head, self.lm_head_f)
+        return self._linear(final_h, self.lm_head, self.lm_head_f,
+                            self.lm_head_f8)
 
     def _ffn(self, L: dict, h: torch.Tensor) -> torch.Tensor:
         """Dense SwiGLU FFN (TP row/col-parallel). Mixtral overrides with
         the routed MoE (models/mixtral.py)."""
-        gu = self._linear(h, L["wgu"], L.get("wgu_f"))
+        gu = self._linear(h, L["wgu"], L.get("wgu_f"),
+                          L.get("wgu_f8"))
         act = D.swiglu(gu[:, :self.ffn_local], gu[:, self.ffn_local:])
         return self._tp_all_reduce(
-            self._linear(act, L["wdown"], L.get("wdown_f")))
+            self._linear(act, L["wdown"], L.get("wdown_f"),
+                         L.get("wdown_f8")))
 
     def _gather_kv(self, kc_l: torch.Tensor, vc_l: torch.Tensor,
                    pages: list[int], n: int):
@@ -411,7 +435,7 @@ class LlamaModel:
         last = torch.tensor([offs[i] + lens[i] - 1 for i in range(len(items))],
                             dtype=torch.int64, device=dev)
         return self._linear(final_h.index_select(0, last), self.lm_head,
-                            self.lm_head_f)
+                            self.lm_head_f, self.lm_head_f8)
 
     @torch.no_grad()
     def forward_prefill(self, tokens: torch.Tensor, kv: PagedKVCache,

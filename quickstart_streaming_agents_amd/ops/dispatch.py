@@ -183,6 +183,39 @@ def skinny_linear(x: torch.Tensor, wf: torch.Tensor, n: int,
     return (x.float() @ w.float().T).to(x.dtype)
 
 
+# ---- fp8 weight-only decode GEMM (W8A16, per-channel scales) -------------
+
+def pack_weight_fp8(w: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
+    """[N,K] bf16 -> (fp8-e4m3 fragment-pair-major stream as uint8,
+    per-channel f32 scales).  Layout doc: ops/hip/skinny_gemm_fp8.hip."""
+    n, k = w.shape
+    assert n % 16 == 0 and k % 256 == 0
+    wf = w.float()
+    s = wf.abs().amax(dim=1).clamp(min=1e-12) / 448.0
+    q = (wf / s[:, None]).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    qf = q.view(torch.uint8) \
+        .reshape(n // 16, 16, k // 64, 2, 4, 8) \
+        .permute(0, 2, 1, 4, 3, 5).contiguous()
+    return qf, s.float().contiguous()
+
+
+def unpack_weight_fp8(qf: torch.Tensor, scale: torch.Tensor, n: int,
+                      k: int) -> torch.Tensor:
+    """Inverse of pack_weight_fp8 -> dequantized f32 [N,K] (reference)."""
+    q = qf.reshape(n // 16, k // 64, 16, 4, 2, 8) \
+        .permute(0, 2, 1, 4, 3, 5).reshape(n, k)
+    return q.view(torch.float8_e4m3fn).float() * scale[:, None].float()
+
+
+def skinny_linear_fp8(x: torch.Tensor, qf: torch.Tensor,
+                      scale: torch.Tensor, n: int, k: int) -> torch.Tensor:
+    """x[M,K] bf16 @ dequant(Q)^T via the fp8 weight-streaming kernel."""
+    if x.is_cuda:
+        return ext().skinny_gemm_fp8(x, qf, scale, n, k)
+    w = unpack_weight_fp8(qf, scale, n, k)
+    return (x.float() @ w.T).to(x.dtype)
+
+
 # ---- retrieval / streaming ------------------------------------------------
 
 def topk_cosine(queries: torch.Tensor, docs: torch.Tensor, k: int):

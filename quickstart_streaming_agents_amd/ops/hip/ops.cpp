@@ -73,6 +73,14 @@ extern "C" void qsa_skinny_gemm_probe_launch(const unsigned short*,
                                              unsigned short*, int, int,
                                              long long, long long, int, int,
                                              int, hipStream_t);
+extern "C" void qsa_skinny_gemm_fp8_launch(const unsigned short*,
+                                           const unsigned char*,
+                                           const float*, unsigned short*,
+                                           int, int, long long, long long,
+                                           hipStream_t);
+extern "C" void qsa_skinny_gemm_fp8_probe_launch(
+    const unsigned short*, const unsigned char*, const float*,
+    unsigned short*, int, int, long long, long long, int, int, hipStream_t);
 extern "C" void qsa_topk_launch(const float*, const float*, float*, int*,
                                 float*, int*, int, int, int, int, int,
                                 hipStream_t);
@@ -321,6 +329,41 @@ torch::Tensor skinny_gemm_probe(torch::Tensor a, torch::Tensor wf, long N,
   return out;
 }
 
+torch::Tensor skinny_gemm_fp8(torch::Tensor a, torch::Tensor qf,
+                              torch::Tensor scale, long N, long K) {
+  // C[M,N] = a @ (scale[N] * dequant(Q))^T; Q pre-packed fp8 e4m3 in the
+  // fragment-pair-major stream layout (skinny_gemm_fp8.hip header).
+  CHK_DEV(a); CHK_BF16(a); CHK_CONT(qf);
+  TORCH_CHECK(qf.scalar_type() == torch::kUInt8, "qf must be uint8 (fp8)");
+  TORCH_CHECK(scale.scalar_type() == torch::kFloat32 && scale.is_cuda() &&
+                  scale.is_contiguous() && scale.numel() == N,
+              "scale must be f32 [N] on device");
+  TORCH_CHECK(a.dim() == 2 && a.stride(1) == 1, "a rows must be contiguous");
+  const int M = a.size(0);
+  TORCH_CHECK(M >= 1 && M <= 32, "skinny_gemm_fp8: M in [1,32]");
+  TORCH_CHECK(a.size(1) == K, "K mismatch");
+  TORCH_CHECK(K % 256 == 0 && N % 16 == 0, "K%256==0, N%16==0");
+  TORCH_CHECK(qf.numel() == (long long)N * K, "qf size");
+  auto out = torch::empty({M, (long long)N}, a.options());
+  qsa_skinny_gemm_fp8_launch(u16(a), qf.data_ptr<unsigned char>(),
+                             scale.data_ptr<float>(), u16m(out), M, (int)N,
+                             K, a.stride(0), cur_stream());
+  return out;
+}
+
+torch::Tensor skinny_gemm_fp8_probe(torch::Tensor a, torch::Tensor qf,
+                                    torch::Tensor scale, long N, long K,
+                                    long waves, long nt) {
+  CHK_DEV(a); CHK_BF16(a); CHK_CONT(qf);
+  const int M = a.size(0);
+  auto out = torch::empty({M, (long long)N}, a.options());
+  qsa_skinny_gemm_fp8_probe_launch(u16(a), qf.data_ptr<unsigned char>(),
+                                   scale.data_ptr<float>(), u16m(out), M,
+                                   (int)N, K, a.stride(0), (int)waves,
+                                   (int)nt, cur_stream());
+  return out;
+}
+
 std::vector<torch::Tensor> topk_cosine(torch::Tensor queries,
                                        torch::Tensor docs, long k) {
   CHK_DEV(queries); CHK_CONT(queries); CHK_F32(queries); CHK_F32(docs);
@@ -404,6 +447,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "decode-batch GEMM (M<=32) on the packed weight stream");
   m.def("skinny_gemm_probe", &skinny_gemm_probe,
         "ablation probe: waves/nt/variant sweep");
+  m.def("skinny_gemm_fp8", &skinny_gemm_fp8,
+        "fp8-weight decode GEMM (M<=32): half the weight stream");
+  m.def("skinny_gemm_fp8_probe", &skinny_gemm_fp8_probe,
+        "fp8 ablation probe: waves/nt sweep");
   m.def("topk_cosine", &topk_cosine, "exact cosine top-k over the HBM index");
   m.def("window_agg", &window_agg, "segmented (key, window) count/sum");
   m.def("anomaly_batch", &anomaly_batch, "batched AR+ridge anomaly scorer");

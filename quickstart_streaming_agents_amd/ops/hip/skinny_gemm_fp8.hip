@@ -1,0 +1,165 @@
+// FP8 (OCP e4m3) weight-streaming decode GEMM for CDNA4 (gfx950):
+// C[M,N] = A[M,K] @ (s[N] * Q[N,K])^T, M <= 32, A bf16, Q fp8, C bf16.
+// Serves the K4 agent-LLM decode projections.
+//
+// The bf16 skinny kernel (skinny_gemm.hip) is HBM-bound on the WEIGHT
+// stream (PMC: ~3% MFMA util); its ceiling is W bytes / 6.3 TB/s.  FP8
+// weights HALVE that stream: per-output-channel symmetric quantization
+// (s[n] = max|W[n,:]| / 448, the standard W8A16 weight-only scheme) keeps
+// activations bf16 and applies the scale once in the epilogue — zero
+// inner-loop cost.  Dequant in-kernel:
+//   v_cvt_pk_f32_fp8 (2 fp8 -> 2 f32, exact)
+//   v_perm_b32       (pack the two f32 high halves -> 2 bf16)
+// The f32 -> bf16 TRUNCATION is exact: e4m3 has 3 mantissa bits, bf16
+// has 8, and every e4m3 value (normals and denormals) is exactly
+// representable in bf16 — the dequantized operand is bit-exact, so the
+// MFMA math matches a bf16 kernel running on the dequantized weights.
+//
+// Pack layout [N/16, K/64, 16, 4, 2, 8] ("fragment-pair-major"): one lane
+// reads ONE 16-byte vector covering its 8 fp8 for TWO consecutive
+// 16x16x32 k-fragments -> full-width dwordx4 loads on a byte stream.
+// Gfx950 also has native fp8 MFMA, but using it would force the
+// ACTIVATIONS to fp8 too (non-scaled fp8 MFMA is A8W8); this kernel is
+// bandwidth-bound, so bf16 MFMA at half the weight bytes is the same
+// speed with better numerics.
+#include "common.h"
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x2v = __attribute__((ext_vector_type(2))) float;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+using u32x4 = __attribute__((ext_vector_type(4))) unsigned int;
+
+#define QSA_KCH8 256  // k per wave-iteration: 4 loads, 8 MFMA k-steps
+
+// 8 fp8 bytes (as 2 dwords) -> bf16x8 fragment, exact.
+__device__ __forceinline__ bf16x8 fp8x8_to_bf16x8(unsigned int a,
+                                                  unsigned int b) {
+  f32x2v f01 = __builtin_amdgcn_cvt_pk_f32_fp8(a, false);
+  f32x2v f23 = __builtin_amdgcn_cvt_pk_f32_fp8(a, true);
+  f32x2v f45 = __builtin_amdgcn_cvt_pk_f32_fp8(b, false);
+  f32x2v f67 = __builtin_amdgcn_cvt_pk_f32_fp8(b, true);
+  union { unsigned int u; float f; } u0, u1;
+  unsigned int p[4];
+  u0.f = f01.x; u1.f = f01.y;
+  p[0] = __builtin_amdgcn_perm(u1.u, u0.u, 0x07060302u);
+  u0.f = f23.x; u1.f = f23.y;
+  p[1] = __builtin_amdgcn_perm(u1.u, u0.u, 0x07060302u);
+  u0.f = f45.x; u1.f = f45.y;
+  p[2] = __builtin_amdgcn_perm(u1.u, u0.u, 0x07060302u);
+  u0.f = f67.x; u1.f = f67.y;
+  p[3] = __builtin_amdgcn_perm(u1.u, u0.u, 0x07060302u);
+  union { unsigned int u[4]; bf16x8 v; } out;
+  out.u[0] = p[0]; out.u[1] = p[1]; out.u[2] = p[2]; out.u[3] = p[3];
+  return out.v;
+}
+
+template <int WAVES, bool NT>
+__global__ void __launch_bounds__(WAVES * 64)
+qsa_skinny_gemm_fp8_t(const unsigned short* __restrict__ A,  // [M,K] bf16
+                      const u32x4* __restrict__ Qf,   // packed fp8 stream
+                      const float* __restrict__ scale,     // [N] f32
+                      unsigned short* __restrict__ Cbf,    // [M, N]
+                      int M, int N, long long K, long long lda) {
+  const int nt = blockIdx.x;              // 16-col n-tile
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const long long kchunks = K / QSA_KCH8;  // K % 256 == 0
+
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+
+  // A rows clamp out-of-batch to M-1 (store is guarded; see bf16 kernel)
+  const int arow = lane & 15;
+  const int akoff = (lane >> 4) * 8;
+  const int r0 = min(arow, M - 1);
+  const int r1 = min(16 + arow, M - 1);
+  const unsigned short* a0base = A + (long long)r0 * lda + akoff;
+  const unsigned short* a1base = A + (long long)r1 * lda + akoff;
+
+  // Q stream: 1024-B block per (nt, kb=64k); this lane's 16 B at byte
+  // (lane&15)*64 + (lane>>4)*16 inside the block (uint4 units below).
+  const u32x4* qbase =
+      Qf + ((long long)nt * (K >> 6)) * 64 +
+      (long long)((lane & 15) * 4 + (lane >> 4));
+
+  for (long long c = wave; c < kchunks; c += WAVES) {
+    const long long k0 = c * QSA_KCH8;
+#pragma unroll
+    for (int s = 0; s < 4; ++s) {          // 4 loads x 64 k = 256 k
+      const long long kb = (k0 >> 6) + s;
+      u32x4 q;
+      if (NT) {
+        q = __builtin_nontemporal_load(qbase + kb * 64);
+      } else {
+        q = *reinterpret_cast<const u32x4*>(
+            __builtin_assume_aligned(qbase + kb * 64, 16));
+      }
+      const bf16x8 w0 = fp8x8_to_bf16x8(q.x, q.y);   // k .. +32
+      const bf16x8 w1 = fp8x8_to_bf16x8(q.z, q.w);   // k+32 .. +64
+      const long long ak = k0 + s * 64;
+      const bf16x8 a00 = *reinterpret_cast<const bf16x8*>(a0base + ak);
+      const bf16x8 a10 = *reinterpret_cast<const bf16x8*>(a1base + ak);
+      const bf16x8 a01 = *reinterpret_cast<const bf16x8*>(a0base + ak + 32);
+      const bf16x8 a11 = *reinterpret_cast<const bf16x8*>(a1base + ak + 32);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a00, w0, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a10, w0, acc1, 0, 0, 0);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a01, w1, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a11, w1, acc1, 0, 0, 0);
+    }
+  }
+
+  // ---- cross-wave K-reduction in LDS + per-channel scale ---------------
+  __shared__ float red[WAVES][64][8];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    red[wave][lane][r] = acc0[r];
+    red[wave][lane][4 + r] = acc1[r];
+  }
+  __syncthreads();
+  if (wave == 0) {
+    const int ncol = nt * 16 + (lane & 15);
+    const int mrow = (lane >> 4) * 4;
+    const float s = scale[ncol];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float v0 = 0.f, v1 = 0.f;
+#pragma unroll
+      for (int wv = 0; wv < WAVES; ++wv) {
+        v0 += red[wv][lane][r];
+        v1 += red[wv][lane][4 + r];
+      }
+      const int m0 = mrow + r;
+      if (m0 < M) Cbf[(long long)m0 * N + ncol] = f32_to_bf16(v0 * s);
+      if (16 + m0 < M)
+        Cbf[(long long)(16 + m0) * N + ncol] = f32_to_bf16(v1 * s);
+    }
+  }
+}
+
+extern "C" void qsa_skinny_gemm_fp8_launch(
+    const unsigned short* A, const unsigned char* Qf, const float* scale,
+    unsigned short* Cbf, int M, int N, long long K, long long lda,
+    hipStream_t stream) {
+  hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<8, false>), dim3(N / 16),
+                     dim3(512), 0, stream, A,
+                     reinterpret_cast<const u32x4*>(Qf), scale, Cbf, M, N,
+                     K, lda);
+}
+
+extern "C" void qsa_skinny_gemm_fp8_probe_launch(
+    const unsigned short* A, const unsigned char* Qf, const float* scale,
+    unsigned short* Cbf, int M, int N, long long K, long long lda,
+    int waves, int nt, hipStream_t stream) {
+  dim3 grid(N / 16);
+#define QSA_CASE(W, NTB)                                                  \
+  if (waves == W && nt == (int)NTB) {                                     \
+    hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<W, NTB>), grid,             \
+                       dim3(W * 64), 0, stream, A,                        \
+                       reinterpret_cast<const u32x4*>(Qf), scale, Cbf,   \
+                       M, N, K, lda);                                     \
+    return;                                                               \
+  }
+  QSA_CASE(8, false) QSA_CASE(8, true) QSA_CASE(4, false) QSA_CASE(4, true)
+  QSA_CASE(2, false) QSA_CASE(2, true)
+#undef QSA_CASE
+}

@@ -24,6 +24,7 @@ sources = [
     os.path.join(HIP_DIR, "elementwise.hip"),
     os.path.join(HIP_DIR, "paged_attn.hip"),
     os.path.join(HIP_DIR, "skinny_gemm.hip"),
+    os.path.join(HIP_DIR, "skinny_gemm_fp8.hip"),
     os.path.join(HIP_DIR, "topk_cosine.hip"),
     os.path.join(HIP_DIR, "streaming.hip"),
 ]
@@ -38,7 +39,7 @@ setup(
                  "hand-written CDNA4 HIP kernels, RCCL over xGMI"),
     packages=find_packages(include=["quickstart_streaming_agents_amd*"]),
     package_data={"quickstart_streaming_agents_amd": [
-        "labs/sql/*.sql", "data/*.csv"]},
+        "labs/sql/*.sql", "data/*.csv", "data/*.json"]},
     python_requires=">=3.10",
     entry_points={"console_scripts": [
         "qsa=quickstart_streaming_agents_amd.cli:main"]},

@@ -307,3 +307,62 @@ def test_paged_attention_forced_rescale(ext):
                                  sl.cpu(), 0.088)
     torch.testing.assert_close(out.float().cpu(), ref.float(), atol=4e-2,
                                rtol=4e-2)
+
+
+def test_skinny_gemm_fp8_matches_dequant_ref():
+    """fp8 (e4m3, per-channel W8A16) decode GEMM vs the dequantized fp32
+    reference: the in-kernel cvt_pk_f32_fp8 + perm pack is exact, so the
+    only tolerance is MFMA accumulation order."""
+    from quickstart_streaming_agents_amd.ops import dispatch as D
+    from quickstart_streaming_agents_amd.ops import ext
+    torch.manual_seed(5)
+    for M, N, K in [(24, 4096, 4096), (1, 512, 512), (32, 1024, 256),
+                    (17, 28672, 512), (24, 128, 14336)]:
+        a = torch.randn(M, K, device="cuda:0", dtype=torch.bfloat16) * 0.5
+        w = torch.randn(N, K, device="cuda:0", dtype=torch.bfloat16) * 0.02
+        qf, s = D.pack_weight_fp8(w)
+        out = ext().skinny_gemm_fp8(a, qf, s, N, K)
+        wd = D.unpack_weight_fp8(qf, s, N, K)      # exact dequant (f32)
+        ref = a.float() @ wd.T
+        err = (out.float() - ref).abs().max().item()
+        scale = ref.abs().max().item() + 1e-6
+        assert err / scale < 2e-2, f"M{M} N{N} K{K}: rel err {err/scale}"
+        # quantization error vs the unquantized weights is bounded (e4m3
+        # per-channel: ~0.5-3% on random normal weights)
+        ref_bf = a.float() @ w.float().T
+        qerr = (out.float() - ref_bf).abs().max().item() / \
+            (ref_bf.abs().std().item() + 1e-6)
+        assert qerr < 0.5, f"quantization error blew up: {qerr}"
+
+
+def test_skinny_gemm_fp8_gpu_pack_matches_cpu_pack():
+    from quickstart_streaming_agents_amd.ops import dispatch as D
+    torch.manual_seed(6)
+    w = torch.randn(64, 512, dtype=torch.bfloat16) * 0.02
+    qf_cpu, s_cpu = D.pack_weight_fp8(w)
+    qf_gpu, s_gpu = D.pack_weight_fp8(w.cuda())
+    assert torch.equal(qf_cpu, qf_gpu.cpu())
+    assert torch.allclose(s_cpu, s_gpu.cpu())
+
+
+def test_llama_decode_uses_fp8_path():
+    """The flagship decode path must run the fp8 kernel by default (the
+    driver's native-code check: no silent bf16/rocBLAS fallback)."""
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    cfg = LlamaConfig.preset("tiny")
+    cfg.hidden = 256
+    m = LlamaModel(cfg, device="cuda:0", seed=1)
+    assert m.use_fp8
+    L = m.layers[0]
+    assert "wqkv_f8" in L and "wgu_f8" in L and "wdown_f8" in L
+    assert m.lm_head_f8 is not None
+    # numerics: one decode-shaped _linear call routes through fp8 and
+    # stays close to the bf16 matmul
+    x = torch.randn(8, cfg.hidden, device="cuda:0",
+                    dtype=torch.bfloat16) * 0.5
+    out = m._linear(x, L["wqkv"], None, L["wqkv_f8"])
+    ref = x.float() @ L["wqkv"].float().T
+    rel = (out.float() - ref).abs().max().item() / \
+        (ref.abs().std().item() + 1e-6)
+    assert rel < 0.5
