@@ -159,7 +159,8 @@ def episode(agent: AgentSpec, user_prompt: str,
     """
     policy = policy or ToolCallPolicy()
     transcript: list[str] = []
-    ctx: dict = {"agent": agent, "observations": [], "user_prompt": user_prompt}
+    ctx: dict = {"agent": agent, "observations": [], "calls": [],
+                 "user_prompt": user_prompt}
     consecutive_failures = 0
     tool_calls = 0
     trace: list[dict] = []
@@ -170,7 +171,8 @@ def episode(agent: AgentSpec, user_prompt: str,
         if tool_schemas is not None:
             from ..models.grammar import build_turn_grammar
             allowed = agent.tools.allowed_tools if agent.tools else None
-            grammar = build_turn_grammar(tool_schemas, allowed, prompt)
+            grammar = build_turn_grammar(tool_schemas, allowed, prompt,
+                                         history=ctx["calls"])
             text = yield ("llm", prompt, max_new_tokens, grammar)
         else:
             text = yield ("llm", prompt, max_new_tokens)
@@ -185,6 +187,7 @@ def episode(agent: AgentSpec, user_prompt: str,
             transcript.append(f"[thought] {action.note or last_text[:200]}")
             continue
         if isinstance(action, ToolCall):
+            ctx["calls"].append((action.name, action.arguments))
             if agent.tools is not None and not agent.tools.allows(action.name):
                 result = f"__error__ tool {action.name} not allowed"
             else:

@@ -94,13 +94,22 @@ def _candidates_for_param(name: str, text: str) -> list:
 
 
 def build_turn_grammar(tool_schemas: dict[str, dict], allowed: tuple | None,
-                       text: str, max_options: int = 8) -> TurnGrammar:
+                       text: str, max_options: int = 8,
+                       history: list[tuple[str, dict]] | None = None,
+                       ) -> TurnGrammar:
     """Candidate actions for one turn.
 
     tool_schemas: {tool name -> MCP inputSchema dict} (tools/list).
     allowed: the agent's CREATE TOOL allowed_tools filter (None = all).
     text: conversation so far (system prompt + user prompt + transcript).
+    history: (tool, arguments) calls already made this episode — an exact
+    repeat is dropped from the candidate set (standard function-calling
+    dedup: re-issuing an identical call yields no new information), so an
+    episode's length is bounded by its distinct actions, not only by
+    max_iterations.
     """
+    seen = {(t, json.dumps(a, sort_keys=True))
+            for t, a in (history or [])}
     options: list[ActionOption] = []
     for name, schema in tool_schemas.items():
         if allowed and name not in allowed:
@@ -123,14 +132,18 @@ def build_turn_grammar(tool_schemas: dict[str, dict], allowed: tuple | None,
         # multi-candidate param fans out, the rest take their first
         fan_param = next((p for p, c in per_param.items() if len(c) > 1),
                          None)
+        cands: list[ActionOption] = []
         if fan_param is None:
-            options.append(ActionOption(
+            cands.append(ActionOption(
                 name, {p: c[0] for p, c in per_param.items()}))
         else:
             for v in per_param[fan_param]:
                 args = {p: (v if p == fan_param else c[0])
                         for p, c in per_param.items()}
-                options.append(ActionOption(name, args))
+                cands.append(ActionOption(name, args))
+        for o in cands:
+            if (o.tool, json.dumps(o.arguments, sort_keys=True)) not in seen:
+                options.append(o)
     return TurnGrammar(options=options[:max_options], allow_finish=True)
 
 

@@ -230,3 +230,20 @@ def test_free_turn_respects_valid_vocab(tok):
     eng.run_to_completion()
     for t in seq.out_tokens:
         assert (tok._BYTE0 <= t < tok.n_tokens) or t == tok.EOS
+
+
+def test_grammar_dedup_bounds_episode_length(tok):
+    """An exact repeat of an already-made call is dropped from the
+    candidate set, so episodes end within distinct-actions + 1 turns
+    instead of spinning to max_iterations."""
+    schemas = _schemas()
+    text = "fetch http://a.test/x please"
+    g0 = build_turn_grammar(schemas, ("http_get",), text)
+    assert len(g0.options) == 1
+    g1 = build_turn_grammar(schemas, ("http_get",), text,
+                            history=[("http_get", g0.options[0].arguments)])
+    assert g1.options == [] and g1.allow_finish
+    # engine level: every episode ends well under the iteration cap
+    results, _ = _run_lab1_episodes(tok, n=10, seed=5, max_iters=10)
+    assert all(r.status == "SUCCESS" for r in results)
+    assert max(r.iterations for r in results) <= 5
