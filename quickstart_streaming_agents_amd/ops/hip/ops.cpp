@@ -80,7 +80,8 @@ extern "C" void qsa_skinny_gemm_fp8_launch(const unsigned short*,
                                            hipStream_t);
 extern "C" void qsa_skinny_gemm_fp8_probe_launch(
     const unsigned short*, const unsigned char*, const float*,
-    unsigned short*, int, int, long long, long long, int, int, hipStream_t);
+    unsigned short*, int, int, long long, long long, int, int, int,
+    hipStream_t);
 extern "C" void qsa_topk_launch(const float*, const float*, float*, int*,
                                 float*, int*, int, int, int, int, int,
                                 hipStream_t);
@@ -353,14 +354,14 @@ torch::Tensor skinny_gemm_fp8(torch::Tensor a, torch::Tensor qf,
 
 torch::Tensor skinny_gemm_fp8_probe(torch::Tensor a, torch::Tensor qf,
                                     torch::Tensor scale, long N, long K,
-                                    long waves, long nt) {
+                                    long waves, long tiles, long nt) {
   CHK_DEV(a); CHK_BF16(a); CHK_CONT(qf);
   const int M = a.size(0);
   auto out = torch::empty({M, (long long)N}, a.options());
   qsa_skinny_gemm_fp8_probe_launch(u16(a), qf.data_ptr<unsigned char>(),
                                    scale.data_ptr<float>(), u16m(out), M,
                                    (int)N, K, a.stride(0), (int)waves,
-                                   (int)nt, cur_stream());
+                                   (int)tiles, (int)nt, cur_stream());
   return out;
 }
 
@@ -450,7 +451,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm_fp8", &skinny_gemm_fp8,
         "fp8-weight decode GEMM (M<=32): half the weight stream");
   m.def("skinny_gemm_fp8_probe", &skinny_gemm_fp8_probe,
-        "fp8 ablation probe: waves/nt sweep");
+        "fp8 ablation probe: waves/tiles/nt sweep");
   m.def("topk_cosine", &topk_cosine, "exact cosine top-k over the HBM index");
   m.def("window_agg", &window_agg, "segmented (key, window) count/sum");
   m.def("anomaly_batch", &anomaly_batch, "batched AR+ridge anomaly scorer");

@@ -53,20 +53,29 @@ __device__ __forceinline__ bf16x8 fp8x8_to_bf16x8(unsigned int a,
   return out.v;
 }
 
-template <int WAVES, bool NT>
+// TILES n-tiles (16 cols each) per workgroup: the A (activation) loads —
+// L2 traffic that rivals the fp8 W stream at 16 cols/WG — are loaded once
+// per k-step and reused across all TILES MFMA streams.  TILES=4 for the
+// big-N projections (wgu/lm_head), 1 for the small ones (grid must still
+// exceed 256 CUs).
+template <int WAVES, int TILES, bool NT>
 __global__ void __launch_bounds__(WAVES * 64)
 qsa_skinny_gemm_fp8_t(const unsigned short* __restrict__ A,  // [M,K] bf16
                       const u32x4* __restrict__ Qf,   // packed fp8 stream
                       const float* __restrict__ scale,     // [N] f32
                       unsigned short* __restrict__ Cbf,    // [M, N]
                       int M, int N, long long K, long long lda) {
-  const int nt = blockIdx.x;              // 16-col n-tile
+  const int nt0 = blockIdx.x * TILES;     // first 16-col n-tile
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const long long kchunks = K / QSA_KCH8;  // K % 256 == 0
 
-  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f};
-  f32x4 acc1 = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc0[TILES], acc1[TILES];
+#pragma unroll
+  for (int t = 0; t < TILES; ++t) {
+    acc0[t] = {0.f, 0.f, 0.f, 0.f};
+    acc1[t] = {0.f, 0.f, 0.f, 0.f};
+  }
 
   // A rows clamp out-of-batch to M-1 (store is guarded; see bf16 kernel)
   const int arow = lane & 15;
@@ -78,60 +87,76 @@ qsa_skinny_gemm_fp8_t(const unsigned short* __restrict__ A,  // [M,K] bf16
 
   // Q stream: 1024-B block per (nt, kb=64k); this lane's 16 B at byte
   // (lane&15)*64 + (lane>>4)*16 inside the block (uint4 units below).
-  const u32x4* qbase =
-      Qf + ((long long)nt * (K >> 6)) * 64 +
-      (long long)((lane & 15) * 4 + (lane >> 4));
+  const u32x4* qbase[TILES];
+#pragma unroll
+  for (int t = 0; t < TILES; ++t)
+    qbase[t] = Qf + ((long long)(nt0 + t) * (K >> 6)) * 64 +
+               (long long)((lane & 15) * 4 + (lane >> 4));
 
   for (long long c = wave; c < kchunks; c += WAVES) {
     const long long k0 = c * QSA_KCH8;
 #pragma unroll
-    for (int s = 0; s < 4; ++s) {          // 4 loads x 64 k = 256 k
+    for (int s = 0; s < 4; ++s) {     // 4 k-steps x 64 k = 256 k
       const long long kb = (k0 >> 6) + s;
-      u32x4 q;
-      if (NT) {
-        q = __builtin_nontemporal_load(qbase + kb * 64);
-      } else {
-        q = *reinterpret_cast<const u32x4*>(
-            __builtin_assume_aligned(qbase + kb * 64, 16));
+      u32x4 q[TILES];
+#pragma unroll
+      for (int t = 0; t < TILES; ++t) {
+        if (NT) {
+          q[t] = __builtin_nontemporal_load(qbase[t] + kb * 64);
+        } else {
+          q[t] = *reinterpret_cast<const u32x4*>(
+              __builtin_assume_aligned(qbase[t] + kb * 64, 16));
+        }
       }
-      const bf16x8 w0 = fp8x8_to_bf16x8(q.x, q.y);   // k .. +32
-      const bf16x8 w1 = fp8x8_to_bf16x8(q.z, q.w);   // k+32 .. +64
       const long long ak = k0 + s * 64;
       const bf16x8 a00 = *reinterpret_cast<const bf16x8*>(a0base + ak);
       const bf16x8 a10 = *reinterpret_cast<const bf16x8*>(a1base + ak);
       const bf16x8 a01 = *reinterpret_cast<const bf16x8*>(a0base + ak + 32);
       const bf16x8 a11 = *reinterpret_cast<const bf16x8*>(a1base + ak + 32);
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a00, w0, acc0, 0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a10, w0, acc1, 0, 0, 0);
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a01, w1, acc0, 0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a11, w1, acc1, 0, 0, 0);
+#pragma unroll
+      for (int t = 0; t < TILES; ++t) {
+        const bf16x8 w0 = fp8x8_to_bf16x8(q[t].x, q[t].y);   // k .. +32
+        const bf16x8 w1 = fp8x8_to_bf16x8(q[t].z, q[t].w);   // +32 .. +64
+        acc0[t] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a00, w0, acc0[t], 0, 0, 0);
+        acc1[t] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a10, w0, acc1[t], 0, 0, 0);
+        acc0[t] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a01, w1, acc0[t], 0, 0, 0);
+        acc1[t] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a11, w1, acc1[t], 0, 0, 0);
+      }
     }
   }
 
   // ---- cross-wave K-reduction in LDS + per-channel scale ---------------
   __shared__ float red[WAVES][64][8];
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    red[wave][lane][r] = acc0[r];
-    red[wave][lane][4 + r] = acc1[r];
-  }
-  __syncthreads();
-  if (wave == 0) {
-    const int ncol = nt * 16 + (lane & 15);
-    const int mrow = (lane >> 4) * 4;
-    const float s = scale[ncol];
+  for (int t = 0; t < TILES; ++t) {
+    if (t > 0) __syncthreads();   // reuse the LDS slab per tile
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float v0 = 0.f, v1 = 0.f;
+      red[wave][lane][r] = acc0[t][r];
+      red[wave][lane][4 + r] = acc1[t][r];
+    }
+    __syncthreads();
+    if (wave == 0) {
+      const int ncol = (nt0 + t) * 16 + (lane & 15);
+      const int mrow = (lane >> 4) * 4;
+      const float s = scale[ncol];
 #pragma unroll
-      for (int wv = 0; wv < WAVES; ++wv) {
-        v0 += red[wv][lane][r];
-        v1 += red[wv][lane][4 + r];
+      for (int r = 0; r < 4; ++r) {
+        float v0 = 0.f, v1 = 0.f;
+#pragma unroll
+        for (int wv = 0; wv < WAVES; ++wv) {
+          v0 += red[wv][lane][r];
+          v1 += red[wv][lane][4 + r];
+        }
+        const int m0 = mrow + r;
+        if (m0 < M) Cbf[(long long)m0 * N + ncol] = f32_to_bf16(v0 * s);
+        if (16 + m0 < M)
+          Cbf[(long long)(16 + m0) * N + ncol] = f32_to_bf16(v1 * s);
       }
-      const int m0 = mrow + r;
-      if (m0 < M) Cbf[(long long)m0 * N + ncol] = f32_to_bf16(v0 * s);
-      if (16 + m0 < M)
-        Cbf[(long long)(16 + m0) * N + ncol] = f32_to_bf16(v1 * s);
     }
   }
 }
@@ -140,26 +165,41 @@ extern "C" void qsa_skinny_gemm_fp8_launch(
     const unsigned short* A, const unsigned char* Qf, const float* scale,
     unsigned short* Cbf, int M, int N, long long K, long long lda,
     hipStream_t stream) {
-  hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<8, false>), dim3(N / 16),
-                     dim3(512), 0, stream, A,
-                     reinterpret_cast<const u32x4*>(Qf), scale, Cbf, M, N,
-                     K, lda);
+  const u32x4* Q = reinterpret_cast<const u32x4*>(Qf);
+  if (N % 64 == 0 && N / 64 >= 256) {
+    // big-N (wgu / lm_head): 4 tiles amortize the A re-read 4x and the
+    // grid still oversubscribes the 256 CUs
+    hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<8, 4, false>), dim3(N / 64),
+                       dim3(512), 0, stream, A, Q, scale, Cbf, M, N, K,
+                       lda);
+  } else if (K >= 8192) {
+    // deep-K small-N (wdown): 16 waves per WG keep more loads in flight
+    // at 1 WG/CU
+    hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<16, 1, false>), dim3(N / 16),
+                       dim3(1024), 0, stream, A, Q, scale, Cbf, M, N, K,
+                       lda);
+  } else {
+    hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<8, 1, false>), dim3(N / 16),
+                       dim3(512), 0, stream, A, Q, scale, Cbf, M, N, K,
+                       lda);
+  }
 }
 
 extern "C" void qsa_skinny_gemm_fp8_probe_launch(
     const unsigned short* A, const unsigned char* Qf, const float* scale,
     unsigned short* Cbf, int M, int N, long long K, long long lda,
-    int waves, int nt, hipStream_t stream) {
-  dim3 grid(N / 16);
-#define QSA_CASE(W, NTB)                                                  \
-  if (waves == W && nt == (int)NTB) {                                     \
-    hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<W, NTB>), grid,             \
-                       dim3(W * 64), 0, stream, A,                        \
-                       reinterpret_cast<const u32x4*>(Qf), scale, Cbf,   \
+    int waves, int tiles, int nt, hipStream_t stream) {
+#define QSA_CASE(W, T, NTB)                                               \
+  if (waves == W && tiles == T && nt == (int)NTB) {                       \
+    hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<W, T, NTB>),                \
+                       dim3(N / (16 * T)), dim3(W * 64), 0, stream, A,    \
+                       reinterpret_cast<const u32x4*>(Qf), scale, Cbf,    \
                        M, N, K, lda);                                     \
     return;                                                               \
   }
-  QSA_CASE(8, false) QSA_CASE(8, true) QSA_CASE(4, false) QSA_CASE(4, true)
-  QSA_CASE(2, false) QSA_CASE(2, true)
+  QSA_CASE(8, 1, false) QSA_CASE(8, 1, true) QSA_CASE(8, 2, false)
+  QSA_CASE(8, 4, false) QSA_CASE(8, 4, true) QSA_CASE(8, 8, false)
+  QSA_CASE(16, 1, false) QSA_CASE(16, 2, false) QSA_CASE(16, 4, false)
+  QSA_CASE(4, 4, false) QSA_CASE(4, 8, false) QSA_CASE(2, 8, false)
 #undef QSA_CASE
 }

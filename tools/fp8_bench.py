@@ -52,6 +52,22 @@ def main():
         t_roc = timeit(lambda: torch.nn.functional.linear(a, w))
         t_bf = timeit(lambda: ext().skinny_gemm(a, wf, N, K))
         t_f8 = timeit(lambda: ext().skinny_gemm_fp8(a, qf, s, N, K))
+        if os.environ.get("QSA_FP8_SWEEP") == "1":
+            for waves, tiles, ntl in [(8, 1, 0), (8, 1, 1), (8, 2, 0),
+                                      (8, 4, 0), (8, 4, 1), (8, 8, 0),
+                                      (16, 1, 0), (16, 2, 0), (16, 4, 0),
+                                      (4, 4, 0), (4, 8, 0), (2, 8, 0)]:
+                if N % (16 * tiles):
+                    continue
+                try:
+                    t = timeit(lambda: ext().skinny_gemm_fp8_probe(
+                        a, qf, s, N, K, waves, tiles, ntl), reps=20)
+                    print(f"    probe {name} w{waves} t{tiles} nt{ntl}: "
+                          f"{t:.1f} us ({(N*K)/(t*1e-6)/1e9:.0f} GB/s)",
+                          file=sys.stderr)
+                except Exception as e:
+                    print(f"    probe {name} w{waves} t{tiles}: {e}",
+                          file=sys.stderr)
         # correctness spot check
         ref = a.float() @ D.unpack_weight_fp8(qf, s, N, K).T
         out = ext().skinny_gemm_fp8(a, qf, s, N, K).float()
