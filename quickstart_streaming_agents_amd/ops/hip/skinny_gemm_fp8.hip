@@ -165,20 +165,24 @@ extern "C" void qsa_skinny_gemm_fp8_launch(
     const unsigned short* A, const unsigned char* Qf, const float* scale,
     unsigned short* Cbf, int M, int N, long long K, long long lda,
     hipStream_t stream) {
+  // Per-shape winners from the measured sweep on MI355X
+  // (profiles/fp8_decode_gemm.md): huge N wants more tiles + fewer
+  // waves (lm_head w2t8 = 4.6 TB/s fp8 bytes), mid N wants t4, qkv-size
+  // t2, small/deep-K t1.
   const u32x4* Q = reinterpret_cast<const u32x4*>(Qf);
-  if (N % 64 == 0 && N / 64 >= 256) {
-    // big-N (wgu / lm_head): 4 tiles amortize the A re-read 4x and the
-    // grid still oversubscribes the 256 CUs
+  if (N % 128 == 0 && N >= 65536) {          // lm_head class
+    hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<2, 8, false>), dim3(N / 128),
+                       dim3(128), 0, stream, A, Q, scale, Cbf, M, N, K,
+                       lda);
+  } else if (N % 64 == 0 && N >= 16384) {    // wgu class
     hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<8, 4, false>), dim3(N / 64),
                        dim3(512), 0, stream, A, Q, scale, Cbf, M, N, K,
                        lda);
-  } else if (K >= 8192) {
-    // deep-K small-N (wdown): 16 waves per WG keep more loads in flight
-    // at 1 WG/CU
-    hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<16, 1, false>), dim3(N / 16),
-                       dim3(1024), 0, stream, A, Q, scale, Cbf, M, N, K,
+  } else if (N % 32 == 0 && N >= 6144 && K < 8192) {   // qkv class
+    hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<8, 2, false>), dim3(N / 32),
+                       dim3(512), 0, stream, A, Q, scale, Cbf, M, N, K,
                        lda);
-  } else {
+  } else {                                   // wo / wdown class
     hipLaunchKernelGGL((qsa_skinny_gemm_fp8_t<8, 1, false>), dim3(N / 16),
                        dim3(512), 0, stream, A, Q, scale, Cbf, M, N, K,
                        lda);
