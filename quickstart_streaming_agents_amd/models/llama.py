@@ -400,7 +400,7 @@ class LlamaModel:
                 # gather/pad/score materialization
                 attn = D.ext().paged_attn_prefill(
                     q, kv.k[li], kv.v[li], bt_t, qb_item_t, qb_pos0_t,
-                    off_t, start_t, len_t, self.scale)
+                    off_t, start_t, len_t, self.scale, True)
             else:
                 # padded q: [nb*nmax, n_q, D] -> [nb*KVH, R*nmax, D]
                 q_pad = torch.zeros(nb * nmax, self.n_q, c.d_head,

@@ -118,12 +118,16 @@ class Engine:
         self.running: list[Sequence] = []
         self.stats = EngineStats()
         # hipGraph-captured decode step (launch-bound otherwise: ~300
-        # kernel/GEMM launches per step across 32 layers).  TP ranks stay
-        # eager: RCCL all-reduce inside hipGraph capture is not exercised
-        # by the CPU test matrix, so correctness-first until measured.
-        self.use_graph = torch.cuda.is_available() and \
-            getattr(model, "tp_size", 1) == 1
+        # kernel/GEMM launches per step across 32 layers).  TP ranks
+        # capture too — the per-layer RCCL all-reduces are recorded into
+        # the graph (validated against eager on hardware,
+        # tests/test_gpu_models.py rccl-in-graph test); if capture fails
+        # on a given stack the engine falls back to eager decode.
+        # TP NOTE: grammar decision sampling draws torch.rand — TP ranks
+        # must share a seed so sampled decisions agree across the group.
+        self.use_graph = torch.cuda.is_available()
         self._graph = None
+        self._graph_failed = False
         self._gbuf: dict = {}
         # free-text sampling mask: tokens outside [lo, hi) (plus EOS) are
         # never emitted — keeps random-init decode inside the tokenizer's
@@ -182,9 +186,24 @@ class Engine:
     # ---- hipGraph decode -------------------------------------------------
     MAX_RUN = 512  # on-device token-history depth per graph run
 
-    def _ensure_graph(self) -> None:
+    def _ensure_graph(self) -> bool:
         if self._graph is not None:
-            return
+            return True
+        if self._graph_failed:
+            return False
+        try:
+            self._capture_graph()
+            return True
+        except Exception:
+            # capture failed (e.g. an op that refuses stream capture on
+            # this stack): permanent eager fallback, correctness first
+            self._graph_failed = True
+            self.use_graph = False
+            self._graph = None
+            torch.cuda.synchronize()
+            return False
+
+    def _capture_graph(self) -> None:
         dev = self.model.device
         B = self.max_batch
         cap = (self.max_seq_len + 63) // 64
@@ -233,10 +252,47 @@ class Engine:
         self._graph = g
         self._gbuf = gb
 
+    def _decode_run_eager(self, batch: list[Sequence], run: int) -> None:
+        """Eager fallback for a graph run (capture unavailable)."""
+        for _ in range(run):
+            live = [s for s in batch if not s.done]
+            if not live:
+                break
+            for s in live:
+                self.kv.extend(s.seq_id, len(s.prompt) + len(s.out_tokens))
+            self._decode_step_eager(live)
+
+    def _decode_step_eager(self, batch: list[Sequence]) -> None:
+        dev = self.model.device
+        tokens = torch.tensor([s.out_tokens[-1] for s in batch],
+                              dtype=torch.int64, device=dev)
+        positions = torch.tensor(
+            [len(s.prompt) + len(s.out_tokens) - 1 for s in batch],
+            dtype=torch.int32, device=dev)
+        seq_ids = [s.seq_id for s in batch]
+        bt = self.kv.block_table(seq_ids)
+        sl = self.kv.seq_lens_tensor(seq_ids)
+        logits = self.model.forward_decode(tokens, self.kv, bt, sl,
+                                           positions)
+        nxt = self._sample(logits).tolist()
+        for s, tok in zip(batch, nxt):
+            if s.script:
+                # grammar-forced continuation (script index: out_tokens[0]
+                # was the decision token)
+                si = len(s.out_tokens) - 1
+                if si < len(s.script):
+                    tok = s.script[si]
+            s.out_tokens.append(int(tok))
+            self._maybe_finish(s)
+        self.stats.decode_tokens += len(batch)
+        self.stats.decode_steps += 1
+
     def _decode_run_graph(self, batch: list[Sequence], run: int) -> None:
         """Run `run` decode steps for `batch` with zero host round trips:
         pages are pre-extended, the graph self-feeds, one sync at the end."""
-        self._ensure_graph()
+        if not self._ensure_graph():
+            self._decode_run_eager(batch, run)
+            return
         gb = self._gbuf
         n = len(batch)
         dev = self.model.device
@@ -428,37 +484,14 @@ class Engine:
         if not batch:
             self._retire()
             return len(self.running) + len(self.pending)
-        dev = self.model.device
         for s in batch:
             self.kv.extend(s.seq_id, len(s.prompt) + len(s.out_tokens))
         if self.use_graph:
-            # one-step graph run (appends + finishes internally)
+            # one-step graph run (appends + finishes internally; falls
+            # back to eager if capture is unavailable)
             self._decode_run_graph(batch, 1)
-            self._retire()
-            return len(self.running) + len(self.pending)
         else:
-            tokens = torch.tensor([s.out_tokens[-1] for s in batch],
-                                  dtype=torch.int64, device=dev)
-            positions = torch.tensor(
-                [len(s.prompt) + len(s.out_tokens) - 1 for s in batch],
-                dtype=torch.int32, device=dev)
-            seq_ids = [s.seq_id for s in batch]
-            bt = self.kv.block_table(seq_ids)
-            sl = self.kv.seq_lens_tensor(seq_ids)
-            logits = self.model.forward_decode(tokens, self.kv, bt, sl,
-                                               positions)
-            nxt = self._sample(logits).tolist()
-        for s, tok in zip(batch, nxt):
-            if s.script:
-                # grammar-forced continuation (script index: out_tokens[0]
-                # was the decision token)
-                si = len(s.out_tokens) - 1
-                if si < len(s.script):
-                    tok = s.script[si]
-            s.out_tokens.append(int(tok))
-            self._maybe_finish(s)
-        self.stats.decode_tokens += len(batch)
-        self.stats.decode_steps += 1
+            self._decode_step_eager(batch)
         self._retire()
         return len(self.running) + len(self.pending)
 

@@ -50,7 +50,7 @@ extern "C" void qsa_kv_append_launch(const unsigned short*,
 extern "C" void qsa_paged_attn_prefill_launch(
     const unsigned short*, const unsigned short*, const unsigned short*,
     const int*, const int*, const int*, const int*, const int*, const int*,
-    unsigned short*, float, int, int, int, int, int, long long,
+    unsigned short*, float, int, int, int, int, int, long long, int,
     hipStream_t);
 extern "C" void qsa_rope_kv_append_launch(unsigned short*,
                                           const unsigned short*,
@@ -237,7 +237,8 @@ torch::Tensor paged_attn_prefill(torch::Tensor q, torch::Tensor kc,
                                  torch::Tensor qb_item, torch::Tensor qb_pos0,
                                  torch::Tensor item_off,
                                  torch::Tensor item_start,
-                                 torch::Tensor item_len, double scale) {
+                                 torch::Tensor item_len, double scale,
+                                 bool causal) {
   CHK_DEV(q); CHK_BF16(q); CHK_BF16(kc); CHK_BF16(vc);
   CHK_CONT(kc); CHK_CONT(vc); CHK_I32(block_table); CHK_CONT(block_table);
   CHK_I32(qb_item); CHK_I32(qb_pos0); CHK_I32(item_off);
@@ -258,7 +259,7 @@ torch::Tensor paged_attn_prefill(torch::Tensor q, torch::Tensor kc,
       qb_item.data_ptr<int>(), qb_pos0.data_ptr<int>(),
       item_off.data_ptr<int>(), item_start.data_ptr<int>(),
       item_len.data_ptr<int>(), u16m(out), (float)scale, QB, QH, KVH, npmax,
-      D, q.stride(0), cur_stream());
+      D, q.stride(0), causal ? 1 : 0, cur_stream());
   return out;
 }
 

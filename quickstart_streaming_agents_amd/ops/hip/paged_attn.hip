@@ -413,8 +413,12 @@ extern "C" void qsa_kv_scatter_launch(const unsigned short* knew,
 //
 // Host passes per-q-block maps (item id, first q position) plus per-item
 // row offset / cached start / new length and the padded page table.
+//
+// CAUSAL=false is the ENCODER path (K1, models/encoder.py): bidirectional
+// attention over the full item — every row attends to every valid key —
+// with the same varlen layout and paged K/V fragment reads.
 // ---------------------------------------------------------------------------
-template <int D>
+template <int D, bool CAUSAL>
 __global__ void __launch_bounds__(256)
 qsa_paged_attn_prefill(const unsigned short* __restrict__ q,   // [T, QH, D] strided
                        const unsigned short* __restrict__ kc,
@@ -465,8 +469,10 @@ qsa_paged_attn_prefill(const unsigned short* __restrict__ q,   // [T, QH, D] str
   for (int dt = 0; dt < DTILES; ++dt)
     o_acc[dt] = (f32x4_a){0.f, 0.f, 0.f, 0.f};
 
-  // causal horizon: the block's LAST row attends to start+pos0+15 (abs)
-  const int max_abs = start + min(pos0 + 15, n_i - 1);
+  // attention horizon: causal -> the block's LAST row's position;
+  // bidirectional -> the whole item
+  const int max_abs = CAUSAL ? start + min(pos0 + 15, n_i - 1)
+                             : start + n_i - 1;
   const int npages = (max_abs + QSA_PAGE) / QSA_PAGE;  // ceil(max_abs+1/64)
   const int* btab = block_table + (long long)item * npmax;
 
@@ -503,7 +509,7 @@ qsa_paged_attn_prefill(const unsigned short* __restrict__ q,   // [T, QH, D] str
       for (int r = 0; r < 4; ++r) {
         const int kpos = pt * 16 + hi * 4 + r;
         const int kabs = pi * QSA_PAGE + kpos;
-        const bool ok = (kpos < nvalid) && (kabs <= qabs) &&
+        const bool ok = (kpos < nvalid) && (!CAUSAL || kabs <= qabs) &&
                         (pos0 + col < n_i);
         float v = ok ? sc[pt][r] * scale : -3.0e38f;
         sc[pt][r] = v;
@@ -604,17 +610,19 @@ extern "C" void qsa_paged_attn_prefill_launch(
     const unsigned short* vc, const int* block_table, const int* qb_item,
     const int* qb_pos0, const int* item_off, const int* item_start,
     const int* item_len, unsigned short* out, float scale, int QB, int QH,
-    int KVH, int npmax, int D, long long qstride, hipStream_t stream) {
+    int KVH, int npmax, int D, long long qstride, int causal,
+    hipStream_t stream) {
   const long long waves = (long long)QB * QH;
   dim3 grid((unsigned)((waves + 3) / 4));
-  if (D == 128)
-    hipLaunchKernelGGL((qsa_paged_attn_prefill<128>), grid, dim3(256), 0,
-                       stream, q, kc, vc, block_table, qb_item, qb_pos0,
-                       item_off, item_start, item_len, out, scale, QB, QH,
-                       KVH, npmax, qstride);
-  else if (D == 64)
-    hipLaunchKernelGGL((qsa_paged_attn_prefill<64>), grid, dim3(256), 0,
-                       stream, q, kc, vc, block_table, qb_item, qb_pos0,
-                       item_off, item_start, item_len, out, scale, QB, QH,
-                       KVH, npmax, qstride);
+#define QSA_PREFILL_CASE(DD, CC)                                          \
+  if (D == DD && causal == (int)CC) {                                     \
+    hipLaunchKernelGGL((qsa_paged_attn_prefill<DD, CC>), grid, dim3(256), \
+                       0, stream, q, kc, vc, block_table, qb_item,        \
+                       qb_pos0, item_off, item_start, item_len, out,      \
+                       scale, QB, QH, KVH, npmax, qstride);               \
+    return;                                                               \
+  }
+  QSA_PREFILL_CASE(128, true) QSA_PREFILL_CASE(128, false)
+  QSA_PREFILL_CASE(64, true) QSA_PREFILL_CASE(64, false)
+#undef QSA_PREFILL_CASE
 }
