@@ -140,9 +140,12 @@ class EmbeddingEncoder:
         use_hip = self._use_hip and have_ext()
         if use_hip:
             maps = self._attn_maps(n, L, lens)
-            kc = torch.empty(maps["n_pages"], c.n_heads, c.d_head // 8, 64,
+            # zeros, not empty: the attention kernel's MFMAs read FULL
+            # 64-slot pages before masking; garbage V in never-scattered
+            # tail slots would 0*inf=NaN the PV accumulate
+            kc = torch.zeros(maps["n_pages"], c.n_heads, c.d_head // 8, 64,
                              8, dtype=self.dtype, device=self.device)
-            vc = torch.empty(maps["n_pages"], c.n_heads, c.d_head, 64,
+            vc = torch.zeros(maps["n_pages"], c.n_heads, c.d_head, 64,
                              dtype=self.dtype, device=self.device)
         else:
             # per-row score limits: row = (seq, head, qpos) -> seq len

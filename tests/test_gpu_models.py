@@ -371,13 +371,15 @@ def test_rccl_all_reduce_inside_hipgraph():
                            tp_size=2, tp_group=dist.group.WORLD)
         prompts = [[3, 7, 11, 19], [2, 9, 4, 6, 8]]
 
-        eng_g = Engine(model, max_batch=4, max_seq_len=256)
+        # max_batch == n prompts: the graph decodes exactly the same
+        # GEMM shapes as eager (padding rows change bf16 numerics)
+        eng_g = Engine(model, max_batch=2, max_seq_len=256)
         assert eng_g.use_graph
         out_g = eng_g.generate_batch([list(p) for p in prompts], [8, 8])
         assert eng_g._graph is not None, \
             "graph capture must succeed with RCCL in the captured region"
 
-        eng_e = Engine(model, max_batch=4, max_seq_len=256)
+        eng_e = Engine(model, max_batch=2, max_seq_len=256)
         eng_e.use_graph = False
         out_e = eng_e.generate_batch([list(p) for p in prompts], [8, 8])
         assert out_g == out_e
