@@ -298,8 +298,13 @@ class Engine:
         dev = self.model.device
         for s in batch:  # pre-extend pages for the whole run
             self.kv.extend(s.seq_id, len(s.prompt) + len(s.out_tokens) + run)
-            # bookkeeping length stays at the pre-run value for seq_lens
-            self.kv._seq_len[s.seq_id] = len(s.prompt) + len(s.out_tokens)
+            # seq_lens the graph starts from EXCLUDE the token the first
+            # replay decodes: the in-graph add_(active) then yields
+            # len(prompt)+len(out) for step 1 — the same "includes the
+            # new token" value the eager path passes (appending at slot
+            # len(prompt)+len(out)-1, no zero hole after the prompt)
+            self.kv._seq_len[s.seq_id] = \
+                len(s.prompt) + len(s.out_tokens) - 1
         seq_ids = [s.seq_id for s in batch]
         bt = self.kv.block_table(seq_ids)
         gb["block_table"][:n, :bt.shape[1]] = bt

@@ -421,25 +421,27 @@ def test_rccl_collectives_on_hardware():
             dist.destroy_process_group()
 
 
-def test_encoder_mfma_attention_matches_cpu_reference():
+def test_encoder_mfma_attention_matches_bmm_reference():
     """K1 encoder bidirectional varlen MFMA attention (CAUSAL=false
-    paged_attn_prefill) vs the CPU padded-bmm reference: same weights,
-    same texts, cosine-identical embeddings."""
+    paged_attn_prefill) vs the padded-bmm reference ON THE SAME DEVICE
+    AND DTYPE — only the attention implementation differs (an f32-CPU
+    end-to-end comparison through 12 random layers diverges chaotically
+    from bf16 rounding alone and tests nothing).  VERDICT r1 item 7."""
     from quickstart_streaming_agents_amd.models.encoder import (
         EmbeddingEncoder, EncoderConfig)
     cfg = EncoderConfig(n_layers=4)
-    gpu = EmbeddingEncoder(cfg, device="cuda:0", seed=2)
-    cpu = EmbeddingEncoder(cfg, device="cpu", dtype=torch.float32, seed=2)
+    enc = EmbeddingEncoder(cfg, device="cuda:0", seed=2)
     texts = ["How do I create a Flink table?",
              "a much longer document about insurance claims in Naples "
              "Florida with anomalous totals across a six hour window " * 4,
              "x",
              "boats dispatched to the French Quarter surge zone"]
-    out_g = gpu.embed_batch(texts)
-    out_c = cpu.embed_batch(texts)
-    assert out_g.shape == out_c.shape == (4, 1536)
+    out_hip = enc.embed_batch(texts)
+    enc._use_hip = False
+    out_bmm = enc.embed_batch(texts)
+    assert out_hip.shape == out_bmm.shape == (4, 1536)
     for i in range(4):
-        cos = float(out_g[i] @ out_c[i])
+        cos = float(out_hip[i] @ out_bmm[i])
         assert cos > 0.98, f"text {i}: cosine {cos}"
 
 
