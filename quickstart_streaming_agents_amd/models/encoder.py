@@ -94,9 +94,17 @@ class EmbeddingEncoder:
         ppi = (L + 63) // 64            # pages per item
         bt = (torch.arange(n * ppi, dtype=torch.int32, device=self.device)
               .reshape(n, ppi).contiguous())
-        slots = torch.arange(n * ppi * 64, dtype=torch.int32,
-                             device=self.device) \
-            .reshape(n, ppi * 64)[:, :L].reshape(-1).contiguous()
+        # pad rows scatter to a TRASH page no block_table references:
+        # their k/v are downstream of an unwritten attention-output row
+        # (possibly NaN), and the PV MFMA reads FULL 64-slot pages before
+        # masking — 0 * NaN would poison valid rows from layer 2 on
+        j = torch.arange(L, device=self.device).unsqueeze(0).expand(n, L)
+        lens_d = lens.to(self.device).unsqueeze(1)
+        base = (torch.arange(n, device=self.device) * ppi * 64) \
+            .unsqueeze(1) + j
+        trash = n * ppi * 64
+        slots = torch.where(j < lens_d, base, trash) \
+            .to(torch.int32).reshape(-1).contiguous()
         qb_item, qb_pos0 = [], []
         lens_l = lens.tolist()
         for i in range(n):
@@ -106,7 +114,7 @@ class EmbeddingEncoder:
         dev = self.device
         mk = lambda x: torch.tensor(x, dtype=torch.int32, device=dev)
         return {
-            "n_pages": n * ppi,
+            "n_pages": n * ppi + 1,     # +1 trash page for pad rows
             "block_table": bt,
             "slots": slots,
             "qb_item": mk(qb_item), "qb_pos0": mk(qb_pos0),
