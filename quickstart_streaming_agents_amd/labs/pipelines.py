@@ -132,20 +132,43 @@ class Lab1PriceMatchPolicy:
             f"${our_price:.2f}; sent a price match email.")
 
 
-def lab1_enriched_orders(broker: Broker, state_ttl_ms: int = 3_600_000) -> list[dict]:
+def lab1_enriched_orders(broker: Broker, state_ttl_ms: int = 3_600_000,
+                         use_gpu: bool | None = None) -> list[dict]:
     """orders |><| customers |><| products with 1-h state TTL
-    (LAB1-Walkthrough.md:119-131)."""
-    customers = TTLTable(lambda r: r["customer_id"], ttl_ms=state_ttl_ms)
-    products = TTLTable(lambda r: r["product_id"], ttl_ms=state_ttl_ms)
-    for _, c in AvroConsumer(broker, "customers", schemas.CUSTOMERS).poll():
-        customers.upsert(c, c.get("updated_at", 0))
-    for _, p in AvroConsumer(broker, "products", schemas.PRODUCTS).poll():
-        products.upsert(p, p.get("updated_at", 0))
+    (LAB1-Walkthrough.md:119-131).
+
+    On a GPU this runs the K8 columnar hash-join path (runtime/joins
+    GpuTTLTable: HBM-resident latest-per-key tables built by
+    ops/hip/hash_join.hip, TTL applied at probe time); on CPU the dict
+    reference with identical semantics."""
+    import torch
+    if use_gpu is None:
+        from ..ops import have_ext
+        use_gpu = torch.cuda.is_available() and have_ext()
+    cust_rows = [c for _, c in
+                 AvroConsumer(broker, "customers", schemas.CUSTOMERS).poll()]
+    prod_rows = [p for _, p in
+                 AvroConsumer(broker, "products", schemas.PRODUCTS).poll()]
     orders = [o for _, o in AvroConsumer(broker, "orders", schemas.ORDERS).poll()]
-    enriched = enrich_join(
-        orders, lambda r: r["order_ts"],
-        [(customers, lambda r: r["customer_id"], None),
-         (products, lambda r: r["product_id"], None)])
+    if use_gpu:
+        from ..runtime.joins import enrich_join_columnar
+        enriched = enrich_join_columnar(
+            orders, lambda r: r["order_ts"],
+            [(cust_rows, "customer_id", "updated_at", "customer_id",
+              state_ttl_ms),
+             (prod_rows, "product_id", "updated_at", "product_id",
+              state_ttl_ms)])
+    else:
+        customers = TTLTable(lambda r: r["customer_id"], ttl_ms=state_ttl_ms)
+        products = TTLTable(lambda r: r["product_id"], ttl_ms=state_ttl_ms)
+        for c in cust_rows:
+            customers.upsert(c, c.get("updated_at", 0))
+        for p in prod_rows:
+            products.upsert(p, p.get("updated_at", 0))
+        enriched = enrich_join(
+            orders, lambda r: r["order_ts"],
+            [(customers, lambda r: r["customer_id"], None),
+             (products, lambda r: r["product_id"], None)])
     # orders.price is the order price; keep product list price separate
     for row in enriched:
         row["order_price"] = row["price"]

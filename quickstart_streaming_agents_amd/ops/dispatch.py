@@ -232,3 +232,30 @@ def window_agg(ts: torch.Tensor, key: torch.Tensor,
         return ext().window_agg(ts, key, value, t0, win_ms, nwin, nkeys)
     from .cpu_ref import window_agg_ref
     return window_agg_ref(ts, key, value, t0, win_ms, nwin, nkeys)
+
+
+# ---- GPU hash join (K8) ---------------------------------------------------
+
+def hash_build(keys: torch.Tensor, ts: torch.Tensor):
+    """(i64 keys, i64 event ts) -> opaque latest-per-key table.
+    GPU: (tkeys, tpay) HBM tensors; CPU: a dict of key -> (ts, row)."""
+    if _cuda(keys):
+        return tuple(ext().hash_build(keys, ts))
+    table: dict = {}
+    for i, (k, t) in enumerate(zip(keys.tolist(), ts.tolist())):
+        prev = table.get(k)
+        if prev is None or (t, i) >= prev:
+            table[k] = (t, i)
+    return table
+
+
+def hash_probe(table, keys: torch.Tensor, min_ts: int) -> torch.Tensor:
+    """Probe -> i32 row indices (-1 = absent or event ts < min_ts)."""
+    if isinstance(table, tuple):
+        return ext().hash_probe(table[0], table[1], keys, int(min_ts))
+    out = torch.full((keys.numel(),), -1, dtype=torch.int32)
+    for i, k in enumerate(keys.tolist()):
+        hit = table.get(k)
+        if hit is not None and hit[0] >= min_ts:
+            out[i] = hit[1]
+    return out

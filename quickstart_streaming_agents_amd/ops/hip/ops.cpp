@@ -82,6 +82,14 @@ extern "C" void qsa_skinny_gemm_fp8_probe_launch(
     const unsigned short*, const unsigned char*, const float*,
     unsigned short*, int, int, long long, long long, int, int, int,
     hipStream_t);
+extern "C" void qsa_hash_build_launch(const long long*, const long long*,
+                                      unsigned long long*,
+                                      unsigned long long*, int, unsigned int,
+                                      hipStream_t);
+extern "C" void qsa_hash_probe_launch(const long long*,
+                                      const unsigned long long*,
+                                      const unsigned long long*, int*, int,
+                                      unsigned int, long long, hipStream_t);
 extern "C" void qsa_topk_launch(const float*, const float*, float*, int*,
                                 float*, int*, int, int, int, int, int,
                                 hipStream_t);
@@ -331,6 +339,46 @@ torch::Tensor skinny_gemm_probe(torch::Tensor a, torch::Tensor wf, long N,
   return out;
 }
 
+std::vector<torch::Tensor> hash_build(torch::Tensor keys,
+                                      torch::Tensor ts) {
+  // latest-event-time-per-key hash table in HBM (K8 streaming join state)
+  CHK_DEV(keys);
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64 && keys.is_contiguous(),
+              "keys must be contiguous i64");
+  TORCH_CHECK(ts.scalar_type() == torch::kInt64 && ts.is_contiguous() &&
+                  ts.numel() == keys.numel(), "ts must be i64 like keys");
+  const long long n = keys.numel();
+  TORCH_CHECK(n < (1 << 24), "hash_build: <= 2^24 rows per batch");
+  long long cap = 64;
+  while (cap < 2 * n) cap <<= 1;
+  auto opts = keys.options();
+  auto tkeys = torch::full({cap}, -1, opts);   // all-ones bits == EMPTY
+  auto tpay = torch::zeros({cap}, opts);
+  qsa_hash_build_launch((const long long*)keys.data_ptr<int64_t>(),
+                        (const long long*)ts.data_ptr<int64_t>(),
+                        (unsigned long long*)tkeys.data_ptr<int64_t>(),
+                        (unsigned long long*)tpay.data_ptr<int64_t>(),
+                        (int)n, (unsigned int)(cap - 1), cur_stream());
+  return {tkeys, tpay};
+}
+
+torch::Tensor hash_probe(torch::Tensor tkeys, torch::Tensor tpay,
+                         torch::Tensor keys, long long min_ts) {
+  CHK_DEV(keys);
+  TORCH_CHECK(keys.scalar_type() == torch::kInt64 && keys.is_contiguous(),
+              "keys must be contiguous i64");
+  const long long m = keys.numel();
+  const long long cap = tkeys.numel();
+  TORCH_CHECK((cap & (cap - 1)) == 0, "table capacity must be pow2");
+  auto out = torch::empty({m}, keys.options().dtype(torch::kInt32));
+  qsa_hash_probe_launch((const long long*)keys.data_ptr<int64_t>(),
+                        (const unsigned long long*)tkeys.data_ptr<int64_t>(),
+                        (const unsigned long long*)tpay.data_ptr<int64_t>(),
+                        out.data_ptr<int>(), (int)m,
+                        (unsigned int)(cap - 1), min_ts, cur_stream());
+  return out;
+}
+
 torch::Tensor skinny_gemm_fp8(torch::Tensor a, torch::Tensor qf,
                               torch::Tensor scale, long N, long K) {
   // C[M,N] = a @ (scale[N] * dequant(Q))^T; Q pre-packed fp8 e4m3 in the
@@ -454,6 +502,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm_fp8_probe", &skinny_gemm_fp8_probe,
         "fp8 ablation probe: waves/tiles/nt sweep");
   m.def("topk_cosine", &topk_cosine, "exact cosine top-k over the HBM index");
+  m.def("hash_build", &hash_build,
+        "build latest-per-key hash table from (keys, event ts)");
+  m.def("hash_probe", &hash_probe,
+        "probe the hash table with TTL cutoff -> row indices or -1");
   m.def("window_agg", &window_agg, "segmented (key, window) count/sum");
   m.def("anomaly_batch", &anomaly_batch, "batched AR+ridge anomaly scorer");
   register_avro(m);

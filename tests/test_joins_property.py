@@ -60,3 +60,57 @@ def test_interval_join_matches_nested_loop(lefts, rights, lo, hi):
     # left columns win on collision (claim row overrides anomaly row)
     for m in got:
         assert m["lts"] == L[m["li"]]["lts"]
+
+
+def test_enrich_join_columnar_matches_dict_reference():
+    """The K8 columnar join path (CPU fallback of the GPU hash table)
+    must match enrich_join/TTLTable on lab1-shaped data, including
+    duplicate keys (latest event time wins) and TTL expiry."""
+    from quickstart_streaming_agents_amd.runtime.joins import (
+        TTLTable, enrich_join, enrich_join_columnar)
+    H = 3_600_000
+    custs = [
+        {"customer_id": "c1", "customer_email": "old@x.com", "updated_at": 10},
+        {"customer_id": "c1", "customer_email": "new@x.com", "updated_at": 20},
+        {"customer_id": "c2", "customer_email": "b@x.com", "updated_at": 15},
+        {"customer_id": "c3", "customer_email": "stale@x.com", "updated_at": 0},
+    ]
+    prods = [{"product_id": "p1", "product_name": "widget", "updated_at": 5}]
+    orders = [
+        {"order_id": "o1", "customer_id": "c1", "product_id": "p1",
+         "order_ts": 100},
+        {"order_id": "o2", "customer_id": "c2", "product_id": "p1",
+         "order_ts": 200},
+        {"order_id": "o3", "customer_id": "cX", "product_id": "p1",
+         "order_ts": 300},                      # missing dim -> held back
+        {"order_id": "o4", "customer_id": "c3", "product_id": "p1",
+         "order_ts": H + 1},                    # c3 expired at this ts
+    ]
+    ct = TTLTable(lambda r: r["customer_id"], ttl_ms=H)
+    pt = TTLTable(lambda r: r["product_id"], ttl_ms=H)
+    for c in custs:
+        ct.upsert(c, c["updated_at"])
+    for p in prods:
+        pt.upsert(p, p["updated_at"])
+    ref = enrich_join(orders, lambda r: r["order_ts"],
+                      [(ct, lambda r: r["customer_id"], None),
+                       (pt, lambda r: r["product_id"], None)])
+    got = enrich_join_columnar(
+        orders, lambda r: r["order_ts"],
+        [(custs, "customer_id", "updated_at", "customer_id", H),
+         (prods, "product_id", "updated_at", "product_id", H)])
+    assert got == ref
+    assert [r["order_id"] for r in got] == ["o1", "o2"]
+    assert got[0]["customer_email"] == "new@x.com"  # latest ts wins
+
+
+def test_lab1_enriched_orders_gpu_flag_matches_cpu():
+    """lab1_enriched_orders(use_gpu=True) on CPU fallback == dict path."""
+    from quickstart_streaming_agents_amd.labs import datagen, pipelines
+    from quickstart_streaming_agents_amd.wire import Broker
+    b1, b2 = Broker(), Broker()
+    datagen.publish_lab1(b1)
+    datagen.publish_lab1(b2)
+    a = pipelines.lab1_enriched_orders(b1, use_gpu=False)
+    b = pipelines.lab1_enriched_orders(b2, use_gpu=True)  # CPU fallback
+    assert a == b and len(a) >= 1
