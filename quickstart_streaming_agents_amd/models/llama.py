@@ -214,13 +214,25 @@ class LlamaModel:
         weight-streaming skinny GEMM — fp8 (W8A16, half the weight
         stream) when packed, else bf16; everything else rocBLAS."""
         n, k = w.shape
-        if x.is_cuda and x.shape[0] <= 32:
-            if wf8 is not None:
+        M = x.shape[0]
+        if x.is_cuda and wf8 is not None:
+            if M <= 32:
                 return D.skinny_linear_fp8(x, wf8[0], wf8[1], n, k)
-            if wf is not None and n * k <= 6144 * 4096:
-                # bf16 skinny wins only the small decode shapes (qkv/wo);
-                # rocBLAS is at the bandwidth floor for the big-N ones
-                return D.skinny_linear(x, wf, n, k)
+            # batched decode (continuous-batching steps, M <= 256): the
+            # fp8 stream GEMM beats rocBLAS bf16 on the weight-bound
+            # shapes; lm_head at these M is COMPUTE-bound (fp8 weights
+            # don't lift the bf16 MFMA rate) so it stays rocBLAS
+            if M <= 256 and n % 64 == 0 and n <= 32768:
+                wgs = n // 64
+                splitk = 1 if wgs >= 256 else (2 if wgs >= 128 else 4)
+                if k % (64 * splitk) == 0:
+                    return D.gemm_fp8_batch(x, wf8[0], wf8[1], n, k,
+                                            splitk)
+        if x.is_cuda and M <= 32 and wf is not None and \
+                n * k <= 6144 * 4096:
+            # bf16 skinny wins only the small decode shapes (qkv/wo);
+            # rocBLAS is at the bandwidth floor for the big-N ones
+            return D.skinny_linear(x, wf, n, k)
         return F.linear(x, w)
 
     def _tp_all_reduce(self, t: torch.Tensor) -> torch.Tensor:

@@ -207,6 +207,16 @@ def unpack_weight_fp8(qf: torch.Tensor, scale: torch.Tensor, n: int,
     return q.view(torch.float8_e4m3fn).float() * scale[:, None].float()
 
 
+def gemm_fp8_batch(x: torch.Tensor, qf: torch.Tensor,
+                   scale: torch.Tensor, n: int, k: int,
+                   splitk: int = 1) -> torch.Tensor:
+    """Batched-M (<=256) fp8 weight-stream GEMM (decode batches)."""
+    if x.is_cuda:
+        return ext().gemm_fp8_batch(x, qf, scale, n, k, splitk)
+    w = unpack_weight_fp8(qf, scale, n, k)
+    return (x.float() @ w.T).to(x.dtype)
+
+
 def skinny_linear_fp8(x: torch.Tensor, qf: torch.Tensor,
                       scale: torch.Tensor, n: int, k: int) -> torch.Tensor:
     """x[M,K] bf16 @ dequant(Q)^T via the fp8 weight-streaming kernel."""

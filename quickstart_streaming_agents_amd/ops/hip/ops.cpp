@@ -82,6 +82,11 @@ extern "C" void qsa_skinny_gemm_fp8_probe_launch(
     const unsigned short*, const unsigned char*, const float*,
     unsigned short*, int, int, long long, long long, int, int, int,
     hipStream_t);
+extern "C" void qsa_gemm_fp8_batch_launch(const unsigned short*,
+                                          const unsigned char*,
+                                          const float*, unsigned short*,
+                                          float*, int, int, long long,
+                                          long long, int, hipStream_t);
 extern "C" void qsa_hash_build_launch(const long long*, const long long*,
                                       unsigned long long*,
                                       unsigned long long*, int, unsigned int,
@@ -339,6 +344,37 @@ torch::Tensor skinny_gemm_probe(torch::Tensor a, torch::Tensor wf, long N,
   return out;
 }
 
+torch::Tensor gemm_fp8_batch(torch::Tensor a, torch::Tensor qf,
+                             torch::Tensor scale, long N, long K,
+                             long splitk) {
+  // C[M,N] = a @ (scale * dequant(Q))^T for decode batches 32 < M <= 256
+  CHK_DEV(a); CHK_BF16(a); CHK_CONT(qf);
+  TORCH_CHECK(qf.scalar_type() == torch::kUInt8, "qf must be uint8 (fp8)");
+  TORCH_CHECK(scale.scalar_type() == torch::kFloat32 && scale.is_cuda() &&
+                  scale.is_contiguous() && scale.numel() == N, "scale");
+  TORCH_CHECK(a.dim() == 2 && a.stride(1) == 1, "a rows must be contiguous");
+  const int M = a.size(0);
+  TORCH_CHECK(M >= 1 && M <= 256, "gemm_fp8_batch: M in [1,256]");
+  TORCH_CHECK(a.size(1) == K, "K mismatch");
+  TORCH_CHECK(N % 64 == 0, "N % 64 == 0");
+  TORCH_CHECK(splitk == 1 || splitk == 2 || splitk == 4, "splitk 1/2/4");
+  TORCH_CHECK(K % (64 * splitk) == 0, "K % (64*splitk) == 0");
+  TORCH_CHECK(qf.numel() == (long long)N * K, "qf size");
+  auto out = torch::empty({(long long)M, (long long)N}, a.options());
+  torch::Tensor ws;
+  float* wsp = nullptr;
+  if (splitk > 1) {
+    ws = torch::empty({splitk, (long long)M, (long long)N},
+                      a.options().dtype(torch::kFloat32));
+    wsp = ws.data_ptr<float>();
+  }
+  qsa_gemm_fp8_batch_launch(u16(a), qf.data_ptr<unsigned char>(),
+                            scale.data_ptr<float>(), u16m(out), wsp, M,
+                            (int)N, K, a.stride(0), (int)splitk,
+                            cur_stream());
+  return out;
+}
+
 std::vector<torch::Tensor> hash_build(torch::Tensor keys,
                                       torch::Tensor ts) {
   // latest-event-time-per-key hash table in HBM (K8 streaming join state)
@@ -499,6 +535,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "ablation probe: waves/nt/variant sweep");
   m.def("skinny_gemm_fp8", &skinny_gemm_fp8,
         "fp8-weight decode GEMM (M<=32): half the weight stream");
+  m.def("gemm_fp8_batch", &gemm_fp8_batch,
+        "batched-M (<=256) fp8 weight-stream GEMM with optional split-K");
   m.def("skinny_gemm_fp8_probe", &skinny_gemm_fp8_probe,
         "fp8 ablation probe: waves/tiles/nt sweep");
   m.def("topk_cosine", &topk_cosine, "exact cosine top-k over the HBM index");

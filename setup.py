@@ -26,6 +26,7 @@ sources = [
     os.path.join(HIP_DIR, "skinny_gemm.hip"),
     os.path.join(HIP_DIR, "skinny_gemm_fp8.hip"),
     os.path.join(HIP_DIR, "hash_join.hip"),
+    os.path.join(HIP_DIR, "gemm_fp8_batch.hip"),
     os.path.join(HIP_DIR, "topk_cosine.hip"),
     os.path.join(HIP_DIR, "streaming.hip"),
 ]

@@ -472,3 +472,48 @@ def test_lab1_gpu_join_pipeline_matches_cpu():
     cpu = pipelines.lab1_enriched_orders(b1, use_gpu=False)
     gpu = pipelines.lab1_enriched_orders(b2, use_gpu=True)
     assert cpu == gpu and len(cpu) >= 1
+
+
+def test_gemm_fp8_batch_matches_dequant_ref():
+    """Batched-M fp8 weight-stream GEMM (LDS-staged A, optional split-K)
+    vs the dequantized fp32 reference, across the llama decode shapes and
+    ragged M values."""
+    from quickstart_streaming_agents_amd.ops import dispatch as D
+    from quickstart_streaming_agents_amd.ops import ext
+    torch.manual_seed(7)
+    cases = [
+        (192, 6144, 4096, 4),    # qkv at bench batch (split-K)
+        (192, 4096, 4096, 4),    # wo
+        (192, 28672, 4096, 1),   # wgu
+        (192, 4096, 14336, 4),   # wdown
+        (33, 4096, 4096, 2),     # just past the skinny cutover
+        (256, 1024, 1024, 1),    # full M tile
+        (100, 2048, 512, 2),     # ragged M, small K
+    ]
+    for M, N, K, splitk in cases:
+        a = torch.randn(M, K, device="cuda:0", dtype=torch.bfloat16) * 0.5
+        w = torch.randn(N, K, device="cuda:0", dtype=torch.bfloat16) * 0.02
+        qf, s = D.pack_weight_fp8(w)
+        out = ext().gemm_fp8_batch(a, qf, s, N, K, splitk)
+        wd = D.unpack_weight_fp8(qf, s, N, K)
+        ref = a.float() @ wd.T
+        err = (out.float() - ref).abs().max().item()
+        rel = err / (ref.abs().max().item() + 1e-6)
+        assert rel < 2e-2, f"M{M} N{N} K{K} sk{splitk}: rel {rel}"
+
+
+def test_llama_batched_decode_routes_fp8():
+    """_linear at decode-batch M routes through the batched fp8 kernel
+    and matches the bf16 matmul within quantization error."""
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    cfg = LlamaConfig.preset("tiny")
+    m = LlamaModel(cfg, device="cuda:0", seed=2)
+    L = m.layers[0]
+    x = torch.randn(192, cfg.hidden, device="cuda:0",
+                    dtype=torch.bfloat16) * 0.5
+    out = m._linear(x, L["wgu"], None, L["wgu_f8"])
+    ref = x.float() @ L["wgu"].float().T
+    rel = (out.float() - ref).abs().max().item() / \
+        (ref.abs().std().item() + 1e-6)
+    assert rel < 0.5
