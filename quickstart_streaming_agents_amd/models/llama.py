@@ -218,11 +218,15 @@ class LlamaModel:
         if x.is_cuda and wf8 is not None:
             if M <= 32:
                 return D.skinny_linear_fp8(x, wf8[0], wf8[1], n, k)
-            # batched decode (continuous-batching steps, M <= 256): the
-            # fp8 stream GEMM beats rocBLAS bf16 on the weight-bound
-            # shapes; lm_head at these M is COMPUTE-bound (fp8 weights
-            # don't lift the bf16 MFMA rate) so it stays rocBLAS
-            if M <= 256 and n % 64 == 0 and n <= 32768:
+            # 32 < M <= 256 (continuous-batching decode): the batched fp8
+            # kernel (ops/hip/gemm_fp8_batch.hip) is measured SLOWER than
+            # rocBLAS bf16 at these M on MI355X — the problem is L2-/
+            # compute-bound, not weight-stream-bound, so halved weight
+            # bytes don't pay (profiles/fp8_decode_gemm.md).  Keep it
+            # opt-in for further tuning; rocBLAS is the measured winner.
+            import os as _os
+            if _os.environ.get("QSA_FP8_BATCH") == "1" and M <= 256 \
+                    and n % 64 == 0 and n <= 32768:
                 wgs = n // 64
                 splitk = 1 if wgs >= 256 else (2 if wgs >= 128 else 4)
                 if k % (64 * splitk) == 0:

@@ -55,7 +55,7 @@ __device__ __forceinline__ bf16x8 qsa_gb_cvt(unsigned int a,
   return out.v;
 }
 
-template <int M_FRAGS, int SPLITK>
+template <int M_FRAGS, int SPLITK, int NT>
 __global__ void __launch_bounds__(256)
 qsa_gemm_fp8_batch(const unsigned short* __restrict__ A,  // [M,K] bf16
                    const u32x4* __restrict__ Qf,  // fp8 fragment stream
@@ -67,21 +67,32 @@ qsa_gemm_fp8_batch(const unsigned short* __restrict__ A,  // [M,K] bf16
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int tid = threadIdx.x;
-  const int nt16 = blockIdx.x * 4 + wave;  // this wave's 16-col n-tile
+  // this wave's NT 16-col n-tiles (NT=2 halves the per-WG A re-read,
+  // the L2-bandwidth bound at 64 output columns per workgroup)
+  const int nt16 = blockIdx.x * 4 * NT + wave * NT;
   const long long ks = (long long)blockIdx.y * (K / SPLITK);
   const long long ke = ks + K / SPLITK;
 
-  __shared__ unsigned short atile[R * QSA_GB_PAD];
+  // double-buffered A tile: iteration kb computes on buf[kb&1] while
+  // the NEXT tile's global loads (issued before the compute) land in
+  // registers and drain to buf[(kb+1)&1] after the MFMA sweep — one
+  // barrier per k-step, memory latency hidden under the MFMA stream
+  __shared__ unsigned short atile[2][R * QSA_GB_PAD];
 
-  f32x4 acc[M_FRAGS];
+  f32x4 acc[NT][M_FRAGS];
 #pragma unroll
-  for (int m = 0; m < M_FRAGS; ++m) acc[m] = (f32x4){0.f, 0.f, 0.f, 0.f};
+  for (int t = 0; t < NT; ++t)
+#pragma unroll
+    for (int m = 0; m < M_FRAGS; ++m)
+      acc[t][m] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  // W stream base for this wave's n-tile (16 B per 64-k block; layout
+  // W stream bases for this wave's n-tiles (16 B per 64-k block; layout
   // identical to skinny_gemm_fp8.hip)
-  const u32x4* qbase =
-      Qf + ((long long)nt16 * (K >> 6)) * 64 +
-      (long long)((lane & 15) * 4 + (lane >> 4));
+  const u32x4* qbase[NT];
+#pragma unroll
+  for (int t = 0; t < NT; ++t)
+    qbase[t] = Qf + ((long long)(nt16 + t) * (K >> 6)) * 64 +
+               (long long)((lane & 15) * 4 + (lane >> 4));
 
   // cooperative A fill map: chunk c covers row c/8, 16 B at col (c%8)*8
   constexpr int CHUNKS = R * 8;            // 16-B chunks per 64-k tile
@@ -89,60 +100,124 @@ qsa_gemm_fp8_batch(const unsigned short* __restrict__ A,  // [M,K] bf16
 
   const int col = lane & 15;
   const int hi = lane >> 4;
+  const int frow = tid >> 3;               // this thread's fill row
+  const int fc8 = (tid & 7) * 8;           // and 16-B column offset
+  // (PER_T chunks per thread: rows frow + i*32, same 16-B column)
 
-  for (long long kb = ks >> 6; kb < (ke >> 6); ++kb) {
-    const long long k0 = kb << 6;
-    // ---- fill A tile [R][64] (rows beyond M clamp to M-1) -------------
+  const long long kb0 = ks >> 6, kb1 = ke >> 6;
+  // prologue: fill buf0 with tile kb0, and take kb0's W fragment
+  {
+    const long long k0 = kb0 << 6;
 #pragma unroll
     for (int i = 0; i < PER_T; ++i) {
-      const int c = tid + i * 256;
-      const int row = c >> 3;
-      const int c8 = (c & 7) * 8;
+      const int row = frow + i * 32;
       const int arow = min(row, M - 1);
       const bf16x8 v = *reinterpret_cast<const bf16x8*>(
-          A + (long long)arow * lda + k0 + c8);
-      *reinterpret_cast<bf16x8*>(&atile[row * QSA_GB_PAD + c8]) = v;
+          A + (long long)arow * lda + k0 + fc8);
+      *reinterpret_cast<bf16x8*>(&atile[0][row * QSA_GB_PAD + fc8]) = v;
     }
-    __syncthreads();
+  }
+  u32x4 q[NT];
+#pragma unroll
+  for (int t = 0; t < NT; ++t)
+    q[t] = __builtin_nontemporal_load(qbase[t] + kb0 * 64);
+  __syncthreads();
+
+  for (long long kb = kb0; kb < kb1; ++kb) {
+    const int cur = (int)((kb - kb0) & 1);
+    // ---- issue NEXT tile's global loads (A and W, no wait) ------------
+    bf16x8 stage[PER_T];
+    u32x4 q_next[NT];
+    const bool have_next = (kb + 1) < kb1;
+    if (have_next) {
+      const long long kn = (kb + 1) << 6;
+#pragma unroll
+      for (int t = 0; t < NT; ++t)
+        q_next[t] = __builtin_nontemporal_load(qbase[t] + (kb + 1) * 64);
+#pragma unroll
+      for (int i = 0; i < PER_T; ++i) {
+        const int arow = min(frow + i * 32, M - 1);
+        stage[i] = *reinterpret_cast<const bf16x8*>(
+            A + (long long)arow * lda + kn + fc8);
+      }
+    }
     // ---- this wave's W fragments + MFMA sweep over the m-frags --------
-    const u32x4 q = __builtin_nontemporal_load(qbase + kb * 64);
-    const bf16x8 w0 = qsa_gb_cvt(q.x, q.y);   // k0 .. +32
-    const bf16x8 w1 = qsa_gb_cvt(q.z, q.w);   // k0+32 .. +64
+    // 2-deep software pipeline on the LDS reads: fragment m's reads are
+    // issued two MFMA pairs ahead of their use, so the ~50-cycle LDS
+    // latency hides under the matrix pipe instead of serializing
+    // read -> waitcnt -> MFMA per fragment (the hipcc default here)
+    bf16x8 w0[NT], w1[NT];
+#pragma unroll
+    for (int t = 0; t < NT; ++t) {
+      w0[t] = qsa_gb_cvt(q[t].x, q[t].y);     // k0 .. +32
+      w1[t] = qsa_gb_cvt(q[t].z, q[t].w);     // k0+32 .. +64
+    }
+    const unsigned short* abase = &atile[cur][col * QSA_GB_PAD + hi * 8];
+#define QSA_GB_RD0(m) \
+  (*reinterpret_cast<const bf16x8*>(abase + (m) * 16 * QSA_GB_PAD))
+#define QSA_GB_RD1(m) \
+  (*reinterpret_cast<const bf16x8*>(abase + (m) * 16 * QSA_GB_PAD + 32))
+    bf16x8 p0[3], p1[3];
+    p0[0] = QSA_GB_RD0(0);
+    p1[0] = QSA_GB_RD1(0);
+    if (M_FRAGS > 1) {
+      p0[1] = QSA_GB_RD0(1);
+      p1[1] = QSA_GB_RD1(1);
+    }
 #pragma unroll
     for (int m = 0; m < M_FRAGS; ++m) {
-      const unsigned short* arow0 =
-          &atile[(m * 16 + col) * QSA_GB_PAD + hi * 8];
-      const bf16x8 a0 = *reinterpret_cast<const bf16x8*>(arow0);
-      const bf16x8 a1 = *reinterpret_cast<const bf16x8*>(arow0 + 32);
-      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, w0, acc[m],
-                                                       0, 0, 0);
-      acc[m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, w1, acc[m],
-                                                       0, 0, 0);
+      if (m + 2 < M_FRAGS) {
+        p0[(m + 2) % 3] = QSA_GB_RD0(m + 2);
+        p1[(m + 2) % 3] = QSA_GB_RD1(m + 2);
+      }
+#pragma unroll
+      for (int t = 0; t < NT; ++t) {
+        acc[t][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            p0[m % 3], w0[t], acc[t][m], 0, 0, 0);
+        acc[t][m] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            p1[m % 3], w1[t], acc[t][m], 0, 0, 0);
+      }
+    }
+#undef QSA_GB_RD0
+#undef QSA_GB_RD1
+    // ---- drain staged loads into the other buffer ---------------------
+    if (have_next) {
+#pragma unroll
+      for (int i = 0; i < PER_T; ++i) {
+        const int row = frow + i * 32;
+        *reinterpret_cast<bf16x8*>(
+            &atile[cur ^ 1][row * QSA_GB_PAD + fc8]) = stage[i];
+      }
+#pragma unroll
+      for (int t = 0; t < NT; ++t) q[t] = q_next[t];
     }
     __syncthreads();
   }
 
   // ---- epilogue: C tile row = m*16 + hi*4 + r, col = lane&15 ----------
-  const int ncol = nt16 * 16 + col;
-  if (SPLITK == 1) {
-    const float s = scale[ncol];
 #pragma unroll
-    for (int m = 0; m < M_FRAGS; ++m) {
+  for (int t = 0; t < NT; ++t) {
+    const int ncol = (nt16 + t) * 16 + col;
+    if (SPLITK == 1) {
+      const float s = scale[ncol];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = m * 16 + hi * 4 + r;
-        if (row < M)
-          Cbf[(long long)row * N + ncol] = f32_to_bf16(acc[m][r] * s);
+      for (int m = 0; m < M_FRAGS; ++m) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = m * 16 + hi * 4 + r;
+          if (row < M)
+            Cbf[(long long)row * N + ncol] = f32_to_bf16(acc[t][m][r] * s);
+        }
       }
-    }
-  } else {
-    float* slab = ws + (long long)blockIdx.y * M * N;
+    } else {
+      float* slab = ws + (long long)blockIdx.y * M * N;
 #pragma unroll
-    for (int m = 0; m < M_FRAGS; ++m) {
+      for (int m = 0; m < M_FRAGS; ++m) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = m * 16 + hi * 4 + r;
-        if (row < M) slab[(long long)row * N + ncol] = acc[m][r];
+        for (int r = 0; r < 4; ++r) {
+          const int row = m * 16 + hi * 4 + r;
+          if (row < M) slab[(long long)row * N + ncol] = acc[t][m][r];
+        }
       }
     }
   }
@@ -165,16 +240,20 @@ extern "C" void qsa_gemm_fp8_batch_launch(
     long long lda, int splitk, hipStream_t stream) {
   const u32x4* Q = reinterpret_cast<const u32x4*>(Qf);
   const int mf = (M + 15) / 16;
-#define QSA_CASE(MF, SK)                                                  \
-  if (mf <= MF && splitk == SK) {                                         \
-    hipLaunchKernelGGL((qsa_gemm_fp8_batch<MF, SK>),                      \
-                       dim3(N / 64, SK), dim3(256), 0, stream, A, Q,      \
-                       scale, Cbf, ws, M, N, K, lda);                     \
+  const int nt = (N % 128 == 0 && mf <= 12) ? 2 : 1;
+#define QSA_CASE(MF, SK, NTT)                                             \
+  if (mf <= MF && splitk == SK && nt == NTT) {                            \
+    hipLaunchKernelGGL((qsa_gemm_fp8_batch<MF, SK, NTT>),                 \
+                       dim3(N / (64 * NTT), SK), dim3(256), 0, stream,    \
+                       A, Q, scale, Cbf, ws, M, N, K, lda);               \
     goto reduce;                                                          \
   }
-  QSA_CASE(4, 1) QSA_CASE(8, 1) QSA_CASE(12, 1) QSA_CASE(16, 1)
-  QSA_CASE(4, 2) QSA_CASE(8, 2) QSA_CASE(12, 2) QSA_CASE(16, 2)
-  QSA_CASE(4, 4) QSA_CASE(8, 4) QSA_CASE(12, 4) QSA_CASE(16, 4)
+  QSA_CASE(4, 1, 2) QSA_CASE(8, 1, 2) QSA_CASE(12, 1, 2)
+  QSA_CASE(4, 2, 2) QSA_CASE(8, 2, 2) QSA_CASE(12, 2, 2)
+  QSA_CASE(4, 4, 2) QSA_CASE(8, 4, 2) QSA_CASE(12, 4, 2)
+  QSA_CASE(4, 1, 1) QSA_CASE(8, 1, 1) QSA_CASE(12, 1, 1) QSA_CASE(16, 1, 1)
+  QSA_CASE(4, 2, 1) QSA_CASE(8, 2, 1) QSA_CASE(12, 2, 1) QSA_CASE(16, 2, 1)
+  QSA_CASE(4, 4, 1) QSA_CASE(8, 4, 1) QSA_CASE(12, 4, 1) QSA_CASE(16, 4, 1)
 #undef QSA_CASE
   return;  // unsupported combination (host validates)
 reduce:
