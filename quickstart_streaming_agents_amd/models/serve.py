@@ -53,6 +53,7 @@ class EngineStats:
     cached_prefix_tokens: int = 0
     decode_tokens: int = 0
     decode_steps: int = 0
+    jump_forward_tokens: int = 0
     prefills: int = 0
     prefill_batches: int = 0
     wall_s: float = 0.0
@@ -106,6 +107,10 @@ class Engine:
         # 0 = greedy decisions (degenerate under random-init weights:
         # every episode makes the same choice).
         self.decision_temperature = float(decision_temperature)
+        # jump-forward on fully-forced grammar segments (see _admit);
+        # QSA_NO_JUMP_FORWARD=1 restores serial decode for comparison
+        self.jump_forward = _os.environ.get(
+            "QSA_NO_JUMP_FORWARD", "") != "1"
         self.max_batch = max_batch
         self.max_seq_len = max_seq_len
         self.eos_id = eos_id  # None -> run to max_new_tokens (random weights)
@@ -462,6 +467,20 @@ class Engine:
             seq.out_tokens.append(int(tok))
             if seq.branches is not None:
                 seq.script = seq.branches.get(int(tok))
+                if seq.script is not None and self.jump_forward:
+                    # jump-forward decoding (the SGLang fast-forward on
+                    # grammar-forced segments): once the sampled decision
+                    # commits to a tool branch, EVERY remaining token of
+                    # the turn is grammar-forced — no model choice is
+                    # left, and the forced tokens' KV is never reused
+                    # (the next turn's prompt carries the OBSERVATION,
+                    # not the raw call text, and retire rolls the cache
+                    # back to the prompt) — so the engine emits the
+                    # script directly instead of serial decode steps.
+                    room = seq.max_new_tokens - 1
+                    seq.out_tokens.extend(seq.script[:room])
+                    self.stats.jump_forward_tokens += min(
+                        len(seq.script), room)
             self.running.append(seq)
             self._maybe_finish(seq)
 

@@ -247,3 +247,17 @@ def test_grammar_dedup_bounds_episode_length(tok):
     results, _ = _run_lab1_episodes(tok, n=10, seed=5, max_iters=10)
     assert all(r.status == "SUCCESS" for r in results)
     assert max(r.iterations for r in results) <= 5
+
+
+def test_jump_forward_matches_serial_decode(tok, monkeypatch):
+    """Jump-forward (grammar-forced segments emitted without serial
+    decode; SGLang-style fast-forward) must produce the same episode
+    results as full serial decoding — the forced tokens' values are
+    grammar-determined and their KV is never reused."""
+    with StubMcpServer() as srv:
+        r_jf, _ = _run_lab1_episodes(tok, n=8, seed=13, server=srv)
+        monkeypatch.setenv("QSA_NO_JUMP_FORWARD", "1")
+        r_sd, _ = _run_lab1_episodes(tok, n=8, seed=13, server=srv)
+    assert [(r.status, r.iterations, r.tool_calls, r.response)
+            for r in r_jf] == \
+        [(r.status, r.iterations, r.tool_calls, r.response) for r in r_sd]
