@@ -26,7 +26,6 @@ from dataclasses import dataclass, field
 import torch
 
 from .llama import LlamaModel
-from .tokenizer import HashTokenizer
 
 
 @dataclass
