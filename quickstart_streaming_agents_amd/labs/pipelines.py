@@ -271,7 +271,17 @@ def lab2_build_index(broker: Broker, embedder) -> VectorIndex:
     docs = [d for _, d in AvroConsumer(broker, "documents",
                                        schemas.DOCUMENTS).poll()]
     index = VectorIndex()
-    index.add_documents(docs, embedder)
+    need_embed = [d for d in docs if d.get("embedding") is None]
+    if need_embed and hasattr(embedder, "embed_batch"):
+        # embed batches and index-add overlap on separate HIP streams
+        # (runtime/streams.py; SURVEY 2.5 operator-pipeline row)
+        from ..runtime.streams import pipelined_embed_index
+        pipelined_embed_index(embedder, index, need_embed)
+        pre = [d for d in docs if d.get("embedding") is not None]
+        if pre:
+            index.add_documents(pre, embedder)
+    else:
+        index.add_documents(docs, embedder)
     return index
 
 

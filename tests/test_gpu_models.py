@@ -493,3 +493,39 @@ def test_full_size_llama3_8b_engine():
     assert all(len(o) == 12 for o in out_g)
     # all tokens inside the vocab
     assert all(0 <= t < 128256 for o in out_g for t in o)
+
+
+def test_stream_pipeline_gpu_matches_sequential():
+    """StreamPipeline on real HIP streams (one per stage, event-chained)
+    == sequential execution; lab2 index build through the pipelined path
+    == sequential add_documents."""
+    from quickstart_streaming_agents_amd.runtime.streams import (
+        StreamPipeline, pipelined_embed_index)
+    torch.manual_seed(0)
+    w1 = torch.randn(256, 256, device="cuda:0")
+    w2 = torch.randn(256, 256, device="cuda:0")
+    batches = [torch.randn(64, 256, device="cuda:0") for _ in range(12)]
+    pipe = StreamPipeline([lambda x: x @ w1,
+                           lambda x: torch.relu(x),
+                           lambda x: x @ w2])
+    assert pipe.use_streams
+    got = pipe.run(batches)
+    for b, g in zip(batches, got):
+        ref = torch.relu(b @ w1) @ w2
+        torch.testing.assert_close(g, ref)
+
+    from quickstart_streaming_agents_amd.labs import datagen
+    from quickstart_streaming_agents_amd.models.encoder import (
+        EmbeddingEncoder, EncoderConfig)
+    from quickstart_streaming_agents_amd.vector.index import VectorIndex
+    enc = EmbeddingEncoder(EncoderConfig(n_layers=2), device="cuda:0",
+                           seed=3)
+    docs = [{"document_id": f"d{i}", "chunk": c["chunk"]}
+            for i, c in enumerate(datagen.lab2_documents(n_chunks=24))]
+    seq = VectorIndex()
+    seq.add_documents([dict(d) for d in docs], enc)
+    pip = VectorIndex()
+    pipelined_embed_index(enc, pip, [dict(d) for d in docs], batch_size=5)
+    q = enc.embed_batch(["anomaly windows in flink"])[0]
+    assert [h.document_id for h in seq.search(q, 3)] == \
+        [h.document_id for h in pip.search(q, 3)]
