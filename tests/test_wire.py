@@ -310,3 +310,42 @@ def test_keyed_produce_consume_roundtrip():
     # without a key schema the raw wire bytes come back untouched
     raw = AvroConsumer(broker, "t", vs).poll()
     assert isinstance(raw[0][0].key, (bytes, bytearray))
+
+
+def test_native_codec_thread_stress():
+    """Race-detection smoke for the NATIVE code paths (SURVEY §5 aux row):
+    16 threads hammer one shared AvroCodec (serialize+deserialize) and
+    per-thread codecs concurrently; every round-trip must be exact.
+    (The C++ codec must be stateless per call — this catches shared
+    mutable state the way the reference's hand-rolled locks were tested.)"""
+    import threading
+
+    from quickstart_streaming_agents_amd.ops import have_ext
+    if not have_ext():
+        pytest.skip("extension not built")
+    from quickstart_streaming_agents_amd.labs.schemas import ORDERS
+    from quickstart_streaming_agents_amd.ops import ext
+    from quickstart_streaming_agents_amd.wire.avro import Schema
+    codec = ext().AvroCodec(Schema(ORDERS).defn)
+    rows = [{"order_id": f"o{i}", "customer_id": f"c{i % 7}",
+             "product_id": f"p{i % 17}", "price": float(i) + 0.5,
+             "order_ts": i * 1000} for i in range(200)]
+    errors: list = []
+
+    def worker(tid: int):
+        try:
+            for rep in range(30):
+                for i, r in enumerate(rows):
+                    raw = bytes(codec.serialize(5, r))
+                    sid, back = codec.deserialize(raw)
+                    assert sid == 5 and back == r, (tid, rep, i)
+        except Exception as e:   # pragma: no cover
+            errors.append(e)
+
+    threads = [threading.Thread(target=worker, args=(t,))
+               for t in range(16)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    assert not errors, errors
