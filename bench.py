@@ -139,7 +139,14 @@ def main():
                        max_new_tokens=args.decode_tokens,
                        tool_schemas=tool_schemas)
                for o in orders]
-        results = run_episodes(eps, llm, tool_fn)
+        if args.stub_llm:
+            results = run_episodes(eps, llm, tool_fn)
+        else:
+            # event-driven continuous batching: turns join the running
+            # decode batch as tool I/O completes (no round barriers)
+            from quickstart_streaming_agents_amd.agents.schedule import \
+                run_episodes_continuous
+            results = run_episodes_continuous(eps, llm, tool_fn)
         assert len(results) == len(orders)
         if record:
             all_results.extend(results)

@@ -88,3 +88,66 @@ def _safe_tool(tool: Callable[[str, dict], str], name: str, args: dict) -> str:
         return tool(name, args)
     except Exception as e:
         return f"__error__ {e}"
+
+
+def run_episodes_continuous(episodes: list[Generator], llm,
+                            tool: Callable[[str, dict], str],
+                            max_tool_workers: int = 16,
+                            decode_chunk: int = 16) -> list[EpisodeResult]:
+    """Event-driven episode scheduler over the continuous-batching engine.
+
+    `run_episodes` advances episodes in lockstep ROUNDS: every episode's
+    turn N completes before any turn N+1 starts, so late rounds decode
+    tiny batches (the engine is continuous but the driver isn't).  This
+    scheduler is truly continuous: episode turns are submitted the
+    moment they are ready (episode start, tool-future completion), the
+    engine runs in short admit+decode slices (Engine.run_chunk), and
+    completed turns re-enter their episodes immediately — a finish turn
+    of one episode decodes in the same batch as turn-2 prefills of
+    another, and per-decision latency stops being quantized to rounds.
+
+    Requires the conversation-aware EngineLLM (submit_turn/pop_finished).
+    """
+    import time as _time
+    results: dict[int, EpisodeResult] = {}
+    started: dict[int, float] = {}
+    tool_futs: dict[int, object] = {}
+    n = len(episodes)
+
+    def _advance(idx: int, send_value: Any, pool) -> None:
+        try:
+            req = episodes[idx].send(send_value)
+        except StopIteration as stop:
+            results[idx] = stop.value
+            if stop.value is not None:
+                stop.value.latency_s = _time.perf_counter() - started[idx]
+            llm.release(idx)
+            return
+        if req[0] == "llm":
+            grammar = req[3] if len(req) > 3 else None
+            llm.submit_turn(idx, req[1], req[2], grammar)
+        elif req[0] == "tool":
+            tool_futs[idx] = pool.submit(_safe_tool, tool, req[1], req[2])
+        else:
+            raise ValueError(f"unknown request {req[0]!r}")
+
+    eng = llm.engine
+    with ThreadPoolExecutor(max_workers=max_tool_workers) as pool:
+        for idx in range(n):
+            started[idx] = _time.perf_counter()
+            _advance(idx, None, pool)
+        while len(results) < n:
+            # tool completions first: their turns join this slice's admit
+            done_futs = [i for i, f in tool_futs.items() if f.done()]
+            for i in done_futs:
+                f = tool_futs.pop(i)
+                _advance(i, f.result(), pool)
+            if eng.pending or eng.running:
+                eng.run_chunk(decode_chunk)
+                for cid, text in llm.pop_finished():
+                    _advance(cid, text, pool)
+            elif tool_futs:
+                # nothing on the GPU: block briefly on tool I/O
+                _time.sleep(0.0005)
+    llm.release_all()
+    return [results[i] for i in range(n)]

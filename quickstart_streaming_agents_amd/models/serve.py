@@ -120,6 +120,7 @@ class Engine:
         self._next_id = 0
         self.pending: list[Sequence] = []
         self.running: list[Sequence] = []
+        self.finished: list[Sequence] = []   # drained by run_chunk callers
         self.stats = EngineStats()
         # hipGraph-captured decode step (launch-bound otherwise: ~300
         # kernel/GEMM launches per step across 32 layers).  TP ranks
@@ -498,7 +499,35 @@ class Engine:
                 # roll back to the prompt: the next turn extends the prompt,
                 # not the decoded tokens
                 self.kv.truncate(s.seq_id, len(s.prompt))
+        self.finished.extend(done)
         self.running = [s for s in self.running if not s.done]
+
+    def run_chunk(self, max_run: int = 16) -> int:
+        """One continuous-batching slice: admit whatever fits, then up to
+        `max_run` decode steps (one graph run).  Returns remaining work.
+        The event-driven scheduler (agents/schedule.py
+        run_episodes_continuous) interleaves these slices with tool I/O
+        completions, so late turns join the running batch instead of
+        waiting for a global round barrier."""
+        while self.pending:
+            before = len(self.pending)
+            self._admit()
+            if len(self.pending) == before:
+                break
+        batch = [s for s in self.running if not s.done]
+        if batch:
+            run = min(min(s.max_new_tokens - len(s.out_tokens)
+                          for s in batch), max_run, self.MAX_RUN)
+            if run > 0:
+                if self.use_graph:
+                    self._decode_run_graph(batch, run)
+                else:
+                    self._decode_run_eager(batch, run)
+        self._retire()
+        if self.pending and not self.running and not batch:
+            raise MemoryError("decode stalled: pending prompts cannot be "
+                              "admitted (KV pages exhausted?)")
+        return len(self.running) + len(self.pending)
 
     def step(self) -> int:
         """Admit + one decode step; returns remaining work count."""
@@ -519,6 +548,7 @@ class Engine:
         return len(self.running) + len(self.pending)
 
     def run_to_completion(self) -> None:
+        self.finished.clear()   # batch callers don't drain it
         t0 = time.perf_counter()
         if self.use_graph:
             while self.pending or self.running:
@@ -580,6 +610,7 @@ class EngineLLM:
         self.engine = engine
         self.tokenizer = tokenizer
         self._convs: dict = {}
+        self._by_seq: dict = {}
 
     def __call__(self, prompts: list[str], max_new_tokens: list[int],
                  conv_ids: list | None = None,
@@ -602,9 +633,37 @@ class EngineLLM:
         self.engine.run_to_completion()
         return [self.tokenizer.decode(s.out_tokens) for s in seqs]
 
+    # ---- event-driven (continuous) scheduling interface ----------------
+    def submit_turn(self, conv_id, prompt: str, max_new_tokens: int,
+                    grammar=None) -> None:
+        """Queue one agent turn WITHOUT running the engine — the
+        continuous scheduler (agents/schedule.py) interleaves engine
+        slices with tool I/O so turns join the running batch as they
+        become ready."""
+        from .grammar import compile_grammar
+        constraint = compile_grammar(grammar, self.tokenizer)             if grammar is not None else None
+        prev = self._convs.get(conv_id)
+        enc = self.tokenizer.encode(prompt)
+        seq = self.engine.submit(enc, max_new_tokens, continue_from=prev,
+                                 keep_alive=True, constraint=constraint)
+        self._convs[conv_id] = seq
+        self._by_seq[id(seq)] = conv_id
+
+    def pop_finished(self) -> list[tuple]:
+        """Drain (conv_id, decoded text) for turns completed since the
+        last call (engine.finished is filled by retire)."""
+        out = []
+        for seq in self.engine.finished:
+            cid = self._by_seq.pop(id(seq), None)
+            if cid is not None:
+                out.append((cid, self.tokenizer.decode(seq.out_tokens)))
+        self.engine.finished.clear()
+        return out
+
     def release(self, conv_id) -> None:
         seq = self._convs.pop(conv_id, None)
         if seq is not None:
+            self._by_seq.pop(id(seq), None)
             self.engine.release(seq)
 
     def release_all(self) -> None:

@@ -261,3 +261,48 @@ def test_jump_forward_matches_serial_decode(tok, monkeypatch):
     assert [(r.status, r.iterations, r.tool_calls, r.response)
             for r in r_jf] == \
         [(r.status, r.iterations, r.tool_calls, r.response) for r in r_sd]
+
+
+def test_continuous_scheduler_completes_episodes(tok):
+    """run_episodes_continuous (event-driven: turns join the running
+    batch as tool futures land, Engine.run_chunk slices) completes every
+    episode within caps, with real tool calls and per-decision
+    latencies; deterministic across runs."""
+    from quickstart_streaming_agents_amd.agents.schedule import \
+        run_episodes_continuous
+
+    def run_once(srv, seed):
+        torch.manual_seed(seed)
+        eng = _engine(tok)
+        llm = EngineLLM(eng, tok)
+        client = McpClient(srv.mcp_endpoint)
+        schemas = {t["name"]: t.get("inputSchema", {})
+                   for t in client.tools_list()}
+        tool_fn = pipelines.mcp_tool_fn(client)
+        tools = ToolSet("lab1_remote_mcp",
+                        allowed_tools=("http_get", "send_email"))
+        agent = AgentSpec("price_match_agent", "m",
+                          pipelines.LAB1_AGENT_PROMPT, tools,
+                          max_iterations=6, max_consecutive_failures=2)
+        products = datagen.lab1_products()
+        eps = []
+        for i in range(10):
+            p = products[i % len(products)]
+            o = {"order_id": f"ORD-{i:04d}",
+                 "product_name": p["product_name"],
+                 "order_price": p["price"],
+                 "customer_email": f"u{i}@example.com"}
+            eps.append(episode(
+                agent, pipelines.lab1_user_prompt(
+                    o, f"{srv.base_url}/competitor", o["customer_email"]),
+                max_new_tokens=24, tool_schemas=schemas))
+        return run_episodes_continuous(eps, llm, tool_fn)
+
+    with StubMcpServer() as srv:
+        r1 = run_once(srv, 21)
+    assert len(r1) == 10
+    assert all(r.status in ("SUCCESS", "FAILED") for r in r1)
+    assert all(r.iterations <= 6 for r in r1)
+    assert all(r.latency_s > 0 for r in r1)
+    assert sum(r.tool_calls for r in r1) >= 1
+    assert any(r.status == "SUCCESS" for r in r1)
