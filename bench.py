@@ -50,6 +50,9 @@ def parse_args():
     p.add_argument("--decode-tokens", type=int, default=64,
                    help="decode tokens per agent LLM turn")
     p.add_argument("--max-seq-len", type=int, default=2048)
+    p.add_argument("--decode-chunk", type=int, default=16,
+                   help="decode steps per engine slice in the continuous "
+                        "scheduler")
     p.add_argument("--stub-llm", action="store_true",
                    help="CPU contract check: stub LLM instead of the GPU engine")
     return p.parse_args()
@@ -146,7 +149,8 @@ def main():
             # decode batch as tool I/O completes (no round barriers)
             from quickstart_streaming_agents_amd.agents.schedule import \
                 run_episodes_continuous
-            results = run_episodes_continuous(eps, llm, tool_fn)
+            results = run_episodes_continuous(
+                eps, llm, tool_fn, decode_chunk=args.decode_chunk)
         assert len(results) == len(orders)
         if record:
             all_results.extend(results)
