@@ -50,7 +50,7 @@ def parse_args():
     p.add_argument("--decode-tokens", type=int, default=64,
                    help="decode tokens per agent LLM turn")
     p.add_argument("--max-seq-len", type=int, default=2048)
-    p.add_argument("--decode-chunk", type=int, default=16,
+    p.add_argument("--decode-chunk", type=int, default=8,
                    help="decode steps per engine slice in the continuous "
                         "scheduler")
     p.add_argument("--stub-llm", action="store_true",

@@ -93,7 +93,7 @@ def _safe_tool(tool: Callable[[str, dict], str], name: str, args: dict) -> str:
 def run_episodes_continuous(episodes: list[Generator], llm,
                             tool: Callable[[str, dict], str],
                             max_tool_workers: int = 16,
-                            decode_chunk: int = 16) -> list[EpisodeResult]:
+                            decode_chunk: int = 8) -> list[EpisodeResult]:
     """Event-driven episode scheduler over the continuous-batching engine.
 
     `run_episodes` advances episodes in lockstep ROUNDS: every episode's
