@@ -237,7 +237,15 @@ def drive_episode(ep: Generator, llm: Callable[[str, int], str],
         req = ep.send(None)
         while True:
             if req[0] == "llm":
-                result = llm(req[1], req[2])
+                if len(req) > 3 and req[3] is not None:
+                    # grammar-capable backends take the TurnGrammar;
+                    # plain 2-arg callables (stub LLMs) decode free
+                    try:
+                        result = llm(req[1], req[2], req[3])
+                    except TypeError:
+                        result = llm(req[1], req[2])
+                else:
+                    result = llm(req[1], req[2])
             elif req[0] == "tool":
                 try:
                     result = tool(req[1], req[2])

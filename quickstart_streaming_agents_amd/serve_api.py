@@ -248,14 +248,24 @@ def build_lab_app(device: str = "cpu", model: str | None = None,
     if 2 in labs:
         dep.datagen(2)
     server = StubMcpServer().start()
-    tool_fn = mcp_tool_fn(McpClient(server.mcp_endpoint))
+    client = McpClient(server.mcp_endpoint)
+    tool_fn = mcp_tool_fn(client)
+    # tool schemas -> per-turn grammars: served agents are model-driven
+    # (grammar-constrained tool-call decisions, models/grammar.py)
+    tool_schemas = {t["name"]: t.get("inputSchema", {})
+                    for t in client.tools_list()}
     llm_batch = dep.llm()
 
     def make_agent_fn(spec):
+        def call_llm(p, t, g=None):
+            if g is not None and hasattr(llm_batch, "submit_turn"):
+                return llm_batch([p], [t], None, [g])[0]   # EngineLLM
+            return llm_batch([p], [t])[0]
+
         def run(prompt: str, record_key=None):
             return drive_episode(
-                episode(spec, prompt),
-                lambda p, t: llm_batch([p], [t])[0], tool_fn)
+                episode(spec, prompt, tool_schemas=tool_schemas),
+                call_llm, tool_fn)
         return run
 
     agents = {name: make_agent_fn(dep.catalog.agent_spec(name))
