@@ -134,6 +134,7 @@ class Engine:
         self._graph = None
         self._graph_failed = False
         self._gbuf: dict = {}
+        self._side_stream = None   # prefill stream for overlapped chunks
         # free-text sampling mask: tokens outside [lo, hi) (plus EOS) are
         # never emitted — keeps random-init decode inside the tokenizer's
         # trained vocab so outputs are real text (models/grammar.py)
@@ -298,6 +299,12 @@ class Engine:
         if not self._ensure_graph():
             self._decode_run_eager(batch, run)
             return
+        self._graph_run_begin(batch, run)
+        self._graph_run_end(batch, run)
+
+    def _graph_run_begin(self, batch: list[Sequence], run: int) -> None:
+        """Fill the graph buffers and queue `run` replays (async on the
+        current stream — no host sync; _graph_run_end collects)."""
         gb = self._gbuf
         n = len(batch)
         dev = self.model.device
@@ -345,12 +352,16 @@ class Engine:
         gb["ctr"].zero_()
         if _TIMING:
             torch.cuda.synchronize()
-            _t0 = time.perf_counter()
+            self._t_graph0 = time.perf_counter()
         for _ in range(run):
             self._graph.replay()
+
+    def _graph_run_end(self, batch: list[Sequence], run: int) -> None:
+        gb = self._gbuf
+        n = len(batch)
         if _TIMING:
             torch.cuda.synchronize()
-            self.stats.decode_s += time.perf_counter() - _t0
+            self.stats.decode_s += time.perf_counter() - self._t_graph0
         hist = gb["hist"][:run, :n].t().tolist()  # one sync
         for s, toks in zip(batch, hist):
             new = [int(t) for t in toks]
@@ -503,26 +514,47 @@ class Engine:
         self.running = [s for s in self.running if not s.done]
 
     def run_chunk(self, max_run: int = 16) -> int:
-        """One continuous-batching slice: admit whatever fits, then up to
-        `max_run` decode steps (one graph run).  Returns remaining work.
-        The event-driven scheduler (agents/schedule.py
-        run_episodes_continuous) interleaves these slices with tool I/O
-        completions, so late turns join the running batch instead of
-        waiting for a global round barrier."""
-        while self.pending:
-            before = len(self.pending)
-            self._admit()
-            if len(self.pending) == before:
-                break
+        """One continuous-batching slice: queue up to `max_run` decode
+        steps for the running batch (one graph run, async), admit
+        pending prompts WHILE the graph replays (prefill runs on a side
+        HIP stream, overlapping the decode — the two touch disjoint
+        sequences and KV pages), then collect the decode results.
+        Returns remaining work count.  The event-driven scheduler
+        (agents/schedule.py run_episodes_continuous) interleaves these
+        slices with tool I/O completions, so late turns join the running
+        batch instead of waiting for a global round barrier."""
         batch = [s for s in self.running if not s.done]
-        if batch:
+        launched = 0
+        use_overlap = self.use_graph and batch and self._ensure_graph()
+        if use_overlap:
             run = min(min(s.max_new_tokens - len(s.out_tokens)
                           for s in batch), max_run, self.MAX_RUN)
             if run > 0:
-                if self.use_graph:
-                    self._decode_run_graph(batch, run)
-                else:
-                    self._decode_run_eager(batch, run)
+                for s in batch:
+                    self.kv.extend(s.seq_id,
+                                   len(s.prompt) + len(s.out_tokens))
+                self._graph_run_begin(batch, run)
+                launched = run
+        # admissions overlap the queued replays: _admit's own sampling
+        # sync drains only the side stream
+        while self.pending:
+            before = len(self.pending)
+            if launched:
+                if self._side_stream is None:
+                    self._side_stream = torch.cuda.Stream()
+                with torch.cuda.stream(self._side_stream):
+                    self._admit()
+            else:
+                self._admit()
+            if len(self.pending) == before:
+                break
+        if launched:
+            self._graph_run_end(batch, launched)
+        elif batch:
+            run = min(min(s.max_new_tokens - len(s.out_tokens)
+                          for s in batch), max_run, self.MAX_RUN)
+            if run > 0:
+                self._decode_run_eager(batch, run)
         self._retire()
         if self.pending and not self.running and not batch:
             raise MemoryError("decode stalled: pending prompts cannot be "
