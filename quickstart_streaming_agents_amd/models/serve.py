@@ -514,47 +514,31 @@ class Engine:
         self.running = [s for s in self.running if not s.done]
 
     def run_chunk(self, max_run: int = 16) -> int:
-        """One continuous-batching slice: queue up to `max_run` decode
-        steps for the running batch (one graph run, async), admit
-        pending prompts WHILE the graph replays (prefill runs on a side
-        HIP stream, overlapping the decode — the two touch disjoint
-        sequences and KV pages), then collect the decode results.
-        Returns remaining work count.  The event-driven scheduler
+        """One continuous-batching slice: admit whatever fits, then up
+        to `max_run` decode steps (one graph run).  Returns remaining
+        work count.  The event-driven scheduler
         (agents/schedule.py run_episodes_continuous) interleaves these
         slices with tool I/O completions, so late turns join the running
         batch instead of waiting for a global round barrier."""
-        batch = [s for s in self.running if not s.done]
-        launched = 0
-        use_overlap = self.use_graph and batch and self._ensure_graph()
-        if use_overlap:
-            run = min(min(s.max_new_tokens - len(s.out_tokens)
-                          for s in batch), max_run, self.MAX_RUN)
-            if run > 0:
-                for s in batch:
-                    self.kv.extend(s.seq_id,
-                                   len(s.prompt) + len(s.out_tokens))
-                self._graph_run_begin(batch, run)
-                launched = run
-        # admissions overlap the queued replays: _admit's own sampling
-        # sync drains only the side stream
+        # NOTE: an overlapped variant (admissions on a side HIP stream
+        # while the decode graph replays) was measured to hang
+        # NON-deterministically at bench scale — the prefill's H2D /
+        # allocator traffic races the replaying graph's memory pool on
+        # this stack — so admissions and decode run sequentially.
         while self.pending:
             before = len(self.pending)
-            if launched:
-                if self._side_stream is None:
-                    self._side_stream = torch.cuda.Stream()
-                with torch.cuda.stream(self._side_stream):
-                    self._admit()
-            else:
-                self._admit()
+            self._admit()
             if len(self.pending) == before:
                 break
-        if launched:
-            self._graph_run_end(batch, launched)
-        elif batch:
+        batch = [s for s in self.running if not s.done]
+        if batch:
             run = min(min(s.max_new_tokens - len(s.out_tokens)
                           for s in batch), max_run, self.MAX_RUN)
             if run > 0:
-                self._decode_run_eager(batch, run)
+                if self.use_graph:
+                    self._decode_run_graph(batch, run)
+                else:
+                    self._decode_run_eager(batch, run)
         self._retire()
         if self.pending and not self.running and not batch:
             raise MemoryError("decode stalled: pending prompts cannot be "
