@@ -306,3 +306,21 @@ def test_continuous_scheduler_completes_episodes(tok):
     assert all(r.latency_s > 0 for r in r1)
     assert sum(r.tool_calls for r in r1) >= 1
     assert any(r.status == "SUCCESS" for r in r1)
+
+
+def test_bpe_round_trip_fuzz(tok):
+    """Property: decode(encode(s)) == s for arbitrary unicode strings."""
+    import random
+    rng = random.Random(7)
+    pools = [
+        lambda: chr(rng.randrange(32, 127)),
+        lambda: chr(rng.randrange(0x80, 0x2FF)),
+        lambda: chr(rng.randrange(0x4E00, 0x4F00)),   # CJK
+        lambda: rng.choice("\n\t  \U0001F600"),
+    ]
+    for _ in range(200):
+        s = "".join(rng.choice(pools)() for _ in range(rng.randrange(0, 60)))
+        assert tok.decode(tok.encode(s, bos=False)) == s
+    # specials embedded mid-string survive too
+    s = "a<|finish|>b<|tool_3|>\nc"
+    assert tok.decode(tok.encode(s, bos=False)) == s
