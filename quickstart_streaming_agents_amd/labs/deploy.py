@@ -178,10 +178,13 @@ class Deployment:
         policies: dict = {}
         indexes: dict = {}
         tool_fn = None
+        tool_schemas = None
         if mcp_server is not None:
             from ..agents.mcp import McpClient
-            tool_fn = pipelines.mcp_tool_fn(
-                McpClient(mcp_server.mcp_endpoint))
+            client = McpClient(mcp_server.mcp_endpoint)
+            tool_fn = pipelines.mcp_tool_fn(client)
+            tool_schemas = {t["name"]: t.get("inputSchema", {})
+                            for t in client.tools_list()}
         if lab == 1:
             url = f"{mcp_server.base_url}/competitor" if mcp_server else ""
             email = "customer@example.com"
@@ -260,7 +263,8 @@ class Deployment:
         return SqlExecutor(self.catalog, self.broker, schemas=topic_schemas,
                            embedder=self.embedder, indexes=indexes,
                            llm_batch=self.llm(), tool_fn=tool_fn,
-                           bindings=bindings, agent_policies=policies)
+                           bindings=bindings, agent_policies=policies,
+                           tool_schemas=tool_schemas)
 
     def run_sql(self, lab: int, mcp_server=None) -> list[dict]:
         """Run one lab end-to-end through the GENERIC SQL executor (the

@@ -491,7 +491,8 @@ class SqlExecutor:
                  llm_batch=None, tool_fn=None,
                  bindings: dict[str, Callable] | None = None,
                  agent_policies: dict[str, Callable] | None = None,
-                 max_new_tokens: int = 96, tracer=None):
+                 max_new_tokens: int = 96, tracer=None,
+                 tool_schemas: dict | None = None):
         self.catalog = catalog
         self.broker = broker
         self.schemas = schemas or {}
@@ -501,6 +502,10 @@ class SqlExecutor:
         self.tool_fn = tool_fn
         self.ev = Evaluator(bindings)
         self.agent_policies = agent_policies or {}
+        # MCP tool schemas (tools/list): agents WITHOUT an installed
+        # scripted policy run model-driven via grammar-constrained
+        # decoding (models/grammar.py), like the bench and serving paths
+        self.tool_schemas = tool_schemas
         self.max_new_tokens = max_new_tokens
         if tracer is None:
             from ..runtime.trace import Tracer
@@ -934,8 +939,10 @@ class SqlExecutor:
             prompt = str(self.ev.eval(args[1], r))
             policy = policy_fn(lambda n, _r=r: _r.resolve(n)) \
                 if policy_fn else None
+            schemas = self.tool_schemas if policy is None else None
             eps.append(episode(spec, prompt, policy=policy, debug=debug,
-                               max_new_tokens=self.max_new_tokens))
+                               max_new_tokens=self.max_new_tokens,
+                               tool_schemas=schemas))
         if self.llm_batch is None:
             raise SqlExecError("no LLM configured for AI_RUN_AGENT")
         results = run_episodes(eps, self.llm_batch,

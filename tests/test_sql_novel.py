@@ -188,3 +188,46 @@ def test_cli_sql_lab_context_runs_ai_statements(monkeypatch, capsys):
     assert rc == 0
     out = capsys.readouterr().out
     assert "smoke: 1 rows" in out
+
+
+def test_ai_run_agent_model_driven_without_policy():
+    """AI_RUN_AGENT with NO installed policy runs the model-driven
+    grammar path when the executor has MCP tool schemas: decisions come
+    from the (stub-deterministic) llm_batch's grammar support or fall
+    back to free decode; episodes still complete within caps."""
+    from quickstart_streaming_agents_amd.agents.mcp import (McpClient,
+                                                            StubMcpServer)
+    from quickstart_streaming_agents_amd.labs import pipelines
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+    with StubMcpServer() as srv:
+        client = McpClient(srv.mcp_endpoint)
+        schemas = {t["name"]: t.get("inputSchema", {})
+                   for t in client.tools_list()}
+        cat = Catalog()
+        cat.execute("""
+        CREATE CONNECTION `mcp` WITH ('type'='MCP_SERVER',
+                                      'endpoint'='stub://local');
+        CREATE MODEL m INPUT (prompt STRING) OUTPUT (response STRING)
+          WITH ('provider'='local');
+        CREATE TOOL t USING CONNECTION `mcp`
+          WITH ('type'='mcp', 'allowed_tools'='http_get');
+        CREATE AGENT a USING MODEL m USING PROMPT 'fetch the page'
+          USING TOOLS t WITH ('max_iterations'='4');
+        CREATE TABLE src (q STRING);
+        CREATE TABLE out AS
+          SELECT s.q AS q, agent_result.status AS status
+          FROM src s, LATERAL TABLE(AI_RUN_AGENT('a', s.q))
+               AS agent_result(status, response);
+        """)
+        broker = Broker()
+        broker.create_topic("src")
+        broker.topics["src"].append(
+            {"q": f"please fetch {srv.base_url}/competitor"}, partition=0)
+        ex = SqlExecutor(cat, broker, llm_batch=pipelines.StubLLM(),
+                         tool_fn=pipelines.mcp_tool_fn(client),
+                         tool_schemas=schemas)
+        rows = ex.run_table("out")
+        assert len(rows) == 1
+        assert rows[0]["status"] in ("SUCCESS", "FAILED")
