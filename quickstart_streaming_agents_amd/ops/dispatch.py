@@ -248,9 +248,13 @@ def window_agg(ts: torch.Tensor, key: torch.Tensor,
 
 def hash_build(keys: torch.Tensor, ts: torch.Tensor):
     """(i64 keys, i64 event ts) -> opaque latest-per-key table.
-    GPU: (tkeys, tpay) HBM tensors; CPU: a dict of key -> (ts, row)."""
+    GPU: (tkeys, tpay) HBM tensors; CPU: a dict of key -> (ts, row).
+    Event times clamp to [0, 2^40): the GPU payload packs (ts << 24 |
+    row), so negative or far-future timestamps would corrupt ordering —
+    clamping keeps CPU/GPU semantics identical."""
+    ts = ts.clamp(min=0, max=(1 << 40) - 1)
     if _cuda(keys):
-        return tuple(ext().hash_build(keys, ts))
+        return tuple(ext().hash_build(keys, ts.contiguous()))
     table: dict = {}
     for i, (k, t) in enumerate(zip(keys.tolist(), ts.tolist())):
         prev = table.get(k)
