@@ -49,6 +49,17 @@ class StreamPipeline:
             return out
 
         S = len(self.stages)
+
+        def _record(v, stream) -> None:
+            # a tensor crossing streams must be pinned to the consuming
+            # stream, or the caching allocator may recycle its memory
+            # once the PRODUCING stream's work retires
+            if torch.is_tensor(v) and v.is_cuda:
+                v.record_stream(stream)
+            elif isinstance(v, (list, tuple)):
+                for x in v:
+                    _record(x, stream)
+
         # in_flight[s] = (value, event) waiting to enter stage s+1
         results: list[Any] = []
         # software pipeline: advance the deepest stages first each tick
@@ -73,6 +84,7 @@ class StreamPipeline:
                     slots[s - 1] = None
                     with torch.cuda.stream(self.streams[s]):
                         self.streams[s].wait_event(ev)
+                        _record(v, self.streams[s])
                         v2 = self.stages[s](v)
                         ev2 = torch.cuda.Event()
                         ev2.record(self.streams[s])
@@ -82,6 +94,7 @@ class StreamPipeline:
                 v, ev = slots[S - 1]
                 slots[S - 1] = None
                 ev.synchronize()
+                _record(v, torch.cuda.current_stream())
                 results.append(v)
                 done += 1
         return results
