@@ -157,13 +157,16 @@ class Deployment:
     FINAL_TABLE = {1: "price_match_results", 2: "search_results_response",
                    3: "completed_actions", 4: "claims_reviewed"}
 
-    def sql_executor(self, lab: int, mcp_server=None):
+    def sql_executor(self, lab: int, mcp_server=None,
+                     scripted_policies: bool = True):
         """Build a generic SqlExecutor for one lab's catalog: the same
         statements labs/sql/*.sql declare, executed by sql/exec.py instead
         of the hand-fused pipelines.  Prompt expressions the reference
         inlines as giant CONCATs are supplied as named bindings; scripted
-        episode policies (the deterministic CPU stand-ins for a real
-        tool-calling LLM) come from agent_policies."""
+        episode policies (the deterministic content stand-ins the lab
+        CONTRACT tests assert against) come from agent_policies.
+        scripted_policies=False drops them: AI_RUN_AGENT then runs the
+        model-driven grammar path (like the bench and serving API)."""
         from ..sql.exec import SqlExecutor
         from . import schemas
 
@@ -260,6 +263,8 @@ class Deployment:
                         f"Question: {res('query')}\nAnswer:")
 
             bindings["rag_prompt"] = _rag
+        if not scripted_policies:
+            policies = {}
         return SqlExecutor(self.catalog, self.broker, schemas=topic_schemas,
                            embedder=self.embedder, indexes=indexes,
                            llm_batch=self.llm(), tool_fn=tool_fn,

@@ -529,3 +529,23 @@ def test_stream_pipeline_gpu_matches_sequential():
     q = enc.embed_batch(["anomaly windows in flink"])[0]
     assert [h.document_id for h in seq.search(q, 3)] == \
         [h.document_id for h in pip.search(q, 3)]
+
+
+def test_generic_sql_lab1_model_driven_on_engine():
+    """The GENERIC SQL executor runs lab1 end-to-end on the GPU engine
+    with MODEL-DRIVEN agents (no scripted policies): grammar-constrained
+    decisions through the real decode path, one result row per enriched
+    order, status from the episode machine."""
+    import torch as T
+    T.manual_seed(99)
+    from quickstart_streaming_agents_amd.agents.mcp import StubMcpServer
+    from quickstart_streaming_agents_amd.labs.deploy import Deployment
+    dep = Deployment(labs=(1,), device="cuda:0", model="tiny")
+    dep.datagen(1)
+    with StubMcpServer() as srv:
+        ex = dep.sql_executor(1, mcp_server=srv, scripted_policies=False)
+        rows = ex.run_table("price_match_results")
+    assert len(rows) >= 1
+    assert all(r["agent_status"] in ("SUCCESS", "FAILED") for r in rows)
+    ids = [r["order_id"] for r in rows]
+    assert len(ids) == len(set(ids))
