@@ -93,6 +93,8 @@ class Engine:
                  temperature: float = 0.0,
                  valid_vocab: tuple[int, int] | None = None,
                  decision_temperature: float = 1.0):
+        if torch.cuda.is_available():
+            _enable_tunableop()   # load the pre-tuned GEMM picks
         self.model = model
         # temperature 0 = greedy (the deterministic benchmark contract);
         # > 0 samples via the Gumbel-argmax trick, which stays a single
