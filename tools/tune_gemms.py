@@ -5,13 +5,25 @@ import torch
 t = torch.cuda.tunable
 t.enable(True)
 t.tuning_enable(True)
-# keep per-shape tuning bounded
-t.set_max_tuning_duration(100)      # ms per candidate set
-t.set_max_tuning_iterations(30)
+t.set_max_tuning_duration(50)
+t.set_max_tuning_iterations(10)
 
 dev = "cuda:0"
 SHAPES = [(6144, 4096), (4096, 4096), (28672, 4096), (4096, 14336)]
-BUCKETS = [4096, 8192, 12288, 16384, 24576, 40960, 49152, 57344]
+BUCKETS = [16384, 24576, 40960, 49152]
+OUT = "gpurun_out/tuned_new.csv"
+
+
+def dump():
+    vals = t.get_validators()
+    res = t.get_results()
+    with open(OUT, "w") as fh:
+        for k, v in vals:
+            fh.write(f"Validator,{k},{v}\n")
+        for op, params, sol, ms in res:
+            fh.write(f"{op},{params},{sol},{ms}\n")
+
+
 t0 = time.perf_counter()
 for M in BUCKETS:
     for N, K in SHAPES:
@@ -20,6 +32,7 @@ for M in BUCKETS:
         for _ in range(3):
             torch.nn.functional.linear(x, w)
         torch.cuda.synchronize()
-        print(f"tuned {N}x{M}x{K}  ({time.perf_counter()-t0:.0f}s)", flush=True)
-t.write_file("gpurun_out/tuned_new.csv")
-print("wrote gpurun_out/tuned_new.csv")
+        dump()   # persist incrementally (timeout-safe)
+        print(f"tuned {N}x{M}x{K}  ({time.perf_counter()-t0:.0f}s)",
+              flush=True)
+print("done")
