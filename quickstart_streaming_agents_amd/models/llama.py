@@ -359,6 +359,30 @@ class LlamaModel:
         use_hip_prefill = str(dev).startswith("cuda") and \
             self.cfg.d_head in (64, 128) and self.n_q % 4 == 0 and \
             _os.environ.get("QSA_NO_HIP_PREFILL") != "1"
+        # optional token-dim bucketing (QSA_PREFILL_PAD=1): pad the
+        # concatenated batch to a bucket with TunableOp-tuned GEMM
+        # entries (data/tunableop_mi355x.csv).  Pad rows run the
+        # projections but are NEVER scattered into the KV cache (the
+        # scatter uses sliced [:T] views) and no attention q-block maps
+        # cover them, so their garbage stays confined row-wise.
+        T_pad = T
+        if use_hip_prefill and _os.environ.get("QSA_PREFILL_PAD") == "1":
+            for b in (4096, 8192, 16384, 24576, 32768, 40960, 49152,
+                      57344, 65536):
+                if T <= b:
+                    T_pad = b
+                    break
+            else:
+                T_pad = T
+            if T_pad > T:
+                padn = T_pad - T
+                tokens = torch.cat([
+                    tokens, torch.zeros(padn, dtype=torch.int64,
+                                        device=dev)])
+                positions = torch.cat([
+                    positions,
+                    torch.zeros(padn, dtype=torch.int32, device=dev)
+                ]).contiguous()
         if use_hip_prefill:
             # varlen flash prefill over the paged cache: per-q-block maps
             # (one WAVE per 16 q rows x head in the kernel)
@@ -408,9 +432,9 @@ class LlamaModel:
                 h = D.rmsnorm_residual(mlp_out, res, L["attn_norm"],
                                        c.norm_eps)
             qkv = F.linear(h, L["wqkv"], L.get("bqkv"))
-            q, k, v = self._split_qkv(qkv, T)
+            q, k, v = self._split_qkv(qkv, T_pad)
             D.rope_inplace(q, k, self.rope_cos, self.rope_sin, positions)
-            D.kv_scatter(k, v, kv.k[li], kv.v[li], slots)
+            D.kv_scatter(k[:T], v[:T], kv.k[li], kv.v[li], slots)
             if use_hip_prefill:
                 # ONE varlen flash kernel: streams K/V pages directly, no
                 # gather/pad/score materialization
