@@ -145,6 +145,7 @@ class GpuTTLTable:
                  ttl_ms: int | None = None, device: str | None = None):
         import torch
         self.rows = rows
+        self.ts_col = ts_col
         self.ttl_ms = ttl_ms
         dev = device or ("cuda:0" if torch.cuda.is_available() else "cpu")
         keys = _key_hash([r[key_col] for r in rows]).to(dev)
@@ -159,17 +160,27 @@ class GpuTTLTable:
 
     def probe(self, probe_keys: list, now_ms: int) -> list[dict | None]:
         """Latest un-expired dimension row per probe key (None = miss)."""
-        import torch
-        if self.table is None:
+        hits = self.probe_rows(probe_keys)
+        if self.ttl_ms is None:
+            return hits
+        cutoff = now_ms - self.ttl_ms
+        return [h if h is not None and self._ts(h) >= cutoff else None
+                for h in hits]
+
+    def probe_rows(self, probe_keys: list) -> list[dict | None]:
+        """Latest dimension row per key with NO TTL filter (one batched
+        kernel probe); callers with per-row event times TTL-filter on the
+        host from the hit row's own timestamp — identical semantics to a
+        per-row-cutoff probe since the table keeps only the latest row."""
+        if self.table is None or not probe_keys:
             return [None] * len(probe_keys)
-        if not probe_keys:
-            return []
-        cutoff = (now_ms - self.ttl_ms) if self.ttl_ms is not None \
-            else -(1 << 62)
         pk = _key_hash(probe_keys).to(self.device)
-        rows_idx = self._D.hash_probe(self.table, pk, cutoff)
+        rows_idx = self._D.hash_probe(self.table, pk, -(1 << 62))
         return [self.rows[i] if i >= 0 else None
                 for i in rows_idx.cpu().tolist()]
+
+    def _ts(self, row: dict) -> int:
+        return int(row.get(self.ts_col, 0) or 0) if self.ts_col else 0
 
 
 def enrich_join_columnar(stream: list[dict], ts_fn, dims: list[tuple],
@@ -184,20 +195,16 @@ def enrich_join_columnar(stream: list[dict], ts_fn, dims: list[tuple],
     keep = [dict(r) for r in stream]
     ok = [True] * len(stream)
     for table, probe_col in tables:
-        hits_by_row = {}
         idx = [i for i in range(len(stream)) if ok[i]]
-        # per-row TTL cutoff uses each stream row's own event time: probe
-        # in groups of identical now (replays are chronological, so this
-        # is typically a handful of batches)
-        by_now: dict[int, list[int]] = {}
-        for i in idx:
-            by_now.setdefault(int(ts_fn(stream[i])), []).append(i)
-        for now, rows_i in sorted(by_now.items()):
-            hits = table.probe([stream[i][probe_col] for i in rows_i], now)
-            for i, hit in zip(rows_i, hits):
-                hits_by_row[i] = hit
-        for i in idx:
-            hit = hits_by_row.get(i)
+        # ONE batched probe (no cutoff), then per-row TTL filtering on
+        # the host against each stream row's own event time — identical
+        # semantics to per-row-cutoff probes since the table keeps only
+        # the latest row per key
+        hits = table.probe_rows([stream[i][probe_col] for i in idx])
+        ttl = table.ttl_ms
+        for i, hit in zip(idx, hits):
+            if hit is not None and ttl is not None and                     int(ts_fn(stream[i])) - table._ts(hit) > ttl:
+                hit = None
             if hit is None:
                 ok[i] = False
             else:
