@@ -47,3 +47,39 @@ def test_no_spills_or_scratch_on_hot_kernels(stats):
             assert st.get("sgpr_spill_count", 0) == 0, name
             assert st.get("private_segment_fixed_size", 0) == 0, \
                 f"{name}: scratch allocated"
+
+
+def test_round2_kernels_present_and_clean(stats):
+    """Round-2 kernels: fp8 GEMMs issue MFMA + the fp8 dequant path
+    (v_cvt_pk_f32_fp8); the hash-join kernels exist; none spill."""
+    for frag in ("skinny_gemm_fp8", "gemm_fp8_batch"):
+        hits = _named(stats, frag)
+        for name, st in hits.items():
+            assert st["mfma"] >= 8, f"{name}: only {st['mfma']} v_mfma"
+            assert st.get("vgpr_spill_count", 0) == 0, name
+            assert st.get("private_segment_fixed_size", 0) == 0, name
+    for frag in ("hash_build", "hash_probe", "splitk_reduce"):
+        for name, st in _named(stats, frag).items():
+            assert st.get("vgpr_spill_count", 0) == 0, name
+
+
+def test_fp8_kernels_use_hw_dequant(stats):
+    """The fp8 GEMMs must dequantize with the hardware cvt_pk_f32_fp8
+    path, not a software bit-twiddling fallback."""
+    import subprocess
+    import tempfile
+
+    import isa_check
+    so = isa_check.find_so()
+    with tempfile.TemporaryDirectory() as wd:
+        cos = isa_check.extract_hsacos(so, wd)
+        found = False
+        for co in cos:
+            dis = subprocess.run(
+                [os.path.join(os.path.dirname(isa_check.BUNDLER),
+                              "llvm-objdump"), "-d", co],
+                capture_output=True, text=True).stdout
+            if "cvt_pk_f32_fp8" in dis:
+                found = True
+                break
+        assert found, "no v_cvt_pk_f32_fp8 in any gfx950 code object"
