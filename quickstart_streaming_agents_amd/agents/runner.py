@@ -187,7 +187,6 @@ def episode(agent: AgentSpec, user_prompt: str,
             transcript.append(f"[thought] {action.note or last_text[:200]}")
             continue
         if isinstance(action, ToolCall):
-            ctx["calls"].append((action.name, action.arguments))
             if agent.tools is not None and not agent.tools.allows(action.name):
                 result = f"__error__ tool {action.name} not allowed"
             else:
@@ -207,6 +206,10 @@ def episode(agent: AgentSpec, user_prompt: str,
                 transcript.append(f"[tool {action.name} failed] {result}")
             else:
                 consecutive_failures = 0
+                # only SUCCESSFUL calls leave the grammar's candidate set
+                # (an identical retry of a FAILED call is legitimate and
+                # bounded by max_consecutive_failures)
+                ctx["calls"].append((action.name, action.arguments))
                 ctx["observations"].append((action.name, result))
                 if debug:
                     trace.append({"iteration": it, "tool": action.name,

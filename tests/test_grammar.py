@@ -324,3 +324,36 @@ def test_bpe_round_trip_fuzz(tok):
     # specials embedded mid-string survive too
     s = "a<|finish|>b<|tool_3|>\nc"
     assert tok.decode(tok.encode(s, bos=False)) == s
+
+
+def test_continuous_scheduler_failure_caps(tok):
+    """Tool failures under the continuous scheduler: an always-failing
+    tool trips max_consecutive_failures and the episode FAILS with the
+    abort message (the reference's cap semantics,
+    LAB1-Walkthrough.md:177-180).  A scripted always-call policy makes
+    the cap path deterministic; the LLM turns still run the real engine."""
+    from quickstart_streaming_agents_amd.agents.runner import ToolCall
+    from quickstart_streaming_agents_amd.agents.schedule import \
+        run_episodes_continuous
+    torch.manual_seed(3)
+    eng = _engine(tok)
+    llm = EngineLLM(eng, tok)
+    agent = AgentSpec("a", "m", "do the thing",
+                      ToolSet("t", allowed_tools=("http_get",)),
+                      max_iterations=8, max_consecutive_failures=2)
+
+    def broken_tool(name, args):
+        raise RuntimeError("connection refused")
+
+    def always_call(text, it, ctx):
+        return ToolCall("http_get", {"url": "http://dead.test/x"})
+
+    eps = [episode(agent, f"fetch number {i}", policy=always_call,
+                   max_new_tokens=12) for i in range(5)]
+    results = run_episodes_continuous(eps, llm, broken_tool)
+    assert len(results) == 5
+    for r in results:
+        assert r.status == "FAILED"
+        assert "2 consecutive" in r.response
+        assert r.iterations == 2 and r.tool_calls == 2
+        assert r.latency_s > 0
