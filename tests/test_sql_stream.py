@@ -345,3 +345,35 @@ def test_state_ttl_bounds_join_buffers():
         pipe2.advance()
     stats2 = {s["sink"]: s for s in pipe2.stats()}["joined"]
     assert stats2["join_buffered"] == 50 and stats2["join_evicted"] == 0
+
+
+def test_lab4_streaming_matches_batch(mcp):
+    """Lab4 (6-h TUMBLE + anomaly + interval join + 14-d state TTL +
+    agent + 4x REGEXP_EXTRACT) through the incremental executor, chunked
+    arrival == batch execution; Naples-only contract holds."""
+    ref = Deployment(labs=(4,), device="cpu")
+    ref.datagen(4)
+    batch_rows = ref.run_sql(4, mcp_server=mcp)
+
+    dep = Deployment(labs=(4,), device="cpu")
+    dep.datagen(4)
+    chunks = _chunked_records(dep, "claims", 6)
+    pipe = StreamingPipeline(dep.sql_executor(4, mcp_server=mcp))
+    incr: list = []
+    for ch in chunks:
+        _replay(dep, "claims", ch)
+        incr += pipe.advance()["claims_reviewed"]
+    incr += pipe.finish()["claims_reviewed"]
+
+    assert len(incr) == len(batch_rows) >= 1
+    assert sorted(r["claim_id"] for r in incr) == \
+        sorted(r["claim_id"] for r in batch_rows)
+    # content contract: the reviewed claims are the Naples-anomaly ones
+    # (claims_reviewed keeps the claim columns the CTAS projects)
+    naples = {k for k in incr[0] if "city" in k.lower()}
+    if naples:
+        col = sorted(naples)[0]
+        assert all(r[col] == "Naples" for r in incr)
+    vt = {r["verdict"] for r in incr}
+    assert vt <= {"APPROVE", "APPROVE_PARTIAL", "REQUEST_DOCS",
+                  "DENY_INELIGIBLE", "DENY_FRAUD"}
