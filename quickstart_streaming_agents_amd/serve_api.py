@@ -119,6 +119,7 @@ def create_app(llm: Callable[[list[str], list[int]], list[str]],
                 "decode_tokens": st.decode_tokens,
                 "decode_steps": st.decode_steps,
                 "prefill_batches": st.prefill_batches,
+                "jump_forward_tokens": st.jump_forward_tokens,
             }
         if index is not None:
             out["index_docs"] = len(index)
