@@ -15,11 +15,14 @@ engine (models/serve.py) gangs the "llm" requests into continuous-batching
 decode steps while "tool" requests run on an I/O thread pool — the GPU never
 waits on MCP round trips.
 
-Action selection is pluggable: ``ToolCallPolicy`` parses tool-call syntax
-from model output (for instruction-tuned weights); the lab harnesses install
-deterministic policies (labs/pipelines.py) because the north-star benchmark
-runs random-init weights — the LLM decode is real compute, the act decisions
-are scripted and documented as such.
+Action selection is MODEL-DRIVEN by default: episodes constructed with
+`tool_schemas` attach a per-turn grammar (models/grammar.py) and the
+engine's constrained decode lets the model choose the action (tool call
+vs finish) even under random-init weights; ``ToolCallPolicy`` parses the
+emitted ``TOOL_CALL {json}`` back into actions.  Scripted policies remain
+pluggable for the lab CONTENT-contract tests (labs/pipelines.py), where
+the assertions are about deterministic output text (French Quarter /
+Naples / verdict enums), not loop shape.
 """
 
 from __future__ import annotations

@@ -15,7 +15,9 @@ sequence alive between turns, the engine rolls the cache back to the shared
 prompt prefix and prefills only the delta (observation text) — turn N of an
 episode costs O(new tokens), not O(whole transcript).
 
-Greedy sampling keeps the benchmark deterministic.
+Sampling: free text decodes greedily (deterministic); grammar decision
+tokens sample from the model's masked distribution under a fixed seed
+(models/grammar.py) — the whole benchmark stays reproducible.
 """
 
 from __future__ import annotations
