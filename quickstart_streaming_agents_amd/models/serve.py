@@ -68,10 +68,11 @@ _TUNABLE_DONE = False
 
 
 def _enable_tunableop() -> None:
-    """Load the pre-tuned hipBLASLt/rocBLAS GEMM picks for the decode and
-    prefill shapes (PyTorch TunableOp results, generated once on MI355X:
-    13% faster decode steps).  Tuning itself stays OFF — unknown shapes
-    fall back to the default heuristics; any failure falls back silently."""
+    """Load the pre-tuned hipBLASLt/rocBLAS GEMM picks for the decode
+    and prefill shapes (PyTorch TunableOp results generated on MI355X by
+    tools/tune_gemms.py; +5% on the flagship bench measured A/B).
+    Tuning itself stays OFF — unknown shapes fall back to the default
+    heuristics; any failure falls back silently."""
     global _TUNABLE_DONE
     if _TUNABLE_DONE or _os.environ.get("QSA_NO_TUNABLEOP") == "1":
         return
@@ -138,7 +139,6 @@ class Engine:
         self._graph = None
         self._graph_failed = False
         self._gbuf: dict = {}
-        self._side_stream = None   # prefill stream for overlapped chunks
         # free-text sampling mask: tokens outside [lo, hi) (plus EOS) are
         # never emitted — keeps random-init decode inside the tokenizer's
         # trained vocab so outputs are real text (models/grammar.py)
