@@ -301,3 +301,68 @@ def test_murmur2_matches_java_kafka_client():
     }
     for k, signed in java.items():
         assert murmur2(k) == (signed & 0x7FFFFFFF), k
+
+
+def _dp_continuous_worker(rank, world):
+    """Per-rank mini flagship step: tiny engine + grammar episodes under
+    the CONTINUOUS scheduler, then the bench's MAX-over-ranks reduction
+    — the same shape the driver's multi-GPU bench launch runs."""
+    import torch
+    import torch.distributed as dist
+    from quickstart_streaming_agents_amd.agents.mcp import (McpClient,
+                                                            StubMcpServer)
+    from quickstart_streaming_agents_amd.agents.runner import (AgentSpec,
+                                                               ToolSet,
+                                                               episode)
+    from quickstart_streaming_agents_amd.agents.schedule import \
+        run_episodes_continuous
+    from quickstart_streaming_agents_amd.labs import datagen, pipelines
+    from quickstart_streaming_agents_amd.models.llama import (LlamaConfig,
+                                                              LlamaModel)
+    from quickstart_streaming_agents_amd.models.serve import Engine, EngineLLM
+    from quickstart_streaming_agents_amd.models.tokenizer import BpeTokenizer
+    dist.init_process_group("gloo")
+    torch.manual_seed(1234 + rank)
+    tok = BpeTokenizer(vocab_size=4096)
+    cfg = LlamaConfig.preset("tiny")
+    cfg.vocab_size = 4096
+    model = LlamaModel(cfg, device="cpu", dtype=torch.float32, seed=0)
+    eng = Engine(model, max_batch=8, max_seq_len=1024, eos_id=tok.EOS,
+                 valid_vocab=(tok._BYTE0, tok.n_tokens))
+    llm = EngineLLM(eng, tok)
+    with StubMcpServer() as srv:
+        client = McpClient(srv.mcp_endpoint)
+        schemas = {t["name"]: t.get("inputSchema", {})
+                   for t in client.tools_list()}
+        tool_fn = pipelines.mcp_tool_fn(client)
+        agent = AgentSpec("price_match_agent", "m",
+                          pipelines.LAB1_AGENT_PROMPT,
+                          ToolSet("t", allowed_tools=("http_get",
+                                                      "send_email")),
+                          max_iterations=5, max_consecutive_failures=2)
+        prods = datagen.lab1_products()
+        eps = []
+        for i in range(6):
+            p = prods[(i + rank) % len(prods)]
+            o = {"order_id": f"ORD-{rank}-{i}",
+                 "product_name": p["product_name"],
+                 "order_price": p["price"],
+                 "customer_email": f"u{rank}{i}@example.com"}
+            eps.append(episode(agent, pipelines.lab1_user_prompt(
+                o, f"{srv.base_url}/competitor", o["customer_email"]),
+                max_new_tokens=16, tool_schemas=schemas))
+        results = run_episodes_continuous(eps, llm, tool_fn)
+    import time
+    t = torch.tensor([float(rank) + 1.0], dtype=torch.float64)
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)   # bench's MAX-elapsed shape
+    dist.destroy_process_group()
+    return (len(results), sum(r.status == "SUCCESS" for r in results),
+            float(t.item()))
+
+
+def test_dp_continuous_scheduler_two_ranks():
+    results = spawn_world(_dp_continuous_worker, world=2)
+    for rank in (0, 1):
+        n, ok, mx = results[rank]
+        assert n == 6
+        assert mx == 2.0                      # reduction saw both ranks
