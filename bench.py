@@ -101,7 +101,14 @@ def main():
             "order_ts": i * 1000,
         })
 
-    server = StubMcpServer().start()
+    # pin the stub server's port (it appears in the prompts via the
+    # competitor URL): identical prompts -> identical token streams ->
+    # reproducible decisions across runs; fall back to an ephemeral port
+    # if the pinned one is taken
+    try:
+        server = StubMcpServer(port=18230 + local_rank).start()
+    except OSError:
+        server = StubMcpServer().start()
     client = McpClient(server.mcp_endpoint)
     tool_fn = pipelines.mcp_tool_fn(client)
     competitor_url = f"{server.base_url}/competitor"
