@@ -357,3 +357,32 @@ def test_continuous_scheduler_failure_caps(tok):
         assert "2 consecutive" in r.response
         assert r.iterations == 2 and r.tool_calls == 2
         assert r.latency_s > 0
+
+
+def test_engine_kv_pages_conserved_random_workload(tok):
+    """Property: after any mix of grammar turns, keep-alive
+    conversations, releases and jump-forwards, every KV page returns to
+    the free list (no page leaks across the continuous scheduler)."""
+    import random
+    rng = random.Random(5)
+    torch.manual_seed(5)
+    eng = _engine(tok, max_batch=6)
+    llm = EngineLLM(eng, tok)
+    total = eng.kv.free_pages
+    from quickstart_streaming_agents_amd.models.grammar import (ActionOption,
+                                                                TurnGrammar)
+    for round_i in range(4):
+        convs = [f"c{round_i}-{j}" for j in range(rng.randrange(1, 5))]
+        for turn in range(rng.randrange(1, 4)):
+            prompts, maxtoks, gs = [], [], []
+            for c in convs:
+                prompts.append(f"turn {turn} of {c} " + "x " * rng.randrange(0, 40))
+                maxtoks.append(rng.randrange(4, 20))
+                gs.append(TurnGrammar(options=[ActionOption(
+                    "http_get", {"url": f"http://t/{c}/{turn}"})])
+                    if rng.random() < 0.5 else None)
+            llm(prompts, maxtoks, convs, gs)
+        for c in convs:
+            llm.release(c)
+        assert eng.kv.free_pages == total, f"leak after round {round_i}"
+    assert not eng.running and not eng.pending

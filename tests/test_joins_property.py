@@ -114,3 +114,70 @@ def test_lab1_enriched_orders_gpu_flag_matches_cpu():
     a = pipelines.lab1_enriched_orders(b1, use_gpu=False)
     b = pipelines.lab1_enriched_orders(b2, use_gpu=True)  # CPU fallback
     assert a == b and len(a) >= 1
+
+
+def test_streaming_join_random_interleavings_match_batch():
+    """Property: for ANY arrival interleaving of the three lab1-shaped
+    topics (including duplicate sends), the streaming two-sided join
+    cascade emits exactly the matched combinations a batch join of the
+    same multiset produces — each pair exactly once (the r1 advisor bug
+    class)."""
+    import itertools
+    import random
+
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.sql.stream import StreamingPipeline
+    from quickstart_streaming_agents_amd.wire import Broker
+
+    rng = random.Random(11)
+    for trial in range(8):
+        orders = [{"o": f"o{i}", "c": f"c{rng.randrange(3)}",
+                   "p": f"p{rng.randrange(3)}", "ts": i}
+                  for i in range(rng.randrange(1, 7))]
+        custs = [{"c": f"c{i}", "cv": f"C{i}-{trial}"} for i in range(3)]
+        prods = [{"p": f"p{i}", "pv": f"P{i}-{trial}"} for i in range(3)]
+        events = [("orders", o) for o in orders] + \
+                 [("custs", c) for c in custs] + \
+                 [("prods", p) for p in prods]
+        # duplicate one dimension record (a second copy must re-match)
+        dup = rng.choice(custs)
+        events.append(("custs", dict(dup)))
+        rng.shuffle(events)
+
+        cat = Catalog()
+        cat.execute("""
+        CREATE TABLE joined AS
+          SELECT o.o AS o, c.cv AS cv, p.pv AS pv
+          FROM orders o JOIN custs c ON o.c = c.c
+                        JOIN prods p ON o.p = p.p;
+        """)
+        broker = Broker()
+        for t in ("orders", "custs", "prods"):
+            broker.create_topic(t)
+        pipe = StreamingPipeline(SqlExecutor(cat, broker))
+        got = []
+        for topic, rec in events:
+            broker.topics[topic].append(dict(rec), partition=0)
+            if rng.random() < 0.6:           # advance at random points
+                got += pipe.advance()["joined"]
+        got += pipe.advance()["joined"]
+
+        # batch reference over the multiset: every (order, cust-copy,
+        # prod) key-match combination appears exactly once
+        n_copies = {}
+        for t, r in events:
+            if t == "custs":
+                n_copies[r["c"]] = n_copies.get(r["c"], 0) + 1
+        want = []
+        for o in orders:
+            for c in custs:
+                if c["c"] != o["c"]:
+                    continue
+                for _ in range(n_copies.get(c["c"], 1) if c["c"] == dup["c"]
+                               else 1):
+                    for p in prods:
+                        if p["p"] == o["p"]:
+                            want.append((o["o"], c["cv"], p["pv"]))
+        assert sorted((r["o"], r["cv"], r["pv"]) for r in got) == \
+            sorted(want), f"trial {trial}"
