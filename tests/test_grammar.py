@@ -386,3 +386,15 @@ def test_engine_kv_pages_conserved_random_workload(tok):
             llm.release(c)
         assert eng.kv.free_pages == total, f"leak after round {round_i}"
     assert not eng.running and not eng.pending
+
+
+def test_bpe_round_trip_hypothesis(tok):
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=300, deadline=None)
+    @given(st.text(max_size=80))
+    def check(s):
+        assert tok.decode(tok.encode(s, bos=False)) == s
+
+    check()

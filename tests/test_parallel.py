@@ -366,3 +366,44 @@ def test_dp_continuous_scheduler_two_ranks():
         n, ok, mx = results[rank]
         assert n == 6
         assert mx == 2.0                      # reduction saw both ranks
+
+
+def test_murmur2_hypothesis_vs_reference():
+    """Property: the partitioner's murmur2 == a direct transcription of
+    org.apache.kafka.common.utils.Utils.murmur2 on arbitrary bytes."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from quickstart_streaming_agents_amd.parallel.stream_shard import murmur2
+
+    def ref(data: bytes) -> int:
+        length = len(data)
+        m, r = 0x5BD1E995, 24
+        h = (0x9747B28C ^ length) & 0xFFFFFFFF
+        for i in range(length // 4):
+            k = int.from_bytes(data[i * 4:i * 4 + 4], "little")
+            k = (k * m) & 0xFFFFFFFF
+            k ^= k >> r
+            k = (k * m) & 0xFFFFFFFF
+            h = (h * m) & 0xFFFFFFFF
+            h ^= k
+        base = length & ~3
+        rem = length % 4
+        if rem >= 3:
+            h ^= data[base + 2] << 16
+        if rem >= 2:
+            h ^= data[base + 1] << 8
+        if rem >= 1:
+            h ^= data[base]
+            h = (h * m) & 0xFFFFFFFF
+        h ^= h >> 13
+        h = (h * m) & 0xFFFFFFFF
+        h ^= h >> 15
+        return h & 0x7FFFFFFF
+
+    @settings(max_examples=300, deadline=None)
+    @given(st.binary(max_size=64))
+    def check(b):
+        assert murmur2(b) == ref(b)
+
+    check()
