@@ -93,7 +93,8 @@ def _safe_tool(tool: Callable[[str, dict], str], name: str, args: dict) -> str:
 def run_episodes_continuous(episodes: list[Generator], llm,
                             tool: Callable[[str, dict], str],
                             max_tool_workers: int = 16,
-                            decode_chunk: int = 8) -> list[EpisodeResult]:
+                            decode_chunk: int = 8,
+                            tracer=None) -> list[EpisodeResult]:
     """Event-driven episode scheduler over the continuous-batching engine.
 
     `run_episodes` advances episodes in lockstep ROUNDS: every episode's
@@ -107,6 +108,8 @@ def run_episodes_continuous(episodes: list[Generator], llm,
     another, and per-decision latency stops being quantized to rounds.
 
     Requires the conversation-aware EngineLLM (submit_turn/pop_finished).
+    tracer: optional runtime.trace.Tracer — one span per episode
+    (submit -> finish wall time, iterations/tool calls in meta).
     """
     import time as _time
     results: dict[int, EpisodeResult] = {}
@@ -121,6 +124,15 @@ def run_episodes_continuous(episodes: list[Generator], llm,
             results[idx] = stop.value
             if stop.value is not None:
                 stop.value.latency_s = _time.perf_counter() - started[idx]
+                if tracer is not None and tracer.enabled:
+                    from ..runtime.trace import Span
+                    tracer.spans.append(Span(
+                        stage=f"episode[{idx}]", t0=started[idx],
+                        dt=stop.value.latency_s, records_in=1,
+                        records_out=1,
+                        meta={"status": stop.value.status,
+                              "iterations": stop.value.iterations,
+                              "tool_calls": stop.value.tool_calls}))
             llm.release(idx)
             return
         if req[0] == "llm":

@@ -398,3 +398,23 @@ def test_bpe_round_trip_hypothesis(tok):
         assert tok.decode(tok.encode(s, bos=False)) == s
 
     check()
+
+
+def test_continuous_scheduler_traces_episodes(tok):
+    from quickstart_streaming_agents_amd.agents.schedule import \
+        run_episodes_continuous
+    from quickstart_streaming_agents_amd.runtime.trace import Tracer
+    torch.manual_seed(2)
+    eng = _engine(tok)
+    llm = EngineLLM(eng, tok)
+    agent = AgentSpec("a", "m", "answer briefly", None, max_iterations=3)
+    eps = [episode(agent, f"question {i}", max_new_tokens=8)
+           for i in range(4)]
+    tr = Tracer("bench")
+    results = run_episodes_continuous(eps, llm, lambda n, a: "",
+                                      tracer=tr)
+    assert len(results) == 4
+    ep_spans = [s for s in tr.spans if s.stage.startswith("episode[")]
+    assert len(ep_spans) == 4
+    assert all(s.dt > 0 and s.meta["status"] in ("SUCCESS", "FAILED")
+               for s in ep_spans)
