@@ -418,3 +418,30 @@ def test_continuous_scheduler_traces_episodes(tok):
     assert len(ep_spans) == 4
     assert all(s.dt > 0 and s.meta["status"] in ("SUCCESS", "FAILED")
                for s in ep_spans)
+
+
+def test_engine_edge_cases(tok):
+    """Engine boundary behavior: prompt at the sequence budget clamps
+    max_new to >= 1; run_chunk raises when a prompt can never be
+    admitted; out-of-vocab decode ids render to nothing."""
+    torch.manual_seed(0)
+    eng = _engine(tok, max_batch=2)
+    # prompt fills the whole budget: still decodes exactly 1 token
+    long_prompt = list(range(16, 16 + 4000))
+    seq = eng.submit(long_prompt, 64)
+    assert len(seq.prompt) == eng.max_seq_len - 1
+    assert seq.max_new_tokens == 1
+    eng.run_to_completion()
+    assert len(seq.out_tokens) == 1
+
+    # a prompt that cannot fit the KV pool -> MemoryError, not a hang
+    small = _engine(tok, max_batch=2)
+    small.kv = small.model.new_kv_cache(4)     # 4 pages = 256 tokens
+    big = small.submit(list(range(16, 16 + 500)), 8)
+    import pytest as _pytest
+    with _pytest.raises(MemoryError):
+        small.run_chunk()
+
+    # decode ids beyond the trained vocab render to nothing (random
+    # weights can emit them when no vocab mask is set)
+    assert tok.decode([tok.n_tokens + 5, 999999]) == ""
