@@ -323,10 +323,11 @@ class StreamingQuery:
             return self._finish_rows(self._window_close(panes))
         new_left = self._new_source_rows(self.stream_table)
         self._observe(self.stream_table, new_left)
-        new_right = {name: self._new_source_rows(name)
-                     for name, _ in self.joins}
-        for name, ds in new_right.items():
-            self._observe(name, ds)
+        new_right: dict[str, list[dict]] = {}
+        for name, _ in self.joins:
+            if name not in new_right:   # self-joins share one fetch
+                new_right[name] = self._new_source_rows(name)
+                self._observe(name, new_right[name])
         stamp = self._stream_time
         rows = [_Row({self.stream_alias: d}) for d in new_left]
         finished: list[_Row] = []

@@ -377,3 +377,29 @@ def test_lab4_streaming_matches_batch(mcp):
     vt = {r["verdict"] for r in incr}
     assert vt <= {"APPROVE", "APPROVE_PARTIAL", "REQUEST_DOCS",
                   "DENY_INELIGIBLE", "DENY_FRAUD"}
+
+
+def test_streaming_self_join_same_table_twice():
+    """A table joined twice in one FROM must feed BOTH stages (the
+    per-advance fetch is shared, not consumed twice)."""
+    from quickstart_streaming_agents_amd.sql.catalog import Catalog
+    from quickstart_streaming_agents_amd.sql.exec import SqlExecutor
+    from quickstart_streaming_agents_amd.wire import Broker
+    cat = Catalog()
+    cat.execute("""
+    CREATE TABLE pairs AS
+      SELECT e.id AS id, a.v AS av, b.v AS bv
+      FROM ev e JOIN dim a ON e.ka = a.k
+                JOIN dim b ON e.kb = b.k;
+    """)
+    broker = Broker()
+    broker.create_topic("ev")
+    broker.create_topic("dim")
+    pipe = StreamingPipeline(SqlExecutor(cat, broker))
+    broker.topics["dim"].append({"k": "x", "v": 1}, partition=0)
+    broker.topics["dim"].append({"k": "y", "v": 2}, partition=0)
+    broker.topics["ev"].append({"id": 7, "ka": "x", "kb": "y"},
+                               partition=0)
+    out = pipe.advance()["pairs"]
+    assert out == [{"id": 7, "av": 1, "bv": 2}]
+    assert pipe.advance()["pairs"] == []
