@@ -403,3 +403,35 @@ def test_streaming_self_join_same_table_twice():
     out = pipe.advance()["pairs"]
     assert out == [{"id": 7, "av": 1, "bv": 2}]
     assert pipe.advance()["pairs"] == []
+
+
+def test_lab1_join_snapshot_restore_with_ttl_stamps(mcp):
+    """Snapshot/restore round-trips the NEW join-buffer format
+    (stream-time stamps + eviction counters + per-partition offsets +
+    stream_time): restore mid-stream, finish identically."""
+    ref = Deployment(labs=(1,), device="cpu")
+    ref.datagen(1)
+    want = sorted(r["order_id"] for r in ref.run_sql(1, mcp_server=mcp))
+
+    dep = Deployment(labs=(1,), device="cpu")
+    dep.datagen(1)
+    orders = _chunked_records(dep, "orders", 2)
+    customers = _chunked_records(dep, "customers", 1)[0]
+    products = _chunked_records(dep, "products", 1)[0]
+    pipe1 = StreamingPipeline(dep.sql_executor(1, mcp_server=mcp))
+    got: list = []
+    _replay(dep, "customers", customers)
+    _replay(dep, "orders", orders[0])
+    got += pipe1.advance()["price_match_results"]   # no products yet
+    snap = json.loads(json.dumps(pipe1.snapshot()))
+
+    pipe2 = StreamingPipeline(dep.sql_executor(1, mcp_server=mcp))
+    pipe2.restore(snap)
+    q = pipe2.by_sink["price_match_results"]
+    assert sum(st.size() for _, st in q.joins) > 0   # buffers restored
+    _replay(dep, "products", products)
+    got += pipe2.advance()["price_match_results"]
+    for ch in orders[1:]:
+        _replay(dep, "orders", ch)
+        got += pipe2.advance()["price_match_results"]
+    assert sorted(r["order_id"] for r in got) == want
