@@ -427,7 +427,7 @@ def test_lab1_join_snapshot_restore_with_ttl_stamps(mcp):
 
     pipe2 = StreamingPipeline(dep.sql_executor(1, mcp_server=mcp))
     pipe2.restore(snap)
-    q = pipe2.by_sink["price_match_results"]
+    q = pipe2.by_sink["enriched_orders"]
     assert sum(st.size() for _, st in q.joins) > 0   # buffers restored
     _replay(dep, "products", products)
     got += pipe2.advance()["price_match_results"]
